@@ -1,0 +1,106 @@
+"""Training engine: the per-step hot loop.
+
+step:
+    forward (NHWC bf16/fp16 kernels)
+    fused softmax-CE loss
+    backward — per-param post-accumulate hooks fill flat-gradient buckets;
+               full buckets all-reduce (AVG) on the side HIP stream
+    reducer.finalize() — compute stream waits bucket events
+    fused SGD over the flat buffers (one kernel)
+"""
+from __future__ import annotations
+
+import time
+
+import torch
+
+from .flags import DistEnv
+from .models import build_model
+from .ops import functional as OF
+from .parallel.comm import make_communicator, NullComm
+from .parallel.ddp import GradReducer
+from .parallel.flat import FlatParams
+from .parallel.optim import FusedSGD
+from .parallel.sync_bn import convert_sync_batchnorm
+from .utils.amp import GradScaler
+
+_DTYPES = {"fp32": torch.float32, "bf16": torch.bfloat16,
+           "fp16": torch.float16}
+
+
+class Trainer:
+    def __init__(self, args, env: DistEnv | None = None):
+        self.args = args
+        self.env = env or DistEnv.from_env(getattr(args, "local_rank", None))
+        dev = args.device
+        if dev == "auto":
+            dev = "cuda" if torch.cuda.is_available() else "cpu"
+        if dev == "cuda":
+            torch.cuda.set_device(self.env.local_rank)
+            self.device = torch.device("cuda", self.env.local_rank)
+        else:
+            self.device = torch.device("cpu")
+        self.compute_dtype = _DTYPES[args.dtype]
+
+        torch.manual_seed(args.seed)  # same init on all ranks (+ broadcast)
+
+        num_classes = 10 if args.data == "synthetic-cifar10" else 1000
+        self.model = build_model(args.model, num_classes, self.compute_dtype)
+        self.model.to(self.device)
+
+        self.comm = make_communicator(self.env, self.device, args.backend)
+        if args.sync_bn and self.env.world_size > 1:
+            bn_comm = self.comm.dup()
+            self.model = convert_sync_batchnorm(self.model, bn_comm)
+
+        self.flat = FlatParams(self.model.named_parameters(),
+                               device=self.device)
+        comm_dtype = None
+        if args.grad_comm_dtype == "bf16":
+            comm_dtype = torch.bfloat16
+        elif args.grad_comm_dtype == "fp16":
+            comm_dtype = torch.float16
+        self.reducer = GradReducer(self.flat, self.comm,
+                                   bucket_cap_mb=args.bucket_cap_mb,
+                                   overlap=not args.no_overlap,
+                                   comm_dtype=comm_dtype)
+        self.reducer.broadcast_params(src=0)
+        self.optimizer = FusedSGD(self.flat, lr=args.lr,
+                                  momentum=args.momentum,
+                                  weight_decay=args.weight_decay)
+        self.scaler = GradScaler(enabled=(args.amp and args.dtype == "fp16"))
+        self.epoch = 0
+        self.global_step = 0
+        self._fault = self._parse_fault(args.fault_inject)
+
+    @staticmethod
+    def _parse_fault(spec: str):
+        # "step:N:rank:R" -> (N, R)
+        if not spec:
+            return None
+        parts = spec.split(":")
+        return int(parts[1]), int(parts[3])
+
+    def train_step(self, images, labels) -> torch.Tensor:
+        """One optimization step; returns the (device) scalar loss."""
+        if self._fault and self.global_step == self._fault[0] \
+                and self.env.rank == self._fault[1]:
+            raise RuntimeError(
+                f"fault injection: step {self.global_step} rank {self.env.rank}")
+        self.optimizer.zero_grad()
+        self.reducer.prepare()
+        logits = self.model(images)
+        loss = OF.softmax_cross_entropy(logits, labels,
+                                        self.args.label_smoothing)
+        self.scaler.scale(loss).backward()
+        self.reducer.finalize()
+        skip = self.scaler.unscale_and_check(self.flat.grad, self.comm)
+        if not skip:
+            self.optimizer.step()
+        self.global_step += 1
+        return loss.detach()
+
+    def state(self):
+        return dict(model=self.model, optimizer=self.optimizer,
+                    scaler=self.scaler, epoch=self.epoch,
+                    step=self.global_step)

@@ -1,0 +1,460 @@
+// Python binding for the gfx950 kernels + the RCCL communicator.
+// This is the only torch-facing translation unit; the .hip files are pure
+// CDNA4 device code exposed as extern "C" launchers.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>  // ROCm-native torch GPU context
+
+#include <hip/hip_runtime_api.h>
+#include <rccl/rccl.h>
+
+#include <vector>
+
+// ---- extern "C" launchers (see the .hip files) ---------------------------
+extern "C" {
+void pdta_relu_fwd(const void*, void*, long, int, hipStream_t);
+void pdta_relu_bwd_mask(const void*, const void*, void*, long, int,
+                        hipStream_t);
+void pdta_add_relu(const void*, const void*, void*, long, int, hipStream_t);
+void pdta_bn_stats(const void*, float*, float*, float*, float*, long, int,
+                   int, hipStream_t);
+void pdta_bn_apply(const void*, void*, const float*, const float*,
+                   const float*, const float*, long, int, int, int,
+                   hipStream_t);
+void pdta_bn_bwd_reduce(const void*, const void*, const float*, float*,
+                        float*, long, int, int, hipStream_t);
+void pdta_bn_bwd_elemt(const void*, const void*, void*, const float*,
+                       const float*, const float*, const float*,
+                       const float*, float, long, int, int, hipStream_t);
+void pdta_maxpool_fwd(const void*, void*, uint8_t*, int, int, int, int, int,
+                      int, int, int, int, int, hipStream_t);
+void pdta_maxpool_bwd(const void*, const uint8_t*, void*, int, int, int, int,
+                      int, int, int, int, int, int, hipStream_t);
+void pdta_gap_fwd(const void*, void*, int, long, int, int, hipStream_t);
+void pdta_gap_bwd(const void*, void*, int, long, int, int, hipStream_t);
+void pdta_fused_sgd(float*, const float*, float*, long, float, float, float,
+                    float, int, float, hipStream_t);
+void pdta_unscale_check(float*, long, float, float*, hipStream_t);
+void pdta_ce_fwd(const void*, const long*, float*, float*, int, int, float,
+                 int, hipStream_t);
+void pdta_ce_bwd(const void*, const long*, const float*, const float*, void*,
+                 int, int, float, float, int, hipStream_t);
+void pdta_gemm_nt(const void*, const void*, void*, const float*, int, int,
+                  int, int, int, hipStream_t);
+void pdta_colsum_f32(const void*, float*, long, int, int, hipStream_t);
+void pdta_conv_fwd(const void*, const void*, void*, const void*, int, int,
+                   int, int, int, int, int, int, int, int, int, int,
+                   hipStream_t);
+void pdta_conv_bwd_data(const void*, const void*, void*, const void*, int,
+                        int, int, int, int, int, int, int, int, int, int,
+                        int, hipStream_t);
+void pdta_conv_bwd_weight(const void*, const void*, float*, const void*, int,
+                          int, int, int, int, int, int, int, int, int, int,
+                          int, hipStream_t);
+void pdta_stem_fwd(const void*, const void*, void*, int, int, int, int, int,
+                   int, int, int, int, int, int, int, hipStream_t);
+void pdta_stem_bwd_weight(const void*, const void*, float*, int, int, int,
+                          int, int, int, int, int, int, int, int, int,
+                          hipStream_t);
+}
+
+namespace {
+
+int dt_of(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kFloat: return 0;
+    case at::kBFloat16: return 1;
+    case at::kHalf: return 2;
+    default: TORCH_CHECK(false, "unsupported dtype ", t.scalar_type());
+  }
+}
+
+hipStream_t cur_stream() {
+  return (hipStream_t)at::hip::getCurrentHIPStreamMasqueradingAsCUDA()
+      .stream();
+}
+
+#define CHECK_IN(t)                                            \
+  TORCH_CHECK(t.is_cuda(), #t " must be on GPU");              \
+  TORCH_CHECK(t.is_contiguous(), #t " must be contiguous")
+
+// 256-B zero page for out-of-bounds im2col taps (per device, lazy)
+const void* zero_page(const at::Tensor& like) {
+  static at::Tensor zp;
+  if (!zp.defined() || zp.device() != like.device())
+    zp = at::zeros({256}, like.options().dtype(at::kByte));
+  return zp.data_ptr();
+}
+
+int conv_out(int h, int k, int stride, int pad) {
+  return (h + 2 * pad - k) / stride + 1;
+}
+
+// ---- elementwise ---------------------------------------------------------
+at::Tensor relu_fwd(at::Tensor x) {
+  CHECK_IN(x);
+  auto y = at::empty_like(x);
+  pdta_relu_fwd(x.data_ptr(), y.data_ptr(), x.numel(), dt_of(x),
+                cur_stream());
+  return y;
+}
+
+at::Tensor relu_bwd_mask(at::Tensor dy, at::Tensor y) {
+  CHECK_IN(dy);
+  CHECK_IN(y);
+  auto dx = at::empty_like(dy);
+  pdta_relu_bwd_mask(dy.data_ptr(), y.data_ptr(), dx.data_ptr(), dy.numel(),
+                     dt_of(dy), cur_stream());
+  return dx;
+}
+
+at::Tensor add_relu_fwd(at::Tensor a, at::Tensor b) {
+  CHECK_IN(a);
+  CHECK_IN(b);
+  auto y = at::empty_like(a);
+  pdta_add_relu(a.data_ptr(), b.data_ptr(), y.data_ptr(), a.numel(),
+                dt_of(a), cur_stream());
+  return y;
+}
+
+// ---- batchnorm -----------------------------------------------------------
+std::vector<at::Tensor> bn_stats(at::Tensor x) {
+  CHECK_IN(x);
+  int C = x.size(3);
+  long rows = x.numel() / C;
+  auto opts = x.options().dtype(at::kFloat);
+  auto sum = at::zeros({C}, opts), sumsq = at::zeros({C}, opts);
+  auto mean = at::empty({C}, opts), var = at::empty({C}, opts);
+  pdta_bn_stats(x.data_ptr(), sum.data_ptr<float>(), sumsq.data_ptr<float>(),
+                mean.data_ptr<float>(), var.data_ptr<float>(), rows, C,
+                dt_of(x), cur_stream());
+  return {mean, var};
+}
+
+at::Tensor bn_apply(at::Tensor x, at::Tensor weight, at::Tensor bias,
+                    at::Tensor mean, at::Tensor invstd, bool relu) {
+  CHECK_IN(x);
+  int C = x.size(3);
+  long rows = x.numel() / C;
+  auto y = at::empty_like(x);
+  auto w = weight.to(at::kFloat).contiguous();
+  auto b = bias.to(at::kFloat).contiguous();
+  auto m = mean.contiguous(), is = invstd.contiguous();
+  pdta_bn_apply(x.data_ptr(), y.data_ptr(), w.data_ptr<float>(),
+                b.data_ptr<float>(), m.data_ptr<float>(),
+                is.data_ptr<float>(), rows, C, relu ? 1 : 0, dt_of(x),
+                cur_stream());
+  return y;
+}
+
+std::vector<at::Tensor> bn_bwd_reduce(at::Tensor x, at::Tensor dy,
+                                      at::Tensor mean, at::Tensor invstd) {
+  CHECK_IN(x);
+  CHECK_IN(dy);
+  int C = x.size(3);
+  long rows = x.numel() / C;
+  auto opts = x.options().dtype(at::kFloat);
+  auto sum_dy = at::zeros({C}, opts), sum_dy_xmu = at::zeros({C}, opts);
+  pdta_bn_bwd_reduce(x.data_ptr(), dy.data_ptr(),
+                     mean.contiguous().data_ptr<float>(),
+                     sum_dy.data_ptr<float>(), sum_dy_xmu.data_ptr<float>(),
+                     rows, C, dt_of(x), cur_stream());
+  return {sum_dy, sum_dy_xmu};
+}
+
+at::Tensor bn_bwd_elemt(at::Tensor x, at::Tensor dy, at::Tensor mean,
+                        at::Tensor invstd, at::Tensor weight,
+                        at::Tensor sum_dy, at::Tensor sum_dy_xmu,
+                        double count) {
+  CHECK_IN(x);
+  CHECK_IN(dy);
+  int C = x.size(3);
+  long rows = x.numel() / C;
+  auto dx = at::empty_like(x);
+  auto w = weight.to(at::kFloat).contiguous();
+  pdta_bn_bwd_elemt(x.data_ptr(), dy.data_ptr(), dx.data_ptr(),
+                    w.data_ptr<float>(), mean.contiguous().data_ptr<float>(),
+                    invstd.contiguous().data_ptr<float>(),
+                    sum_dy.contiguous().data_ptr<float>(),
+                    sum_dy_xmu.contiguous().data_ptr<float>(), (float)count,
+                    rows, C, dt_of(x), cur_stream());
+  return dx;
+}
+
+// ---- pooling -------------------------------------------------------------
+std::vector<at::Tensor> maxpool_fwd(at::Tensor x, int64_t k, int64_t stride,
+                                    int64_t pad) {
+  CHECK_IN(x);
+  int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  int P = conv_out(H, k, stride, pad), Q = conv_out(W, k, stride, pad);
+  auto y = at::empty({N, P, Q, C}, x.options());
+  auto idx = at::empty({N, P, Q, C}, x.options().dtype(at::kByte));
+  pdta_maxpool_fwd(x.data_ptr(), y.data_ptr(), idx.data_ptr<uint8_t>(), N, H,
+                   W, C, P, Q, k, stride, pad, dt_of(x), cur_stream());
+  return {y, idx};
+}
+
+at::Tensor maxpool_bwd(at::Tensor dy, at::Tensor idx, int64_t H, int64_t W,
+                       int64_t k, int64_t stride, int64_t pad) {
+  CHECK_IN(dy);
+  CHECK_IN(idx);
+  int N = dy.size(0), P = dy.size(1), Q = dy.size(2), C = dy.size(3);
+  auto dx = at::empty({N, (long)H, (long)W, (long)C}, dy.options());
+  pdta_maxpool_bwd(dy.data_ptr(), idx.data_ptr<uint8_t>(), dx.data_ptr(), N,
+                   H, W, C, P, Q, k, stride, pad, dt_of(dy), cur_stream());
+  return dx;
+}
+
+at::Tensor avgpool_global_fwd(at::Tensor x) {
+  CHECK_IN(x);
+  int N = x.size(0), C = x.size(3);
+  long HW = (long)x.size(1) * x.size(2);
+  auto y = at::empty({N, C}, x.options());
+  pdta_gap_fwd(x.data_ptr(), y.data_ptr(), N, HW, C, dt_of(x), cur_stream());
+  return y;
+}
+
+at::Tensor avgpool_global_bwd(at::Tensor dy, int64_t H, int64_t W) {
+  CHECK_IN(dy);
+  int N = dy.size(0), C = dy.size(1);
+  auto dx = at::empty({N, H, W, C}, dy.options());
+  pdta_gap_bwd(dy.data_ptr(), dx.data_ptr(), N, (long)H * W, C, dt_of(dy),
+               cur_stream());
+  return dx;
+}
+
+// ---- optimizer / amp -----------------------------------------------------
+void fused_sgd(at::Tensor p, at::Tensor g, at::Tensor m, double lr,
+               double mu, double wd, double damp, bool nesterov, double gs) {
+  CHECK_IN(p);
+  CHECK_IN(g);
+  pdta_fused_sgd(p.data_ptr<float>(), g.data_ptr<float>(),
+                 m.numel() ? m.data_ptr<float>() : nullptr, p.numel(), lr,
+                 mu, wd, damp, nesterov ? 1 : 0, gs, cur_stream());
+}
+
+at::Tensor unscale_check(at::Tensor g, double inv) {
+  CHECK_IN(g);
+  auto found = at::zeros({1}, g.options().dtype(at::kFloat));
+  pdta_unscale_check(g.data_ptr<float>(), g.numel(), inv,
+                     found.data_ptr<float>(), cur_stream());
+  return found;
+}
+
+// ---- loss ----------------------------------------------------------------
+std::vector<at::Tensor> softmax_ce_fwd(at::Tensor logits, at::Tensor labels,
+                                       double eps) {
+  CHECK_IN(logits);
+  CHECK_IN(labels);
+  int B = logits.size(0), n = logits.size(1);
+  auto opts = logits.options().dtype(at::kFloat);
+  auto loss = at::empty({B}, opts), lse = at::empty({B}, opts);
+  pdta_ce_fwd(logits.data_ptr(), labels.data_ptr<long>(),
+              loss.data_ptr<float>(), lse.data_ptr<float>(), B, n, eps,
+              dt_of(logits), cur_stream());
+  return {loss, lse};
+}
+
+at::Tensor softmax_ce_bwd(at::Tensor logits, at::Tensor labels,
+                          at::Tensor lse, double eps, double scale,
+                          at::Tensor gout) {
+  CHECK_IN(logits);
+  auto d = at::empty_like(logits);
+  pdta_ce_bwd(logits.data_ptr(), labels.data_ptr<long>(),
+              lse.data_ptr<float>(), gout.data_ptr<float>(), d.data_ptr(),
+              logits.size(0), logits.size(1), eps, scale, dt_of(logits),
+              cur_stream());
+  return d;
+}
+
+// ---- GEMM ----------------------------------------------------------------
+at::Tensor gemm_nt(at::Tensor a, at::Tensor b,
+                   c10::optional<at::Tensor> bias, bool out_f32) {
+  CHECK_IN(a);
+  CHECK_IN(b);
+  int M = a.size(0), K = a.size(1), N = b.size(0);
+  TORCH_CHECK(b.size(1) == K, "gemm_nt: K mismatch");
+  TORCH_CHECK(K % 64 == 0, "gemm_nt: K must be a multiple of 64 (pad)");
+  auto y = at::empty({M, N},
+                     out_f32 ? a.options().dtype(at::kFloat) : a.options());
+  const float* bp = nullptr;
+  at::Tensor bf;
+  if (bias.has_value() && bias->numel()) {
+    bf = bias->to(at::kFloat).contiguous();
+    bp = bf.data_ptr<float>();
+  }
+  pdta_gemm_nt(a.data_ptr(), b.data_ptr(), y.data_ptr(), bp, M, N, K,
+               dt_of(a), out_f32 ? 1 : 0, cur_stream());
+  return y;
+}
+
+at::Tensor colsum_f32(at::Tensor dy) {
+  CHECK_IN(dy);
+  auto db = at::zeros({dy.size(1)}, dy.options().dtype(at::kFloat));
+  pdta_colsum_f32(dy.data_ptr(), db.data_ptr<float>(), dy.size(0),
+                  dy.size(1), dt_of(dy), cur_stream());
+  return db;
+}
+
+// ---- conv ----------------------------------------------------------------
+at::Tensor conv2d_fwd(at::Tensor x, at::Tensor wt, int64_t stride,
+                      int64_t pad) {
+  // x [N,H,W,C]; wt pack [Kout,R,S,C]
+  CHECK_IN(x);
+  CHECK_IN(wt);
+  int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  int Kout = wt.size(0), R = wt.size(1), S = wt.size(2);
+  TORCH_CHECK(wt.size(3) == C, "conv: channel mismatch");
+  int P = conv_out(H, R, stride, pad), Q = conv_out(W, S, stride, pad);
+  auto y = at::empty({N, P, Q, Kout}, x.options());
+  if (C % 64 == 0) {
+    pdta_conv_fwd(x.data_ptr(), wt.data_ptr(), y.data_ptr(), zero_page(x), N,
+                  H, W, C, P, Q, Kout, R, S, stride, pad, dt_of(x),
+                  cur_stream());
+  } else {
+    TORCH_CHECK(Kout % 8 == 0, "stem conv needs Kout % 8 == 0");
+    pdta_stem_fwd(x.data_ptr(), wt.data_ptr(), y.data_ptr(), N, H, W, C, P,
+                  Q, Kout, R, S, stride, pad, dt_of(x), cur_stream());
+  }
+  return y;
+}
+
+at::Tensor conv2d_bwd_data(at::Tensor dy, at::Tensor w2, int64_t H,
+                           int64_t W, int64_t stride, int64_t pad) {
+  // dy [N,P,Q,Kout]; w2 pack [R,S,C,Kout]
+  CHECK_IN(dy);
+  CHECK_IN(w2);
+  int N = dy.size(0), P = dy.size(1), Q = dy.size(2), Kout = dy.size(3);
+  int R = w2.size(0), S = w2.size(1), C = w2.size(2);
+  TORCH_CHECK(C % 64 == 0, "bwd_data: C must be a multiple of 64 "
+                           "(the stem conv never needs dx)");
+  TORCH_CHECK(Kout % 64 == 0, "bwd_data: Kout must be a multiple of 64");
+  auto dx = at::empty({N, H, W, (long)C}, dy.options());
+  pdta_conv_bwd_data(dy.data_ptr(), w2.data_ptr(), dx.data_ptr(),
+                     zero_page(dy), N, H, W, C, P, Q, Kout, R, S, stride,
+                     pad, dt_of(dy), cur_stream());
+  return dx;
+}
+
+at::Tensor conv2d_bwd_weight(at::Tensor x, at::Tensor dy, int64_t R,
+                             int64_t S, int64_t stride, int64_t pad) {
+  CHECK_IN(x);
+  CHECK_IN(dy);
+  int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  int P = dy.size(1), Q = dy.size(2), Kout = dy.size(3);
+  auto dw = at::zeros({Kout, C, R, S}, x.options().dtype(at::kFloat));
+  if (C % 64 == 0) {
+    pdta_conv_bwd_weight(x.data_ptr(), dy.data_ptr(), dw.data_ptr<float>(),
+                         zero_page(x), N, H, W, C, P, Q, Kout, R, S, stride,
+                         pad, dt_of(x), cur_stream());
+  } else {
+    pdta_stem_bwd_weight(x.data_ptr(), dy.data_ptr(), dw.data_ptr<float>(),
+                         N, H, W, C, P, Q, Kout, R, S, stride, pad, dt_of(x),
+                         cur_stream());
+  }
+  return dw;
+}
+
+// ---- RCCL ----------------------------------------------------------------
+#define RCCL_CHECK(cmd)                                                  \
+  do {                                                                   \
+    ncclResult_t r = (cmd);                                              \
+    TORCH_CHECK(r == ncclSuccess, "RCCL error: ", ncclGetErrorString(r)); \
+  } while (0)
+
+at::Tensor rccl_unique_id() {
+  ncclUniqueId id;
+  RCCL_CHECK(ncclGetUniqueId(&id));
+  auto t = at::empty({(long)sizeof(ncclUniqueId)},
+                     at::TensorOptions().dtype(at::kByte));
+  memcpy(t.data_ptr(), &id, sizeof(id));
+  return t;
+}
+
+int64_t rccl_comm_init(at::Tensor uid, int64_t rank, int64_t world) {
+  TORCH_CHECK(uid.numel() >= (long)sizeof(ncclUniqueId));
+  ncclUniqueId id;
+  memcpy(&id, uid.contiguous().data_ptr(), sizeof(id));
+  ncclComm_t comm;
+  RCCL_CHECK(ncclCommInitRank(&comm, world, id, rank));
+  return (int64_t)(uintptr_t)comm;
+}
+
+ncclDataType_t nccl_dt(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kFloat: return ncclFloat32;
+    case at::kBFloat16: return ncclBfloat16;
+    case at::kHalf: return ncclFloat16;
+    case at::kByte: return ncclUint8;
+    case at::kLong: return ncclInt64;
+    default: TORCH_CHECK(false, "rccl: unsupported dtype");
+  }
+}
+
+void rccl_allreduce(at::Tensor t, int64_t comm, int64_t stream, int op) {
+  CHECK_IN(t);
+  RCCL_CHECK(ncclAllReduce(t.data_ptr(), t.data_ptr(), t.numel(),
+                           nccl_dt(t), (ncclRedOp_t)op,
+                           (ncclComm_t)(uintptr_t)comm,
+                           (hipStream_t)(uintptr_t)stream));
+}
+
+void rccl_allreduce_avg(at::Tensor t, int64_t comm, int64_t stream) {
+  rccl_allreduce(t, comm, stream, (int)ncclAvg);
+}
+
+void rccl_allreduce_sum(at::Tensor t, int64_t comm, int64_t stream) {
+  rccl_allreduce(t, comm, stream, (int)ncclSum);
+}
+
+void rccl_broadcast(at::Tensor t, int64_t src, int64_t comm, int64_t stream) {
+  CHECK_IN(t);
+  RCCL_CHECK(ncclBroadcast(t.data_ptr(), t.data_ptr(), t.numel(), nccl_dt(t),
+                           (int)src, (ncclComm_t)(uintptr_t)comm,
+                           (hipStream_t)(uintptr_t)stream));
+}
+
+void rccl_allgather(at::Tensor inp, at::Tensor out, int64_t comm,
+                    int64_t stream) {
+  CHECK_IN(inp);
+  CHECK_IN(out);
+  RCCL_CHECK(ncclAllGather(inp.data_ptr(), out.data_ptr(), inp.numel(),
+                           nccl_dt(inp), (ncclComm_t)(uintptr_t)comm,
+                           (hipStream_t)(uintptr_t)stream));
+}
+
+void rccl_comm_destroy(int64_t comm) {
+  ncclCommDestroy((ncclComm_t)(uintptr_t)comm);
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("relu_fwd", &relu_fwd);
+  m.def("relu_bwd_mask", &relu_bwd_mask);
+  m.def("add_relu_fwd", &add_relu_fwd);
+  m.def("bn_stats", &bn_stats);
+  m.def("bn_apply", &bn_apply);
+  m.def("bn_bwd_reduce", &bn_bwd_reduce);
+  m.def("bn_bwd_elemt", &bn_bwd_elemt);
+  m.def("maxpool_fwd", &maxpool_fwd);
+  m.def("maxpool_bwd", &maxpool_bwd);
+  m.def("avgpool_global_fwd", &avgpool_global_fwd);
+  m.def("avgpool_global_bwd", &avgpool_global_bwd);
+  m.def("fused_sgd", &fused_sgd);
+  m.def("unscale_check", &unscale_check);
+  m.def("softmax_ce_fwd", &softmax_ce_fwd);
+  m.def("softmax_ce_bwd", &softmax_ce_bwd);
+  m.def("gemm_nt", &gemm_nt, py::arg("a"), py::arg("b"),
+        py::arg("bias") = c10::nullopt, py::arg("out_f32") = false);
+  m.def("colsum_f32", &colsum_f32);
+  m.def("conv2d_fwd", &conv2d_fwd);
+  m.def("conv2d_bwd_data", &conv2d_bwd_data);
+  m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
+  m.def("rccl_unique_id", &rccl_unique_id);
+  m.def("rccl_comm_init", &rccl_comm_init);
+  m.def("rccl_allreduce_avg", &rccl_allreduce_avg);
+  m.def("rccl_allreduce_sum", &rccl_allreduce_sum);
+  m.def("rccl_broadcast", &rccl_broadcast);
+  m.def("rccl_allgather", &rccl_allgather);
+  m.def("rccl_comm_destroy", &rccl_comm_destroy);
+}

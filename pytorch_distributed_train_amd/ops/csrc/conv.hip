@@ -1,0 +1,589 @@
+// Implicit-GEMM convolution for gfx950 (SURVEY.md §2d conv rows; replaces
+// MIOpen's conv fwd / bwd-data / bwd-weights with NHWC-native MFMA
+// kernels — the north-star mandatory hand-written op).
+//
+// All three directions are GEMM-shaped with K contiguous on both operands
+// ("nt" shape, like gemm.hip) thanks to two weight packs:
+//   fwd:      y[m=npq][kout]  = sum_{rsc} im2col(x)[m][rsc] * Wt[kout][rsc]
+//             Wt pack = [Kout][R,S,C] (c innermost)   -> both K-contiguous
+//   bwd-data: dx[m=nhw][c]    = sum_{rs,ko} dyg[m][ko] * W2[r,s,c][ko]
+//             W2 pack = [R,S,C][Kout]                 -> both K-contiguous
+//   bwd-wei:  dW[kout][c]@rs  = sum_{npq} dy[row][kout] * xg[row][c]
+//             transposed LDS fragment reads, fp32 atomics, split-K over npq
+// The im2col/dy gathers happen in the per-lane SOURCE address of
+// global_load_lds; out-of-bounds taps load from a zero page (128 B of
+// zeros) so no lane predication is needed. C and Kout are multiples of 64
+// in every non-stem layer, so a 64-wide k-chunk never crosses an (r,s)
+// boundary. The C=3 stem (7x7/s2) gets direct kernels (stem_fwd/stem_wrw);
+// its dx is never needed (network input).
+#include "common.h"
+#include "mfma_tile.h"
+
+namespace {
+
+constexpr int TPB = 256;
+
+// ---------------------------------------------------------------- fwd --
+template <typename T>
+__global__ __launch_bounds__(TPB) void igemm_fwd_k(
+    const T* __restrict__ x, const T* __restrict__ wt, T* __restrict__ y,
+    const T* __restrict__ zero_page, int Nn, int H, int W, int C, int P,
+    int Q, int Kout, int R, int S, int stride, int pad, int tiles_n) {
+  __shared__ __attribute__((aligned(16))) char lds[2][2][128 * 64 * 2];
+
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  int tm = bid / tiles_n, tn = bid % tiles_n;
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  int wm = wave >> 1, wn = wave & 1;
+  long M = (long)Nn * P * Q;
+  int m0 = tm * 128, n0 = tn * 128;
+  int RSC = R * S * C;
+
+  // staging geometry for this thread's 4 A-rows (and 4 B-rows)
+  int row_in_chunk = lane >> 3;
+  int gcol16 = (lane & 7);  // swizzle applied per-row below
+  int a_n[4], a_h0[4], a_w0[4], a_swz[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int row = wave * 32 + i * 8 + row_in_chunk;
+    long m = m0 + row;
+    if (m > M - 1) m = M - 1;
+    int q = (int)(m % Q);
+    int p = (int)((m / Q) % P);
+    a_n[i] = (int)(m / ((long)P * Q));
+    a_h0[i] = p * stride - pad;
+    a_w0[i] = q * stride - pad;
+    a_swz[i] = gcol16 ^ (row & 7);
+  }
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  int cchunks = C / 64;
+  int ktiles = R * S * cchunks;
+
+  auto stage = [&](int kt, int buf) {
+    int tap = kt / cchunks, ct = kt % cchunks;
+    int r = tap / S, ss = tap % S;
+    // A: gathered x rows (or zero page)
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int h = a_h0[i] + r, w = a_w0[i] + ss;
+      const char* gptr;
+      if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
+        gptr = reinterpret_cast<const char*>(x) +
+               ((((long)a_n[i] * H + h) * W + w) * C + ct * 64) * 2 +
+               a_swz[i] * 16;
+      else
+        gptr = reinterpret_cast<const char*>(zero_page);
+      glds16(gptr, (char*)lds[buf][0] + (wave * 32 + i * 8) * 128);
+    }
+    // B: Wt rows (kout), k = tap*C + ct*64 contiguous
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int row = wave * 32 + i * 8 + row_in_chunk;
+      int ko = n0 + row;
+      if (ko > Kout - 1) ko = Kout - 1;
+      const char* gptr = reinterpret_cast<const char*>(wt) +
+                         ((long)ko * RSC + tap * C + ct * 64) * 2 +
+                         (gcol16 ^ (row & 7)) * 16;
+      glds16(gptr, (char*)lds[buf][1] + (wave * 32 + i * 8) * 128);
+    }
+  };
+
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  int cur = 0;
+  for (int kt = 0; kt < ktiles; ++kt) {
+    if (kt + 1 < ktiles) stage(kt + 1, cur ^ 1);
+    const char* la = lds[cur][0];
+    const char* lb = lds[cur][1];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      s16x8 af[4], bf[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        af[i] = frag_read128(la, wm * 64 + i * 16 + (lane & 15),
+                             (lane >> 4) + kk * 4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        bf[j] = frag_read128(lb, wn * 64 + j * 16 + (lane & 15),
+                             (lane >> 4) + kk * 4);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = MfmaOp<T>::run(af[i], bf[j], acc[i][j]);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    long rbase = m0 + wm * 64 + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = n0 + wn * 64 + j * 16 + (lane & 15);
+      if (col >= Kout) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = rbase + r;
+        if (row < M) y[row * Kout + col] = from_f32<T>(acc[i][j][r]);
+      }
+    }
+  }
+}
+
+// ----------------------------------------------------------- bwd-data --
+template <typename T>
+__global__ __launch_bounds__(TPB) void igemm_bwd_data_k(
+    const T* __restrict__ dy, const T* __restrict__ w2, T* __restrict__ dx,
+    const T* __restrict__ zero_page, int Nn, int H, int W, int C, int P,
+    int Q, int Kout, int R, int S, int stride, int pad, int tiles_n) {
+  __shared__ __attribute__((aligned(16))) char lds[2][2][128 * 64 * 2];
+
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  int tm = bid / tiles_n, tn = bid % tiles_n;
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  int wm = wave >> 1, wn = wave & 1;
+  long M = (long)Nn * H * W;
+  int m0 = tm * 128, n0 = tn * 128;
+
+  int row_in_chunk = lane >> 3;
+  int gcol16 = (lane & 7);
+  int a_n[4], a_h[4], a_w[4], a_swz[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int row = wave * 32 + i * 8 + row_in_chunk;
+    long m = m0 + row;
+    if (m > M - 1) m = M - 1;
+    a_w[i] = (int)(m % W);
+    a_h[i] = (int)((m / W) % H);
+    a_n[i] = (int)(m / ((long)H * W));
+    a_swz[i] = gcol16 ^ (row & 7);
+  }
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  int kchunks = Kout / 64;
+  int ktiles = R * S * kchunks;
+
+  auto stage = [&](int kt, int buf) {
+    int tap = kt / kchunks, kot = kt % kchunks;
+    int r = tap / S, ss = tap % S;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int pn = a_h[i] + pad - r;
+      int qn = a_w[i] + pad - ss;
+      int p = pn / stride, q = qn / stride;
+      bool valid = pn >= 0 && qn >= 0 && pn % stride == 0 &&
+                   qn % stride == 0 && p < P && q < Q;
+      const char* gptr;
+      if (valid)
+        gptr = reinterpret_cast<const char*>(dy) +
+               ((((long)a_n[i] * P + p) * Q + q) * Kout + kot * 64) * 2 +
+               a_swz[i] * 16;
+      else
+        gptr = reinterpret_cast<const char*>(zero_page);
+      glds16(gptr, (char*)lds[buf][0] + (wave * 32 + i * 8) * 128);
+    }
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int row = wave * 32 + i * 8 + row_in_chunk;
+      int c = n0 + row;
+      if (c > C - 1) c = C - 1;
+      const char* gptr = reinterpret_cast<const char*>(w2) +
+                         ((((long)tap) * C + c) * Kout + kot * 64) * 2 +
+                         (gcol16 ^ (row & 7)) * 16;
+      glds16(gptr, (char*)lds[buf][1] + (wave * 32 + i * 8) * 128);
+    }
+  };
+
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  int cur = 0;
+  for (int kt = 0; kt < ktiles; ++kt) {
+    if (kt + 1 < ktiles) stage(kt + 1, cur ^ 1);
+    const char* la = lds[cur][0];
+    const char* lb = lds[cur][1];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      s16x8 af[4], bf[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        af[i] = frag_read128(la, wm * 64 + i * 16 + (lane & 15),
+                             (lane >> 4) + kk * 4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        bf[j] = frag_read128(lb, wn * 64 + j * 16 + (lane & 15),
+                             (lane >> 4) + kk * 4);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = MfmaOp<T>::run(af[i], bf[j], acc[i][j]);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    long rbase = m0 + wm * 64 + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = n0 + wn * 64 + j * 16 + (lane & 15);
+      if (col >= C) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = rbase + r;
+        if (row < M) dx[row * C + col] = from_f32<T>(acc[i][j][r]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------- bwd-weight --
+// dW[kout][c] at one (r,s): 64x64 tile per block, split-K over npq rows
+// (64-row LDS tiles, transposed fragment reads), fp32 atomicAdd into the
+// torch-layout dW [Kout,C,R,S] (no repack needed). Rows beyond the image
+// or invalid taps stage from the zero page (adds 0).
+template <typename T>
+__global__ __launch_bounds__(TPB) void igemm_wrw_k(
+    const T* __restrict__ x, const T* __restrict__ dy,
+    float* __restrict__ dw, const T* __restrict__ zero_page, int Nn, int H,
+    int W, int C, int P, int Q, int Kout, int R, int S, int stride, int pad,
+    long rows_per_chunk, int ko_tiles, int c_tiles) {
+  __shared__ __attribute__((aligned(16))) char lds[2][2][64 * 64 * 2];
+
+  int bid = blockIdx.x;
+  int tap = bid / (ko_tiles * c_tiles);
+  int rem = bid % (ko_tiles * c_tiles);
+  int ko_t = rem / c_tiles, c_t = rem % c_tiles;
+  int r = tap / S, ss = tap % S;
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  int wm = wave >> 1, wn = wave & 1;  // 2x2 waves over 64(ko) x 64(c)
+
+  long NPQ = (long)Nn * P * Q;
+  long r0 = (long)blockIdx.y * rows_per_chunk;
+  long r1 = min(NPQ, r0 + rows_per_chunk);
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  int row_in_chunk = lane >> 3;
+  int col16 = lane & 7;
+
+  // stage 64 rows of dy [row][64 kout] and gathered x [row][64 c]; linear
+  // LDS (no swizzle): the fragment reads are transposed scalars anyway.
+  auto stage = [&](long rowbase, int buf) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      long row = rowbase + wave * 16 + i * 8 + row_in_chunk;
+      const char* gdy;
+      const char* gx;
+      if (row < r1) {
+        gdy = reinterpret_cast<const char*>(dy) +
+              (row * Kout + ko_t * 64) * 2 + col16 * 16;
+        int q = (int)(row % Q);
+        int p = (int)((row / Q) % P);
+        int n = (int)(row / ((long)P * Q));
+        int h = p * stride - pad + r, w = q * stride - pad + ss;
+        if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
+          gx = reinterpret_cast<const char*>(x) +
+               ((((long)n * H + h) * W + w) * C + c_t * 64) * 2 + col16 * 16;
+        else
+          gx = reinterpret_cast<const char*>(zero_page);
+      } else {
+        gdy = reinterpret_cast<const char*>(zero_page);
+        gx = reinterpret_cast<const char*>(zero_page);
+      }
+      int ldsrow = wave * 16 + i * 8;
+      glds16(gdy, (char*)lds[buf][0] + ldsrow * 128);
+      glds16(gx, (char*)lds[buf][1] + ldsrow * 128);
+    }
+  };
+
+  stage(r0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  int cur = 0;
+  for (long rb = r0; rb < r1; rb += 64) {
+    if (rb + 64 < r1) stage(rb + 64, cur ^ 1);
+    const char* ldy = lds[cur][0];
+    const char* lx = lds[cur][1];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      // transposed fragment reads: element (row, col) at row*128 + col*2
+      s16x8 af[2], bf[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        int ko_local = wm * 32 + i * 16 + (lane & 15);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int row = kk * 32 + (lane >> 4) * 8 + j;
+          af[i][j] = *reinterpret_cast<const short*>(
+              ldy + row * 128 + ko_local * 2);
+        }
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        int c_local = wn * 32 + i * 16 + (lane & 15);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int row = kk * 32 + (lane >> 4) * 8 + j;
+          bf[i][j] = *reinterpret_cast<const short*>(
+              lx + row * 128 + c_local * 2);
+        }
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = MfmaOp<T>::run(af[i], bf[j], acc[i][j]);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    int kbase = ko_t * 64 + wm * 32 + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int c = c_t * 64 + wn * 32 + j * 16 + (lane & 15);
+      if (c >= C) continue;
+#pragma unroll
+      for (int rg = 0; rg < 4; ++rg) {
+        int ko = kbase + rg;
+        if (ko < Kout)
+          atomicAdd(&dw[(((long)ko * C + c) * R + r) * S + ss],
+                    acc[i][j][rg]);
+      }
+    }
+  }
+}
+
+// ------------------------------------------------------------- stem ----
+// Direct 7x7/s2 C=3 forward: weights LDS-cached, 8 kout per thread.
+template <typename T>
+__global__ void stem_fwd_k(const T* __restrict__ x, const T* __restrict__ wt,
+                           T* __restrict__ y, int Nn, int H, int W, int Cin,
+                           int P, int Q, int Kout, int R, int S, int stride,
+                           int pad) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  T* wl = reinterpret_cast<T*>(smem);  // [R*S*Cin][Kout]
+  int rsc = R * S * Cin;
+  for (int i = threadIdx.x; i < rsc * Kout; i += blockDim.x)
+    wl[i] = wt[i];
+  __syncthreads();
+
+  int kg = threadIdx.x % (Kout / 8);        // kout group (8 each)
+  int pos_per_block = blockDim.x / (Kout / 8);
+  long m = (long)blockIdx.x * pos_per_block + threadIdx.x / (Kout / 8);
+  long M = (long)Nn * P * Q;
+  if (m >= M) return;
+  int q = (int)(m % Q);
+  int p = (int)((m / Q) % P);
+  int n = (int)(m / ((long)P * Q));
+  int h0 = p * stride - pad, w0 = q * stride - pad;
+
+  float acc[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) acc[j] = 0.f;
+  for (int rr = 0; rr < R; ++rr) {
+    int h = h0 + rr;
+    if ((unsigned)h >= (unsigned)H) continue;
+    for (int sc = 0; sc < S; ++sc) {
+      int w = w0 + sc;
+      if ((unsigned)w >= (unsigned)W) continue;
+      const T* xp = x + (((long)n * H + h) * W + w) * Cin;
+      for (int c = 0; c < Cin; ++c) {
+        float xv = to_f32(xp[c]);
+        const T* wp = wl + ((rr * S + sc) * Cin + c) * Kout + kg * 8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[j] += xv * to_f32(wp[j]);
+      }
+    }
+  }
+  T* yp = y + m * Kout + kg * 8;
+  Vec8<T> out;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) out.v[j] = from_f32<T>(acc[j]);
+  store8(yp, out);
+}
+
+// Stem bwd-weight: block stages 4 rows of (x window [rsc], dy [Kout]) at a
+// time; thread (kout = t%Kout, rsc strided) accumulates, atomicAdd at end.
+template <typename T>
+__global__ void stem_wrw_k(const T* __restrict__ x, const T* __restrict__ dy,
+                           float* __restrict__ dw, int Nn, int H, int W,
+                           int Cin, int P, int Q, int Kout, int R, int S,
+                           int stride, int pad, long rows_per_chunk) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  int rsc = R * S * Cin;
+  constexpr int RB = 4;
+  float* xw = reinterpret_cast<float*>(smem);           // [RB][rsc]
+  float* dyl = xw + RB * rsc;                           // [RB][Kout]
+
+  int ko = threadIdx.x % Kout;
+  int jbase = threadIdx.x / Kout;              // rsc stride group
+  int jstride = blockDim.x / Kout;
+  long NPQ = (long)Nn * P * Q;
+  long r0 = (long)blockIdx.x * rows_per_chunk;
+  long r1 = min(NPQ, r0 + rows_per_chunk);
+
+  // accumulators: ceil(rsc / jstride) per thread
+  float acc[64];
+  int nacc = (rsc + jstride - 1) / jstride;
+  for (int i = 0; i < nacc; ++i) acc[i] = 0.f;
+
+  for (long rb = r0; rb < r1; rb += RB) {
+    int nrows = (int)min((long)RB, r1 - rb);
+    // cooperative stage
+    for (int i = threadIdx.x; i < nrows * (rsc + Kout); i += blockDim.x) {
+      int rowi = i / (rsc + Kout), e = i % (rsc + Kout);
+      long row = rb + rowi;
+      int q = (int)(row % Q);
+      int p = (int)((row / Q) % P);
+      int n = (int)(row / ((long)P * Q));
+      if (e < rsc) {
+        int c = e % Cin, t = e / Cin;
+        int sc = t % S, rr = t / S;
+        int h = p * stride - pad + rr, w = q * stride - pad + sc;
+        float v = 0.f;
+        if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
+          v = to_f32(x[(((long)n * H + h) * W + w) * Cin + c]);
+        xw[rowi * rsc + e] = v;
+      } else {
+        int k = e - rsc;
+        dyl[rowi * Kout + k] = to_f32(dy[row * Kout + k]);
+      }
+    }
+    __syncthreads();
+    for (int rowi = 0; rowi < nrows; ++rowi) {
+      float g = dyl[rowi * Kout + ko];
+      for (int i = 0; i < nacc; ++i) {
+        int e = jbase + i * jstride;
+        if (e < rsc) acc[i] += g * xw[rowi * rsc + e];
+      }
+    }
+    __syncthreads();
+  }
+  for (int i = 0; i < nacc; ++i) {
+    int e = jbase + i * jstride;
+    if (e < rsc) {
+      int c = e % Cin, t = e / Cin;
+      int sc = t % S, rr = t / S;
+      // dW torch layout [Kout, Cin, R, S]
+      atomicAdd(&dw[(((long)ko * Cin + c) * R + rr) * S + sc], acc[i]);
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+void pdta_conv_fwd(const void* x, const void* wt, void* y,
+                   const void* zero_page, int Nn, int H, int W, int C, int P,
+                   int Q, int Kout, int R, int S, int stride, int pad,
+                   int dt, hipStream_t s) {
+  long M = (long)Nn * P * Q;
+  int tiles_m = (int)((M + 127) / 128), tiles_n = (Kout + 127) / 128;
+  dim3 grid(tiles_m * tiles_n);
+  PDTA_DISPATCH_DTYPE16(dt, T, {
+    igemm_fwd_k<T><<<grid, TPB, 0, s>>>((const T*)x, (const T*)wt, (T*)y,
+                                        (const T*)zero_page, Nn, H, W, C, P,
+                                        Q, Kout, R, S, stride, pad, tiles_n);
+  });
+}
+
+void pdta_conv_bwd_data(const void* dy, const void* w2, void* dx,
+                        const void* zero_page, int Nn, int H, int W, int C,
+                        int P, int Q, int Kout, int R, int S, int stride,
+                        int pad, int dt, hipStream_t s) {
+  long M = (long)Nn * H * W;
+  int tiles_m = (int)((M + 127) / 128), tiles_n = (C + 127) / 128;
+  dim3 grid(tiles_m * tiles_n);
+  PDTA_DISPATCH_DTYPE16(dt, T, {
+    igemm_bwd_data_k<T><<<grid, TPB, 0, s>>>(
+        (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H, W, C,
+        P, Q, Kout, R, S, stride, pad, tiles_n);
+  });
+}
+
+// dw fp32 [Kout,C,R,S] must be ZEROED by caller.
+void pdta_conv_bwd_weight(const void* x, const void* dy, float* dw,
+                          const void* zero_page, int Nn, int H, int W, int C,
+                          int P, int Q, int Kout, int R, int S, int stride,
+                          int pad, int dt, hipStream_t s) {
+  long NPQ = (long)Nn * P * Q;
+  int ko_tiles = (Kout + 63) / 64, c_tiles = (C + 63) / 64;
+  int tiles = ko_tiles * c_tiles * R * S;
+  long splitk = 2048 / tiles;
+  if (splitk < 1) splitk = 1;
+  if (splitk > 64) splitk = 64;
+  long rows_per_chunk = ((NPQ + splitk - 1) / splitk + 63) / 64 * 64;
+  long chunks = (NPQ + rows_per_chunk - 1) / rows_per_chunk;
+  dim3 grid(tiles, (unsigned)chunks);
+  PDTA_DISPATCH_DTYPE16(dt, T, {
+    igemm_wrw_k<T><<<grid, TPB, 0, s>>>(
+        (const T*)x, (const T*)dy, dw, (const T*)zero_page, Nn, H, W, C, P,
+        Q, Kout, R, S, stride, pad, rows_per_chunk, ko_tiles, c_tiles);
+  });
+}
+
+void pdta_stem_fwd(const void* x, const void* wt, void* y, int Nn, int H,
+                   int W, int Cin, int P, int Q, int Kout, int R, int S,
+                   int stride, int pad, int dt, hipStream_t s) {
+  long M = (long)Nn * P * Q;
+  int pos_per_block = 256 / (Kout / 8);
+  long grid = (M + pos_per_block - 1) / pos_per_block;
+  int lds_bytes = R * S * Cin * Kout * 2;
+  PDTA_DISPATCH_DTYPE16(dt, T, {
+    stem_fwd_k<T><<<(unsigned)grid, 256, lds_bytes, s>>>(
+        (const T*)x, (const T*)wt, (T*)y, Nn, H, W, Cin, P, Q, Kout, R, S,
+        stride, pad);
+  });
+}
+
+// dw fp32 [Kout,Cin,R,S] zeroed by caller.
+void pdta_stem_bwd_weight(const void* x, const void* dy, float* dw, int Nn,
+                          int H, int W, int Cin, int P, int Q, int Kout,
+                          int R, int S, int stride, int pad, int dt,
+                          hipStream_t s) {
+  long NPQ = (long)Nn * P * Q;
+  int rsc = R * S * Cin;
+  long chunks = 512;
+  long rows_per_chunk = (NPQ + chunks - 1) / chunks;
+  if (rows_per_chunk < 4) rows_per_chunk = 4;
+  chunks = (NPQ + rows_per_chunk - 1) / rows_per_chunk;
+  int lds_bytes = 4 * (rsc + Kout) * (int)sizeof(float);
+  PDTA_DISPATCH_DTYPE16(dt, T, {
+    stem_wrw_k<T><<<(unsigned)chunks, 256, lds_bytes, s>>>(
+        (const T*)x, (const T*)dy, dw, Nn, H, W, Cin, P, Q, Kout, R, S,
+        stride, pad, rows_per_chunk);
+  });
+}
+
+}  // extern "C"

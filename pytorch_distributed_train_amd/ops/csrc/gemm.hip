@@ -1,0 +1,177 @@
+// MFMA GEMM for gfx950 (SURVEY.md §2d FC/GEMM row; the "transformer MFMA
+// GEMM path" of BASELINE config #5 starts here).
+//
+// Shape convention: "nt" — C[M,N] = A[M,K] · B[N,K]^T with BOTH operands
+// row-major and K contiguous (torch Linear weight is already [N,K]; the
+// "nn"/"tn" variants are expressed as nt over cheaply-packed transposed
+// operands — see ops/functional.py). K must be a multiple of 64 (python
+// pads); M and N tails are clamp-loaded and mask-stored.
+//
+// Structure (cdna_hip_programming.md §5, the "minimum 2-phase" schedule):
+//   - 128x128 output tile, BK=64, 4 waves (256 threads), 64x64 per wave,
+//     4x4 fragments of v_mfma_f32_16x16x32_{bf16,f16}, fp32 accumulate
+//   - global->LDS staging via __builtin_amdgcn_global_load_lds width 16
+//     (the compiler never auto-emits it), double-buffered
+//   - LDS XOR swizzle col16 ^= (row&7) applied on the *source* address
+//     (glds writes lane-linear — guide §5.4 rule 21) and on fragment reads
+//   - XCD-aware bijective blockIdx swizzle (T1) for L2 locality
+#include "common.h"
+#include "mfma_tile.h"
+
+namespace {
+
+constexpr int BM = 128, BN = 128, BK = 64;
+constexpr int TPB = 256;  // 4 waves, 2x2 wave grid of 64x64 sub-tiles
+
+// stage one 128x64 operand tile: 4 waves x 4 glds, each glds = 64 lanes x
+// 16 B = 8 rows of 128 B, lane-linear dest; source address carries the
+// swizzle (rule 21); rows clamp to row_max (tails masked at the store).
+template <typename T>
+DEV_INLINE void stage_tile(const T* __restrict__ src, long ld,
+                           int row0, int row_max, char* lds_base, int wave,
+                           int lane) {
+  int row_in_chunk = lane >> 3;
+  int col16 = lane & 7;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int row = wave * 32 + i * 8 + row_in_chunk;
+    int gcol16 = col16 ^ (row & 7);
+    int grow = row0 + row;
+    if (grow > row_max) grow = row_max;
+    const char* gptr = reinterpret_cast<const char*>(src) +
+                       ((long)grow * ld) * 2 + gcol16 * 16;
+    glds16(gptr, lds_base + (wave * 32 + i * 8) * 128);
+  }
+}
+
+
+template <typename T, typename OutT, bool BIAS>
+__global__ __launch_bounds__(TPB) void gemm_nt_k(
+    const T* __restrict__ A, const T* __restrict__ B, OutT* __restrict__ C,
+    const float* __restrict__ bias, int M, int N, int K, int tiles_n) {
+  __shared__ __attribute__((aligned(16))) char lds[2][2][BM * BK * 2];
+
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  int tm = bid / tiles_n, tn = bid % tiles_n;
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  int wm = wave >> 1, wn = wave & 1;  // 2x2 wave grid
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  int m0 = tm * BM, n0 = tn * BN;
+  int ktiles = K / BK;
+
+  stage_tile<T>(A, K, m0, M - 1, lds[0][0], wave, lane);
+  stage_tile<T>(B, K, n0, N - 1, lds[0][1], wave, lane);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  int cur = 0;
+  for (int kt = 0; kt < ktiles; ++kt) {
+    if (kt + 1 < ktiles) {
+      stage_tile<T>(A + (long)(kt + 1) * BK, K, m0, M - 1, lds[cur ^ 1][0],
+                    wave, lane);
+      stage_tile<T>(B + (long)(kt + 1) * BK, K, n0, N - 1, lds[cur ^ 1][1],
+                    wave, lane);
+    }
+    const char* la = lds[cur][0];
+    const char* lb = lds[cur][1];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {  // two K=32 steps per BK=64
+      s16x8 af[4], bf[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int row = wm * 64 + i * 16 + (lane & 15);
+        af[i] = frag_read128(la, row, (lane >> 4) + kk * 4);
+      }
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int row = wn * 64 + j * 16 + (lane & 15);
+        bf[j] = frag_read128(lb, row, (lane >> 4) + kk * 4);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = MfmaOp<T>::run(af[i], bf[j], acc[i][j]);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // epilogue: acc[i][j] reg r -> C[m0+wm*64+i*16+(lane>>4)*4+r]
+  //                              [n0+wn*64+j*16+(lane&15)]
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int rbase = m0 + wm * 64 + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = n0 + wn * 64 + j * 16 + (lane & 15);
+      if (col >= N) continue;
+      float badd = BIAS ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = rbase + r;
+        if (row < M)
+          C[(long)row * N + col] = from_f32<OutT>(acc[i][j][r] + badd);
+      }
+    }
+  }
+}
+
+// column sum: db[n] = sum_m dy[m][n]  (linear bias grad)
+template <typename T>
+__global__ void colsum_f32_k(const T* __restrict__ dy, float* __restrict__ db,
+                             long M, int N) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= N) return;
+  long r0 = blockIdx.y, rs = gridDim.y;
+  float acc = 0.f;
+  for (long r = r0; r < M; r += rs) acc += to_f32(dy[r * N + c]);
+  atomicAdd(&db[c], acc);
+}
+
+}  // namespace
+
+extern "C" {
+
+// C[M,N] = A[M,K] B[N,K]^T (+bias); K % 64 == 0. out_f32: C is fp32.
+void pdta_gemm_nt(const void* A, const void* B, void* C, const float* bias,
+                  int M, int N, int K, int dt, int out_f32, hipStream_t s) {
+  int tiles_m = (M + BM - 1) / BM, tiles_n = (N + BN - 1) / BN;
+  dim3 grid(tiles_m * tiles_n);
+  PDTA_DISPATCH_DTYPE16(dt, T, {
+    if (out_f32) {
+      if (bias)
+        gemm_nt_k<T, float, true><<<grid, TPB, 0, s>>>(
+            (const T*)A, (const T*)B, (float*)C, bias, M, N, K, tiles_n);
+      else
+        gemm_nt_k<T, float, false><<<grid, TPB, 0, s>>>(
+            (const T*)A, (const T*)B, (float*)C, nullptr, M, N, K, tiles_n);
+    } else {
+      if (bias)
+        gemm_nt_k<T, T, true><<<grid, TPB, 0, s>>>(
+            (const T*)A, (const T*)B, (T*)C, bias, M, N, K, tiles_n);
+      else
+        gemm_nt_k<T, T, false><<<grid, TPB, 0, s>>>(
+            (const T*)A, (const T*)B, (T*)C, nullptr, M, N, K, tiles_n);
+    }
+  });
+}
+
+// db (fp32 [N], zeroed by caller) += column sums of dy [M,N]
+void pdta_colsum_f32(const void* dy, float* db, long M, int N, int dt,
+                     hipStream_t s) {
+  dim3 block(256);
+  dim3 grid((N + 255) / 256, (unsigned)min((long)512, M));
+  PDTA_DISPATCH_DTYPE16(dt, T, {
+    colsum_f32_k<T><<<grid, block, 0, s>>>((const T*)dy, db, M, N);
+  });
+}
+
+}  // extern "C"

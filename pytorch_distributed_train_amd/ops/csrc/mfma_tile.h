@@ -1,0 +1,53 @@
+// Shared MFMA tile machinery for the GEMM-shaped kernels (gemm.hip,
+// conv.hip): fragment types, 16x16x32 MFMA dispatch, swizzled LDS
+// addressing, global_load_lds staging, XCD-aware block swizzle.
+#pragma once
+#include "common.h"
+
+using bf16 = __hip_bfloat16;
+using f16 = __half;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(8))) short s16x8;
+
+template <typename T> struct MfmaOp;
+template <> struct MfmaOp<bf16> {
+  static DEV_INLINE f32x4 run(s16x8 a, s16x8 b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+  }
+};
+template <> struct MfmaOp<f16> {
+  static DEV_INLINE f32x4 run(s16x8 a, s16x8 b, f32x4 c) {
+    typedef __attribute__((ext_vector_type(8))) _Float16 f16x8;
+    union U { s16x8 s; f16x8 h; };
+    U ua, ub;
+    ua.s = a;
+    ub.s = b;
+    return __builtin_amdgcn_mfma_f32_16x16x32_f16(ua.h, ub.h, c, 0, 0, 0);
+  }
+};
+
+// LDS operand tile [rows][64 k-elements] of 2-B dtype: 128-B rows with the
+// bank-conflict XOR swizzle on the 16-B chunk index (T2; write side is the
+// pre-swizzled glds SOURCE address — guide §5.4 rule 21).
+DEV_INLINE int lds_off128(int row, int col16) {
+  return row * 128 + (col16 ^ (row & 7)) * 16;
+}
+
+DEV_INLINE s16x8 frag_read128(const char* lds, int row, int col16) {
+  return *reinterpret_cast<const s16x8*>(lds + lds_off128(row, col16));
+}
+
+// one glds wave-instruction: 64 lanes x 16 B -> 1 KiB at a wave-uniform
+// LDS base (lane-linear); per-lane global source address.
+DEV_INLINE void glds16(const void* gsrc, void* lds_dst) {
+  typedef const __attribute__((address_space(1))) uint32_t* gp_t;
+  typedef __attribute__((address_space(3))) uint32_t* lp_t;
+  __builtin_amdgcn_global_load_lds((gp_t)gsrc, (lp_t)lds_dst, 16, 0, 0);
+}
+
+// bijective XCD swizzle (T1): contiguous chunk of the grid per XCD
+DEV_INLINE int xcd_swizzle(int bid, int nwg) {
+  int q = nwg / 8, r = nwg % 8;
+  int xcd = bid % 8, idx = bid / 8;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+}

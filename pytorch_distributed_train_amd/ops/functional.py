@@ -1,0 +1,432 @@
+"""Autograd ops over NHWC activations.
+
+Every op has two implementations:
+  - CUDA (ROCm) tensors -> the in-tree gfx950 HIP extension (backend.ext()),
+    which raises loudly if not built — no silent eager fallback on GPU;
+  - CPU tensors -> a plain PyTorch fp32 reference. This is the numerics
+    oracle the GPU kernels are tested against (tests/test_kernels_gpu.py)
+    and what the gloo world_size=2 CPU plumbing config runs on.
+
+Layout contract: activations are [N, H, W, C] contiguous ("NHWC"); C is
+innermost so channel vectors are contiguous for MFMA K-dim staging and BN
+reductions. Conv weights: master fp32 [K, C, R, S] (torch checkpoint layout,
+SURVEY.md §5.4 checkpoint parity), packed compute copy [R, S, C, K] in the
+compute dtype. Linear weights: master fp32 [N, K] (torch layout), packed =
+same layout cast to the compute dtype (the MFMA GEMM consumes B^T = [N, K]).
+
+Reference behavior being rebuilt: MIOpen conv, ATen
+batch_norm kernels, fused log-softmax+NLL, maxpool/avgpool — re-designed
+as CDNA4 kernels, not ported.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+
+from .backend import ext
+
+
+def _nhwc_to_nchw(x):
+    return x.permute(0, 3, 1, 2).contiguous()
+
+
+def _nchw_to_nhwc(x):
+    return x.permute(0, 2, 3, 1).contiguous()
+
+
+def conv_out_dim(h: int, k: int, stride: int, pad: int) -> int:
+    return (h + 2 * pad - k) // stride + 1
+
+
+# ---------------------------------------------------------------------------
+# Convolution (implicit GEMM on GPU)
+# ---------------------------------------------------------------------------
+
+class Conv2dNHWC(torch.autograd.Function):
+    """y[n,p,q,k] = sum_{r,s,c} x[n, p*st-pad+r, q*st-pad+s, c] * w[k,c,r,s].
+
+    GPU: implicit-GEMM MFMA kernels (csrc/conv.hip). Two compute-dtype
+    weight packs keep every direction K-contiguous ("nt"-shaped):
+      wt [K,R,S,C] for forward, w2 [R,S,C,K] for backward-data (see
+      conv.hip header). Gradient is returned for the master fp32 weight
+    directly in its [K,C,R,S] layout (the wrw kernel scatters there).
+    """
+
+    @staticmethod
+    def forward(ctx, x, weight, wt, w2, stride: int, pad: int):
+        ctx.stride, ctx.pad = stride, pad
+        ctx.save_for_backward(x, weight, w2)
+        if x.is_cuda:
+            return ext().conv2d_fwd(x, wt, stride, pad)
+        # CPU reference (fp32)
+        w = weight.float()
+        y = F.conv2d(_nhwc_to_nchw(x).float(), w, stride=stride, padding=pad)
+        return _nchw_to_nhwc(y).to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, w2 = ctx.saved_tensors
+        stride, pad = ctx.stride, ctx.pad
+        dy = dy.contiguous()
+        need_dx = ctx.needs_input_grad[0]
+        need_dw = ctx.needs_input_grad[1]
+        dx = dw = None
+        if x.is_cuda:
+            if need_dx:
+                dx = ext().conv2d_bwd_data(dy, w2, x.shape[1], x.shape[2],
+                                           stride, pad)
+            if need_dw:
+                dw = ext().conv2d_bwd_weight(x, dy, weight.shape[2],
+                                             weight.shape[3], stride, pad)
+        else:
+            x_nchw = _nhwc_to_nchw(x).float()
+            dy_nchw = _nhwc_to_nchw(dy).float()
+            w = weight.float()
+            if need_dx:
+                dx = torch.nn.grad.conv2d_input(x_nchw.shape, w, dy_nchw,
+                                                stride=stride, padding=pad)
+                dx = _nchw_to_nhwc(dx).to(x.dtype)
+            if need_dw:
+                dw = torch.nn.grad.conv2d_weight(x_nchw, w.shape, dy_nchw,
+                                                 stride=stride, padding=pad)
+        return dx, dw, None, None, None, None
+
+
+def conv2d_nhwc(x, weight, wt, w2, stride: int = 1, pad: int = 0):
+    return Conv2dNHWC.apply(x, weight, wt, w2, stride, pad)
+
+
+# ---------------------------------------------------------------------------
+# BatchNorm (+ fused ReLU)
+# ---------------------------------------------------------------------------
+
+def _bn_cpu_stats(x):
+    x32 = x.float()
+    n = x.shape[0] * x.shape[1] * x.shape[2]
+    mean = x32.sum(dim=(0, 1, 2)) / n
+    var = x32.pow(2).sum(dim=(0, 1, 2)) / n - mean * mean
+    return mean, var.clamp_(min=0.0)
+
+
+class BatchNorm2dNHWCFn(torch.autograd.Function):
+    """Training-mode BN over [N,H,W,C] with optional fused ReLU.
+
+    Decomposed exactly as the kernels are (stats -> apply; bwd_reduce ->
+    bwd_elemt) so SyncBatchNorm reuses the same kernels with an all-gather
+    between stats and apply.
+    """
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, mean, invstd, fuse_relu: bool,
+                comm=None, count_override=None):
+        # mean/invstd are per-channel fp32, computed by the module (local
+        # stats for BN, cross-rank combined for SyncBN). For SyncBN, `comm`
+        # all-reduces the backward sums and `count_override` is the GLOBAL
+        # element count (all_reduce of [sum_dy, sum_dy_xmu]).
+        if x.is_cuda:
+            y = ext().bn_apply(x, weight, bias, mean, invstd, fuse_relu)
+        else:
+            x32 = x.float()
+            xhat = (x32 - mean) * invstd
+            y = xhat * weight.float() + bias.float()
+            if fuse_relu:
+                y = y.clamp_(min=0.0)
+            y = y.to(x.dtype)
+        ctx.fuse_relu = fuse_relu
+        ctx.comm = comm
+        ctx.count_override = count_override
+        ctx.save_for_backward(x, weight, mean, invstd, y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, mean, invstd, y = ctx.saved_tensors
+        fuse_relu = ctx.fuse_relu
+        dy = dy.contiguous()
+        n = x.shape[0] * x.shape[1] * x.shape[2]
+        if ctx.count_override is not None:
+            n = ctx.count_override
+        if x.is_cuda:
+            if fuse_relu:
+                dy = ext().relu_bwd_mask(dy, y)
+            sum_dy, sum_dy_xmu = ext().bn_bwd_reduce(x, dy, mean, invstd)
+            # param grads from LOCAL sums (DDP averages them afterwards,
+            # matching torch SyncBN's split of responsibilities)
+            dweight = sum_dy_xmu * invstd
+            dbias = sum_dy.clone() if ctx.comm is not None else sum_dy
+            if ctx.comm is not None:
+                packed = torch.cat([sum_dy, sum_dy_xmu])
+                ctx.comm.allreduce_sum(packed).sync_stream()
+                c = sum_dy.numel()
+                sum_dy, sum_dy_xmu = packed[:c], packed[c:]
+            dx = ext().bn_bwd_elemt(x, dy, mean, invstd, weight,
+                                    sum_dy.contiguous(),
+                                    sum_dy_xmu.contiguous(), float(n))
+        else:
+            dy32 = dy.float()
+            if fuse_relu:
+                dy32 = dy32 * (y.float() > 0)
+            x32 = x.float()
+            xmu = x32 - mean
+            sum_dy = dy32.sum(dim=(0, 1, 2))
+            sum_dy_xmu = (dy32 * xmu).sum(dim=(0, 1, 2))
+            dweight = sum_dy_xmu * invstd
+            dbias = sum_dy.clone() if ctx.comm is not None else sum_dy
+            if ctx.comm is not None:
+                packed = torch.cat([sum_dy, sum_dy_xmu])
+                ctx.comm.allreduce_sum(packed).wait()
+                c = sum_dy.numel()
+                sum_dy, sum_dy_xmu = packed[:c], packed[c:]
+            w32 = weight.float()
+            dx = (dy32 - sum_dy / n - xmu * invstd * invstd * sum_dy_xmu / n)
+            dx = (dx * w32 * invstd).to(x.dtype)
+        return dx, dweight, dbias, None, None, None, None, None
+
+
+def bn_stats(x):
+    """Per-channel (mean, var) over N*H*W in fp32. GPU: reduction kernel."""
+    if x.is_cuda:
+        return ext().bn_stats(x)
+    return _bn_cpu_stats(x)
+
+
+def bn_train(x, weight, bias, mean, invstd, fuse_relu=False, comm=None,
+             count=None):
+    return BatchNorm2dNHWCFn.apply(x, weight, bias, mean, invstd, fuse_relu,
+                                   comm, count)
+
+
+def bn_infer(x, weight, bias, running_mean, running_var, eps, fuse_relu=False):
+    invstd = (running_var.float() + eps).rsqrt()
+    if x.is_cuda:
+        return ext().bn_apply(x, weight, bias, running_mean.float(), invstd,
+                              fuse_relu)
+    x32 = x.float()
+    y = (x32 - running_mean) * invstd * weight.float() + bias.float()
+    if fuse_relu:
+        y = y.clamp_(min=0.0)
+    return y.to(x.dtype)
+
+
+# ---------------------------------------------------------------------------
+# Elementwise: relu, residual add+relu
+# ---------------------------------------------------------------------------
+
+class ReLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        y = ext().relu_fwd(x) if x.is_cuda else x.clamp(min=0)
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        if y.is_cuda:
+            return ext().relu_bwd_mask(dy.contiguous(), y)
+        return dy * (y > 0)
+
+
+class AddReLU(torch.autograd.Function):
+    """y = relu(a + b) — the residual join, one kernel instead of two."""
+
+    @staticmethod
+    def forward(ctx, a, b):
+        if a.is_cuda:
+            y = ext().add_relu_fwd(a, b)
+        else:
+            y = (a + b).clamp_(min=0)
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        if y.is_cuda:
+            d = ext().relu_bwd_mask(dy.contiguous(), y)
+        else:
+            d = dy * (y > 0)
+        return d, d
+
+
+def relu(x):
+    return ReLU.apply(x)
+
+
+def add_relu(a, b):
+    return AddReLU.apply(a, b)
+
+
+# ---------------------------------------------------------------------------
+# Pooling
+# ---------------------------------------------------------------------------
+
+class MaxPool2dNHWC(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, kernel: int, stride: int, pad: int):
+        ctx.kernel, ctx.stride, ctx.pad = kernel, stride, pad
+        ctx.in_shape = x.shape
+        if x.is_cuda:
+            y, idx = ext().maxpool_fwd(x, kernel, stride, pad)
+            ctx.save_for_backward(idx)
+            ctx.gpu = True
+            return y
+        x_nchw = _nhwc_to_nchw(x).float()
+        y, idx = F.max_pool2d(x_nchw, kernel, stride, pad, return_indices=True)
+        ctx.save_for_backward(idx)
+        ctx.gpu = False
+        return _nchw_to_nhwc(y).to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        n, h, w, c = ctx.in_shape
+        dy = dy.contiguous()
+        if ctx.gpu:
+            dx = ext().maxpool_bwd(dy, idx, h, w, ctx.kernel, ctx.stride,
+                                   ctx.pad)
+            return dx, None, None, None
+        dy_nchw = _nhwc_to_nchw(dy).float()
+        # scatter_ADD (not max_unpool, which overwrites duplicate winners of
+        # overlapping windows) to match true max_pool backward semantics
+        nn_, cc = dy_nchw.shape[0], dy_nchw.shape[1]
+        dx = torch.zeros(nn_, cc, h * w, dtype=dy_nchw.dtype)
+        dx.scatter_add_(2, idx.view(nn_, cc, -1), dy_nchw.view(nn_, cc, -1))
+        dx = dx.view(nn_, cc, h, w)
+        return _nchw_to_nhwc(dx).to(dy.dtype), None, None, None
+
+
+class GlobalAvgPoolNHWC(torch.autograd.Function):
+    """[N,H,W,C] -> [N,C] mean over H,W."""
+
+    @staticmethod
+    def forward(ctx, x):
+        ctx.in_shape = x.shape
+        if x.is_cuda:
+            return ext().avgpool_global_fwd(x)
+        return x.float().mean(dim=(1, 2)).to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        n, h, w, c = ctx.in_shape
+        dy = dy.contiguous()
+        if dy.is_cuda:
+            return ext().avgpool_global_bwd(dy, h, w)
+        dx = (dy.float() / (h * w)).view(n, 1, 1, c).expand(n, h, w, c)
+        return dx.to(dy.dtype).contiguous()
+
+
+def maxpool2d(x, kernel=3, stride=2, pad=1):
+    return MaxPool2dNHWC.apply(x, kernel, stride, pad)
+
+
+def global_avgpool(x):
+    return GlobalAvgPoolNHWC.apply(x)
+
+
+# ---------------------------------------------------------------------------
+# Linear (MFMA GEMM)
+# ---------------------------------------------------------------------------
+
+def _pad_cols64(t):
+    """Zero-pad the last dim to a multiple of 64 (GEMM K-dim contract)."""
+    k = t.shape[-1]
+    r = (-k) % 64
+    return t if r == 0 else F.pad(t, (0, r))
+
+class LinearFn(torch.autograd.Function):
+    """y = x @ W^T + b; x [M,K] compute dtype, W master fp32 [N,K]."""
+
+    @staticmethod
+    def forward(ctx, x, weight, w_packed, bias):
+        ctx.save_for_backward(x, weight, w_packed)
+        ctx.has_bias = bias is not None
+        if x.is_cuda:
+            return ext().gemm_nt(x, w_packed, bias)
+        y = x.float() @ weight.float().t()
+        if bias is not None:
+            y = y + bias.float()
+        return y.to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, w_packed = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = dw = db = None
+        if x.is_cuda:
+            # every variant is the nt kernel over cheap repacks; K-dims are
+            # zero-padded to multiples of 64 (zero rows/cols add nothing)
+            if ctx.needs_input_grad[0]:
+                dyp = _pad_cols64(dy)
+                wtp = _pad_cols64(w_packed.t().contiguous())  # [K_in, O]
+                dx = ext().gemm_nt(dyp, wtp)                  # [M, K_in]
+            if ctx.needs_input_grad[1]:
+                dyt = _pad_cols64(dy.t().contiguous())        # [O, M]
+                xt = _pad_cols64(x.t().contiguous())          # [K_in, M]
+                dw = ext().gemm_nt(dyt, xt, None, True)       # fp32 [O, K_in]
+            if ctx.has_bias and ctx.needs_input_grad[3]:
+                db = ext().colsum_f32(dy)                     # [M,N] -> [N]
+        else:
+            dy32, x32, w32 = dy.float(), x.float(), weight.float()
+            if ctx.needs_input_grad[0]:
+                dx = (dy32 @ w32).to(x.dtype)
+            if ctx.needs_input_grad[1]:
+                dw = dy32.t() @ x32
+            if ctx.has_bias and ctx.needs_input_grad[3]:
+                db = dy32.sum(dim=0)
+        return dx, dw, None, db
+
+
+def linear(x, weight, w_packed, bias=None):
+    return LinearFn.apply(x, weight, w_packed, bias)
+
+
+# ---------------------------------------------------------------------------
+# Fused softmax cross-entropy
+# ---------------------------------------------------------------------------
+
+class SoftmaxCrossEntropy(torch.autograd.Function):
+    """Mean CE over the batch; fused log-softmax + NLL fwd and
+    dlogits = (softmax - onehot)/B bwd (SURVEY.md §2d softmax-CE row)."""
+
+    @staticmethod
+    def forward(ctx, logits, labels, label_smoothing: float):
+        ctx.label_smoothing = label_smoothing
+        if logits.is_cuda:
+            loss_per, lse = ext().softmax_ce_fwd(logits, labels,
+                                                 label_smoothing)
+            ctx.save_for_backward(logits, labels, lse)
+            return loss_per.mean()
+        l32 = logits.float()
+        lse = torch.logsumexp(l32, dim=1)
+        nll = lse - l32.gather(1, labels.view(-1, 1)).squeeze(1)
+        if label_smoothing > 0.0:
+            eps = label_smoothing
+            smooth = lse - l32.mean(dim=1)
+            nll = (1 - eps) * nll + eps * smooth
+        ctx.save_for_backward(logits, labels, lse)
+        return nll.mean()
+
+    @staticmethod
+    def backward(ctx, grad):
+        logits, labels, lse = ctx.saved_tensors
+        eps = ctx.label_smoothing
+        b = logits.shape[0]
+        # grad is a device scalar tensor; never .item() it (host sync in the
+        # hot loop) — the kernel multiplies by it on device.
+        if logits.is_cuda:
+            dlogits = ext().softmax_ce_bwd(logits, labels, lse, eps, 1.0 / b,
+                                           grad.to(torch.float32))
+        else:
+            l32 = logits.float()
+            p = torch.exp(l32 - lse.unsqueeze(1))
+            n = l32.shape[1]
+            tgt = F.one_hot(labels, n).float()
+            if eps > 0.0:
+                tgt = (1 - eps) * tgt + eps / n
+            dlogits = ((p - tgt) * (grad.float() / b)).to(logits.dtype)
+        return dlogits, None, None
+
+
+def softmax_cross_entropy(logits, labels, label_smoothing: float = 0.0):
+    return SoftmaxCrossEntropy.apply(logits, labels, label_smoothing)

@@ -1,0 +1,208 @@
+"""Collective communicator abstraction.
+
+Data plane on GPU: our own RCCL binding (ops/_hip_ops owns the ncclComm_t;
+collectives are enqueued on caller-chosen HIP streams) — RCCL's topology
+detection drives the 7-link xGMI all-to-all pattern. Control plane /
+bootstrap and the CPU plumbing config: torch.distributed with gloo
+(init_method env://), kept cleanly behind this interface (no dual GPU
+paths: RcclComm is the only GPU communicator).
+
+Collective surface mirrors what the reference exercises (SURVEY.md §2b
+row 12): all_reduce, broadcast, all_gather_into_tensor, barrier.
+"""
+from __future__ import annotations
+
+import os
+
+import torch
+import torch.distributed as torch_dist
+
+from ..flags import DistEnv
+
+
+class Work:
+    def wait(self) -> None:  # host-side wait (CPU backends)
+        pass
+
+    def sync_stream(self, stream=None) -> None:
+        """Make `stream` (default: current) wait for completion, device-side."""
+        self.wait()
+
+
+class _DistWork(Work):
+    def __init__(self, work, post=None):
+        self._work = work
+        self._post = post
+
+    def wait(self) -> None:
+        if self._work is not None:
+            self._work.wait()
+        if self._post is not None:
+            self._post()
+            self._post = None
+
+
+class _EventWork(Work):
+    def __init__(self, event: "torch.cuda.Event"):
+        self._event = event
+
+    def wait(self) -> None:
+        self._event.synchronize()
+
+    def sync_stream(self, stream=None) -> None:
+        if stream is None:
+            stream = torch.cuda.current_stream()
+        stream.wait_event(self._event)
+
+
+class NullComm:
+    """world_size == 1: every collective is a no-op."""
+
+    rank = 0
+    world_size = 1
+
+    def allreduce_avg(self, tensor, stream=None) -> Work:
+        return Work()
+
+    def allreduce_sum(self, tensor, stream=None) -> Work:
+        return Work()
+
+    def broadcast(self, tensor, src: int = 0, stream=None) -> Work:
+        return Work()
+
+    def allgather_into(self, out, inp, stream=None) -> Work:
+        out.copy_(inp)
+        return Work()
+
+    def barrier(self) -> None:
+        pass
+
+    def dup(self) -> "NullComm":
+        return self
+
+
+class TorchDistComm:
+    """gloo via torch.distributed — CPU plumbing config (BASELINE config #1)
+    and rendezvous substrate for RcclComm bootstrap."""
+
+    def __init__(self, env: DistEnv, backend: str = "gloo"):
+        if not torch_dist.is_initialized():
+            os.environ.setdefault("MASTER_ADDR", env.master_addr)
+            os.environ.setdefault("MASTER_PORT", str(env.master_port))
+            torch_dist.init_process_group(
+                backend=backend, init_method="env://",
+                rank=env.rank, world_size=env.world_size)
+        self.rank = torch_dist.get_rank()
+        self.world_size = torch_dist.get_world_size()
+
+    def allreduce_sum(self, tensor, stream=None) -> Work:
+        w = torch_dist.all_reduce(tensor, op=torch_dist.ReduceOp.SUM,
+                                  async_op=True)
+        return _DistWork(w)
+
+    def allreduce_avg(self, tensor, stream=None) -> Work:
+        w = torch_dist.all_reduce(tensor, op=torch_dist.ReduceOp.SUM,
+                                  async_op=True)
+        ws = self.world_size
+        return _DistWork(w, post=lambda: tensor.div_(ws))
+
+    def broadcast(self, tensor, src: int = 0, stream=None) -> Work:
+        return _DistWork(torch_dist.broadcast(tensor, src=src, async_op=True))
+
+    def allgather_into(self, out, inp, stream=None) -> Work:
+        return _DistWork(
+            torch_dist.all_gather_into_tensor(out, inp, async_op=True))
+
+    def barrier(self) -> None:
+        torch_dist.barrier()
+
+    def dup(self) -> "TorchDistComm":
+        return self  # gloo enqueue is host-ordered; one PG suffices on CPU
+
+
+class RcclComm:
+    """RCCL over xGMI; the extension owns the ncclComm_t.
+
+    Bootstrap: ncclUniqueId from rank 0, exchanged over the gloo control
+    plane; ncclCommInitRank on the local device. Collectives enqueue on the
+    caller's HIP stream and complete device-side (sync_stream), so the
+    reducer can overlap them with backward on a side stream.
+    """
+
+    def __init__(self, env: DistEnv, control: TorchDistComm, device):
+        from ..ops.backend import ext
+        self._ext = ext()
+        self.rank = env.rank
+        self.world_size = env.world_size
+        self.device = device
+        uid = torch.zeros(128, dtype=torch.uint8)  # NCCL_UNIQUE_ID_BYTES
+        if self.rank == 0:
+            uid.copy_(self._ext.rccl_unique_id())
+        control.broadcast(uid, src=0).wait()
+        torch.cuda.set_device(device)
+        self._comm = self._ext.rccl_comm_init(uid, self.rank, self.world_size)
+        self._control = control
+
+    def _stream_ptr(self, stream):
+        s = stream if stream is not None else torch.cuda.current_stream()
+        return s.cuda_stream
+
+    def _run(self, fn, tensor, stream, *args) -> Work:
+        fn(tensor, *args, self._comm, self._stream_ptr(stream))
+        ev = torch.cuda.Event()
+        ev.record(stream if stream is not None
+                  else torch.cuda.current_stream())
+        return _EventWork(ev)
+
+    def allreduce_avg(self, tensor, stream=None) -> Work:
+        return self._run(self._ext.rccl_allreduce_avg, tensor, stream)
+
+    def allreduce_sum(self, tensor, stream=None) -> Work:
+        return self._run(self._ext.rccl_allreduce_sum, tensor, stream)
+
+    def broadcast(self, tensor, src: int = 0, stream=None) -> Work:
+        return self._run(self._ext.rccl_broadcast, tensor, stream, src)
+
+    def allgather_into(self, out, inp, stream=None) -> Work:
+        self._ext.rccl_allgather(inp, out, self._comm,
+                                 self._stream_ptr(stream))
+        ev = torch.cuda.Event()
+        ev.record(stream if stream is not None
+                  else torch.cuda.current_stream())
+        return _EventWork(ev)
+
+    def barrier(self) -> None:
+        self._control.barrier()
+        torch.cuda.synchronize()
+
+    def dup(self) -> "RcclComm":
+        """A second communicator (own ncclComm_t) for collectives that must
+        not share enqueue ordering with the reducer's (e.g. SyncBN on the
+        compute stream while buckets fly on the side stream)."""
+        clone = object.__new__(RcclComm)
+        clone._ext = self._ext
+        clone.rank, clone.world_size = self.rank, self.world_size
+        clone.device = self.device
+        uid = torch.zeros(128, dtype=torch.uint8)
+        if self.rank == 0:
+            uid.copy_(self._ext.rccl_unique_id())
+        self._control.broadcast(uid, src=0).wait()
+        clone._comm = self._ext.rccl_comm_init(uid, self.rank,
+                                               self.world_size)
+        clone._control = self._control
+        return clone
+
+
+def make_communicator(env: DistEnv, device, backend: str = "auto"):
+    """backend: auto|rccl|gloo. auto = rccl on cuda devices, gloo on cpu."""
+    if env.world_size <= 1:
+        return NullComm()
+    dev = torch.device(device)
+    if backend == "auto":
+        backend = "rccl" if dev.type == "cuda" else "gloo"
+    control = TorchDistComm(env, backend="gloo")
+    if backend == "gloo":
+        return control
+    if backend == "rccl":
+        return RcclComm(env, control, dev)
+    raise ValueError(f"unknown backend {backend}")

@@ -1,0 +1,168 @@
+"""Bucketed gradient reducer + DDP wrapper.
+
+Re-designs torch's C++ Reducer (reducer.hpp
+semantics) around the flat gradient buffer:
+
+  - buckets are contiguous slices of FlatParams.grad (params laid out in
+    reverse registration order, so arrival order ~= flat order; no
+    rebuild-by-arrival pass is needed and there is no bucket copy at all),
+  - each param carries a post-accumulate-grad hook; when the last param of
+    a bucket arrives, the bucket slice is all-reduced (AVG) — on GPU on a
+    dedicated side HIP stream that waits on an event recorded on the
+    compute stream, overlapping communication with the rest of backward
+    (SURVEY.md §3c),
+  - finalize() makes the compute stream wait on every bucket's completion
+    event before optimizer.step(), and flushes any bucket whose params
+    produced no gradient this iteration (grads are zeroed, so flushing
+    all-reduces zeros — rank-consistent by construction),
+  - bucket_cap is tuned for 7-link xGMI: fewer, larger buckets than the
+    NVSwitch-era 25 MiB default so each RCCL reduce-scatter/all-gather
+    keeps per-peer chunks well above the latency floor.
+
+Optional gradient compression (comm_dtype=bf16): the bucket slice is cast
+into a persistent bf16 shadow buffer, reduced there, and cast back —
+halves xGMI bytes at fp32-master accuracy (FP16CompressCommHook analog,
+default_comm_hooks.hpp:25 behavior).
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from .flat import FlatParams
+from .comm import NullComm
+
+
+class _Bucket:
+    __slots__ = ("index", "start", "end", "param_ids", "pending", "work")
+
+    def __init__(self, index, start, end, param_ids):
+        self.index = index
+        self.start = start
+        self.end = end
+        self.param_ids = param_ids
+        self.pending = 0
+        self.work = None
+
+
+class GradReducer:
+    def __init__(self, flat: FlatParams, comm, bucket_cap_mb: int = 50,
+                 overlap: bool = True, comm_dtype=None):
+        self.flat = flat
+        self.comm = comm
+        self.overlap = overlap and not isinstance(comm, NullComm)
+        self.comm_dtype = comm_dtype
+        self.enabled = not isinstance(comm, NullComm)
+
+        cap_elems = max(1, bucket_cap_mb) * (1 << 20) // 4  # fp32 elements
+        self.buckets: list[_Bucket] = []
+        self._param_bucket = {}
+        ids, start = [], 0
+        for i in range(flat.num_params):
+            ids.append(i)
+            end = flat.offsets[i] + flat.sizes[i]
+            nxt = flat.offsets[i + 1] if i + 1 < flat.num_params else end
+            if nxt - start >= cap_elems or i == flat.num_params - 1:
+                b = _Bucket(len(self.buckets), start, nxt, ids)
+                for j in ids:
+                    self._param_bucket[j] = b
+                self.buckets.append(b)
+                ids, start = [], nxt
+        # pad last bucket to the full flat length (alignment tail)
+        if self.buckets:
+            self.buckets[-1].end = flat.total
+
+        self._comm_stream = None
+        self._shadow = None
+        if self.enabled and flat.master.is_cuda:
+            self._comm_stream = torch.cuda.Stream(device=flat.device)
+            if comm_dtype is not None and comm_dtype != torch.float32:
+                self._shadow = torch.zeros(flat.total, dtype=comm_dtype,
+                                           device=flat.device)
+
+        self._hooks = []
+        if self.enabled:
+            for i, p in enumerate(flat.params):
+                self._hooks.append(p.register_post_accumulate_grad_hook(
+                    self._make_hook(i)))
+        self._iteration_open = False
+
+    # ------------------------------------------------------------------
+    def prepare(self) -> None:
+        """Call before each backward (resets bucket arrival counts)."""
+        if not self.enabled:
+            return
+        for b in self.buckets:
+            b.pending = len(b.param_ids)
+            b.work = None
+        self._iteration_open = True
+
+    def _make_hook(self, param_id: int):
+        def hook(_param):
+            if not self._iteration_open:
+                return
+            b = self._param_bucket[param_id]
+            b.pending -= 1
+            if b.pending == 0 and self.overlap:
+                self._launch(b)
+        return hook
+
+    def _launch(self, b: _Bucket) -> None:
+        view = self.flat.grad[b.start:b.end]
+        if self._comm_stream is not None:
+            ev = torch.cuda.Event()
+            ev.record()  # compute (current) stream: grads of b are ready
+            self._comm_stream.wait_event(ev)
+            if self._shadow is not None:
+                shadow = self._shadow[b.start:b.end]
+                with torch.cuda.stream(self._comm_stream):
+                    shadow.copy_(view)
+                w = self.comm.allreduce_avg(shadow, stream=self._comm_stream)
+                with torch.cuda.stream(self._comm_stream):
+                    view.copy_(shadow)
+                ev2 = torch.cuda.Event()
+                ev2.record(self._comm_stream)
+                from .comm import _EventWork
+                b.work = _EventWork(ev2)
+            else:
+                b.work = self.comm.allreduce_avg(view,
+                                                 stream=self._comm_stream)
+        else:
+            b.work = self.comm.allreduce_avg(view)
+
+    def finalize(self) -> None:
+        """After backward, before optimizer.step()."""
+        if not self.enabled:
+            return
+        self._iteration_open = False
+        for b in self.buckets:
+            if b.work is None:  # not launched (overlap off or unused params)
+                self._launch(b)
+        for b in self.buckets:
+            b.work.sync_stream()  # device-side wait on GPU, host wait on CPU
+            b.work = None
+
+    def broadcast_params(self, src: int = 0) -> None:
+        """Init-time coalesced parameter broadcast
+        (nn/parallel/distributed.py:1012 analog — one flat broadcast)."""
+        if self.enabled:
+            # sync_stream: host wait on CPU backends, device-side stream
+            # dependency on GPU (params ready before first forward kernel)
+            self.comm.broadcast(self.flat.master, src=src).sync_stream()
+
+
+class DistributedDataParallel(nn.Module):
+    """API-parity wrapper: model = DDP(model, flat, comm).forward(...)"""
+
+    def __init__(self, module: nn.Module, flat: FlatParams, comm,
+                 bucket_cap_mb: int = 50, overlap: bool = True,
+                 comm_dtype=None):
+        super().__init__()
+        self.module = module
+        self.reducer = GradReducer(flat, comm, bucket_cap_mb, overlap,
+                                   comm_dtype)
+        self.reducer.broadcast_params(src=0)
+
+    def forward(self, *args, **kwargs):
+        self.reducer.prepare()
+        return self.module(*args, **kwargs)

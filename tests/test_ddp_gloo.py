@@ -1,0 +1,112 @@
+"""Multi-process CPU integration (BASELINE config #1): gloo world_size=2.
+
+Oracle: with SyncBatchNorm, DDP gradients on ranks holding
+the two halves of a global batch must equal the single-process gradients
+on the full batch; with identical per-rank batches, plain-BN DDP gradients
+equal the single-process gradients (all-reduce AVG of identical grads).
+"""
+import os
+import socket
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from pytorch_distributed_train_amd.flags import build_parser
+
+pytestmark = pytest.mark.timeout(300)
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _make_args(extra=()):
+    return build_parser().parse_args([
+        "--model", "resnet18", "--data", "synthetic-cifar10",
+        "--dtype", "fp32", "--device", "cpu", "--backend", "gloo",
+        "--batch-size", "4", "--seed", "11", *extra])
+
+
+def _global_batch(n=8):
+    g = torch.Generator().manual_seed(99)
+    x = torch.randn(n, 3, 32, 32, generator=g)
+    y = torch.randint(0, 10, (n,), generator=g)
+    return x, y
+
+
+def _worker(rank, world, port, sync_bn, identical_batches, out_dir):
+    os.environ.update(RANK=str(rank), LOCAL_RANK=str(rank),
+                      WORLD_SIZE=str(world), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    from pytorch_distributed_train_amd.engine import Trainer
+    extra = ["--sync-bn"] if sync_bn else []
+    trainer = Trainer(_make_args(extra))
+    x, y = _global_batch()
+    if identical_batches:
+        xb, yb = x[:4], y[:4]
+    else:
+        xb, yb = x[rank * 4:(rank + 1) * 4], y[rank * 4:(rank + 1) * 4]
+    trainer.model.train()
+    loss = trainer.train_step(xb, yb)
+    torch.save({"grad": trainer.flat.grad.clone(),
+                "master": trainer.flat.master.clone(),
+                "loss": loss},
+               os.path.join(out_dir, f"rank{rank}.pt"))
+    import torch.distributed as dist
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def _single_process_grads(x, y, sync_semantics):
+    from pytorch_distributed_train_amd.engine import Trainer
+    trainer = Trainer(_make_args())
+    trainer.model.train()
+    trainer.train_step(x, y)
+    return trainer.flat.grad.clone(), trainer.flat.master.clone()
+
+
+def _run_workers(sync_bn, identical_batches, tmp_path):
+    port = _free_port()
+    ctx = mp.get_context("spawn")
+    procs = []
+    for r in range(2):
+        p = ctx.Process(target=_worker,
+                        args=(r, 2, port, sync_bn, identical_batches,
+                              str(tmp_path)))
+        p.start()
+        procs.append(p)
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0, f"worker failed rc={p.exitcode}"
+    r0 = torch.load(tmp_path / "rank0.pt", weights_only=False)
+    r1 = torch.load(tmp_path / "rank1.pt", weights_only=False)
+    return r0, r1
+
+
+def test_ddp_identical_batches_plain_bn(tmp_path):
+    r0, r1 = _run_workers(sync_bn=False, identical_batches=True,
+                          tmp_path=tmp_path)
+    # identical inputs -> identical grads; AVG all-reduce must be a no-op
+    assert torch.allclose(r0["grad"], r1["grad"], atol=1e-6)
+    x, y = _global_batch()
+    ref_grad, ref_master = _single_process_grads(x[:4], y[:4], False)
+    assert torch.allclose(r0["grad"], ref_grad, atol=1e-5), \
+        (r0["grad"] - ref_grad).abs().max()
+    # one SGD step on identical grads -> identical params across ranks
+    assert torch.allclose(r0["master"], r1["master"], atol=1e-6)
+    assert torch.allclose(r0["master"], ref_master, atol=1e-5)
+
+
+def test_ddp_split_batch_syncbn_matches_single_process(tmp_path):
+    r0, r1 = _run_workers(sync_bn=True, identical_batches=False,
+                          tmp_path=tmp_path)
+    assert torch.allclose(r0["grad"], r1["grad"], atol=1e-6)
+    x, y = _global_batch()
+    ref_grad, _ = _single_process_grads(x, y, True)
+    assert torch.allclose(r0["grad"], ref_grad, atol=3e-5), \
+        (r0["grad"] - ref_grad).abs().max()

@@ -1,0 +1,62 @@
+"""Gang launcher: env contract, restart-on-failure, exit codes."""
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+OK_SCRIPT = """
+import json, os, sys
+out = sys.argv[-1]
+with open(os.path.join(out, f"w{os.environ['RANK']}.json"), "w") as f:
+    json.dump({k: os.environ[k] for k in
+               ("RANK", "LOCAL_RANK", "WORLD_SIZE", "MASTER_ADDR",
+                "MASTER_PORT")}, f)
+# legacy --local_rank argv must be present unless --use-env
+assert any(a.startswith("--local_rank=") for a in sys.argv[1:]), sys.argv
+"""
+
+FAIL_ONCE_SCRIPT = """
+import os, sys
+out = sys.argv[-1]
+marker = os.path.join(out, "attempt_" + os.environ["RANK"])
+first = not os.path.exists(marker)
+open(marker, "w").close()
+if first and os.environ["RANK"] == "1":
+    sys.exit(7)   # first gang attempt: rank 1 dies
+"""
+
+
+def _launch(script_body, tmp_path, extra=()):
+    script = tmp_path / "worker.py"
+    script.write_text(script_body)
+    cmd = [sys.executable, "-m", "pytorch_distributed_train_amd.launch",
+           "--nproc_per_node=2", "--master_port=29612",
+           "--monitor-interval=0.05", *extra, str(script), str(tmp_path)]
+    return subprocess.run(cmd, cwd=REPO, capture_output=True, text=True,
+                          timeout=120)
+
+
+def test_launcher_env_contract(tmp_path):
+    r = _launch(OK_SCRIPT, tmp_path)
+    assert r.returncode == 0, r.stderr
+    import json
+    for rank in (0, 1):
+        env = json.load(open(tmp_path / f"w{rank}.json"))
+        assert env["RANK"] == str(rank)
+        assert env["LOCAL_RANK"] == str(rank)
+        assert env["WORLD_SIZE"] == "2"
+        assert env["MASTER_ADDR"] == "127.0.0.1"
+
+
+def test_launcher_restarts_gang(tmp_path):
+    r = _launch(FAIL_ONCE_SCRIPT, tmp_path, extra=("--max-restarts=2",))
+    assert r.returncode == 0, r.stderr
+    assert "restarting gang (1/2)" in r.stderr
+
+
+def test_launcher_exhausts_restarts(tmp_path):
+    script = "import sys; sys.exit(3)"
+    r = _launch(script, tmp_path, extra=("--max-restarts=1",))
+    assert r.returncode == 3
+    assert "exhausted" in r.stderr

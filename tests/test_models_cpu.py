@@ -1,0 +1,101 @@
+"""CPU numerics: our NHWC model vs the pure-torch NCHW twin.
+
+The CPU path of ops.functional IS the oracle for the GPU kernels; this
+test closes the loop by checking the composed model against stock torch.
+"""
+import torch
+import pytest
+
+from pytorch_distributed_train_amd.models import build_model
+from pytorch_distributed_train_amd.ops import functional as OF
+from tests.torch_twin import twin_resnet18
+
+torch.manual_seed(0)
+
+
+def test_state_dict_layout_matches_torchvision_style():
+    ours = build_model("resnet18", 10, torch.float32)
+    twin = twin_resnet18(10)
+    ours_sd = ours.state_dict()
+    twin_sd = twin.state_dict()
+    assert set(ours_sd.keys()) == set(twin_sd.keys())
+    for k in ours_sd:
+        assert ours_sd[k].shape == twin_sd[k].shape, k
+
+
+def test_forward_matches_twin():
+    ours = build_model("resnet18", 10, torch.float32)
+    twin = twin_resnet18(10)
+    twin.load_state_dict(ours.state_dict())
+    ours.eval()
+    twin.eval()
+    x = torch.randn(2, 3, 32, 32)
+    with torch.no_grad():
+        y_ours = ours(x)
+        y_twin = twin(x)
+    assert torch.allclose(y_ours, y_twin, atol=1e-4, rtol=1e-4), \
+        (y_ours - y_twin).abs().max()
+
+
+def test_train_step_grads_match_twin():
+    ours = build_model("resnet18", 10, torch.float32)
+    twin = twin_resnet18(10)
+    twin.load_state_dict(ours.state_dict())
+    ours.train()
+    twin.train()
+    x = torch.randn(4, 3, 32, 32)
+    labels = torch.randint(0, 10, (4,))
+
+    loss_o = OF.softmax_cross_entropy(ours(x), labels)
+    loss_o.backward()
+    loss_t = torch.nn.functional.cross_entropy(twin(x), labels)
+    loss_t.backward()
+
+    assert torch.allclose(loss_o, loss_t, atol=1e-5), (loss_o, loss_t)
+    ours_g = {n: p.grad for n, p in ours.named_parameters()}
+    for n, p in twin.named_parameters():
+        assert n in ours_g
+        # fp32 accumulation-order noise scales with grad magnitude; compare
+        # against the tensor's max |grad|
+        tol = 1e-3 * p.grad.abs().max().clamp(min=1.0)
+        err = (ours_g[n] - p.grad).abs().max()
+        assert err <= tol, (n, err, tol)
+    # running stats updated identically
+    ours_b = dict(ours.named_buffers())
+    for n, b in twin.named_buffers():
+        assert torch.allclose(ours_b[n].float(), b.float(),
+                              atol=1e-5, rtol=1e-4), n
+
+
+def test_loss_decreases_few_steps():
+    from pytorch_distributed_train_amd.parallel.flat import FlatParams
+    from pytorch_distributed_train_amd.parallel.optim import FusedSGD
+    model = build_model("resnet18", 10, torch.float32)
+    model.train()
+    flat = FlatParams(model.named_parameters())
+    opt = FusedSGD(flat, lr=0.05, momentum=0.9)
+    x = torch.randn(8, 3, 32, 32)
+    labels = torch.randint(0, 10, (8,))
+    losses = []
+    for _ in range(8):
+        opt.zero_grad()
+        loss = OF.softmax_cross_entropy(model(x), labels)
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0], losses
+
+
+def test_label_smoothing_matches_torch():
+    logits = torch.randn(6, 13, requires_grad=True)
+    labels = torch.randint(0, 13, (6,))
+    l_ours = OF.softmax_cross_entropy(logits, labels, 0.1)
+    l_torch = torch.nn.functional.cross_entropy(logits, labels,
+                                                label_smoothing=0.1)
+    assert torch.allclose(l_ours, l_torch, atol=1e-6)
+    g_ours = torch.autograd.grad(l_ours, logits, retain_graph=True)[0]
+    logits2 = logits.detach().requires_grad_()
+    l2 = torch.nn.functional.cross_entropy(logits2, labels,
+                                           label_smoothing=0.1)
+    g_torch = torch.autograd.grad(l2, logits2)[0]
+    assert torch.allclose(g_ours, g_torch, atol=1e-6)

@@ -1,0 +1,91 @@
+#!/usr/bin/env python3
+"""train.py — reference-style entrypoint.
+
+Same launch contract as the reference (torch.distributed.launch-style):
+
+  # our launcher:
+  python -m pytorch_distributed_train_amd.launch --nproc_per_node=8 \
+      train.py --model resnet50 --dtype bf16 --batch-size 256
+  # or torchrun:
+  python -m torch.distributed.run --nproc-per-node 8 \
+      --master-addr 127.0.0.1 train.py --model resnet50
+
+Reads RANK / LOCAL_RANK / WORLD_SIZE / MASTER_ADDR / MASTER_PORT (or legacy
+--local_rank argv), trains on synthetic data, prints rank-0 loss/images/sec,
+checkpoints on rank 0, resumes with --resume.
+"""
+from __future__ import annotations
+
+import os
+import sys
+
+import torch
+
+from pytorch_distributed_train_amd.flags import build_parser, DistEnv
+from pytorch_distributed_train_amd.engine import Trainer
+from pytorch_distributed_train_amd.data import (DistributedSampler, Loader,
+                                                make_dataset)
+from pytorch_distributed_train_amd.utils.checkpoint import (load_checkpoint,
+                                                            save_checkpoint)
+from pytorch_distributed_train_amd.utils.metrics import RankLogger, StepTimer
+
+
+def main(argv=None):
+    args = build_parser().parse_args(argv)
+    env = DistEnv.from_env(args.local_rank)
+    trainer = Trainer(args, env)
+
+    dataset = make_dataset(args.data, args.num_samples, seed=args.seed)
+    sampler = DistributedSampler(dataset, env.world_size, env.rank,
+                                 shuffle=True, seed=args.seed)
+    loader = Loader(dataset, sampler, args.batch_size,
+                    pin_memory=(trainer.device.type == "cuda"))
+
+    start_epoch = 0
+    if args.resume:
+        state = load_checkpoint(args.resume, trainer.model, trainer.optimizer,
+                                trainer.scaler, map_location=trainer.device)
+        start_epoch = state.get("epoch", 0)
+        trainer.global_step = state.get("step", 0)
+        if env.rank == 0:
+            print(f"resumed from {args.resume} at epoch {start_epoch}")
+    # DDP init broadcast keeps ranks identical after resume as well
+    trainer.reducer.broadcast_params(src=0)
+
+    logger = RankLogger(env.rank, os.path.join("gpurun_out", "metrics.jsonl")
+                        if os.path.isdir("gpurun_out") else None)
+    timer = StepTimer()
+
+    ckpt_path = None
+    if args.checkpoint_dir:
+        ckpt_path = os.path.join(args.checkpoint_dir,
+                                 f"{args.model}_checkpoint.pt")
+
+    for epoch in range(start_epoch, args.epochs):
+        trainer.epoch = epoch
+        sampler.set_epoch(epoch)
+        trainer.model.train()
+        timer.start()
+        for step, (images, labels) in enumerate(loader):
+            if args.steps_per_epoch and step >= args.steps_per_epoch:
+                break
+            images = images.to(trainer.device, non_blocking=True)
+            labels = labels.to(trainer.device, non_blocking=True)
+            loss = trainer.train_step(images, labels)
+            ips = timer.tick(args.batch_size * env.world_size)
+            if step % args.print_freq == 0:
+                logger.log(epoch=epoch, step=step, loss=float(loss),
+                           images_per_sec=round(ips, 1), lr=args.lr)
+        if ckpt_path and env.rank == 0 and (
+                epoch == args.epochs - 1 or
+                (args.checkpoint_freq and
+                 (epoch + 1) % args.checkpoint_freq == 0)):
+            save_checkpoint(ckpt_path, epoch + 1, trainer.global_step,
+                            trainer.model, trainer.optimizer, trainer.scaler)
+            logger.log(epoch=epoch, checkpoint=ckpt_path)
+    logger.close()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
