@@ -393,8 +393,12 @@ __global__ void stem_fwd_k(const T* __restrict__ x, const T* __restrict__ wt,
   extern __shared__ __attribute__((aligned(16))) char smem[];
   T* wl = reinterpret_cast<T*>(smem);  // [R*S*Cin][Kout]
   int rsc = R * S * Cin;
-  for (int i = threadIdx.x; i < rsc * Kout; i += blockDim.x)
-    wl[i] = wt[i];
+  // wt arrives [Kout][rsc]; transpose into the [rsc][Kout] LDS image the
+  // inner loop consumes (one-time, weight-sized)
+  for (int i = threadIdx.x; i < rsc * Kout; i += blockDim.x) {
+    int e = i / Kout, k = i % Kout;
+    wl[i] = wt[(long)k * rsc + e];
+  }
   __syncthreads();
 
   int kg = threadIdx.x % (Kout / 8);        // kout group (8 each)
