@@ -78,6 +78,15 @@ class BatchNorm2dNHWC(nn.Module):
 
     def forward(self, x):
         if self.training:
+            if x.is_cuda and self.comm is None:
+                # fast path: stats + running update in one kernel pair
+                mean, invstd = OF.bn_stats_train(
+                    x, self.running_mean, self.running_var, self.momentum,
+                    self.eps)
+                with torch.no_grad():
+                    self.num_batches_tracked += 1
+                return OF.bn_train(x, self.weight, self.bias, mean, invstd,
+                                   self.fuse_relu)
             mean, var, count = self._compute_stats(x)
             with torch.no_grad():
                 self.num_batches_tracked += 1

@@ -15,16 +15,19 @@ void pdta_relu_fwd(const void*, void*, long, int, hipStream_t);
 void pdta_relu_bwd_mask(const void*, const void*, void*, long, int,
                         hipStream_t);
 void pdta_add_relu(const void*, const void*, void*, long, int, hipStream_t);
-void pdta_bn_stats(const void*, float*, float*, float*, float*, long, int,
-                   int, hipStream_t);
+void pdta_bn_stats(const void*, float*, float*, float*, float*, float*,
+                   float, float, int, long, int, int, hipStream_t);
+int pdta_bn_chunks(long, int);
 void pdta_bn_apply(const void*, void*, const float*, const float*,
                    const float*, const float*, long, int, int, int,
                    hipStream_t);
-void pdta_bn_bwd_reduce(const void*, const void*, const float*, float*,
-                        float*, long, int, int, hipStream_t);
-void pdta_bn_bwd_elemt(const void*, const void*, void*, const float*,
+void pdta_bn_bwd_reduce(const void*, const void*, const void*,
+                        const float*, float*, float*, float*, long, int,
+                        int, hipStream_t);
+void pdta_bn_bwd_elemt(const void*, const void*, const void*, void*,
                        const float*, const float*, const float*,
-                       const float*, float, long, int, int, hipStream_t);
+                       const float*, const float*, float, long, int, int,
+                       hipStream_t);
 void pdta_maxpool_fwd(const void*, void*, uint8_t*, int, int, int, int, int,
                       int, int, int, int, int, hipStream_t);
 void pdta_maxpool_bwd(const void*, const uint8_t*, void*, int, int, int, int,
@@ -118,16 +121,36 @@ at::Tensor add_relu_fwd(at::Tensor a, at::Tensor b) {
 
 // ---- batchnorm -----------------------------------------------------------
 std::vector<at::Tensor> bn_stats(at::Tensor x) {
+  // (mean, var) — SyncBN partials / parity path (no running update)
   CHECK_IN(x);
   int C = x.size(3);
   long rows = x.numel() / C;
   auto opts = x.options().dtype(at::kFloat);
-  auto sum = at::zeros({C}, opts), sumsq = at::zeros({C}, opts);
+  int chunks = pdta_bn_chunks(rows, C);
+  auto part = at::empty({(long)chunks * 2 * C}, opts);
   auto mean = at::empty({C}, opts), var = at::empty({C}, opts);
-  pdta_bn_stats(x.data_ptr(), sum.data_ptr<float>(), sumsq.data_ptr<float>(),
-                mean.data_ptr<float>(), var.data_ptr<float>(), rows, C,
-                dt_of(x), cur_stream());
+  pdta_bn_stats(x.data_ptr(), part.data_ptr<float>(), mean.data_ptr<float>(),
+                var.data_ptr<float>(), nullptr, nullptr, 0.f, 0.f, 0, rows,
+                C, dt_of(x), cur_stream());
   return {mean, var};
+}
+
+std::vector<at::Tensor> bn_stats_train(at::Tensor x, at::Tensor running_mean,
+                                       at::Tensor running_var,
+                                       double momentum, double eps) {
+  // (mean, invstd); updates running stats in-kernel (training fast path)
+  CHECK_IN(x);
+  int C = x.size(3);
+  long rows = x.numel() / C;
+  auto opts = x.options().dtype(at::kFloat);
+  int chunks = pdta_bn_chunks(rows, C);
+  auto part = at::empty({(long)chunks * 2 * C}, opts);
+  auto mean = at::empty({C}, opts), invstd = at::empty({C}, opts);
+  pdta_bn_stats(x.data_ptr(), part.data_ptr<float>(), mean.data_ptr<float>(),
+                invstd.data_ptr<float>(), running_mean.data_ptr<float>(),
+                running_var.data_ptr<float>(), momentum, eps, 1, rows, C,
+                dt_of(x), cur_stream());
+  return {mean, invstd};
 }
 
 at::Tensor bn_apply(at::Tensor x, at::Tensor weight, at::Tensor bias,
@@ -147,21 +170,27 @@ at::Tensor bn_apply(at::Tensor x, at::Tensor weight, at::Tensor bias,
 }
 
 std::vector<at::Tensor> bn_bwd_reduce(at::Tensor x, at::Tensor dy,
-                                      at::Tensor mean, at::Tensor invstd) {
+                                      c10::optional<at::Tensor> relu_y,
+                                      at::Tensor mean) {
   CHECK_IN(x);
   CHECK_IN(dy);
   int C = x.size(3);
   long rows = x.numel() / C;
   auto opts = x.options().dtype(at::kFloat);
-  auto sum_dy = at::zeros({C}, opts), sum_dy_xmu = at::zeros({C}, opts);
+  int chunks = pdta_bn_chunks(rows, C);
+  auto part = at::empty({(long)chunks * 2 * C}, opts);
+  auto sum_dy = at::empty({C}, opts), sum_dy_xmu = at::empty({C}, opts);
   pdta_bn_bwd_reduce(x.data_ptr(), dy.data_ptr(),
+                     relu_y.has_value() ? relu_y->data_ptr() : nullptr,
                      mean.contiguous().data_ptr<float>(),
-                     sum_dy.data_ptr<float>(), sum_dy_xmu.data_ptr<float>(),
-                     rows, C, dt_of(x), cur_stream());
+                     part.data_ptr<float>(), sum_dy.data_ptr<float>(),
+                     sum_dy_xmu.data_ptr<float>(), rows, C, dt_of(x),
+                     cur_stream());
   return {sum_dy, sum_dy_xmu};
 }
 
-at::Tensor bn_bwd_elemt(at::Tensor x, at::Tensor dy, at::Tensor mean,
+at::Tensor bn_bwd_elemt(at::Tensor x, at::Tensor dy,
+                        c10::optional<at::Tensor> relu_y, at::Tensor mean,
                         at::Tensor invstd, at::Tensor weight,
                         at::Tensor sum_dy, at::Tensor sum_dy_xmu,
                         double count) {
@@ -171,8 +200,10 @@ at::Tensor bn_bwd_elemt(at::Tensor x, at::Tensor dy, at::Tensor mean,
   long rows = x.numel() / C;
   auto dx = at::empty_like(x);
   auto w = weight.to(at::kFloat).contiguous();
-  pdta_bn_bwd_elemt(x.data_ptr(), dy.data_ptr(), dx.data_ptr(),
-                    w.data_ptr<float>(), mean.contiguous().data_ptr<float>(),
+  pdta_bn_bwd_elemt(x.data_ptr(), dy.data_ptr(),
+                    relu_y.has_value() ? relu_y->data_ptr() : nullptr,
+                    dx.data_ptr(), w.data_ptr<float>(),
+                    mean.contiguous().data_ptr<float>(),
                     invstd.contiguous().data_ptr<float>(),
                     sum_dy.contiguous().data_ptr<float>(),
                     sum_dy_xmu.contiguous().data_ptr<float>(), (float)count,
@@ -433,6 +464,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_bwd_mask", &relu_bwd_mask);
   m.def("add_relu_fwd", &add_relu_fwd);
   m.def("bn_stats", &bn_stats);
+  m.def("bn_stats_train", &bn_stats_train);
   m.def("bn_apply", &bn_apply);
   m.def("bn_bwd_reduce", &bn_bwd_reduce);
   m.def("bn_bwd_elemt", &bn_bwd_elemt);

@@ -1,29 +1,34 @@
 // BatchNorm kernels over [N,H,W,C] (C innermost, C % 64 == 0).
 // Rebuilds torch's batch_norm_stats / batch_norm_elemt /
 // batch_norm_backward_reduce / batch_norm_backward_elemt set
-// (SURVEY.md §2d BatchNorm row) as NHWC-native CDNA4 kernels:
-//   - stats & bwd_reduce: each thread owns ONE channel column and strides
-//     rows (coalesced: consecutive lanes read consecutive channels),
-//     fp32 accumulation, per-block atomicAdd into fp32 [C] buffers;
-//   - apply & bwd_elemt: each thread owns EIGHT channels (16-B bf16
-//     vectors), per-channel affine constants held in registers across the
-//     row loop — pure HBM-rate streaming with optional fused ReLU.
-// The same kernels serve SyncBatchNorm: stats deliver (sum, sumsq) partials
-// that the python layer all-gathers/combines.
+// (SURVEY.md §2d BatchNorm row) as NHWC-native CDNA4 kernels.
+//
+// Reductions are TWO-STAGE with plain stores (no global atomics, no
+// pre-zeroed buffers — at B=256 the per-layer at::zeros launches were ~3%
+// of the step): stage 1 writes per-chunk partials [chunk][2C] (each thread
+// owns one channel column, strides rows — coalesced since consecutive
+// lanes read consecutive channels; LDS combine across the row-parallel
+// threads); stage 2 reduces chunks, computes mean/invstd, and (training
+// path) updates the running stats in the same kernel.
+//
+// The ReLU that precedes BN backward is fused: when `relu_y` is non-null,
+// dy is masked by (y > 0) on the fly in both backward kernels, removing
+// the separate relu_bwd_mask pass over dy.
 #include "common.h"
 
 namespace {
 
 constexpr int TPB = 256;
 
-// -- per-channel reduction: sum and sumsq --------------------------------
+// -- stage 1: per-chunk partial sums (sum, sumsq) -------------------------
 template <typename T>
-__global__ void bn_stats_k(const T* __restrict__ x, float* __restrict__ sum,
-                           float* __restrict__ sumsq, long rows, int C,
-                           int rows_per_block) {
-  int cpb = min(C, TPB);            // channels covered per block
-  int rpar = TPB / cpb;             // rows processed in parallel
-  int c = blockIdx.x * cpb + (int)(threadIdx.x % cpb);
+__global__ void bn_part_k(const T* __restrict__ x, float* __restrict__ part,
+                          long rows, int C, int rows_per_block) {
+  __shared__ float red[2][TPB];
+  int cpb = min(C, TPB);
+  int rpar = TPB / cpb;
+  int lane_c = (int)(threadIdx.x % cpb);
+  int c = blockIdx.x * cpb + lane_c;
   long r0 = (long)blockIdx.y * rows_per_block + threadIdx.x / cpb;
   long r_end = min(rows, (long)(blockIdx.y + 1) * rows_per_block);
   float s = 0.f, ss = 0.f;
@@ -32,27 +37,55 @@ __global__ void bn_stats_k(const T* __restrict__ x, float* __restrict__ sum,
     s += v;
     ss += v * v;
   }
-  atomicAdd(&sum[c], s);
-  atomicAdd(&sumsq[c], ss);
+  // combine the rpar threads sharing this channel
+  red[0][threadIdx.x] = s;
+  red[1][threadIdx.x] = ss;
+  __syncthreads();
+  if (threadIdx.x < cpb) {
+    for (int i = 1; i < rpar; ++i) {
+      s += red[0][threadIdx.x + i * cpb];
+      ss += red[1][threadIdx.x + i * cpb];
+    }
+    float* p = part + ((long)blockIdx.y * C + blockIdx.x * cpb) * 2;
+    p[lane_c] = s;
+    p[cpb + lane_c] = ss;
+  }
 }
 
-__global__ void bn_finalize_k(const float* __restrict__ sum,
-                              const float* __restrict__ sumsq,
-                              float* __restrict__ mean,
-                              float* __restrict__ var, float inv_n, int C) {
+// -- stage 2: reduce chunks -> mean/var (+invstd, +running update) --------
+__global__ void bn_stats_final_k(const float* __restrict__ part, int chunks,
+                                 int C, float inv_n, float unbias,
+                                 float* __restrict__ mean,
+                                 float* __restrict__ var_or_invstd,
+                                 float* __restrict__ running_mean,
+                                 float* __restrict__ running_var,
+                                 float momentum, float eps, int mode) {
+  // mode 0: write var (SyncBN partials path). mode 1: write invstd and
+  // update running stats (plain BN training fast path).
   int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c < C) {
-    float m = sum[c] * inv_n;
-    mean[c] = m;
-    var[c] = fmaxf(sumsq[c] * inv_n - m * m, 0.f);
+  if (c >= C) return;
+  int cpb = min(C, TPB);
+  int cblk = c / cpb, lane_c = c % cpb;
+  float s = 0.f, ss = 0.f;
+  for (int k = 0; k < chunks; ++k) {
+    const float* p = part + ((long)k * C + cblk * cpb) * 2;
+    s += p[lane_c];
+    ss += p[cpb + lane_c];
+  }
+  float m = s * inv_n;
+  float v = fmaxf(ss * inv_n - m * m, 0.f);
+  mean[c] = m;
+  if (mode == 0) {
+    var_or_invstd[c] = v;
+  } else {
+    var_or_invstd[c] = rsqrtf(v + eps);
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+    running_var[c] =
+        (1.f - momentum) * running_var[c] + momentum * v * unbias;
   }
 }
 
 // -- y = (x - mean) * invstd * w + b, optional ReLU ----------------------
-// thread mapping for the streaming kernels: cpg = min(C/8, TPB) threads
-// cover the channel axis (8 channels each), the rest of the block plus
-// gridDim.y stride the row axis; per-thread channel constants live in
-// registers across the whole row loop.
 template <typename T, bool RELU>
 __global__ void bn_apply_k(const T* __restrict__ x, T* __restrict__ y,
                            const float* __restrict__ weight,
@@ -87,36 +120,69 @@ __global__ void bn_apply_k(const T* __restrict__ x, T* __restrict__ y,
   }
 }
 
-// -- backward reduce: sum_dy, sum_dy_xmu ---------------------------------
-template <typename T>
-__global__ void bn_bwd_reduce_k(const T* __restrict__ x,
-                                const T* __restrict__ dy,
-                                const float* __restrict__ mean,
-                                float* __restrict__ sum_dy,
-                                float* __restrict__ sum_dy_xmu, long rows,
-                                int C, int rows_per_block) {
+// -- backward stage 1: per-chunk partial (sum_dy, sum_dy_xmu), with the
+//    preceding-ReLU mask fused when RELU (dy := y>0 ? dy : 0) ------------
+template <typename T, bool RELU>
+__global__ void bn_bwd_part_k(const T* __restrict__ x,
+                              const T* __restrict__ dy,
+                              const T* __restrict__ relu_y,
+                              const float* __restrict__ mean,
+                              float* __restrict__ part, long rows, int C,
+                              int rows_per_block) {
+  __shared__ float red[2][TPB];
   int cpb = min(C, TPB);
   int rpar = TPB / cpb;
-  int c = blockIdx.x * cpb + (int)(threadIdx.x % cpb);
+  int lane_c = (int)(threadIdx.x % cpb);
+  int c = blockIdx.x * cpb + lane_c;
   long r0 = (long)blockIdx.y * rows_per_block + threadIdx.x / cpb;
   long r_end = min(rows, (long)(blockIdx.y + 1) * rows_per_block);
   float m = mean[c];
   float s = 0.f, sx = 0.f;
   for (long r = r0; r < r_end; r += rpar) {
     float g = to_f32(dy[r * C + c]);
+    if (RELU) g = to_f32(relu_y[r * C + c]) > 0.f ? g : 0.f;
     float v = to_f32(x[r * C + c]);
     s += g;
     sx += g * (v - m);
   }
-  atomicAdd(&sum_dy[c], s);
-  atomicAdd(&sum_dy_xmu[c], sx);
+  red[0][threadIdx.x] = s;
+  red[1][threadIdx.x] = sx;
+  __syncthreads();
+  if (threadIdx.x < cpb) {
+    for (int i = 1; i < rpar; ++i) {
+      s += red[0][threadIdx.x + i * cpb];
+      sx += red[1][threadIdx.x + i * cpb];
+    }
+    float* p = part + ((long)blockIdx.y * C + blockIdx.x * cpb) * 2;
+    p[lane_c] = s;
+    p[cpb + lane_c] = sx;
+  }
 }
 
-// -- backward elementwise:
+__global__ void bn_bwd_final_k(const float* __restrict__ part, int chunks,
+                               int C, float* __restrict__ sum_dy,
+                               float* __restrict__ sum_dy_xmu) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  int cpb = min(C, TPB);
+  int cblk = c / cpb, lane_c = c % cpb;
+  float s = 0.f, sx = 0.f;
+  for (int k = 0; k < chunks; ++k) {
+    const float* p = part + ((long)k * C + cblk * cpb) * 2;
+    s += p[lane_c];
+    sx += p[cpb + lane_c];
+  }
+  sum_dy[c] = s;
+  sum_dy_xmu[c] = sx;
+}
+
+// -- backward elementwise (ReLU mask fused when RELU):
 //    dx = w*invstd * (dy - sum_dy/n - (x-mean)*invstd^2*sum_dy_xmu/n)
-template <typename T>
+template <typename T, bool RELU>
 __global__ void bn_bwd_elemt_k(const T* __restrict__ x,
-                               const T* __restrict__ dy, T* __restrict__ dx,
+                               const T* __restrict__ dy,
+                               const T* __restrict__ relu_y,
+                               T* __restrict__ dx,
                                const float* __restrict__ weight,
                                const float* __restrict__ mean,
                                const float* __restrict__ invstd,
@@ -143,20 +209,23 @@ __global__ void bn_bwd_elemt_k(const T* __restrict__ x,
        r += rstride) {
     Vec8<T> g = load8(dy + r * C + c0);
     Vec8<T> v = load8(x + r * C + c0);
+    Vec8<T> yv;
+    if (RELU) yv = load8(relu_y + r * C + c0);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      float o = k1[j] * (to_f32(g.v[j]) - k2[j] -
-                         (to_f32(v.v[j]) - mu[j]) * k3[j]);
+      float gg = to_f32(g.v[j]);
+      if (RELU) gg = to_f32(yv.v[j]) > 0.f ? gg : 0.f;
+      float o = k1[j] * (gg - k2[j] - (to_f32(v.v[j]) - mu[j]) * k3[j]);
       g.v[j] = from_f32<T>(o);
     }
     store8(dx + r * C + c0, g);
   }
 }
 
-static inline dim3 reduce_grid(long rows, int C, int* rows_per_block) {
+static inline dim3 reduce_grid(long rows, int C, int* rows_per_block,
+                               int* chunks) {
   int cpb = C < TPB ? C : TPB;
   int cblocks = (C + cpb - 1) / cpb;
-  // target ~2048 blocks total for full-chip residency
   long want = 2048 / cblocks;
   if (want < 1) want = 1;
   long rpb = (rows + want - 1) / want;
@@ -164,16 +233,17 @@ static inline dim3 reduce_grid(long rows, int C, int* rows_per_block) {
   if (rpb < rpar) rpb = rpar;
   *rows_per_block = (int)rpb;
   long rblocks = (rows + rpb - 1) / rpb;
+  *chunks = (int)rblocks;
   return dim3(cblocks, (unsigned)rblocks);
 }
 
 static inline dim3 ew_grid2(long rows, int C) {
   int c8 = C / 8;
   int cpg = c8 < TPB ? c8 : TPB;
-  int cgb = (c8 + cpg - 1) / cpg;  // blocks along channel axis
+  int cgb = (c8 + cpg - 1) / cpg;
   int rpar = TPB / cpg;
   long rblocks = (rows + rpar - 1) / rpar;
-  long want = 2048 / cgb;  // ~2048 blocks total
+  long want = 2048 / cgb;
   if (want < 1) want = 1;
   if (rblocks > want) rblocks = want;
   return dim3(cgb, (unsigned)rblocks);
@@ -183,16 +253,28 @@ static inline dim3 ew_grid2(long rows, int C) {
 
 extern "C" {
 
-// sum/sumsq/mean/var are fp32 [C]; sum & sumsq must be ZEROED by caller.
-void pdta_bn_stats(const void* x, float* sum, float* sumsq, float* mean,
-                   float* var, long rows, int C, int dt, hipStream_t s) {
-  int rpb;
-  dim3 g = reduce_grid(rows, C, &rpb);
+// part: fp32 workspace [chunks][2C] (uninitialized ok). mode 1: training
+// fast path (invstd out + running stats update); mode 0: (mean, var) out.
+void pdta_bn_stats(const void* x, float* part, float* mean, float* out2,
+                   float* running_mean, float* running_var, float momentum,
+                   float eps, int mode, long rows, int C, int dt,
+                   hipStream_t s) {
+  int rpb, chunks;
+  dim3 g = reduce_grid(rows, C, &rpb, &chunks);
   PDTA_DISPATCH_DTYPE(dt, T, {
-    bn_stats_k<T><<<g, TPB, 0, s>>>((const T*)x, sum, sumsq, rows, C, rpb);
+    bn_part_k<T><<<g, TPB, 0, s>>>((const T*)x, part, rows, C, rpb);
   });
-  bn_finalize_k<<<(C + 255) / 256, 256, 0, s>>>(sum, sumsq, mean, var,
-                                                1.f / (float)rows, C);
+  float unbias = rows > 1 ? (float)rows / (float)(rows - 1) : 1.f;
+  bn_stats_final_k<<<(C + 255) / 256, 256, 0, s>>>(
+      part, chunks, C, 1.f / (float)rows, unbias, mean, out2, running_mean,
+      running_var, momentum, eps, mode);
+}
+
+// chunks needed by the caller to size `part`:
+int pdta_bn_chunks(long rows, int C) {
+  int rpb, chunks;
+  reduce_grid(rows, C, &rpb, &chunks);
+  return chunks;
 }
 
 void pdta_bn_apply(const void* x, void* y, const float* weight,
@@ -209,28 +291,41 @@ void pdta_bn_apply(const void* x, void* y, const float* weight,
   });
 }
 
-// sum_dy / sum_dy_xmu fp32 [C], ZEROED by caller.
-void pdta_bn_bwd_reduce(const void* x, const void* dy, const float* mean,
-                        float* sum_dy, float* sum_dy_xmu, long rows, int C,
-                        int dt, hipStream_t s) {
-  int rpb;
-  dim3 g = reduce_grid(rows, C, &rpb);
+// relu_y: non-null => fuse dy := (relu_y > 0) ? dy : 0.
+void pdta_bn_bwd_reduce(const void* x, const void* dy, const void* relu_y,
+                        const float* mean, float* part, float* sum_dy,
+                        float* sum_dy_xmu, long rows, int C, int dt,
+                        hipStream_t s) {
+  int rpb, chunks;
+  dim3 g = reduce_grid(rows, C, &rpb, &chunks);
   PDTA_DISPATCH_DTYPE(dt, T, {
-    bn_bwd_reduce_k<T><<<g, TPB, 0, s>>>((const T*)x, (const T*)dy, mean,
-                                         sum_dy, sum_dy_xmu, rows, C, rpb);
+    if (relu_y)
+      bn_bwd_part_k<T, true><<<g, TPB, 0, s>>>(
+          (const T*)x, (const T*)dy, (const T*)relu_y, mean, part, rows, C,
+          rpb);
+    else
+      bn_bwd_part_k<T, false><<<g, TPB, 0, s>>>(
+          (const T*)x, (const T*)dy, nullptr, mean, part, rows, C, rpb);
   });
+  bn_bwd_final_k<<<(C + 255) / 256, 256, 0, s>>>(part, chunks, C, sum_dy,
+                                                 sum_dy_xmu);
 }
 
-void pdta_bn_bwd_elemt(const void* x, const void* dy, void* dx,
-                       const float* weight, const float* mean,
+void pdta_bn_bwd_elemt(const void* x, const void* dy, const void* relu_y,
+                       void* dx, const float* weight, const float* mean,
                        const float* invstd, const float* sum_dy,
                        const float* sum_dy_xmu, float count, long rows,
                        int C, int dt, hipStream_t s) {
   dim3 g = ew_grid2(rows, C);
   PDTA_DISPATCH_DTYPE(dt, T, {
-    bn_bwd_elemt_k<T><<<g, TPB, 0, s>>>(
-        (const T*)x, (const T*)dy, (T*)dx, weight, mean, invstd, sum_dy,
-        sum_dy_xmu, 1.f / count, rows, C / 8);
+    if (relu_y)
+      bn_bwd_elemt_k<T, true><<<g, TPB, 0, s>>>(
+          (const T*)x, (const T*)dy, (const T*)relu_y, (T*)dx, weight, mean,
+          invstd, sum_dy, sum_dy_xmu, 1.f / count, rows, C / 8);
+    else
+      bn_bwd_elemt_k<T, false><<<g, TPB, 0, s>>>(
+          (const T*)x, (const T*)dy, nullptr, (T*)dx, weight, mean, invstd,
+          sum_dy, sum_dy_xmu, 1.f / count, rows, C / 8);
   });
 }
 

@@ -142,7 +142,7 @@ __global__ __launch_bounds__(TPB) void igemm_fwd_k(
 }
 
 // ----------------------------------------------------------- bwd-data --
-template <typename T>
+template <typename T, int STRIDE>  // STRIDE: 1/2 compile-time, 0 = runtime
 __global__ __launch_bounds__(TPB) void igemm_bwd_data_k(
     const T* __restrict__ dy, const T* __restrict__ w2, T* __restrict__ dx,
     const T* __restrict__ zero_page, int Nn, int H, int W, int C, int P,
@@ -186,9 +186,22 @@ __global__ __launch_bounds__(TPB) void igemm_bwd_data_k(
     for (int i = 0; i < 4; ++i) {
       int pn = a_h[i] + pad - r;
       int qn = a_w[i] + pad - ss;
-      int p = pn / stride, q = qn / stride;
-      bool valid = pn >= 0 && qn >= 0 && pn % stride == 0 &&
-                   qn % stride == 0 && p < P && q < Q;
+      int p, q;
+      bool valid;
+      if constexpr (STRIDE == 1) {
+        p = pn;
+        q = qn;
+        valid = pn >= 0 && qn >= 0 && p < P && q < Q;
+      } else if constexpr (STRIDE == 2) {
+        p = pn >> 1;
+        q = qn >> 1;
+        valid = pn >= 0 && qn >= 0 && !((pn | qn) & 1) && p < P && q < Q;
+      } else {
+        p = pn / stride;
+        q = qn / stride;
+        valid = pn >= 0 && qn >= 0 && pn % stride == 0 &&
+                qn % stride == 0 && p < P && q < Q;
+      }
       const char* gptr;
       if (valid)
         gptr = reinterpret_cast<const char*>(dy) +
@@ -436,72 +449,73 @@ __global__ void stem_fwd_k(const T* __restrict__ x, const T* __restrict__ wt,
   store8(yp, out);
 }
 
-// Stem bwd-weight: block stages 4 rows of (x window [rsc], dy [Kout]) at a
-// time; thread (kout = t%Kout, rsc strided) accumulates, atomicAdd at end.
+// Stem bwd-weight v2: thread = (kout lane, group of 8 rsc taps); within a
+// wave all 64 lanes share the same rsc taps, so every x load is a
+// hardware broadcast and the dy row is one coalesced 128-B line. Rows are
+// decoded by carry counters (no per-row division); no syncs, no LDS; one
+// atomicAdd per accumulator at the end. (v1 staged 4 rows per
+// __syncthreads and was latency-serialized: 39 ms/call = 40% of the
+// ResNet-50 step — profiles/r50_b256_kernel_stats_r01.md.)
 template <typename T>
 __global__ void stem_wrw_k(const T* __restrict__ x, const T* __restrict__ dy,
                            float* __restrict__ dw, int Nn, int H, int W,
                            int Cin, int P, int Q, int Kout, int R, int S,
                            int stride, int pad, long rows_per_chunk) {
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  int rsc = R * S * Cin;
-  constexpr int RB = 4;
-  float* xw = reinterpret_cast<float*>(smem);           // [RB][rsc]
-  float* dyl = xw + RB * rsc;                           // [RB][Kout]
-
+  constexpr int RPT = 8;  // rsc taps per thread
   int ko = threadIdx.x % Kout;
-  int jbase = threadIdx.x / Kout;              // rsc stride group
-  int jstride = blockDim.x / Kout;
+  int grp = threadIdx.x / Kout;
+  int ngrp = blockDim.x / Kout;
+  int rsc = R * S * Cin;
+
+  int rr[RPT], sc[RPT], cc[RPT];
+  bool ev[RPT];
+  int base = (blockIdx.x * ngrp + grp) * RPT;
+#pragma unroll
+  for (int j = 0; j < RPT; ++j) {
+    int e = base + j;
+    ev[j] = e < rsc;
+    int ee = ev[j] ? e : 0;
+    cc[j] = ee % Cin;
+    int t = ee / Cin;
+    sc[j] = t % S;
+    rr[j] = t / S;
+  }
+
   long NPQ = (long)Nn * P * Q;
-  long r0 = (long)blockIdx.x * rows_per_chunk;
+  long r0 = (long)blockIdx.y * rows_per_chunk;
   long r1 = min(NPQ, r0 + rows_per_chunk);
+  if (r0 >= r1) return;
+  // carry-counter row decode
+  int q = (int)(r0 % Q);
+  int p = (int)((r0 / Q) % P);
+  int n = (int)(r0 / ((long)P * Q));
 
-  // accumulators: ceil(rsc / jstride) per thread
-  float acc[64];
-  int nacc = (rsc + jstride - 1) / jstride;
-  for (int i = 0; i < nacc; ++i) acc[i] = 0.f;
+  float acc[RPT];
+#pragma unroll
+  for (int j = 0; j < RPT; ++j) acc[j] = 0.f;
 
-  for (long rb = r0; rb < r1; rb += RB) {
-    int nrows = (int)min((long)RB, r1 - rb);
-    // cooperative stage
-    for (int i = threadIdx.x; i < nrows * (rsc + Kout); i += blockDim.x) {
-      int rowi = i / (rsc + Kout), e = i % (rsc + Kout);
-      long row = rb + rowi;
-      int q = (int)(row % Q);
-      int p = (int)((row / Q) % P);
-      int n = (int)(row / ((long)P * Q));
-      if (e < rsc) {
-        int c = e % Cin, t = e / Cin;
-        int sc = t % S, rr = t / S;
-        int h = p * stride - pad + rr, w = q * stride - pad + sc;
-        float v = 0.f;
-        if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
-          v = to_f32(x[(((long)n * H + h) * W + w) * Cin + c]);
-        xw[rowi * rsc + e] = v;
-      } else {
-        int k = e - rsc;
-        dyl[rowi * Kout + k] = to_f32(dy[row * Kout + k]);
+  for (long r = r0; r < r1; ++r) {
+    float g = to_f32(dy[r * Kout + ko]);
+    int h0 = p * stride - pad, w0 = q * stride - pad;
+#pragma unroll
+    for (int j = 0; j < RPT; ++j) {
+      int h = h0 + rr[j], w = w0 + sc[j];
+      if (ev[j] && (unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
+        acc[j] += g * to_f32(x[(((long)n * H + h) * W + w) * Cin + cc[j]]);
+    }
+    if (++q == Q) {
+      q = 0;
+      if (++p == P) {
+        p = 0;
+        ++n;
       }
     }
-    __syncthreads();
-    for (int rowi = 0; rowi < nrows; ++rowi) {
-      float g = dyl[rowi * Kout + ko];
-      for (int i = 0; i < nacc; ++i) {
-        int e = jbase + i * jstride;
-        if (e < rsc) acc[i] += g * xw[rowi * rsc + e];
-      }
-    }
-    __syncthreads();
   }
-  for (int i = 0; i < nacc; ++i) {
-    int e = jbase + i * jstride;
-    if (e < rsc) {
-      int c = e % Cin, t = e / Cin;
-      int sc = t % S, rr = t / S;
-      // dW torch layout [Kout, Cin, R, S]
-      atomicAdd(&dw[(((long)ko * Cin + c) * R + rr) * S + sc], acc[i]);
-    }
-  }
+#pragma unroll
+  for (int j = 0; j < RPT; ++j)
+    if (ev[j])
+      atomicAdd(&dw[(((long)ko * Cin + cc[j]) * R + rr[j]) * S + sc[j]],
+                acc[j]);
 }
 
 }  // namespace
@@ -530,9 +544,18 @@ void pdta_conv_bwd_data(const void* dy, const void* w2, void* dx,
   int tiles_m = (int)((M + 127) / 128), tiles_n = (C + 127) / 128;
   dim3 grid(tiles_m * tiles_n);
   PDTA_DISPATCH_DTYPE16(dt, T, {
-    igemm_bwd_data_k<T><<<grid, TPB, 0, s>>>(
-        (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H, W, C,
-        P, Q, Kout, R, S, stride, pad, tiles_n);
+    if (stride == 1)
+      igemm_bwd_data_k<T, 1><<<grid, TPB, 0, s>>>(
+          (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H, W,
+          C, P, Q, Kout, R, S, stride, pad, tiles_n);
+    else if (stride == 2)
+      igemm_bwd_data_k<T, 2><<<grid, TPB, 0, s>>>(
+          (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H, W,
+          C, P, Q, Kout, R, S, stride, pad, tiles_n);
+    else
+      igemm_bwd_data_k<T, 0><<<grid, TPB, 0, s>>>(
+          (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H, W,
+          C, P, Q, Kout, R, S, stride, pad, tiles_n);
   });
 }
 
@@ -544,9 +567,9 @@ void pdta_conv_bwd_weight(const void* x, const void* dy, float* dw,
   long NPQ = (long)Nn * P * Q;
   int ko_tiles = (Kout + 63) / 64, c_tiles = (C + 63) / 64;
   int tiles = ko_tiles * c_tiles * R * S;
-  long splitk = 2048 / tiles;
+  long splitk = 4096 / tiles;
   if (splitk < 1) splitk = 1;
-  if (splitk > 64) splitk = 64;
+  if (splitk > 256) splitk = 256;
   long rows_per_chunk = ((NPQ + splitk - 1) / splitk + 63) / 64 * 64;
   long chunks = (NPQ + rows_per_chunk - 1) / rows_per_chunk;
   dim3 grid(tiles, (unsigned)chunks);
@@ -571,22 +594,25 @@ void pdta_stem_fwd(const void* x, const void* wt, void* y, int Nn, int H,
   });
 }
 
-// dw fp32 [Kout,Cin,R,S] zeroed by caller.
+// dw fp32 [Kout,Cin,R,S] zeroed by caller. Needs Kout <= 256 and
+// 256 % Kout == 0 (stem: Kout = 64).
 void pdta_stem_bwd_weight(const void* x, const void* dy, float* dw, int Nn,
                           int H, int W, int Cin, int P, int Q, int Kout,
                           int R, int S, int stride, int pad, int dt,
                           hipStream_t s) {
   long NPQ = (long)Nn * P * Q;
   int rsc = R * S * Cin;
-  long chunks = 512;
+  int ngrp = 256 / Kout;
+  int bx = (rsc + ngrp * 8 - 1) / (ngrp * 8);
+  long chunks = 2048 / bx;
+  if (chunks < 1) chunks = 1;
   long rows_per_chunk = (NPQ + chunks - 1) / chunks;
-  if (rows_per_chunk < 4) rows_per_chunk = 4;
   chunks = (NPQ + rows_per_chunk - 1) / rows_per_chunk;
-  int lds_bytes = 4 * (rsc + Kout) * (int)sizeof(float);
+  dim3 grid(bx, (unsigned)chunks);
   PDTA_DISPATCH_DTYPE16(dt, T, {
-    stem_wrw_k<T><<<(unsigned)chunks, 256, lds_bytes, s>>>(
-        (const T*)x, (const T*)dy, dw, Nn, H, W, Cin, P, Q, Kout, R, S,
-        stride, pad, rows_per_chunk);
+    stem_wrw_k<T><<<grid, 256, 0, s>>>((const T*)x, (const T*)dy, dw, Nn, H,
+                                       W, Cin, P, Q, Kout, R, S, stride, pad,
+                                       rows_per_chunk);
   });
 }
 

@@ -147,9 +147,8 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
         if ctx.count_override is not None:
             n = ctx.count_override
         if x.is_cuda:
-            if fuse_relu:
-                dy = ext().relu_bwd_mask(dy, y)
-            sum_dy, sum_dy_xmu = ext().bn_bwd_reduce(x, dy, mean, invstd)
+            relu_y = y if fuse_relu else None  # mask fused in the kernels
+            sum_dy, sum_dy_xmu = ext().bn_bwd_reduce(x, dy, relu_y, mean)
             # param grads from LOCAL sums (DDP averages them afterwards,
             # matching torch SyncBN's split of responsibilities)
             dweight = sum_dy_xmu * invstd
@@ -159,7 +158,7 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
                 ctx.comm.allreduce_sum(packed).sync_stream()
                 c = sum_dy.numel()
                 sum_dy, sum_dy_xmu = packed[:c], packed[c:]
-            dx = ext().bn_bwd_elemt(x, dy, mean, invstd, weight,
+            dx = ext().bn_bwd_elemt(x, dy, relu_y, mean, invstd, weight,
                                     sum_dy.contiguous(),
                                     sum_dy_xmu.contiguous(), float(n))
         else:
@@ -188,6 +187,12 @@ def bn_stats(x):
     if x.is_cuda:
         return ext().bn_stats(x)
     return _bn_cpu_stats(x)
+
+
+def bn_stats_train(x, running_mean, running_var, momentum, eps):
+    """(mean, invstd) + in-kernel running-stat update — GPU training fast
+    path (one kernel pair instead of stats + 4 python-side buffer ops)."""
+    return ext().bn_stats_train(x, running_mean, running_var, momentum, eps)
 
 
 def bn_train(x, weight, bias, mean, invstd, fuse_relu=False, comm=None,

@@ -151,17 +151,48 @@ def test_bn_stats_apply_bwd():
     _close(y, ref, what="bn_apply+relu")
 
     dy = _rand(n, h, h, c, seed=17)
-    sum_dy, sum_dy_xmu = ext().bn_bwd_reduce(x, dy, mean, invstd)
+    sum_dy, sum_dy_xmu = ext().bn_bwd_reduce(x, dy, None, mean)
     dy32 = dy.float()
     _close(sum_dy, dy32.sum(dim=(0, 1, 2)), rtol=1e-2, atol=1e-2,
            what="sum_dy")
     _close(sum_dy_xmu, (dy32 * (x32 - mean)).sum(dim=(0, 1, 2)), rtol=1e-2,
            atol=5e-2, what="sum_dy_xmu")
     cnt = float(n * h * h)
-    dx = ext().bn_bwd_elemt(x, dy, mean, invstd, w, sum_dy, sum_dy_xmu, cnt)
+    dx = ext().bn_bwd_elemt(x, dy, None, mean, invstd, w, sum_dy,
+                            sum_dy_xmu, cnt)
     ref_dx = (dy32 - sum_dy / cnt -
               (x32 - mean) * invstd * invstd * sum_dy_xmu / cnt) * w * invstd
     _close(dx, ref_dx, what="bn_bwd_elemt")
+
+    # fused ReLU mask path: mask dy by (y > 0) inside both kernels
+    s_m, sx_m = ext().bn_bwd_reduce(x, dy, y, mean)
+    dym = dy32 * (y.float() > 0)
+    _close(s_m, dym.sum(dim=(0, 1, 2)), rtol=1e-2, atol=1e-2,
+           what="sum_dy masked")
+    _close(sx_m, (dym * (x32 - mean)).sum(dim=(0, 1, 2)), rtol=1e-2,
+           atol=5e-2, what="sum_dy_xmu masked")
+    dx_m = ext().bn_bwd_elemt(x, dy, y, mean, invstd, w, s_m, sx_m, cnt)
+    ref_dx_m = (dym - s_m / cnt -
+                (x32 - mean) * invstd * invstd * sx_m / cnt) * w * invstd
+    _close(dx_m, ref_dx_m, what="bn_bwd_elemt masked")
+
+
+def test_bn_stats_train_updates_running():
+    n, h, c = 2, 8, 64
+    x = _rand(n, h, h, c, seed=30)
+    rm = torch.zeros(c, device=DEV)
+    rv = torch.ones(c, device=DEV)
+    mean, invstd = ext().bn_stats_train(x, rm, rv, 0.1, 1e-5)
+    x32 = x.float()
+    ref_mean = x32.mean(dim=(0, 1, 2))
+    ref_var = x32.var(dim=(0, 1, 2), unbiased=False)
+    _close(mean, ref_mean, rtol=1e-3, atol=1e-3, what="train mean")
+    _close(invstd, (ref_var + 1e-5).rsqrt(), rtol=1e-3, atol=1e-3,
+           what="train invstd")
+    nn = n * h * h
+    _close(rm, 0.1 * ref_mean, rtol=1e-3, atol=1e-3, what="running_mean")
+    _close(rv, 0.9 + 0.1 * ref_var * nn / (nn - 1), rtol=1e-3, atol=1e-3,
+           what="running_var")
 
 
 # ---------------------------------------------------------- elementwise --
