@@ -53,11 +53,8 @@ void pdta_conv_bwd_data(const void*, const void*, void*, const void*, int,
 void pdta_conv_bwd_weight(const void*, const void*, float*, const void*, int,
                           int, int, int, int, int, int, int, int, int, int,
                           int, hipStream_t);
-void pdta_stem_fwd(const void*, const void*, void*, int, int, int, int, int,
-                   int, int, int, int, int, int, int, hipStream_t);
-void pdta_stem_bwd_weight(const void*, const void*, float*, int, int, int,
-                          int, int, int, int, int, int, int, int, int,
-                          hipStream_t);
+void pdta_stem_im2col(const void*, void*, int, int, int, int, int, int,
+                      int, int, int, int, int, int, hipStream_t);
 }
 
 namespace {
@@ -342,9 +339,16 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor wt, int64_t stride,
                   H, W, C, P, Q, Kout, R, S, stride, pad, dt_of(x),
                   cur_stream());
   } else {
-    TORCH_CHECK(Kout % 8 == 0, "stem conv needs Kout % 8 == 0");
-    pdta_stem_fwd(x.data_ptr(), wt.data_ptr(), y.data_ptr(), N, H, W, C, P,
-                  Q, Kout, R, S, stride, pad, dt_of(x), cur_stream());
+    // stem (small C): im2col into the SAME igemm path (1x1 over xcol)
+    int rsc = R * S * C, rsc_pad = (rsc + 63) / 64 * 64;
+    auto xcol = at::empty({(long)N * P * Q, rsc_pad}, x.options());
+    pdta_stem_im2col(x.data_ptr(), xcol.data_ptr(), N, H, W, C, P, Q, R, S,
+                     stride, pad, rsc_pad, dt_of(x), cur_stream());
+    auto wt_pad = at::constant_pad_nd(wt.reshape({Kout, rsc}),
+                                      {0, rsc_pad - rsc}).contiguous();
+    pdta_conv_fwd(xcol.data_ptr(), wt_pad.data_ptr(), y.data_ptr(),
+                  zero_page(x), N, P, Q, rsc_pad, P, Q, Kout, 1, 1, 1, 0,
+                  dt_of(x), cur_stream());
   }
   return y;
 }
@@ -372,17 +376,27 @@ at::Tensor conv2d_bwd_weight(at::Tensor x, at::Tensor dy, int64_t R,
   CHECK_IN(dy);
   int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
   int P = dy.size(1), Q = dy.size(2), Kout = dy.size(3);
-  auto dw = at::zeros({Kout, C, R, S}, x.options().dtype(at::kFloat));
   if (C % 64 == 0) {
+    auto dw = at::zeros({Kout, C, R, S}, x.options().dtype(at::kFloat));
     pdta_conv_bwd_weight(x.data_ptr(), dy.data_ptr(), dw.data_ptr<float>(),
                          zero_page(x), N, H, W, C, P, Q, Kout, R, S, stride,
                          pad, dt_of(x), cur_stream());
-  } else {
-    pdta_stem_bwd_weight(x.data_ptr(), dy.data_ptr(), dw.data_ptr<float>(),
-                         N, H, W, C, P, Q, Kout, R, S, stride, pad, dt_of(x),
-                         cur_stream());
+    return dw;
   }
-  return dw;
+  // stem: im2col + 1x1 igemm wrw, then repack [K][r][s][c] -> [K,C,R,S]
+  int rsc = R * S * C, rsc_pad = (rsc + 63) / 64 * 64;
+  auto xcol = at::empty({(long)N * P * Q, rsc_pad}, x.options());
+  pdta_stem_im2col(x.data_ptr(), xcol.data_ptr(), N, H, W, C, P, Q, R, S,
+                   stride, pad, rsc_pad, dt_of(x), cur_stream());
+  auto dwcol = at::zeros({Kout, rsc_pad}, x.options().dtype(at::kFloat));
+  pdta_conv_bwd_weight(xcol.data_ptr(), dy.data_ptr(),
+                       dwcol.data_ptr<float>(), zero_page(x), N, P, Q,
+                       rsc_pad, P, Q, Kout, 1, 1, 1, 0, dt_of(x),
+                       cur_stream());
+  return dwcol.slice(1, 0, rsc)
+      .view({Kout, R, S, C})
+      .permute({0, 3, 1, 2})
+      .contiguous();
 }
 
 // ---- RCCL ----------------------------------------------------------------

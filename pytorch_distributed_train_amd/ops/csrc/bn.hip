@@ -53,6 +53,9 @@ __global__ void bn_part_k(const T* __restrict__ x, float* __restrict__ part,
 }
 
 // -- stage 2: reduce chunks -> mean/var (+invstd, +running update) --------
+// finalize: block = 64 channels x 4 chunk-slices (full waves even at
+// C=64; the slices LDS-reduce) — a C-threads-only version was one wave on
+// the whole chip and chunk-serial (350 us; this form is ~5 us).
 __global__ void bn_stats_final_k(const float* __restrict__ part, int chunks,
                                  int C, float inv_n, float unbias,
                                  float* __restrict__ mean,
@@ -62,26 +65,39 @@ __global__ void bn_stats_final_k(const float* __restrict__ part, int chunks,
                                  float momentum, float eps, int mode) {
   // mode 0: write var (SyncBN partials path). mode 1: write invstd and
   // update running stats (plain BN training fast path).
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ float red[2][TPB];
+  int c = blockIdx.x * 64 + (int)(threadIdx.x % 64);
+  int slice = threadIdx.x / 64;  // 4 slices
   int cpb = min(C, TPB);
   int cblk = c / cpb, lane_c = c % cpb;
   float s = 0.f, ss = 0.f;
-  for (int k = 0; k < chunks; ++k) {
-    const float* p = part + ((long)k * C + cblk * cpb) * 2;
-    s += p[lane_c];
-    ss += p[cpb + lane_c];
+  if (c < C) {
+    for (int k = slice; k < chunks; k += 4) {
+      const float* p = part + ((long)k * C + cblk * cpb) * 2;
+      s += p[lane_c];
+      ss += p[cpb + lane_c];
+    }
   }
-  float m = s * inv_n;
-  float v = fmaxf(ss * inv_n - m * m, 0.f);
-  mean[c] = m;
-  if (mode == 0) {
-    var_or_invstd[c] = v;
-  } else {
-    var_or_invstd[c] = rsqrtf(v + eps);
-    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
-    running_var[c] =
-        (1.f - momentum) * running_var[c] + momentum * v * unbias;
+  red[0][threadIdx.x] = s;
+  red[1][threadIdx.x] = ss;
+  __syncthreads();
+  if (slice == 0 && c < C) {
+#pragma unroll
+    for (int i = 1; i < 4; ++i) {
+      s += red[0][threadIdx.x + i * 64];
+      ss += red[1][threadIdx.x + i * 64];
+    }
+    float m = s * inv_n;
+    float v = fmaxf(ss * inv_n - m * m, 0.f);
+    mean[c] = m;
+    if (mode == 0) {
+      var_or_invstd[c] = v;
+    } else {
+      var_or_invstd[c] = rsqrtf(v + eps);
+      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+      running_var[c] =
+          (1.f - momentum) * running_var[c] + momentum * v * unbias;
+    }
   }
 }
 
@@ -162,18 +178,31 @@ __global__ void bn_bwd_part_k(const T* __restrict__ x,
 __global__ void bn_bwd_final_k(const float* __restrict__ part, int chunks,
                                int C, float* __restrict__ sum_dy,
                                float* __restrict__ sum_dy_xmu) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ float red[2][TPB];
+  int c = blockIdx.x * 64 + (int)(threadIdx.x % 64);
+  int slice = threadIdx.x / 64;
   int cpb = min(C, TPB);
   int cblk = c / cpb, lane_c = c % cpb;
   float s = 0.f, sx = 0.f;
-  for (int k = 0; k < chunks; ++k) {
-    const float* p = part + ((long)k * C + cblk * cpb) * 2;
-    s += p[lane_c];
-    sx += p[cpb + lane_c];
+  if (c < C) {
+    for (int k = slice; k < chunks; k += 4) {
+      const float* p = part + ((long)k * C + cblk * cpb) * 2;
+      s += p[lane_c];
+      sx += p[cpb + lane_c];
+    }
   }
-  sum_dy[c] = s;
-  sum_dy_xmu[c] = sx;
+  red[0][threadIdx.x] = s;
+  red[1][threadIdx.x] = sx;
+  __syncthreads();
+  if (slice == 0 && c < C) {
+#pragma unroll
+    for (int i = 1; i < 4; ++i) {
+      s += red[0][threadIdx.x + i * 64];
+      sx += red[1][threadIdx.x + i * 64];
+    }
+    sum_dy[c] = s;
+    sum_dy_xmu[c] = sx;
+  }
 }
 
 // -- backward elementwise (ReLU mask fused when RELU):
@@ -265,7 +294,7 @@ void pdta_bn_stats(const void* x, float* part, float* mean, float* out2,
     bn_part_k<T><<<g, TPB, 0, s>>>((const T*)x, part, rows, C, rpb);
   });
   float unbias = rows > 1 ? (float)rows / (float)(rows - 1) : 1.f;
-  bn_stats_final_k<<<(C + 255) / 256, 256, 0, s>>>(
+  bn_stats_final_k<<<(C + 63) / 64, 256, 0, s>>>(
       part, chunks, C, 1.f / (float)rows, unbias, mean, out2, running_mean,
       running_var, momentum, eps, mode);
 }
@@ -307,7 +336,7 @@ void pdta_bn_bwd_reduce(const void* x, const void* dy, const void* relu_y,
       bn_bwd_part_k<T, false><<<g, TPB, 0, s>>>(
           (const T*)x, (const T*)dy, nullptr, mean, part, rows, C, rpb);
   });
-  bn_bwd_final_k<<<(C + 255) / 256, 256, 0, s>>>(part, chunks, C, sum_dy,
+  bn_bwd_final_k<<<(C + 63) / 64, 256, 0, s>>>(part, chunks, C, sum_dy,
                                                  sum_dy_xmu);
 }
 

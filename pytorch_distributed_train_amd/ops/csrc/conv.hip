@@ -397,125 +397,46 @@ __global__ __launch_bounds__(TPB) void igemm_wrw_k(
 }
 
 // ------------------------------------------------------------- stem ----
-// Direct 7x7/s2 C=3 forward: weights LDS-cached, 8 kout per thread.
+// The C=3 stem (7x7/s2) goes through explicit im2col into the SAME MFMA
+// igemm path as every other conv: xcol[npq][rsc_pad] with rsc = R*S*Cin
+// zero-padded to a multiple of 64 (zero columns match zero-padded weight
+// packs, so results are exact). The direct stem kernels this replaces
+// were VALU-issue- and latency-bound (stem wrw alone was 29-40% of the
+// ResNet-50 step — profiles/r50_b256_kernel_stats_r01.md).
 template <typename T>
-__global__ void stem_fwd_k(const T* __restrict__ x, const T* __restrict__ wt,
-                           T* __restrict__ y, int Nn, int H, int W, int Cin,
-                           int P, int Q, int Kout, int R, int S, int stride,
-                           int pad) {
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  T* wl = reinterpret_cast<T*>(smem);  // [R*S*Cin][Kout]
+__global__ void stem_im2col_k(const T* __restrict__ x, T* __restrict__ out,
+                              int Nn, int H, int W, int Cin, int P, int Q,
+                              int R, int S, int stride, int pad,
+                              int rsc_pad) {
+  int g8 = rsc_pad / 8;
+  long total = (long)Nn * P * Q * g8;
+  long item = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long gstride = (long)gridDim.x * blockDim.x;
   int rsc = R * S * Cin;
-  // wt arrives [Kout][rsc]; transpose into the [rsc][Kout] LDS image the
-  // inner loop consumes (one-time, weight-sized)
-  for (int i = threadIdx.x; i < rsc * Kout; i += blockDim.x) {
-    int e = i / Kout, k = i % Kout;
-    wl[i] = wt[(long)k * rsc + e];
-  }
-  __syncthreads();
-
-  int kg = threadIdx.x % (Kout / 8);        // kout group (8 each)
-  int pos_per_block = blockDim.x / (Kout / 8);
-  long m = (long)blockIdx.x * pos_per_block + threadIdx.x / (Kout / 8);
-  long M = (long)Nn * P * Q;
-  if (m >= M) return;
-  int q = (int)(m % Q);
-  int p = (int)((m / Q) % P);
-  int n = (int)(m / ((long)P * Q));
-  int h0 = p * stride - pad, w0 = q * stride - pad;
-
-  float acc[8];
-#pragma unroll
-  for (int j = 0; j < 8; ++j) acc[j] = 0.f;
-  for (int rr = 0; rr < R; ++rr) {
-    int h = h0 + rr;
-    if ((unsigned)h >= (unsigned)H) continue;
-    for (int sc = 0; sc < S; ++sc) {
-      int w = w0 + sc;
-      if ((unsigned)w >= (unsigned)W) continue;
-      const T* xp = x + (((long)n * H + h) * W + w) * Cin;
-      for (int c = 0; c < Cin; ++c) {
-        float xv = to_f32(xp[c]);
-        const T* wp = wl + ((rr * S + sc) * Cin + c) * Kout + kg * 8;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) acc[j] += xv * to_f32(wp[j]);
-      }
-    }
-  }
-  T* yp = y + m * Kout + kg * 8;
-  Vec8<T> out;
-#pragma unroll
-  for (int j = 0; j < 8; ++j) out.v[j] = from_f32<T>(acc[j]);
-  store8(yp, out);
-}
-
-// Stem bwd-weight v2: thread = (kout lane, group of 8 rsc taps); within a
-// wave all 64 lanes share the same rsc taps, so every x load is a
-// hardware broadcast and the dy row is one coalesced 128-B line. Rows are
-// decoded by carry counters (no per-row division); no syncs, no LDS; one
-// atomicAdd per accumulator at the end. (v1 staged 4 rows per
-// __syncthreads and was latency-serialized: 39 ms/call = 40% of the
-// ResNet-50 step — profiles/r50_b256_kernel_stats_r01.md.)
-template <typename T>
-__global__ void stem_wrw_k(const T* __restrict__ x, const T* __restrict__ dy,
-                           float* __restrict__ dw, int Nn, int H, int W,
-                           int Cin, int P, int Q, int Kout, int R, int S,
-                           int stride, int pad, long rows_per_chunk) {
-  constexpr int RPT = 8;  // rsc taps per thread
-  int ko = threadIdx.x % Kout;
-  int grp = threadIdx.x / Kout;
-  int ngrp = blockDim.x / Kout;
-  int rsc = R * S * Cin;
-
-  int rr[RPT], sc[RPT], cc[RPT];
-  bool ev[RPT];
-  int base = (blockIdx.x * ngrp + grp) * RPT;
-#pragma unroll
-  for (int j = 0; j < RPT; ++j) {
-    int e = base + j;
-    ev[j] = e < rsc;
-    int ee = ev[j] ? e : 0;
-    cc[j] = ee % Cin;
-    int t = ee / Cin;
-    sc[j] = t % S;
-    rr[j] = t / S;
-  }
-
-  long NPQ = (long)Nn * P * Q;
-  long r0 = (long)blockIdx.y * rows_per_chunk;
-  long r1 = min(NPQ, r0 + rows_per_chunk);
-  if (r0 >= r1) return;
-  // carry-counter row decode
-  int q = (int)(r0 % Q);
-  int p = (int)((r0 / Q) % P);
-  int n = (int)(r0 / ((long)P * Q));
-
-  float acc[RPT];
-#pragma unroll
-  for (int j = 0; j < RPT; ++j) acc[j] = 0.f;
-
-  for (long r = r0; r < r1; ++r) {
-    float g = to_f32(dy[r * Kout + ko]);
+  for (; item < total; item += gstride) {
+    int grp = (int)(item % g8);
+    long row = item / g8;
+    int q = (int)(row % Q);
+    int p = (int)((row / Q) % P);
+    int n = (int)(row / ((long)P * Q));
     int h0 = p * stride - pad, w0 = q * stride - pad;
+    Vec8<T> v;
 #pragma unroll
-    for (int j = 0; j < RPT; ++j) {
-      int h = h0 + rr[j], w = w0 + sc[j];
-      if (ev[j] && (unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
-        acc[j] += g * to_f32(x[(((long)n * H + h) * W + w) * Cin + cc[j]]);
-    }
-    if (++q == Q) {
-      q = 0;
-      if (++p == P) {
-        p = 0;
-        ++n;
+    for (int j = 0; j < 8; ++j) {
+      int e = grp * 8 + j;
+      float val = 0.f;
+      if (e < rsc) {
+        int c = e % Cin;
+        int t = e / Cin;
+        int sc = t % S, rr = t / S;
+        int h = h0 + rr, w = w0 + sc;
+        if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
+          val = to_f32(x[(((long)n * H + h) * W + w) * Cin + c]);
       }
+      v.v[j] = from_f32<T>(val);
     }
+    store8(out + row * rsc_pad + grp * 8, v);
   }
-#pragma unroll
-  for (int j = 0; j < RPT; ++j)
-    if (ev[j])
-      atomicAdd(&dw[(((long)ko * Cin + cc[j]) * R + rr[j]) * S + sc[j]],
-                acc[j]);
 }
 
 }  // namespace
@@ -580,39 +501,14 @@ void pdta_conv_bwd_weight(const void* x, const void* dy, float* dw,
   });
 }
 
-void pdta_stem_fwd(const void* x, const void* wt, void* y, int Nn, int H,
-                   int W, int Cin, int P, int Q, int Kout, int R, int S,
-                   int stride, int pad, int dt, hipStream_t s) {
-  long M = (long)Nn * P * Q;
-  int pos_per_block = 256 / (Kout / 8);
-  long grid = (M + pos_per_block - 1) / pos_per_block;
-  int lds_bytes = R * S * Cin * Kout * 2;
+void pdta_stem_im2col(const void* x, void* out, int Nn, int H, int W,
+                      int Cin, int P, int Q, int R, int S, int stride,
+                      int pad, int rsc_pad, int dt, hipStream_t s) {
+  long total = (long)Nn * P * Q * (rsc_pad / 8);
   PDTA_DISPATCH_DTYPE16(dt, T, {
-    stem_fwd_k<T><<<(unsigned)grid, 256, lds_bytes, s>>>(
-        (const T*)x, (const T*)wt, (T*)y, Nn, H, W, Cin, P, Q, Kout, R, S,
-        stride, pad);
-  });
-}
-
-// dw fp32 [Kout,Cin,R,S] zeroed by caller. Needs Kout <= 256 and
-// 256 % Kout == 0 (stem: Kout = 64).
-void pdta_stem_bwd_weight(const void* x, const void* dy, float* dw, int Nn,
-                          int H, int W, int Cin, int P, int Q, int Kout,
-                          int R, int S, int stride, int pad, int dt,
-                          hipStream_t s) {
-  long NPQ = (long)Nn * P * Q;
-  int rsc = R * S * Cin;
-  int ngrp = 256 / Kout;
-  int bx = (rsc + ngrp * 8 - 1) / (ngrp * 8);
-  long chunks = 2048 / bx;
-  if (chunks < 1) chunks = 1;
-  long rows_per_chunk = (NPQ + chunks - 1) / chunks;
-  chunks = (NPQ + rows_per_chunk - 1) / rows_per_chunk;
-  dim3 grid(bx, (unsigned)chunks);
-  PDTA_DISPATCH_DTYPE16(dt, T, {
-    stem_wrw_k<T><<<grid, 256, 0, s>>>((const T*)x, (const T*)dy, dw, Nn, H,
-                                       W, Cin, P, Q, Kout, R, S, stride, pad,
-                                       rows_per_chunk);
+    stem_im2col_k<T><<<ew_grid(total, TPB, 1), TPB, 0, s>>>(
+        (const T*)x, (T*)out, Nn, H, W, Cin, P, Q, R, S, stride, pad,
+        rsc_pad);
   });
 }
 
