@@ -251,6 +251,20 @@ __global__ void bn_bwd_elemt_k(const T* __restrict__ x,
   }
 }
 
+
+// -- optional middle reduction: [chunks][2C] -> [16][2C] so the finalize
+// kernels never loop more than 16 chunks (a 2048-chunk finalize at C=64
+// was 95 us; stage1+mid+final is ~15 us total) -----------------------------
+__global__ void bn_mid_k(const float* __restrict__ in,
+                         float* __restrict__ out, int chunks, int twoC) {
+  int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= twoC) return;
+  float s = 0.f;
+  for (int k = blockIdx.y; k < chunks; k += 16)
+    s += in[(long)k * twoC + col];
+  out[(long)blockIdx.y * twoC + col] = s;
+}
+
 static inline dim3 reduce_grid(long rows, int C, int* rows_per_block,
                                int* chunks) {
   int cpb = C < TPB ? C : TPB;
@@ -293,9 +307,18 @@ void pdta_bn_stats(const void* x, float* part, float* mean, float* out2,
   PDTA_DISPATCH_DTYPE(dt, T, {
     bn_part_k<T><<<g, TPB, 0, s>>>((const T*)x, part, rows, C, rpb);
   });
+  const float* fin = part;
+  int fchunks = chunks;
+  if (chunks > 64) {
+    float* part2 = part + (long)chunks * 2 * C;
+    bn_mid_k<<<dim3((2 * C + 255) / 256, 16), 256, 0, s>>>(part, part2,
+                                                           chunks, 2 * C);
+    fin = part2;
+    fchunks = 16;
+  }
   float unbias = rows > 1 ? (float)rows / (float)(rows - 1) : 1.f;
   bn_stats_final_k<<<(C + 63) / 64, 256, 0, s>>>(
-      part, chunks, C, 1.f / (float)rows, unbias, mean, out2, running_mean,
+      fin, fchunks, C, 1.f / (float)rows, unbias, mean, out2, running_mean,
       running_var, momentum, eps, mode);
 }
 
@@ -303,7 +326,7 @@ void pdta_bn_stats(const void* x, float* part, float* mean, float* out2,
 int pdta_bn_chunks(long rows, int C) {
   int rpb, chunks;
   reduce_grid(rows, C, &rpb, &chunks);
-  return chunks;
+  return chunks + 16;  // +16 rows of mid-reduce scratch at the tail
 }
 
 void pdta_bn_apply(const void* x, void* y, const float* weight,
@@ -336,8 +359,17 @@ void pdta_bn_bwd_reduce(const void* x, const void* dy, const void* relu_y,
       bn_bwd_part_k<T, false><<<g, TPB, 0, s>>>(
           (const T*)x, (const T*)dy, nullptr, mean, part, rows, C, rpb);
   });
-  bn_bwd_final_k<<<(C + 63) / 64, 256, 0, s>>>(part, chunks, C, sum_dy,
-                                                 sum_dy_xmu);
+  const float* fin = part;
+  int fchunks = chunks;
+  if (chunks > 64) {
+    float* part2 = part + (long)chunks * 2 * C;
+    bn_mid_k<<<dim3((2 * C + 255) / 256, 16), 256, 0, s>>>(part, part2,
+                                                           chunks, 2 * C);
+    fin = part2;
+    fchunks = 16;
+  }
+  bn_bwd_final_k<<<(C + 63) / 64, 256, 0, s>>>(fin, fchunks, C, sum_dy,
+                                               sum_dy_xmu);
 }
 
 void pdta_bn_bwd_elemt(const void* x, const void* dy, const void* relu_y,

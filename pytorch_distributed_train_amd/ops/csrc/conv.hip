@@ -304,33 +304,36 @@ __global__ __launch_bounds__(TPB) void igemm_wrw_k(
   int row_in_chunk = lane >> 3;
   int col16 = lane & 7;
 
-  // stage 64 rows of dy [row][64 kout] and gathered x [row][64 c]; linear
-  // LDS (no swizzle): the fragment reads are transposed scalars anyway.
+  // stage 64 rows of dy [row][64 kout] and gathered x [row][64 c]; the
+  // source address carries the chunk^=(row&7) swizzle matching the
+  // tr_b16 fragment reads (rule 21: glds dest is lane-linear).
   auto stage = [&](long rowbase, int buf) {
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      long row = rowbase + wave * 16 + i * 8 + row_in_chunk;
+      int ldsrow = wave * 16 + i * 8 + row_in_chunk;
+      int scol16 = col16 ^ (ldsrow & 7);
+      long row = rowbase + ldsrow;
       const char* gdy;
       const char* gx;
       if (row < r1) {
         gdy = reinterpret_cast<const char*>(dy) +
-              (row * Kout + ko_t * 64) * 2 + col16 * 16;
+              (row * Kout + ko_t * 64) * 2 + scol16 * 16;
         int q = (int)(row % Q);
         int p = (int)((row / Q) % P);
         int n = (int)(row / ((long)P * Q));
         int h = p * stride - pad + r, w = q * stride - pad + ss;
         if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
           gx = reinterpret_cast<const char*>(x) +
-               ((((long)n * H + h) * W + w) * C + c_t * 64) * 2 + col16 * 16;
+               ((((long)n * H + h) * W + w) * C + c_t * 64) * 2 +
+               scol16 * 16;
         else
           gx = reinterpret_cast<const char*>(zero_page);
       } else {
         gdy = reinterpret_cast<const char*>(zero_page);
         gx = reinterpret_cast<const char*>(zero_page);
       }
-      int ldsrow = wave * 16 + i * 8;
-      glds16(gdy, (char*)lds[buf][0] + ldsrow * 128);
-      glds16(gx, (char*)lds[buf][1] + ldsrow * 128);
+      glds16(gdy, (char*)lds[buf][0] + (wave * 16 + i * 8) * 128);
+      glds16(gx, (char*)lds[buf][1] + (wave * 16 + i * 8) * 128);
     }
   };
 
@@ -339,34 +342,21 @@ __global__ __launch_bounds__(TPB) void igemm_wrw_k(
   __syncthreads();
 
   int cur = 0;
+  unsigned ldy_base[2] = {lds_addr_of(lds[0][0]), lds_addr_of(lds[1][0])};
+  unsigned lx_base[2] = {lds_addr_of(lds[0][1]), lds_addr_of(lds[1][1])};
   for (long rb = r0; rb < r1; rb += 64) {
     if (rb + 64 < r1) stage(rb + 64, cur ^ 1);
-    const char* ldy = lds[cur][0];
-    const char* lx = lds[cur][1];
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      // transposed fragment reads: element (row, col) at row*128 + col*2
+      // hardware-transposed fragment reads (ds_read_b64_tr_b16)
       s16x8 af[2], bf[2];
 #pragma unroll
-      for (int i = 0; i < 2; ++i) {
-        int ko_local = wm * 32 + i * 16 + (lane & 15);
+      for (int i = 0; i < 2; ++i)
+        af[i] = frag_read_tr(ldy_base[cur], kk * 32, wm * 32 + i * 16,
+                             lane);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          int row = kk * 32 + (lane >> 4) * 8 + j;
-          af[i][j] = *reinterpret_cast<const short*>(
-              ldy + row * 128 + ko_local * 2);
-        }
-      }
-#pragma unroll
-      for (int i = 0; i < 2; ++i) {
-        int c_local = wn * 32 + i * 16 + (lane & 15);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          int row = kk * 32 + (lane >> 4) * 8 + j;
-          bf[i][j] = *reinterpret_cast<const short*>(
-              lx + row * 128 + c_local * 2);
-        }
-      }
+      for (int i = 0; i < 2; ++i)
+        bf[i] = frag_read_tr(lx_base[cur], kk * 32, wn * 32 + i * 16, lane);
 #pragma unroll
       for (int i = 0; i < 2; ++i)
 #pragma unroll

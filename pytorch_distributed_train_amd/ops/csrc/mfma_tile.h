@@ -45,6 +45,46 @@ DEV_INLINE void glds16(const void* gsrc, void* lds_dst) {
   __builtin_amdgcn_global_load_lds((gp_t)gsrc, (lp_t)lds_dst, 16, 0, 0);
 }
 
+// Transposed fragment read via ds_read_b64_tr_b16 (gfx950 hardware
+// transpose; mapping probed on silicon — gpurun_out/tr_probe.log): a
+// 16-lane group's concatenated 8-B fetches (in lane order) form a [4][16]
+// row-major 16-bit matrix and each lane receives its (lane&15) column.
+// Reads an MFMA A/B fragment (lane l: element j = M[m0 + (l&15)]
+// [k0 + (l>>4)*8 + j]) from a [k][m] LDS image with 128-B rows carrying
+// the chunk^=(k&7) 16-B swizzle (write side pre-swizzles the glds source).
+DEV_INLINE s16x8 frag_read_tr(unsigned lds_base, int k0, int m0, int lane) {
+  int g = lane >> 4;
+  int row = (lane & 15) >> 2;  // which of the 4 k-rows this lane fetches
+  int seg = lane & 3;          // which 8-B segment of the 32-B m-span
+  int byte0 = m0 * 2 + seg * 8;
+  int chunk = byte0 >> 4, rem = byte0 & 15;
+  int ka = k0 + g * 8 + row;      // rows for elements j=0..3
+  int kb = ka + 4;                // rows for elements j=4..7
+  unsigned a0 = lds_base + ka * 128 + (((chunk ^ (ka & 7)) << 4) | rem);
+  unsigned a1 = lds_base + kb * 128 + (((chunk ^ (kb & 7)) << 4) | rem);
+  unsigned long long lo, hi;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(lo), "=&v"(hi)
+      : "v"(a0), "v"(a1));
+  __builtin_amdgcn_sched_barrier(0);
+  union {
+    struct { unsigned long long lo, hi; } u;
+    s16x8 v;
+  } cvt;
+  cvt.u.lo = lo;
+  cvt.u.hi = hi;
+  return cvt.v;
+}
+
+// LDS byte offset of a generic pointer to __shared__ memory
+DEV_INLINE unsigned lds_addr_of(const void* p) {
+  return (unsigned)(uintptr_t)(__attribute__((address_space(3))) const void*)
+      p;
+}
+
 // bijective XCD swizzle (T1): contiguous chunk of the grid per XCD
 DEV_INLINE int xcd_swizzle(int bid, int nwg) {
   int q = nwg / 8, r = nwg % 8;
