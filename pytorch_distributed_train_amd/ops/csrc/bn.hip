@@ -260,7 +260,7 @@ __global__ void bn_mid_k(const float* __restrict__ in,
   int col = blockIdx.x * blockDim.x + threadIdx.x;
   if (col >= twoC) return;
   float s = 0.f;
-  for (int k = blockIdx.y; k < chunks; k += 16)
+  for (int k = blockIdx.y; k < chunks; k += 64)
     s += in[(long)k * twoC + col];
   out[(long)blockIdx.y * twoC + col] = s;
 }
@@ -311,10 +311,10 @@ void pdta_bn_stats(const void* x, float* part, float* mean, float* out2,
   int fchunks = chunks;
   if (chunks > 64) {
     float* part2 = part + (long)chunks * 2 * C;
-    bn_mid_k<<<dim3((2 * C + 255) / 256, 16), 256, 0, s>>>(part, part2,
+    bn_mid_k<<<dim3((2 * C + 255) / 256, 64), 256, 0, s>>>(part, part2,
                                                            chunks, 2 * C);
     fin = part2;
-    fchunks = 16;
+    fchunks = 64;
   }
   float unbias = rows > 1 ? (float)rows / (float)(rows - 1) : 1.f;
   bn_stats_final_k<<<(C + 63) / 64, 256, 0, s>>>(
@@ -326,7 +326,7 @@ void pdta_bn_stats(const void* x, float* part, float* mean, float* out2,
 int pdta_bn_chunks(long rows, int C) {
   int rpb, chunks;
   reduce_grid(rows, C, &rpb, &chunks);
-  return chunks + 16;  // +16 rows of mid-reduce scratch at the tail
+  return chunks + 64;  // +64 rows of mid-reduce scratch at the tail
 }
 
 void pdta_bn_apply(const void* x, void* y, const float* weight,
@@ -363,10 +363,10 @@ void pdta_bn_bwd_reduce(const void* x, const void* dy, const void* relu_y,
   int fchunks = chunks;
   if (chunks > 64) {
     float* part2 = part + (long)chunks * 2 * C;
-    bn_mid_k<<<dim3((2 * C + 255) / 256, 16), 256, 0, s>>>(part, part2,
+    bn_mid_k<<<dim3((2 * C + 255) / 256, 64), 256, 0, s>>>(part, part2,
                                                            chunks, 2 * C);
     fin = part2;
-    fchunks = 16;
+    fchunks = 64;
   }
   bn_bwd_final_k<<<(C + 63) / 64, 256, 0, s>>>(fin, fchunks, C, sum_dy,
                                                sum_dy_xmu);

@@ -24,19 +24,25 @@ namespace {
 constexpr int TPB = 256;
 
 // ---------------------------------------------------------------- fwd --
-template <typename T>
+template <typename T, int BNT>  // BNT: output-channel tile (64 or 128)
 __global__ __launch_bounds__(TPB) void igemm_fwd_k(
     const T* __restrict__ x, const T* __restrict__ wt, T* __restrict__ y,
     const T* __restrict__ zero_page, int Nn, int H, int W, int C, int P,
     int Q, int Kout, int R, int S, int stride, int pad, int tiles_n) {
-  __shared__ __attribute__((aligned(16))) char lds[2][2][128 * 64 * 2];
+  // single LDS object (guide §5 trap (a)): A tile 16 KiB + B tile
+  constexpr int BOFF = 128 * 64 * 2;
+  constexpr int NW_N = (BNT == 128) ? 2 : 1;  // waves along n
+  constexpr int NW_M = 4 / NW_N;              // waves along m
+  constexpr int MI = 128 / NW_M / 16;         // m-fragments per wave
+  constexpr int NJ = BNT / NW_N / 16;         // n-fragments per wave
+  __shared__ __attribute__((aligned(16))) char lds[2][BOFF + BNT * 64 * 2];
 
   int bid = xcd_swizzle(blockIdx.x, gridDim.x);
   int tm = bid / tiles_n, tn = bid % tiles_n;
   int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
-  int wm = wave >> 1, wn = wave & 1;
+  int wm = wave / NW_N, wn = wave % NW_N;
   long M = (long)Nn * P * Q;
-  int m0 = tm * 128, n0 = tn * 128;
+  int m0 = tm * 128, n0 = tn * BNT;
   int RSC = R * S * C;
 
   // staging geometry for this thread's 4 A-rows (and 4 B-rows)
@@ -56,11 +62,11 @@ __global__ __launch_bounds__(TPB) void igemm_fwd_k(
     a_swz[i] = gcol16 ^ (row & 7);
   }
 
-  f32x4 acc[4][4];
+  f32x4 acc[MI][NJ];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < MI; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < NJ; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   int cchunks = C / 64;
   int ktiles = R * S * cchunks;
@@ -79,18 +85,18 @@ __global__ __launch_bounds__(TPB) void igemm_fwd_k(
                a_swz[i] * 16;
       else
         gptr = reinterpret_cast<const char*>(zero_page);
-      glds16(gptr, (char*)lds[buf][0] + (wave * 32 + i * 8) * 128);
+      glds16(gptr, (char*)lds[buf] + (wave * 32 + i * 8) * 128);
     }
     // B: Wt rows (kout), k = tap*C + ct*64 contiguous
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      int row = wave * 32 + i * 8 + row_in_chunk;
+    for (int i = 0; i < BNT / 32; ++i) {  // 2 (BNT=64) or 4 (128)
+      int row = wave * (BNT / 4) + i * 8 + row_in_chunk;
       int ko = n0 + row;
       if (ko > Kout - 1) ko = Kout - 1;
       const char* gptr = reinterpret_cast<const char*>(wt) +
                          ((long)ko * RSC + tap * C + ct * 64) * 2 +
                          (gcol16 ^ (row & 7)) * 16;
-      glds16(gptr, (char*)lds[buf][1] + (wave * 32 + i * 8) * 128);
+      glds16(gptr, (char*)lds[buf] + BOFF + (wave * (BNT / 4) + i * 8) * 128);
     }
   };
 
@@ -101,23 +107,23 @@ __global__ __launch_bounds__(TPB) void igemm_fwd_k(
   int cur = 0;
   for (int kt = 0; kt < ktiles; ++kt) {
     if (kt + 1 < ktiles) stage(kt + 1, cur ^ 1);
-    const char* la = lds[cur][0];
-    const char* lb = lds[cur][1];
+    const char* la = lds[cur];
+    const char* lb = lds[cur] + BOFF;
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      s16x8 af[4], bf[4];
+      s16x8 af[MI], bf[NJ];
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
-        af[i] = frag_read128(la, wm * 64 + i * 16 + (lane & 15),
+      for (int i = 0; i < MI; ++i)
+        af[i] = frag_read128(la, wm * (128 / NW_M) + i * 16 + (lane & 15),
                              (lane >> 4) + kk * 4);
 #pragma unroll
-      for (int j = 0; j < 4; ++j)
-        bf[j] = frag_read128(lb, wn * 64 + j * 16 + (lane & 15),
+      for (int j = 0; j < NJ; ++j)
+        bf[j] = frag_read128(lb, wn * (BNT / NW_N) + j * 16 + (lane & 15),
                              (lane >> 4) + kk * 4);
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
+      for (int i = 0; i < MI; ++i)
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
+        for (int j = 0; j < NJ; ++j)
           acc[i][j] = MfmaOp<T>::run(af[i], bf[j], acc[i][j]);
     }
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -126,11 +132,11 @@ __global__ __launch_bounds__(TPB) void igemm_fwd_k(
   }
 
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    long rbase = m0 + wm * 64 + i * 16 + (lane >> 4) * 4;
+  for (int i = 0; i < MI; ++i) {
+    long rbase = m0 + wm * (128 / NW_M) + i * 16 + (lane >> 4) * 4;
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      int col = n0 + wn * 64 + j * 16 + (lane & 15);
+    for (int j = 0; j < NJ; ++j) {
+      int col = n0 + wn * (BNT / NW_N) + j * 16 + (lane & 15);
       if (col >= Kout) continue;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -142,19 +148,24 @@ __global__ __launch_bounds__(TPB) void igemm_fwd_k(
 }
 
 // ----------------------------------------------------------- bwd-data --
-template <typename T, int STRIDE>  // STRIDE: 1/2 compile-time, 0 = runtime
+template <typename T, int STRIDE, int BNT>  // STRIDE 1/2/0(runtime)
 __global__ __launch_bounds__(TPB) void igemm_bwd_data_k(
     const T* __restrict__ dy, const T* __restrict__ w2, T* __restrict__ dx,
     const T* __restrict__ zero_page, int Nn, int H, int W, int C, int P,
     int Q, int Kout, int R, int S, int stride, int pad, int tiles_n) {
-  __shared__ __attribute__((aligned(16))) char lds[2][2][128 * 64 * 2];
+  constexpr int BOFF = 128 * 64 * 2;
+  constexpr int NW_N = (BNT == 128) ? 2 : 1;
+  constexpr int NW_M = 4 / NW_N;
+  constexpr int MI = 128 / NW_M / 16;
+  constexpr int NJ = BNT / NW_N / 16;
+  __shared__ __attribute__((aligned(16))) char lds[2][BOFF + BNT * 64 * 2];
 
   int bid = xcd_swizzle(blockIdx.x, gridDim.x);
   int tm = bid / tiles_n, tn = bid % tiles_n;
   int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
-  int wm = wave >> 1, wn = wave & 1;
+  int wm = wave / NW_N, wn = wave % NW_N;
   long M = (long)Nn * H * W;
-  int m0 = tm * 128, n0 = tn * 128;
+  int m0 = tm * 128, n0 = tn * BNT;
 
   int row_in_chunk = lane >> 3;
   int gcol16 = (lane & 7);
@@ -170,11 +181,11 @@ __global__ __launch_bounds__(TPB) void igemm_bwd_data_k(
     a_swz[i] = gcol16 ^ (row & 7);
   }
 
-  f32x4 acc[4][4];
+  f32x4 acc[MI][NJ];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < MI; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < NJ; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   int kchunks = Kout / 64;
   int ktiles = R * S * kchunks;
@@ -209,17 +220,18 @@ __global__ __launch_bounds__(TPB) void igemm_bwd_data_k(
                a_swz[i] * 16;
       else
         gptr = reinterpret_cast<const char*>(zero_page);
-      glds16(gptr, (char*)lds[buf][0] + (wave * 32 + i * 8) * 128);
+      glds16(gptr, (char*)lds[buf] + (wave * 32 + i * 8) * 128);
     }
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      int row = wave * 32 + i * 8 + row_in_chunk;
+    for (int i = 0; i < BNT / 32; ++i) {
+      int row = wave * (BNT / 4) + i * 8 + row_in_chunk;
       int c = n0 + row;
       if (c > C - 1) c = C - 1;
       const char* gptr = reinterpret_cast<const char*>(w2) +
                          ((((long)tap) * C + c) * Kout + kot * 64) * 2 +
                          (gcol16 ^ (row & 7)) * 16;
-      glds16(gptr, (char*)lds[buf][1] + (wave * 32 + i * 8) * 128);
+      glds16(gptr,
+             (char*)lds[buf] + BOFF + (wave * (BNT / 4) + i * 8) * 128);
     }
   };
 
@@ -230,23 +242,23 @@ __global__ __launch_bounds__(TPB) void igemm_bwd_data_k(
   int cur = 0;
   for (int kt = 0; kt < ktiles; ++kt) {
     if (kt + 1 < ktiles) stage(kt + 1, cur ^ 1);
-    const char* la = lds[cur][0];
-    const char* lb = lds[cur][1];
+    const char* la = lds[cur];
+    const char* lb = lds[cur] + BOFF;
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      s16x8 af[4], bf[4];
+      s16x8 af[MI], bf[NJ];
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
-        af[i] = frag_read128(la, wm * 64 + i * 16 + (lane & 15),
+      for (int i = 0; i < MI; ++i)
+        af[i] = frag_read128(la, wm * (128 / NW_M) + i * 16 + (lane & 15),
                              (lane >> 4) + kk * 4);
 #pragma unroll
-      for (int j = 0; j < 4; ++j)
-        bf[j] = frag_read128(lb, wn * 64 + j * 16 + (lane & 15),
+      for (int j = 0; j < NJ; ++j)
+        bf[j] = frag_read128(lb, wn * (BNT / NW_N) + j * 16 + (lane & 15),
                              (lane >> 4) + kk * 4);
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
+      for (int i = 0; i < MI; ++i)
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
+        for (int j = 0; j < NJ; ++j)
           acc[i][j] = MfmaOp<T>::run(af[i], bf[j], acc[i][j]);
     }
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -255,11 +267,11 @@ __global__ __launch_bounds__(TPB) void igemm_bwd_data_k(
   }
 
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    long rbase = m0 + wm * 64 + i * 16 + (lane >> 4) * 4;
+  for (int i = 0; i < MI; ++i) {
+    long rbase = m0 + wm * (128 / NW_M) + i * 16 + (lane >> 4) * 4;
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      int col = n0 + wn * 64 + j * 16 + (lane & 15);
+    for (int j = 0; j < NJ; ++j) {
+      int col = n0 + wn * (BNT / NW_N) + j * 16 + (lane & 15);
       if (col >= C) continue;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -275,7 +287,7 @@ __global__ __launch_bounds__(TPB) void igemm_bwd_data_k(
 // (64-row LDS tiles, transposed fragment reads), fp32 atomicAdd into the
 // torch-layout dW [Kout,C,R,S] (no repack needed). Rows beyond the image
 // or invalid taps stage from the zero page (adds 0).
-template <typename T>
+template <typename T, bool GATHER>  // GATHER=false: 1x1/s1/p0 (identity)
 __global__ __launch_bounds__(TPB) void igemm_wrw_k(
     const T* __restrict__ x, const T* __restrict__ dy,
     float* __restrict__ dw, const T* __restrict__ zero_page, int Nn, int H,
@@ -307,6 +319,31 @@ __global__ __launch_bounds__(TPB) void igemm_wrw_k(
   // stage 64 rows of dy [row][64 kout] and gathered x [row][64 c]; the
   // source address carries the chunk^=(row&7) swizzle matching the
   // tr_b16 fragment reads (rule 21: glds dest is lane-linear).
+  // incremental row->(n,p,q) decode for the gather (divisions only once)
+  int g_q[2], g_p[2], g_n[2];
+  if (GATHER) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      long row = r0 + wave * 16 + i * 8 + row_in_chunk;
+      g_q[i] = (int)(row % Q);
+      g_p[i] = (int)((row / Q) % P);
+      g_n[i] = (int)(row / ((long)P * Q));
+    }
+  }
+  auto advance = [&](int steps) {
+    if (!GATHER) return;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      g_q[i] += steps;
+      while (g_q[i] >= Q) {
+        g_q[i] -= Q;
+        if (++g_p[i] >= P) {
+          g_p[i] = 0;
+          ++g_n[i];
+        }
+      }
+    }
+  };
   auto stage = [&](long rowbase, int buf) {
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
@@ -318,16 +355,18 @@ __global__ __launch_bounds__(TPB) void igemm_wrw_k(
       if (row < r1) {
         gdy = reinterpret_cast<const char*>(dy) +
               (row * Kout + ko_t * 64) * 2 + scol16 * 16;
-        int q = (int)(row % Q);
-        int p = (int)((row / Q) % P);
-        int n = (int)(row / ((long)P * Q));
-        int h = p * stride - pad + r, w = q * stride - pad + ss;
-        if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
+        if (GATHER) {
+          int h = g_p[i] * stride - pad + r, w = g_q[i] * stride - pad + ss;
+          if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
+            gx = reinterpret_cast<const char*>(x) +
+                 ((((long)g_n[i] * H + h) * W + w) * C + c_t * 64) * 2 +
+                 scol16 * 16;
+          else
+            gx = reinterpret_cast<const char*>(zero_page);
+        } else {
           gx = reinterpret_cast<const char*>(x) +
-               ((((long)n * H + h) * W + w) * C + c_t * 64) * 2 +
-               scol16 * 16;
-        else
-          gx = reinterpret_cast<const char*>(zero_page);
+               (row * C + c_t * 64) * 2 + scol16 * 16;
+        }
       } else {
         gdy = reinterpret_cast<const char*>(zero_page);
         gx = reinterpret_cast<const char*>(zero_page);
@@ -345,7 +384,10 @@ __global__ __launch_bounds__(TPB) void igemm_wrw_k(
   unsigned ldy_base[2] = {lds_addr_of(lds[0][0]), lds_addr_of(lds[1][0])};
   unsigned lx_base[2] = {lds_addr_of(lds[0][1]), lds_addr_of(lds[1][1])};
   for (long rb = r0; rb < r1; rb += 64) {
-    if (rb + 64 < r1) stage(rb + 64, cur ^ 1);
+    if (rb + 64 < r1) {
+      advance(64);
+      stage(rb + 64, cur ^ 1);
+    }
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       // hardware-transposed fragment reads (ds_read_b64_tr_b16)
@@ -398,34 +440,32 @@ __global__ void stem_im2col_k(const T* __restrict__ x, T* __restrict__ out,
                               int Nn, int H, int W, int Cin, int P, int Q,
                               int R, int S, int stride, int pad,
                               int rsc_pad) {
-  int g8 = rsc_pad / 8;
-  long total = (long)Nn * P * Q * g8;
-  long item = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  // thread per output row: decode once, gather the whole R*S*Cin window
+  // (within one r the window row is S*Cin contiguous elements of x)
+  long total = (long)Nn * P * Q;
+  long row = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long gstride = (long)gridDim.x * blockDim.x;
-  int rsc = R * S * Cin;
-  for (; item < total; item += gstride) {
-    int grp = (int)(item % g8);
-    long row = item / g8;
+  int SC = S * Cin;
+  for (; row < total; row += gstride) {
     int q = (int)(row % Q);
     int p = (int)((row / Q) % P);
     int n = (int)(row / ((long)P * Q));
     int h0 = p * stride - pad, w0 = q * stride - pad;
-    Vec8<T> v;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int e = grp * 8 + j;
-      float val = 0.f;
-      if (e < rsc) {
-        int c = e % Cin;
-        int t = e / Cin;
-        int sc = t % S, rr = t / S;
-        int h = h0 + rr, w = w0 + sc;
-        if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
-          val = to_f32(x[(((long)n * H + h) * W + w) * Cin + c]);
+    T* o = out + row * rsc_pad;
+    int e = 0;
+    for (int rr = 0; rr < R; ++rr) {
+      int h = h0 + rr;
+      bool hv = (unsigned)h < (unsigned)H;
+      const T* xr = x + (((long)n * H + h) * W + w0) * Cin;
+      for (int sc = 0; sc < S; ++sc) {
+        int w = w0 + sc;
+        bool v = hv && (unsigned)w < (unsigned)W;
+        for (int c = 0; c < Cin; ++c)
+          o[e + c] = v ? xr[sc * Cin + c] : from_f32<T>(0.f);
+        e += Cin;
       }
-      v.v[j] = from_f32<T>(val);
     }
-    store8(out + row * rsc_pad + grp * 8, v);
+    for (; e < rsc_pad; ++e) o[e] = from_f32<T>(0.f);
   }
 }
 
@@ -438,12 +478,18 @@ void pdta_conv_fwd(const void* x, const void* wt, void* y,
                    int Q, int Kout, int R, int S, int stride, int pad,
                    int dt, hipStream_t s) {
   long M = (long)Nn * P * Q;
-  int tiles_m = (int)((M + 127) / 128), tiles_n = (Kout + 127) / 128;
+  int bnt = (Kout % 128 == 0) ? 128 : 64;
+  int tiles_m = (int)((M + 127) / 128), tiles_n = (Kout + bnt - 1) / bnt;
   dim3 grid(tiles_m * tiles_n);
   PDTA_DISPATCH_DTYPE16(dt, T, {
-    igemm_fwd_k<T><<<grid, TPB, 0, s>>>((const T*)x, (const T*)wt, (T*)y,
-                                        (const T*)zero_page, Nn, H, W, C, P,
-                                        Q, Kout, R, S, stride, pad, tiles_n);
+    if (bnt == 128)
+      igemm_fwd_k<T, 128><<<grid, TPB, 0, s>>>(
+          (const T*)x, (const T*)wt, (T*)y, (const T*)zero_page, Nn, H, W,
+          C, P, Q, Kout, R, S, stride, pad, tiles_n);
+    else
+      igemm_fwd_k<T, 64><<<grid, TPB, 0, s>>>(
+          (const T*)x, (const T*)wt, (T*)y, (const T*)zero_page, Nn, H, W,
+          C, P, Q, Kout, R, S, stride, pad, tiles_n);
   });
 }
 
@@ -452,21 +498,37 @@ void pdta_conv_bwd_data(const void* dy, const void* w2, void* dx,
                         int P, int Q, int Kout, int R, int S, int stride,
                         int pad, int dt, hipStream_t s) {
   long M = (long)Nn * H * W;
-  int tiles_m = (int)((M + 127) / 128), tiles_n = (C + 127) / 128;
+  int bnt = (C % 128 == 0) ? 128 : 64;
+  int tiles_m = (int)((M + 127) / 128), tiles_n = (C + bnt - 1) / bnt;
   dim3 grid(tiles_m * tiles_n);
   PDTA_DISPATCH_DTYPE16(dt, T, {
-    if (stride == 1)
-      igemm_bwd_data_k<T, 1><<<grid, TPB, 0, s>>>(
-          (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H, W,
-          C, P, Q, Kout, R, S, stride, pad, tiles_n);
-    else if (stride == 2)
-      igemm_bwd_data_k<T, 2><<<grid, TPB, 0, s>>>(
-          (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H, W,
-          C, P, Q, Kout, R, S, stride, pad, tiles_n);
-    else
-      igemm_bwd_data_k<T, 0><<<grid, TPB, 0, s>>>(
-          (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H, W,
-          C, P, Q, Kout, R, S, stride, pad, tiles_n);
+    if (bnt == 128) {
+      if (stride == 1)
+        igemm_bwd_data_k<T, 1, 128><<<grid, TPB, 0, s>>>(
+            (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H,
+            W, C, P, Q, Kout, R, S, stride, pad, tiles_n);
+      else if (stride == 2)
+        igemm_bwd_data_k<T, 2, 128><<<grid, TPB, 0, s>>>(
+            (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H,
+            W, C, P, Q, Kout, R, S, stride, pad, tiles_n);
+      else
+        igemm_bwd_data_k<T, 0, 128><<<grid, TPB, 0, s>>>(
+            (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H,
+            W, C, P, Q, Kout, R, S, stride, pad, tiles_n);
+    } else {
+      if (stride == 1)
+        igemm_bwd_data_k<T, 1, 64><<<grid, TPB, 0, s>>>(
+            (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H,
+            W, C, P, Q, Kout, R, S, stride, pad, tiles_n);
+      else if (stride == 2)
+        igemm_bwd_data_k<T, 2, 64><<<grid, TPB, 0, s>>>(
+            (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H,
+            W, C, P, Q, Kout, R, S, stride, pad, tiles_n);
+      else
+        igemm_bwd_data_k<T, 0, 64><<<grid, TPB, 0, s>>>(
+            (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H,
+            W, C, P, Q, Kout, R, S, stride, pad, tiles_n);
+    }
   });
 }
 
@@ -484,10 +546,16 @@ void pdta_conv_bwd_weight(const void* x, const void* dy, float* dw,
   long rows_per_chunk = ((NPQ + splitk - 1) / splitk + 63) / 64 * 64;
   long chunks = (NPQ + rows_per_chunk - 1) / rows_per_chunk;
   dim3 grid(tiles, (unsigned)chunks);
+  bool identity = (R == 1 && S == 1 && stride == 1 && pad == 0);
   PDTA_DISPATCH_DTYPE16(dt, T, {
-    igemm_wrw_k<T><<<grid, TPB, 0, s>>>(
-        (const T*)x, (const T*)dy, dw, (const T*)zero_page, Nn, H, W, C, P,
-        Q, Kout, R, S, stride, pad, rows_per_chunk, ko_tiles, c_tiles);
+    if (identity)
+      igemm_wrw_k<T, false><<<grid, TPB, 0, s>>>(
+          (const T*)x, (const T*)dy, dw, (const T*)zero_page, Nn, H, W, C,
+          P, Q, Kout, R, S, stride, pad, rows_per_chunk, ko_tiles, c_tiles);
+    else
+      igemm_wrw_k<T, true><<<grid, TPB, 0, s>>>(
+          (const T*)x, (const T*)dy, dw, (const T*)zero_page, Nn, H, W, C,
+          P, Q, Kout, R, S, stride, pad, rows_per_chunk, ko_tiles, c_tiles);
   });
 }
 
