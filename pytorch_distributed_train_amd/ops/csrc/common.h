@@ -79,6 +79,24 @@ DEV_INLINE float wave_reduce_max(float x) {
   return x;
 }
 
+// ---- division by launch-invariant divisor (magic numbers; validated
+// exhaustively for n < 2^31 over the divisors this framework uses) -------
+struct MagicDiv {
+  unsigned mul;
+  int shift;
+};
+static inline MagicDiv magic_div(unsigned d) {
+  int s = 0;
+  while ((1u << s) < d) ++s;
+  MagicDiv m;
+  m.shift = 32 + s;
+  m.mul = (unsigned)(((1ull << m.shift) + d - 1) / d);
+  return m;
+}
+DEV_INLINE unsigned mdiv(unsigned n, unsigned mul, int shift) {
+  return (unsigned)(((unsigned long long)n * mul) >> shift);
+}
+
 // ---- grid sizing ---------------------------------------------------------
 static inline int ew_grid(long n, int tpb, int per_thread) {
   long blocks = (n + (long)tpb * per_thread - 1) / ((long)tpb * per_thread);

@@ -330,17 +330,18 @@ __global__ __launch_bounds__(TPB) void igemm_wrw_k(
       g_n[i] = (int)(row / ((long)P * Q));
     }
   }
-  auto advance = [&](int steps) {
+  int adv_dq = 64 % Q, adv_dp = 64 / Q;  // bounded carries per advance
+  auto advance = [&]() {
     if (!GATHER) return;
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      g_q[i] += steps;
-      while (g_q[i] >= Q) {
-        g_q[i] -= Q;
-        if (++g_p[i] >= P) {
-          g_p[i] = 0;
-          ++g_n[i];
-        }
+      g_q[i] += adv_dq;
+      int cy = g_q[i] >= Q;
+      g_q[i] -= cy ? Q : 0;
+      g_p[i] += adv_dp + cy;
+      while (g_p[i] >= P) {
+        g_p[i] -= P;
+        ++g_n[i];
       }
     }
   };
@@ -385,7 +386,7 @@ __global__ __launch_bounds__(TPB) void igemm_wrw_k(
   unsigned lx_base[2] = {lds_addr_of(lds[0][1]), lds_addr_of(lds[1][1])};
   for (long rb = r0; rb < r1; rb += 64) {
     if (rb + 64 < r1) {
-      advance(64);
+      advance();
       stage(rb + 64, cur ^ 1);
     }
 #pragma unroll
@@ -438,34 +439,41 @@ __global__ __launch_bounds__(TPB) void igemm_wrw_k(
 template <typename T>
 __global__ void stem_im2col_k(const T* __restrict__ x, T* __restrict__ out,
                               int Nn, int H, int W, int Cin, int P, int Q,
-                              int R, int S, int stride, int pad,
-                              int rsc_pad) {
-  // thread per output row: decode once, gather the whole R*S*Cin window
-  // (within one r the window row is S*Cin contiguous elements of x)
-  long total = (long)Nn * P * Q;
-  long row = (long)blockIdx.x * blockDim.x + threadIdx.x;
+                              int R, int S, int stride, int pad, int rsc_pad,
+                              unsigned mq, int sq, unsigned mpq, int spq) {
+  // item = (row, 8-element group): consecutive threads write consecutive
+  // 16-B chunks of one row (fully coalesced); row decode via magic
+  // division (runtime-divisor div was ~40% of this kernel's issue slots)
+  int g8 = rsc_pad / 8;
+  long total = (long)Nn * P * Q * g8;
+  long item = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long gstride = (long)gridDim.x * blockDim.x;
-  int SC = S * Cin;
-  for (; row < total; row += gstride) {
-    int q = (int)(row % Q);
-    int p = (int)((row / Q) % P);
-    int n = (int)(row / ((long)P * Q));
-    int h0 = p * stride - pad, w0 = q * stride - pad;
-    T* o = out + row * rsc_pad;
-    int e = 0;
-    for (int rr = 0; rr < R; ++rr) {
-      int h = h0 + rr;
-      bool hv = (unsigned)h < (unsigned)H;
-      const T* xr = x + (((long)n * H + h) * W + w0) * Cin;
-      for (int sc = 0; sc < S; ++sc) {
-        int w = w0 + sc;
-        bool v = hv && (unsigned)w < (unsigned)W;
-        for (int c = 0; c < Cin; ++c)
-          o[e + c] = v ? xr[sc * Cin + c] : from_f32<T>(0.f);
-        e += Cin;
+  int rsc = R * S * Cin;
+  for (; item < total; item += gstride) {
+    int grp = (int)(item % g8);  // g8 is 24 for the stem: strength-reduced?
+    unsigned row = (unsigned)(item / g8);
+    unsigned npq = mdiv(row, mpq, spq);      // row / (P*Q)
+    unsigned rem = row - npq * (unsigned)(P * Q);
+    unsigned p = mdiv(rem, mq, sq);          // rem / Q
+    unsigned q = rem - p * (unsigned)Q;
+    int n = (int)npq;
+    int h0 = (int)p * stride - pad, w0 = (int)q * stride - pad;
+    Vec8<T> v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int e = grp * 8 + j;
+      float val = 0.f;
+      if (e < rsc) {
+        int c = e % Cin;
+        int t = e / Cin;
+        int sc = t % S, rr = t / S;
+        int h = h0 + rr, w = w0 + sc;
+        if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
+          val = to_f32(x[(((long)n * H + h) * W + w) * Cin + c]);
       }
+      v.v[j] = from_f32<T>(val);
     }
-    for (; e < rsc_pad; ++e) o[e] = from_f32<T>(0.f);
+    store8(out + (long)row * rsc_pad + grp * 8, v);
   }
 }
 
@@ -563,10 +571,11 @@ void pdta_stem_im2col(const void* x, void* out, int Nn, int H, int W,
                       int Cin, int P, int Q, int R, int S, int stride,
                       int pad, int rsc_pad, int dt, hipStream_t s) {
   long total = (long)Nn * P * Q * (rsc_pad / 8);
+  MagicDiv mq = magic_div(Q), mpq = magic_div(P * Q);
   PDTA_DISPATCH_DTYPE16(dt, T, {
     stem_im2col_k<T><<<ew_grid(total, TPB, 1), TPB, 0, s>>>(
         (const T*)x, (T*)out, Nn, H, W, Cin, P, Q, R, S, stride, pad,
-        rsc_pad);
+        rsc_pad, mq.mul, mq.shift, mpq.mul, mpq.shift);
   });
 }
 
