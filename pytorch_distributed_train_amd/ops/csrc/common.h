@@ -86,9 +86,16 @@ struct MagicDiv {
   int shift;
 };
 static inline MagicDiv magic_div(unsigned d) {
+  MagicDiv m;
+  if ((d & (d - 1)) == 0) {  // power of two: ceil(2^s/d) = 2^32 truncates
+    int k = 0;
+    while ((1u << k) < d) ++k;
+    m.mul = 0x80000000u;  // n >> k == (n * 2^31) >> (31 + k)
+    m.shift = 31 + k;
+    return m;
+  }
   int s = 0;
   while ((1u << s) < d) ++s;
-  MagicDiv m;
   m.shift = 32 + s;
   m.mul = (unsigned)(((1ull << m.shift) + d - 1) / d);
   return m;
