@@ -20,7 +20,6 @@ class Conv2dNHWC(nn.Module):
     def __init__(self, in_channels: int, out_channels: int, kernel_size: int,
                  stride: int = 1, padding: int = 0, bias: bool = False):
         super().__init__()
-        assert not bias, "ResNet/ViT convs carry no bias (BN/LN follows)"
         self.in_channels = in_channels
         self.out_channels = out_channels
         self.kernel_size = kernel_size
@@ -31,6 +30,10 @@ class Conv2dNHWC(nn.Module):
         # kaiming normal, fan_out, relu — the torchvision resnet init
         fan_out = kernel_size * kernel_size * out_channels
         nn.init.normal_(self.weight, 0, math.sqrt(2.0 / fan_out))
+        if bias:
+            self.bias = nn.Parameter(torch.zeros(out_channels))
+        else:
+            self.register_parameter("bias", None)
 
     def forward(self, x):
         # two compute-dtype packs keep fwd AND bwd-data K-contiguous
@@ -38,7 +41,10 @@ class Conv2dNHWC(nn.Module):
         w = self.weight
         wt = w.permute(0, 2, 3, 1).to(x.dtype).contiguous()
         w2 = w.permute(2, 3, 1, 0).to(x.dtype).contiguous()
-        return OF.conv2d_nhwc(x, w, wt, w2, self.stride, self.padding)
+        y = OF.conv2d_nhwc(x, w, wt, w2, self.stride, self.padding)
+        if self.bias is not None:
+            y = y + self.bias.to(y.dtype)  # broadcast over trailing C
+        return y
 
     def extra_repr(self):
         return (f"{self.in_channels}, {self.out_channels}, "

@@ -1,0 +1,144 @@
+"""ViT-B/16 from scratch (BASELINE config #5: "transformer MFMA GEMM
+path"). State-dict keys follow torchvision VisionTransformer naming
+(class_token, conv_proj, encoder.pos_embedding,
+encoder.layers.encoder_layer_i.{ln_1, self_attention.*, ln_2, mlp.0/3},
+encoder.ln, heads.head) for checkpoint parity.
+
+Compute design (MI355X-first): the patch embedding is a 16x16/s16 conv
+through the same im2col+MFMA igemm path as the ResNet stem; the token
+stream is PADDED from 197 to 256 rows so every GEMM K/M dim is a multiple
+of 64 — padded rows never mix into real rows (softmax masks key columns
+>= 197, padded V rows are zero, LN/MLP are row-local), so results for the
+class token are exact. Attention runs as batched MFMA GEMMs (QK^T in
+fp32) + a fused masked-softmax kernel (csrc/ln.hip).
+"""
+from __future__ import annotations
+
+from collections import OrderedDict
+
+import torch
+import torch.nn as nn
+
+from ..ops import functional as OF
+from .layers import Conv2dNHWC, Linear
+
+SEQ = 197      # 14*14 patches + class token
+SEQ_PAD = 256  # padded to a GEMM-friendly length
+
+
+class LayerNorm(nn.Module):
+    def __init__(self, dim: int, eps: float = 1e-6):
+        super().__init__()
+        self.eps = eps
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.bias = nn.Parameter(torch.zeros(dim))
+
+    def forward(self, x):
+        return OF.layer_norm(x, self.weight, self.bias, self.eps)
+
+
+class MultiheadSelfAttention(nn.Module):
+    """torchvision-compatible parameter layout: in_proj_weight [3d, d],
+    in_proj_bias [3d], out_proj Linear."""
+
+    def __init__(self, dim: int, num_heads: int):
+        super().__init__()
+        self.num_heads = num_heads
+        self.head_dim = dim // num_heads
+        self.in_proj_weight = nn.Parameter(torch.empty(3 * dim, dim))
+        self.in_proj_bias = nn.Parameter(torch.zeros(3 * dim))
+        nn.init.xavier_uniform_(self.in_proj_weight)
+        self.out_proj = Linear(dim, dim)
+
+    def forward(self, x, nvalid: int = SEQ):
+        b, s, d = x.shape
+        h, hd = self.num_heads, self.head_dim
+        packed = self.in_proj_weight.to(x.dtype)
+        qkv = OF.linear(x.reshape(b * s, d), self.in_proj_weight, packed,
+                        self.in_proj_bias)
+        qkv = qkv.view(b, s, 3, h, hd).permute(2, 0, 3, 1, 4)
+        q = qkv[0].reshape(b * h, s, hd).contiguous()
+        k = qkv[1].reshape(b * h, s, hd).contiguous()
+        v = qkv[2].reshape(b * h, s, hd).contiguous()
+        out = OF.attention(q, k, v, nvalid, 1.0 / (hd ** 0.5))
+        out = out.view(b, h, s, hd).permute(0, 2, 1, 3).reshape(b * s, d)
+        return self.out_proj(out.contiguous()).view(b, s, d)
+
+
+class GELU(nn.Module):
+    def forward(self, x):
+        return OF.gelu(x)
+
+
+class EncoderBlock(nn.Module):
+    def __init__(self, dim: int, num_heads: int, mlp_dim: int):
+        super().__init__()
+        self.ln_1 = LayerNorm(dim)
+        self.self_attention = MultiheadSelfAttention(dim, num_heads)
+        self.ln_2 = LayerNorm(dim)
+        self.mlp = nn.Sequential(
+            Linear(dim, mlp_dim), GELU(), nn.Identity(), Linear(mlp_dim, dim))
+
+    def forward(self, x):
+        b, s, d = x.shape
+        x = x + self.self_attention(self.ln_1(x))
+        h = self.ln_2(x).reshape(b * s, d)
+        h = self.mlp[3](self.mlp[1](self.mlp[0](h))).view(b, s, d)
+        return x + h
+
+
+class Encoder(nn.Module):
+    def __init__(self, num_layers: int, dim: int, num_heads: int,
+                 mlp_dim: int):
+        super().__init__()
+        self.pos_embedding = nn.Parameter(
+            torch.empty(1, SEQ, dim).normal_(std=0.02))
+        layers = OrderedDict()
+        for i in range(num_layers):
+            layers[f"encoder_layer_{i}"] = EncoderBlock(dim, num_heads,
+                                                        mlp_dim)
+        self.layers = nn.Sequential(layers)
+        self.ln = LayerNorm(dim)
+
+    def forward(self, x):  # x [B, SEQ, dim]
+        x = x + self.pos_embedding.to(x.dtype)
+        b, s, d = x.shape
+        if s < SEQ_PAD:  # pad token rows for GEMM-friendly shapes
+            pad = torch.zeros(b, SEQ_PAD - s, d, dtype=x.dtype,
+                              device=x.device)
+            x = torch.cat([x, pad], dim=1)
+        x = self.layers(x)
+        return self.ln(x)
+
+
+class VisionTransformer(nn.Module):
+    def __init__(self, image_size=224, patch_size=16, num_layers=12,
+                 num_heads=12, hidden_dim=768, mlp_dim=3072,
+                 num_classes=1000, compute_dtype=None):
+        super().__init__()
+        self.compute_dtype = compute_dtype or torch.float32
+        self.patch_size = patch_size
+        self.hidden_dim = hidden_dim
+        self.class_token = nn.Parameter(torch.zeros(1, 1, hidden_dim))
+        self.conv_proj = Conv2dNHWC(3, hidden_dim, patch_size,
+                                    stride=patch_size, padding=0, bias=True)
+        self.encoder = Encoder(num_layers, hidden_dim, num_heads, mlp_dim)
+        self.heads = nn.Sequential(OrderedDict(
+            [("head", Linear(hidden_dim, num_classes))]))
+        nn.init.zeros_(self.heads.head.weight)
+        nn.init.zeros_(self.heads.head.bias)
+
+    def forward(self, x):  # NCHW input, like the reference's loader feeds
+        b = x.shape[0]
+        x = x.permute(0, 2, 3, 1).to(self.compute_dtype).contiguous()
+        x = self.conv_proj(x)                      # [B, 14, 14, 768]
+        x = x.reshape(b, -1, self.hidden_dim)      # [B, 196, 768]
+        cls = self.class_token.to(x.dtype).expand(b, -1, -1)
+        x = torch.cat([cls, x], dim=1)             # [B, 197, 768]
+        x = self.encoder(x)                        # [B, 256, 768] (padded)
+        return self.heads(x[:, 0])                 # class token -> logits
+
+
+def vit_b_16(num_classes=1000, compute_dtype=None):
+    return VisionTransformer(num_classes=num_classes,
+                             compute_dtype=compute_dtype)

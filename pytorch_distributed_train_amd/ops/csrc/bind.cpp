@@ -55,6 +55,20 @@ void pdta_conv_bwd_weight(const void*, const void*, float*, const void*, int,
                           int, hipStream_t);
 void pdta_stem_im2col(const void*, void*, int, int, int, int, int, int,
                       int, int, int, int, int, int, hipStream_t);
+void pdta_gemm_nt_batched(const void*, const void*, void*, int, int, int,
+                          int, long, long, long, int, int, hipStream_t);
+void pdta_ln_fwd(const void*, void*, const float*, const float*, float*,
+                 float*, long, int, float, int, hipStream_t);
+int pdta_ln_chunks(long);
+void pdta_ln_bwd(const void*, const void*, void*, const float*,
+                 const float*, const float*, float*, float*, float*, long,
+                 int, int, hipStream_t);
+void pdta_gelu_fwd(const void*, void*, long, int, hipStream_t);
+void pdta_gelu_bwd(const void*, const void*, void*, long, int, hipStream_t);
+void pdta_softmax_fwd(const float*, void*, long, int, int, float, int,
+                      hipStream_t);
+void pdta_softmax_bwd(const void*, const void*, float*, long, int, float,
+                      int, hipStream_t);
 }
 
 namespace {
@@ -399,6 +413,99 @@ at::Tensor conv2d_bwd_weight(at::Tensor x, at::Tensor dy, int64_t R,
       .contiguous();
 }
 
+// ---- transformer ops (ViT path) -----------------------------------------
+at::Tensor gemm_nt_batched(at::Tensor a, at::Tensor b, bool out_f32) {
+  // a [bat, M, K], b [bat, N, K] -> [bat, M, N]
+  CHECK_IN(a);
+  CHECK_IN(b);
+  int bat = a.size(0), M = a.size(1), K = a.size(2), N = b.size(1);
+  TORCH_CHECK(b.size(0) == bat && b.size(2) == K && K % 64 == 0,
+              "gemm_nt_batched shape");
+  auto y = at::empty({bat, M, N},
+                     out_f32 ? a.options().dtype(at::kFloat) : a.options());
+  pdta_gemm_nt_batched(a.data_ptr(), b.data_ptr(), y.data_ptr(), M, N, K,
+                       bat, (long)M * K, (long)N * K, (long)M * N, dt_of(a),
+                       out_f32 ? 1 : 0, cur_stream());
+  return y;
+}
+
+std::vector<at::Tensor> ln_fwd(at::Tensor x, at::Tensor w, at::Tensor b,
+                               double eps) {
+  CHECK_IN(x);
+  int d = x.size(-1);
+  long rows = x.numel() / d;
+  auto y = at::empty_like(x);
+  auto opts = x.options().dtype(at::kFloat);
+  auto mean = at::empty({rows}, opts), invstd = at::empty({rows}, opts);
+  auto wf = w.to(at::kFloat).contiguous(), bf = b.to(at::kFloat).contiguous();
+  pdta_ln_fwd(x.data_ptr(), y.data_ptr(), wf.data_ptr<float>(),
+              bf.data_ptr<float>(), mean.data_ptr<float>(),
+              invstd.data_ptr<float>(), rows, d, eps, dt_of(x),
+              cur_stream());
+  return {y, mean, invstd};
+}
+
+std::vector<at::Tensor> ln_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
+                               at::Tensor mean, at::Tensor invstd) {
+  CHECK_IN(x);
+  CHECK_IN(dy);
+  int d = x.size(-1);
+  long rows = x.numel() / d;
+  auto dx = at::empty_like(x);
+  auto opts = x.options().dtype(at::kFloat);
+  int chunks = pdta_ln_chunks(rows);
+  auto part = at::empty({(long)chunks * 2 * d}, opts);
+  auto dw = at::empty({d}, opts), db = at::empty({d}, opts);
+  auto wf = w.to(at::kFloat).contiguous();
+  pdta_ln_bwd(x.data_ptr(), dy.data_ptr(), dx.data_ptr(),
+              wf.data_ptr<float>(), mean.data_ptr<float>(),
+              invstd.data_ptr<float>(), part.data_ptr<float>(),
+              dw.data_ptr<float>(), db.data_ptr<float>(), rows, d, dt_of(x),
+              cur_stream());
+  return {dx, dw, db};
+}
+
+at::Tensor gelu_fwd(at::Tensor x) {
+  CHECK_IN(x);
+  auto y = at::empty_like(x);
+  pdta_gelu_fwd(x.data_ptr(), y.data_ptr(), x.numel(), dt_of(x),
+                cur_stream());
+  return y;
+}
+
+at::Tensor gelu_bwd(at::Tensor x, at::Tensor dy) {
+  CHECK_IN(x);
+  CHECK_IN(dy);
+  auto dx = at::empty_like(x);
+  pdta_gelu_bwd(x.data_ptr(), dy.data_ptr(), dx.data_ptr(), x.numel(),
+                dt_of(x), cur_stream());
+  return dx;
+}
+
+at::Tensor attn_softmax_fwd(at::Tensor scores, int64_t nvalid, double scale,
+                            at::ScalarType out_dtype) {
+  CHECK_IN(scores);  // fp32 [.., n]
+  int n = scores.size(-1);
+  long rows = scores.numel() / n;
+  auto p = at::empty(scores.sizes(), scores.options().dtype(out_dtype));
+  pdta_softmax_fwd(scores.data_ptr<float>(), p.data_ptr(), rows, n, nvalid,
+                   scale, out_dtype == at::kBFloat16 ? 1 : 2, cur_stream());
+  return p;
+}
+
+at::Tensor attn_softmax_bwd(at::Tensor probs, at::Tensor dprobs,
+                            double scale) {
+  CHECK_IN(probs);
+  CHECK_IN(dprobs);
+  int n = probs.size(-1);
+  long rows = probs.numel() / n;
+  auto ds = at::empty(probs.sizes(), probs.options().dtype(at::kFloat));
+  pdta_softmax_bwd(probs.data_ptr(), dprobs.data_ptr(),
+                   ds.data_ptr<float>(), rows, n, scale, dt_of(probs),
+                   cur_stream());
+  return ds;
+}
+
 // ---- RCCL ----------------------------------------------------------------
 #define RCCL_CHECK(cmd)                                                  \
   do {                                                                   \
@@ -496,6 +603,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd", &conv2d_fwd);
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
   m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
+  m.def("gemm_nt_batched", &gemm_nt_batched, py::arg("a"), py::arg("b"),
+        py::arg("out_f32") = false);
+  m.def("ln_fwd", &ln_fwd);
+  m.def("ln_bwd", &ln_bwd);
+  m.def("gelu_fwd", &gelu_fwd);
+  m.def("gelu_bwd", &gelu_bwd);
+  m.def("attn_softmax_fwd", &attn_softmax_fwd);
+  m.def("attn_softmax_bwd", &attn_softmax_bwd);
   m.def("rccl_unique_id", &rccl_unique_id);
   m.def("rccl_comm_init", &rccl_comm_init);
   m.def("rccl_allreduce_avg", &rccl_allreduce_avg);

@@ -48,7 +48,13 @@ DEV_INLINE void stage_tile(const T* __restrict__ src, long ld,
 template <typename T, typename OutT, bool BIAS>
 __global__ __launch_bounds__(TPB) void gemm_nt_k(
     const T* __restrict__ A, const T* __restrict__ B, OutT* __restrict__ C,
-    const float* __restrict__ bias, int M, int N, int K, int tiles_n) {
+    const float* __restrict__ bias, int M, int N, int K, int tiles_n,
+    long batch_sa, long batch_sb, long batch_sc) {
+  // blockIdx.y = batch index (strided batched GEMM; attention uses
+  // batch = B*heads with per-head strides)
+  A += (long)blockIdx.y * batch_sa;
+  B += (long)blockIdx.y * batch_sb;
+  C += (long)blockIdx.y * batch_sc;
   __shared__ __attribute__((aligned(16))) char lds[2][2][BM * BK * 2];
 
   int bid = xcd_swizzle(blockIdx.x, gridDim.x);
@@ -149,18 +155,40 @@ void pdta_gemm_nt(const void* A, const void* B, void* C, const float* bias,
     if (out_f32) {
       if (bias)
         gemm_nt_k<T, float, true><<<grid, TPB, 0, s>>>(
-            (const T*)A, (const T*)B, (float*)C, bias, M, N, K, tiles_n);
+            (const T*)A, (const T*)B, (float*)C, bias, M, N, K, tiles_n,
+            0, 0, 0);
       else
         gemm_nt_k<T, float, false><<<grid, TPB, 0, s>>>(
-            (const T*)A, (const T*)B, (float*)C, nullptr, M, N, K, tiles_n);
+            (const T*)A, (const T*)B, (float*)C, nullptr, M, N, K, tiles_n,
+            0, 0, 0);
     } else {
       if (bias)
         gemm_nt_k<T, T, true><<<grid, TPB, 0, s>>>(
-            (const T*)A, (const T*)B, (T*)C, bias, M, N, K, tiles_n);
+            (const T*)A, (const T*)B, (T*)C, bias, M, N, K, tiles_n, 0, 0,
+            0);
       else
         gemm_nt_k<T, T, false><<<grid, TPB, 0, s>>>(
-            (const T*)A, (const T*)B, (T*)C, nullptr, M, N, K, tiles_n);
+            (const T*)A, (const T*)B, (T*)C, nullptr, M, N, K, tiles_n, 0,
+            0, 0);
     }
+  });
+}
+
+// Batched C[b] = A[b] B[b]^T; per-batch element strides. out_f32 option.
+void pdta_gemm_nt_batched(const void* A, const void* B, void* C, int M,
+                          int N, int K, int batches, long sa, long sb,
+                          long sc, int dt, int out_f32, hipStream_t s) {
+  int tiles_m = (M + BM - 1) / BM, tiles_n = (N + BN - 1) / BN;
+  dim3 grid(tiles_m * tiles_n, batches);
+  PDTA_DISPATCH_DTYPE16(dt, T, {
+    if (out_f32)
+      gemm_nt_k<T, float, false><<<grid, TPB, 0, s>>>(
+          (const T*)A, (const T*)B, (float*)C, nullptr, M, N, K, tiles_n,
+          sa, sb, sc);
+    else
+      gemm_nt_k<T, T, false><<<grid, TPB, 0, s>>>(
+          (const T*)A, (const T*)B, (T*)C, nullptr, M, N, K, tiles_n, sa,
+          sb, sc);
   });
 }
 

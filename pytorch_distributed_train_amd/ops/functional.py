@@ -435,3 +435,129 @@ class SoftmaxCrossEntropy(torch.autograd.Function):
 
 def softmax_cross_entropy(logits, labels, label_smoothing: float = 0.0):
     return SoftmaxCrossEntropy.apply(logits, labels, label_smoothing)
+
+
+# ---------------------------------------------------------------------------
+# Transformer ops (ViT path: LayerNorm, GELU, attention)
+# ---------------------------------------------------------------------------
+
+class LayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps: float):
+        if x.is_cuda:
+            y, mean, invstd = ext().ln_fwd(x, weight, bias, eps)
+        else:
+            x32 = x.float()
+            mean = x32.mean(dim=-1)
+            var = x32.var(dim=-1, unbiased=False)
+            invstd = (var + eps).rsqrt()
+            y = ((x32 - mean.unsqueeze(-1)) * invstd.unsqueeze(-1)
+                 * weight.float() + bias.float()).to(x.dtype)
+        ctx.save_for_backward(x, weight, mean, invstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, mean, invstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        if x.is_cuda:
+            dx, dw, db = ext().ln_bwd(x, dy, weight, mean, invstd)
+        else:
+            d = x.shape[-1]
+            x32, g32, w32 = x.float(), dy.float(), weight.float()
+            xhat = (x32 - mean.unsqueeze(-1)) * invstd.unsqueeze(-1)
+            gw = g32 * w32
+            s1 = gw.mean(dim=-1, keepdim=True)
+            s2 = (gw * xhat).mean(dim=-1, keepdim=True)
+            dx = (invstd.unsqueeze(-1) * (gw - s1 - xhat * s2)).to(x.dtype)
+            dims = tuple(range(x.dim() - 1))
+            dw = (g32 * xhat).sum(dim=dims)
+            db = g32.sum(dim=dims)
+        return dx, dw, db, None
+
+
+def layer_norm(x, weight, bias, eps: float = 1e-6):
+    return LayerNormFn.apply(x, weight, bias, eps)
+
+
+class GELUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.save_for_backward(x)
+        if x.is_cuda:
+            return ext().gelu_fwd(x)
+        return F.gelu(x.float()).to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x,) = ctx.saved_tensors
+        dy = dy.contiguous()
+        if x.is_cuda:
+            return ext().gelu_bwd(x, dy)
+        x32 = x.float().detach().requires_grad_()
+        with torch.enable_grad():
+            y = F.gelu(x32)
+        (dx,) = torch.autograd.grad(y, x32, dy.float())
+        return dx.to(x.dtype)
+
+
+def gelu(x):
+    return GELUFn.apply(x)
+
+
+class AttentionFn(torch.autograd.Function):
+    """Scaled-dot-product attention over padded sequences.
+
+    q,k,v: [bat, S, hd] (S padded to a multiple of 64; hd % 64 == 0);
+    columns >= nvalid are masked out of the softmax, padded v rows are
+    zero, so results for rows < nvalid are exact. GPU path: batched MFMA
+    GEMMs (QK^T fp32) + fused masked-softmax kernel (SURVEY.md §2d ViT
+    attention row).
+    """
+
+    @staticmethod
+    def forward(ctx, q, k, v, nvalid: int, scale: float):
+        if q.is_cuda:
+            scores = ext().gemm_nt_batched(q, k, True)        # fp32 [b,S,S]
+            p = ext().attn_softmax_fwd(scores, nvalid, scale, q.dtype)
+            vt = v.transpose(1, 2).contiguous()               # [b, hd, S]
+            out = ext().gemm_nt_batched(p, vt)                # [b, S, hd]
+        else:
+            s32 = (q.float() @ k.float().transpose(1, 2)) * scale
+            mask = torch.arange(q.shape[1]) >= nvalid
+            s32 = s32.masked_fill(mask.view(1, 1, -1), float("-inf"))
+            p32 = torch.softmax(s32, dim=-1)
+            p = p32.to(q.dtype)
+            out = (p32 @ v.float()).to(q.dtype)
+        ctx.save_for_backward(q, k, v, p)
+        ctx.scale = scale
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, p = ctx.saved_tensors
+        dout = dout.contiguous()
+        scale = ctx.scale
+        if q.is_cuda:
+            dp = ext().gemm_nt_batched(dout, v)               # [b, S, S]
+            ds = ext().attn_softmax_bwd(p, dp, scale).to(q.dtype)
+            kt = k.transpose(1, 2).contiguous()               # [b, hd, S]
+            dq = ext().gemm_nt_batched(ds, kt)                # [b, S, hd]
+            dst = ds.transpose(1, 2).contiguous()
+            qt = q.transpose(1, 2).contiguous()
+            dk = ext().gemm_nt_batched(dst, qt)
+            pt = p.transpose(1, 2).contiguous()
+            dot = dout.transpose(1, 2).contiguous()
+            dv = ext().gemm_nt_batched(pt, dot)
+        else:
+            p32, g32 = p.float(), dout.float()
+            dp = g32 @ v.float().transpose(1, 2)
+            ds = (p32 * (dp - (p32 * dp).sum(-1, keepdim=True)) * scale)
+            dq = (ds @ k.float()).to(q.dtype)
+            dk = (ds.transpose(1, 2) @ q.float()).to(q.dtype)
+            dv = (p32.transpose(1, 2) @ g32).to(q.dtype)
+        return dq, dk, dv, None, None
+
+
+def attention(q, k, v, nvalid: int, scale: float):
+    return AttentionFn.apply(q, k, v, nvalid, scale)
