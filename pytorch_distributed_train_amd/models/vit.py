@@ -136,7 +136,7 @@ class VisionTransformer(nn.Module):
         cls = self.class_token.to(x.dtype).expand(b, -1, -1)
         x = torch.cat([cls, x], dim=1)             # [B, 197, 768]
         x = self.encoder(x)                        # [B, 256, 768] (padded)
-        return self.heads(x[:, 0])                 # class token -> logits
+        return self.heads(x[:, 0].contiguous())    # class token -> logits
 
 
 def vit_b_16(num_classes=1000, compute_dtype=None):
