@@ -97,7 +97,9 @@ def main(argv=None) -> int:
         ms = t_max / b.steps * 1000.0
         total_images = b.batch_size * n_gpus * b.steps
         print(json.dumps({
-            "metric": "images/sec (whole node) ResNet-50 ImageNet-shape",
+            "metric": ("images/sec (whole node) ResNet-50 ImageNet-shape"
+                       if b.model == "resnet50" else
+                       f"images/sec (whole node) {b.model} ImageNet-shape"),
             "value": round(total_images / t_max, 2),
             "unit": "images/sec",
             "n_gpus": n_gpus,

@@ -13,6 +13,7 @@ from __future__ import annotations
 import time
 
 import torch
+from torch.profiler import record_function
 
 from .flags import DistEnv
 from .models import build_model
@@ -23,6 +24,7 @@ from .parallel.flat import FlatParams
 from .parallel.optim import FusedSGD
 from .parallel.sync_bn import convert_sync_batchnorm
 from .utils.amp import GradScaler
+from .utils.watchdog import Watchdog
 
 _DTYPES = {"fp32": torch.float32, "bf16": torch.bfloat16,
            "fp16": torch.float16}
@@ -72,6 +74,9 @@ class Trainer:
         self.epoch = 0
         self.global_step = 0
         self._fault = self._parse_fault(args.fault_inject)
+        self.watchdog = Watchdog(getattr(args, "watchdog_sec", 0.0),
+                                 rank=self.env.rank)
+        self.watchdog.start()
 
     @staticmethod
     def _parse_fault(spec: str):
@@ -89,15 +94,20 @@ class Trainer:
                 f"fault injection: step {self.global_step} rank {self.env.rank}")
         self.optimizer.zero_grad()
         self.reducer.prepare()
-        logits = self.model(images)
-        loss = OF.softmax_cross_entropy(logits, labels,
-                                        self.args.label_smoothing)
-        self.scaler.scale(loss).backward()
-        self.reducer.finalize()
+        with record_function("pdta.forward"):
+            logits = self.model(images)
+            loss = OF.softmax_cross_entropy(logits, labels,
+                                            self.args.label_smoothing)
+        with record_function("pdta.backward"):
+            self.scaler.scale(loss).backward()
+        with record_function("pdta.reduce_finalize"):
+            self.reducer.finalize()
         skip = self.scaler.unscale_and_check(self.flat.grad, self.comm)
         if not skip:
-            self.optimizer.step()
+            with record_function("pdta.optimizer"):
+                self.optimizer.step()
         self.global_step += 1
+        self.watchdog.heartbeat()
         return loss.detach()
 
     def state(self):

@@ -64,6 +64,14 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--label-smoothing", type=float, default=0.0)
     p.add_argument("--amp", action="store_true",
                    help="fp16 loss-scaling path (GradScaler semantics)")
+    p.add_argument("--eval-freq", type=int, default=0,
+                   help="evaluate top-1/top-5 every N epochs (0 = never)")
+    p.add_argument("--eval-batches", type=int, default=10,
+                   help="cap eval batches (synthetic data)")
+    p.add_argument("--watchdog-sec", type=float,
+                   default=float(os.environ.get("PDTA_WATCHDOG_SEC", "0")),
+                   help="abort + dump flight recorder if no training "
+                        "progress for this long (0 = off)")
     p.add_argument("--fault-inject", default="",
                    help="step:N:rank:R -> raise at step N on rank R (testing)")
     return p

@@ -18,6 +18,7 @@ import torch
 import torch.distributed as torch_dist
 
 from ..flags import DistEnv
+from ..utils.watchdog import record_collective
 
 
 class Work:
@@ -96,11 +97,15 @@ class TorchDistComm:
         self.world_size = torch_dist.get_world_size()
 
     def allreduce_sum(self, tensor, stream=None) -> Work:
+        record_collective("gloo.allreduce_sum",
+                          tensor.numel() * tensor.element_size())
         w = torch_dist.all_reduce(tensor, op=torch_dist.ReduceOp.SUM,
                                   async_op=True)
         return _DistWork(w)
 
     def allreduce_avg(self, tensor, stream=None) -> Work:
+        record_collective("gloo.allreduce_avg",
+                          tensor.numel() * tensor.element_size())
         w = torch_dist.all_reduce(tensor, op=torch_dist.ReduceOp.SUM,
                                   async_op=True)
         ws = self.world_size
@@ -148,6 +153,8 @@ class RcclComm:
         return s.cuda_stream
 
     def _run(self, fn, tensor, stream, *args) -> Work:
+        record_collective(fn.__name__,
+                          tensor.numel() * tensor.element_size())
         fn(tensor, *args, self._comm, self._stream_ptr(stream))
         ev = torch.cuda.Event()
         ev.record(stream if stream is not None

@@ -30,6 +30,31 @@ from pytorch_distributed_train_amd.utils.checkpoint import (load_checkpoint,
 from pytorch_distributed_train_amd.utils.metrics import RankLogger, StepTimer
 
 
+def evaluate(trainer, dataset, env, batch_size, max_batches=0):
+    """Top-1/top-5 on a synthetic validation split (rank-local; averaged
+    across ranks by the caller if distributed)."""
+    from pytorch_distributed_train_amd.data import DistributedSampler, Loader
+    sampler = DistributedSampler(dataset, env.world_size, env.rank,
+                                 shuffle=False)
+    loader = Loader(dataset, sampler, batch_size)
+    trainer.model.eval()
+    top1 = top5 = total = 0
+    with torch.no_grad():
+        for i, (images, labels) in enumerate(loader):
+            if max_batches and i >= max_batches:
+                break
+            images = images.to(trainer.device)
+            labels = labels.to(trainer.device)
+            logits = trainer.model(images).float()
+            _, pred = logits.topk(5, dim=1)
+            correct = pred.eq(labels.view(-1, 1))
+            top1 += int(correct[:, 0].sum())
+            top5 += int(correct.any(dim=1).sum())
+            total += labels.numel()
+    trainer.model.train()
+    return (top1 / max(total, 1), top5 / max(total, 1))
+
+
 def main(argv=None):
     args = build_parser().parse_args(argv)
     env = DistEnv.from_env(args.local_rank)
@@ -76,6 +101,10 @@ def main(argv=None):
             if step % args.print_freq == 0:
                 logger.log(epoch=epoch, step=step, loss=float(loss),
                            images_per_sec=round(ips, 1), lr=args.lr)
+        if args.eval_freq and (epoch + 1) % args.eval_freq == 0:
+            t1, t5 = evaluate(trainer, dataset, env, args.batch_size,
+                              max_batches=args.eval_batches)
+            logger.log(epoch=epoch, top1=round(t1, 4), top5=round(t5, 4))
         if ckpt_path and env.rank == 0 and (
                 epoch == args.epochs - 1 or
                 (args.checkpoint_freq and
