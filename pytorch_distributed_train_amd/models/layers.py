@@ -82,7 +82,8 @@ class BatchNorm2dNHWC(nn.Module):
             mean, var = OF.bn_stats(x)
         return mean, var, self._batch_count(x)
 
-    def forward(self, x):
+    def forward(self, x, residual=None):
+        # residual: fused ResNet join y = relu(bn(x) + residual)
         if self.training:
             if x.is_cuda and self.comm is None:
                 # fast path: stats + running update in one kernel pair
@@ -92,7 +93,7 @@ class BatchNorm2dNHWC(nn.Module):
                 with torch.no_grad():
                     self.num_batches_tracked += 1
                 return OF.bn_train(x, self.weight, self.bias, mean, invstd,
-                                   self.fuse_relu)
+                                   self.fuse_relu, residual=residual)
             mean, var, count = self._compute_stats(x)
             with torch.no_grad():
                 self.num_batches_tracked += 1
@@ -104,9 +105,11 @@ class BatchNorm2dNHWC(nn.Module):
             invstd = (var + self.eps).rsqrt()
             return OF.bn_train(x, self.weight, self.bias, mean, invstd,
                                self.fuse_relu, self.comm,
-                               count if self.comm is not None else None)
+                               count if self.comm is not None else None,
+                               residual)
         return OF.bn_infer(x, self.weight, self.bias, self.running_mean,
-                           self.running_var, self.eps, self.fuse_relu)
+                           self.running_var, self.eps, self.fuse_relu,
+                           residual)
 
     def extra_repr(self):
         return (f"{self.num_features}, eps={self.eps}, "

@@ -39,16 +39,17 @@ class BasicBlock(nn.Module):
         self.relu = ReLUNHWC()  # kept for name parity; fused in practice
         self.conv2 = conv3x3(planes, planes)
         self.bn2 = BatchNorm2dNHWC(planes)
+        self.bn2.fuse_relu = True  # fused with the residual join
         self.downsample = downsample
         self.stride = stride
 
     def forward(self, x):
         identity = x
         out = self.bn1(self.conv1(x))  # ReLU fused into bn1
-        out = self.bn2(self.conv2(out))
         if self.downsample is not None:
             identity = self.downsample(x)
-        return OF.add_relu(out, identity)
+        # bn2 + residual add + ReLU in one kernel
+        return self.bn2(self.conv2(out), identity)
 
 
 class Bottleneck(nn.Module):
@@ -64,6 +65,7 @@ class Bottleneck(nn.Module):
         self.bn2.fuse_relu = True
         self.conv3 = conv1x1(planes, planes * self.expansion)
         self.bn3 = BatchNorm2dNHWC(planes * self.expansion)
+        self.bn3.fuse_relu = True  # fused with the residual join
         self.relu = ReLUNHWC()
         self.downsample = downsample
         self.stride = stride
@@ -72,10 +74,10 @@ class Bottleneck(nn.Module):
         identity = x
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
-        out = self.bn3(self.conv3(out))
         if self.downsample is not None:
             identity = self.downsample(x)
-        return OF.add_relu(out, identity)
+        # bn3 + residual add + ReLU in one kernel
+        return self.bn3(self.conv3(out), identity)
 
 
 class ResNet(nn.Module):

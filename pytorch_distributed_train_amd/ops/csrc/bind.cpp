@@ -18,14 +18,14 @@ void pdta_add_relu(const void*, const void*, void*, long, int, hipStream_t);
 void pdta_bn_stats(const void*, float*, float*, float*, float*, float*,
                    float, float, int, long, int, int, hipStream_t);
 int pdta_bn_chunks(long, int);
-void pdta_bn_apply(const void*, void*, const float*, const float*,
-                   const float*, const float*, long, int, int, int,
-                   hipStream_t);
+void pdta_bn_apply(const void*, void*, const void*, const float*,
+                   const float*, const float*, const float*, long, int, int,
+                   int, hipStream_t);
 void pdta_bn_bwd_reduce(const void*, const void*, const void*,
                         const float*, float*, float*, float*, long, int,
                         int, hipStream_t);
 void pdta_bn_bwd_elemt(const void*, const void*, const void*, void*,
-                       const float*, const float*, const float*,
+                       void*, const float*, const float*, const float*,
                        const float*, const float*, float, long, int, int,
                        hipStream_t);
 void pdta_maxpool_fwd(const void*, void*, uint8_t*, int, int, int, int, int,
@@ -165,7 +165,8 @@ std::vector<at::Tensor> bn_stats_train(at::Tensor x, at::Tensor running_mean,
 }
 
 at::Tensor bn_apply(at::Tensor x, at::Tensor weight, at::Tensor bias,
-                    at::Tensor mean, at::Tensor invstd, bool relu) {
+                    at::Tensor mean, at::Tensor invstd, bool relu,
+                    c10::optional<at::Tensor> res) {
   CHECK_IN(x);
   int C = x.size(3);
   long rows = x.numel() / C;
@@ -173,10 +174,11 @@ at::Tensor bn_apply(at::Tensor x, at::Tensor weight, at::Tensor bias,
   auto w = weight.to(at::kFloat).contiguous();
   auto b = bias.to(at::kFloat).contiguous();
   auto m = mean.contiguous(), is = invstd.contiguous();
-  pdta_bn_apply(x.data_ptr(), y.data_ptr(), w.data_ptr<float>(),
-                b.data_ptr<float>(), m.data_ptr<float>(),
-                is.data_ptr<float>(), rows, C, relu ? 1 : 0, dt_of(x),
-                cur_stream());
+  pdta_bn_apply(x.data_ptr(), y.data_ptr(),
+                res.has_value() ? res->data_ptr() : nullptr,
+                w.data_ptr<float>(), b.data_ptr<float>(),
+                m.data_ptr<float>(), is.data_ptr<float>(), rows, C,
+                relu ? 1 : 0, dt_of(x), cur_stream());
   return y;
 }
 
@@ -200,26 +202,31 @@ std::vector<at::Tensor> bn_bwd_reduce(at::Tensor x, at::Tensor dy,
   return {sum_dy, sum_dy_xmu};
 }
 
-at::Tensor bn_bwd_elemt(at::Tensor x, at::Tensor dy,
-                        c10::optional<at::Tensor> relu_y, at::Tensor mean,
-                        at::Tensor invstd, at::Tensor weight,
-                        at::Tensor sum_dy, at::Tensor sum_dy_xmu,
-                        double count) {
+std::vector<at::Tensor> bn_bwd_elemt(at::Tensor x, at::Tensor dy,
+                                     c10::optional<at::Tensor> relu_y,
+                                     at::Tensor mean, at::Tensor invstd,
+                                     at::Tensor weight, at::Tensor sum_dy,
+                                     at::Tensor sum_dy_xmu, double count,
+                                     bool want_dres) {
   CHECK_IN(x);
   CHECK_IN(dy);
   int C = x.size(3);
   long rows = x.numel() / C;
   auto dx = at::empty_like(x);
+  at::Tensor dres;
+  if (want_dres) dres = at::empty_like(x);
   auto w = weight.to(at::kFloat).contiguous();
   pdta_bn_bwd_elemt(x.data_ptr(), dy.data_ptr(),
                     relu_y.has_value() ? relu_y->data_ptr() : nullptr,
-                    dx.data_ptr(), w.data_ptr<float>(),
+                    dx.data_ptr(), want_dres ? dres.data_ptr() : nullptr,
+                    w.data_ptr<float>(),
                     mean.contiguous().data_ptr<float>(),
                     invstd.contiguous().data_ptr<float>(),
                     sum_dy.contiguous().data_ptr<float>(),
                     sum_dy_xmu.contiguous().data_ptr<float>(), (float)count,
                     rows, C, dt_of(x), cur_stream());
-  return dx;
+  if (want_dres) return {dx, dres};
+  return {dx};
 }
 
 // ---- pooling -------------------------------------------------------------
@@ -586,9 +593,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("add_relu_fwd", &add_relu_fwd);
   m.def("bn_stats", &bn_stats);
   m.def("bn_stats_train", &bn_stats_train);
-  m.def("bn_apply", &bn_apply);
+  m.def("bn_apply", &bn_apply, py::arg("x"), py::arg("weight"),
+        py::arg("bias"), py::arg("mean"), py::arg("invstd"),
+        py::arg("relu"), py::arg("res") = c10::nullopt);
   m.def("bn_bwd_reduce", &bn_bwd_reduce);
-  m.def("bn_bwd_elemt", &bn_bwd_elemt);
+  m.def("bn_bwd_elemt", &bn_bwd_elemt, py::arg("x"), py::arg("dy"),
+        py::arg("relu_y"), py::arg("mean"), py::arg("invstd"),
+        py::arg("weight"), py::arg("sum_dy"), py::arg("sum_dy_xmu"),
+        py::arg("count"), py::arg("want_dres") = false);
   m.def("maxpool_fwd", &maxpool_fwd);
   m.def("maxpool_bwd", &maxpool_bwd);
   m.def("avgpool_global_fwd", &avgpool_global_fwd);

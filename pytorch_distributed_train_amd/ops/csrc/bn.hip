@@ -102,8 +102,9 @@ __global__ void bn_stats_final_k(const float* __restrict__ part, int chunks,
 }
 
 // -- y = (x - mean) * invstd * w + b, optional ReLU ----------------------
-template <typename T, bool RELU>
+template <typename T, bool RELU, bool RES>
 __global__ void bn_apply_k(const T* __restrict__ x, T* __restrict__ y,
+                           const T* __restrict__ res,
                            const float* __restrict__ weight,
                            const float* __restrict__ bias,
                            const float* __restrict__ mean,
@@ -126,9 +127,12 @@ __global__ void bn_apply_k(const T* __restrict__ x, T* __restrict__ y,
   for (long r = (long)blockIdx.y * rpar + threadIdx.x / cpg; r < rows;
        r += rstride) {
     Vec8<T> v = load8(x + r * C + c0);
+    Vec8<T> rv;
+    if (RES) rv = load8(res + r * C + c0);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float o = to_f32(v.v[j]) * scale[j] + shift[j];
+      if (RES) o += to_f32(rv.v[j]);
       if (RELU) o = fmaxf(o, 0.f);
       v.v[j] = from_f32<T>(o);
     }
@@ -207,11 +211,11 @@ __global__ void bn_bwd_final_k(const float* __restrict__ part, int chunks,
 
 // -- backward elementwise (ReLU mask fused when RELU):
 //    dx = w*invstd * (dy - sum_dy/n - (x-mean)*invstd^2*sum_dy_xmu/n)
-template <typename T, bool RELU>
+template <typename T, bool RELU, bool DRES>
 __global__ void bn_bwd_elemt_k(const T* __restrict__ x,
                                const T* __restrict__ dy,
                                const T* __restrict__ relu_y,
-                               T* __restrict__ dx,
+                               T* __restrict__ dx, T* __restrict__ dres,
                                const float* __restrict__ weight,
                                const float* __restrict__ mean,
                                const float* __restrict__ invstd,
@@ -238,15 +242,17 @@ __global__ void bn_bwd_elemt_k(const T* __restrict__ x,
        r += rstride) {
     Vec8<T> g = load8(dy + r * C + c0);
     Vec8<T> v = load8(x + r * C + c0);
-    Vec8<T> yv;
+    Vec8<T> yv, dr;
     if (RELU) yv = load8(relu_y + r * C + c0);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float gg = to_f32(g.v[j]);
       if (RELU) gg = to_f32(yv.v[j]) > 0.f ? gg : 0.f;
+      if (DRES) dr.v[j] = from_f32<T>(gg);
       float o = k1[j] * (gg - k2[j] - (to_f32(v.v[j]) - mu[j]) * k3[j]);
       g.v[j] = from_f32<T>(o);
     }
+    if (DRES) store8(dres + r * C + c0, dr);
     store8(dx + r * C + c0, g);
   }
 }
@@ -329,17 +335,31 @@ int pdta_bn_chunks(long rows, int C) {
   return chunks + 64;  // +64 rows of mid-reduce scratch at the tail
 }
 
-void pdta_bn_apply(const void* x, void* y, const float* weight,
-                   const float* bias, const float* mean, const float* invstd,
-                   long rows, int C, int relu, int dt, hipStream_t s) {
+void pdta_bn_apply(const void* x, void* y, const void* res,
+                   const float* weight, const float* bias, const float* mean,
+                   const float* invstd, long rows, int C, int relu, int dt,
+                   hipStream_t s) {
   dim3 g = ew_grid2(rows, C);
   PDTA_DISPATCH_DTYPE(dt, T, {
-    if (relu)
-      bn_apply_k<T, true><<<g, TPB, 0, s>>>((const T*)x, (T*)y, weight, bias,
-                                            mean, invstd, rows, C / 8);
-    else
-      bn_apply_k<T, false><<<g, TPB, 0, s>>>((const T*)x, (T*)y, weight,
-                                             bias, mean, invstd, rows, C / 8);
+    if (res) {
+      if (relu)
+        bn_apply_k<T, true, true><<<g, TPB, 0, s>>>(
+            (const T*)x, (T*)y, (const T*)res, weight, bias, mean, invstd,
+            rows, C / 8);
+      else
+        bn_apply_k<T, false, true><<<g, TPB, 0, s>>>(
+            (const T*)x, (T*)y, (const T*)res, weight, bias, mean, invstd,
+            rows, C / 8);
+    } else {
+      if (relu)
+        bn_apply_k<T, true, false><<<g, TPB, 0, s>>>(
+            (const T*)x, (T*)y, nullptr, weight, bias, mean, invstd, rows,
+            C / 8);
+      else
+        bn_apply_k<T, false, false><<<g, TPB, 0, s>>>(
+            (const T*)x, (T*)y, nullptr, weight, bias, mean, invstd, rows,
+            C / 8);
+    }
   });
 }
 
@@ -373,20 +393,29 @@ void pdta_bn_bwd_reduce(const void* x, const void* dy, const void* relu_y,
 }
 
 void pdta_bn_bwd_elemt(const void* x, const void* dy, const void* relu_y,
-                       void* dx, const float* weight, const float* mean,
-                       const float* invstd, const float* sum_dy,
-                       const float* sum_dy_xmu, float count, long rows,
-                       int C, int dt, hipStream_t s) {
+                       void* dx, void* dres, const float* weight,
+                       const float* mean, const float* invstd,
+                       const float* sum_dy, const float* sum_dy_xmu,
+                       float count, long rows, int C, int dt,
+                       hipStream_t s) {
   dim3 g = ew_grid2(rows, C);
   PDTA_DISPATCH_DTYPE(dt, T, {
-    if (relu_y)
-      bn_bwd_elemt_k<T, true><<<g, TPB, 0, s>>>(
-          (const T*)x, (const T*)dy, (const T*)relu_y, (T*)dx, weight, mean,
+    if (relu_y) {
+      if (dres)
+        bn_bwd_elemt_k<T, true, true><<<g, TPB, 0, s>>>(
+            (const T*)x, (const T*)dy, (const T*)relu_y, (T*)dx, (T*)dres,
+            weight, mean, invstd, sum_dy, sum_dy_xmu, 1.f / count, rows,
+            C / 8);
+      else
+        bn_bwd_elemt_k<T, true, false><<<g, TPB, 0, s>>>(
+            (const T*)x, (const T*)dy, (const T*)relu_y, (T*)dx, nullptr,
+            weight, mean, invstd, sum_dy, sum_dy_xmu, 1.f / count, rows,
+            C / 8);
+    } else {
+      bn_bwd_elemt_k<T, false, false><<<g, TPB, 0, s>>>(
+          (const T*)x, (const T*)dy, nullptr, (T*)dx, nullptr, weight, mean,
           invstd, sum_dy, sum_dy_xmu, 1.f / count, rows, C / 8);
-    else
-      bn_bwd_elemt_k<T, false><<<g, TPB, 0, s>>>(
-          (const T*)x, (const T*)dy, nullptr, (T*)dx, weight, mean, invstd,
-          sum_dy, sum_dy_xmu, 1.f / count, rows, C / 8);
+    }
   });
 }
 

@@ -118,23 +118,29 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, bias, mean, invstd, fuse_relu: bool,
-                comm=None, count_override=None):
+                comm=None, count_override=None, residual=None):
         # mean/invstd are per-channel fp32, computed by the module (local
         # stats for BN, cross-rank combined for SyncBN). For SyncBN, `comm`
         # all-reduces the backward sums and `count_override` is the GLOBAL
         # element count (all_reduce of [sum_dy, sum_dy_xmu]).
+        # `residual` fuses the ResNet join: y = relu(bn(x) + residual) in
+        # one pass (requires fuse_relu).
         if x.is_cuda:
-            y = ext().bn_apply(x, weight, bias, mean, invstd, fuse_relu)
+            y = ext().bn_apply(x, weight, bias, mean, invstd, fuse_relu,
+                               residual)
         else:
             x32 = x.float()
             xhat = (x32 - mean) * invstd
             y = xhat * weight.float() + bias.float()
+            if residual is not None:
+                y = y + residual.float()
             if fuse_relu:
                 y = y.clamp_(min=0.0)
             y = y.to(x.dtype)
         ctx.fuse_relu = fuse_relu
         ctx.comm = comm
         ctx.count_override = count_override
+        ctx.has_residual = residual is not None
         ctx.save_for_backward(x, weight, mean, invstd, y)
         return y
 
@@ -146,6 +152,7 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
         n = x.shape[0] * x.shape[1] * x.shape[2]
         if ctx.count_override is not None:
             n = ctx.count_override
+        dres = None
         if x.is_cuda:
             relu_y = y if fuse_relu else None  # mask fused in the kernels
             sum_dy, sum_dy_xmu = ext().bn_bwd_reduce(x, dy, relu_y, mean)
@@ -158,9 +165,13 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
                 ctx.comm.allreduce_sum(packed).sync_stream()
                 c = sum_dy.numel()
                 sum_dy, sum_dy_xmu = packed[:c], packed[c:]
-            dx = ext().bn_bwd_elemt(x, dy, relu_y, mean, invstd, weight,
-                                    sum_dy.contiguous(),
-                                    sum_dy_xmu.contiguous(), float(n))
+            out = ext().bn_bwd_elemt(x, dy, relu_y, mean, invstd, weight,
+                                     sum_dy.contiguous(),
+                                     sum_dy_xmu.contiguous(), float(n),
+                                     ctx.has_residual)
+            dx = out[0]
+            if ctx.has_residual:
+                dres = out[1]
         else:
             dy32 = dy.float()
             if fuse_relu:
@@ -177,9 +188,11 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
                 c = sum_dy.numel()
                 sum_dy, sum_dy_xmu = packed[:c], packed[c:]
             w32 = weight.float()
+            if ctx.has_residual:
+                dres = dy32.to(x.dtype)
             dx = (dy32 - sum_dy / n - xmu * invstd * invstd * sum_dy_xmu / n)
             dx = (dx * w32 * invstd).to(x.dtype)
-        return dx, dweight, dbias, None, None, None, None, None
+        return (dx, dweight, dbias, None, None, None, None, None, dres)
 
 
 def bn_stats(x):
@@ -196,18 +209,21 @@ def bn_stats_train(x, running_mean, running_var, momentum, eps):
 
 
 def bn_train(x, weight, bias, mean, invstd, fuse_relu=False, comm=None,
-             count=None):
+             count=None, residual=None):
     return BatchNorm2dNHWCFn.apply(x, weight, bias, mean, invstd, fuse_relu,
-                                   comm, count)
+                                   comm, count, residual)
 
 
-def bn_infer(x, weight, bias, running_mean, running_var, eps, fuse_relu=False):
+def bn_infer(x, weight, bias, running_mean, running_var, eps, fuse_relu=False,
+             residual=None):
     invstd = (running_var.float() + eps).rsqrt()
     if x.is_cuda:
         return ext().bn_apply(x, weight, bias, running_mean.float(), invstd,
-                              fuse_relu)
+                              fuse_relu, residual)
     x32 = x.float()
     y = (x32 - running_mean) * invstd * weight.float() + bias.float()
+    if residual is not None:
+        y = y + residual.float()
     if fuse_relu:
         y = y.clamp_(min=0.0)
     return y.to(x.dtype)

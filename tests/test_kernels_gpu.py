@@ -159,7 +159,7 @@ def test_bn_stats_apply_bwd():
            atol=5e-2, what="sum_dy_xmu")
     cnt = float(n * h * h)
     dx = ext().bn_bwd_elemt(x, dy, None, mean, invstd, w, sum_dy,
-                            sum_dy_xmu, cnt)
+                            sum_dy_xmu, cnt)[0]
     ref_dx = (dy32 - sum_dy / cnt -
               (x32 - mean) * invstd * invstd * sum_dy_xmu / cnt) * w * invstd
     _close(dx, ref_dx, what="bn_bwd_elemt")
@@ -171,10 +171,18 @@ def test_bn_stats_apply_bwd():
            what="sum_dy masked")
     _close(sx_m, (dym * (x32 - mean)).sum(dim=(0, 1, 2)), rtol=1e-2,
            atol=5e-2, what="sum_dy_xmu masked")
-    dx_m = ext().bn_bwd_elemt(x, dy, y, mean, invstd, w, s_m, sx_m, cnt)
+    dx_m, dres = ext().bn_bwd_elemt(x, dy, y, mean, invstd, w, s_m, sx_m,
+                                    cnt, True)
     ref_dx_m = (dym - s_m / cnt -
                 (x32 - mean) * invstd * invstd * sx_m / cnt) * w * invstd
     _close(dx_m, ref_dx_m, what="bn_bwd_elemt masked")
+    _close(dres, dym, what="bn_bwd_elemt dres")
+
+    # fused residual: y = relu(bn(x) + res)
+    res = _rand(n, h, h, c, seed=31)
+    yr = ext().bn_apply(x, w, b, mean, invstd, True, res)
+    ref_yr = ((x32 - mean) * invstd * w + b + res.float()).clamp(min=0)
+    _close(yr, ref_yr, what="bn_apply+res+relu")
 
 
 def test_bn_stats_train_updates_running():
