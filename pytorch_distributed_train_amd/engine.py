@@ -71,6 +71,11 @@ class Trainer:
                                   momentum=args.momentum,
                                   weight_decay=args.weight_decay)
         self.scaler = GradScaler(enabled=(args.amp and args.dtype == "fp16"))
+        self.packs = None
+        if self.device.type == "cuda" and self.compute_dtype != torch.float32:
+            from .ops.packs import PackManager
+            self.packs = PackManager(self.model, self.flat,
+                                     self.compute_dtype, self.device)
         self.epoch = 0
         self.global_step = 0
         self._fault = self._parse_fault(args.fault_inject)
@@ -94,6 +99,8 @@ class Trainer:
                 f"fault injection: step {self.global_step} rank {self.env.rank}")
         self.optimizer.zero_grad()
         self.reducer.prepare()
+        if self.packs is not None:
+            self.packs.refresh()  # one kernel repacks every layer's weights
         with record_function("pdta.forward"):
             logits = self.model(images)
             loss = OF.softmax_cross_entropy(logits, labels,

@@ -37,10 +37,16 @@ class Conv2dNHWC(nn.Module):
 
     def forward(self, x):
         # two compute-dtype packs keep fwd AND bwd-data K-contiguous
-        # (conv.hip header): wt [K,R,S,C] for fwd, w2 [R,S,C,K] for bwd-data
+        # (conv.hip header): wt [K,R,S,C] for fwd, w2 [R,S,C,K] for
+        # bwd-data. With a PackManager (ops/packs.py) the packs are pool
+        # views refreshed by one batched kernel per step.
         w = self.weight
-        wt = w.permute(0, 2, 3, 1).to(x.dtype).contiguous()
-        w2 = w.permute(2, 3, 1, 0).to(x.dtype).contiguous()
+        wt = getattr(self, "_pack_wt", None)
+        if wt is None or wt.dtype != x.dtype:
+            wt = w.permute(0, 2, 3, 1).to(x.dtype).contiguous()
+            w2 = w.permute(2, 3, 1, 0).to(x.dtype).contiguous()
+        else:
+            w2 = self._pack_w2
         y = OF.conv2d_nhwc(x, w, wt, w2, self.stride, self.padding)
         if self.bias is not None:
             y = y + self.bias.to(y.dtype)  # broadcast over trailing C
@@ -131,8 +137,12 @@ class Linear(nn.Module):
             self.register_parameter("bias", None)
 
     def forward(self, x):
-        packed = self.weight.to(x.dtype)  # [N,K] = B^T, the MFMA GEMM layout
-        return OF.linear(x, self.weight, packed, self.bias)
+        packed = getattr(self, "_pack_w", None)
+        packed_t = getattr(self, "_pack_wt", None)
+        if packed is None or packed.dtype != x.dtype:
+            packed = self.weight.to(x.dtype)  # [N,K] = B^T (MFMA layout)
+            packed_t = None
+        return OF.linear(x, self.weight, packed, self.bias, packed_t)
 
     def extra_repr(self):
         return (f"in_features={self.in_features}, "

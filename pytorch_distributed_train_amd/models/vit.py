@@ -53,9 +53,13 @@ class MultiheadSelfAttention(nn.Module):
     def forward(self, x, nvalid: int = SEQ):
         b, s, d = x.shape
         h, hd = self.num_heads, self.head_dim
-        packed = self.in_proj_weight.to(x.dtype)
+        packed = getattr(self, "_pack_inproj", None)
+        packed_t = getattr(self, "_pack_inproj_t", None)
+        if packed is None or packed.dtype != x.dtype:
+            packed = self.in_proj_weight.to(x.dtype)
+            packed_t = None
         qkv = OF.linear(x.reshape(b * s, d), self.in_proj_weight, packed,
-                        self.in_proj_bias)
+                        self.in_proj_bias, packed_t)
         qkv = qkv.view(b, s, 3, h, hd).permute(2, 0, 3, 1, 4)
         q = qkv[0].reshape(b * h, s, hd).contiguous()
         k = qkv[1].reshape(b * h, s, hd).contiguous()

@@ -69,6 +69,8 @@ void pdta_softmax_fwd(const float*, void*, long, int, int, float, int,
                       hipStream_t);
 void pdta_softmax_bwd(const void*, const void*, float*, long, int, float,
                       int, hipStream_t);
+void pdta_pack_weights(const float*, void*, const long*, const long*, int,
+                       long, int, hipStream_t);
 }
 
 namespace {
@@ -513,6 +515,15 @@ at::Tensor attn_softmax_bwd(at::Tensor probs, at::Tensor dprobs,
   return ds;
 }
 
+void pack_weights(at::Tensor master, at::Tensor pool, at::Tensor desc,
+                  at::Tensor prefix, int64_t total_chunks) {
+  CHECK_IN(master);
+  CHECK_IN(pool);
+  pdta_pack_weights(master.data_ptr<float>(), pool.data_ptr(),
+                    desc.data_ptr<long>(), prefix.data_ptr<long>(),
+                    desc.size(0), total_chunks, dt_of(pool), cur_stream());
+}
+
 // ---- RCCL ----------------------------------------------------------------
 #define RCCL_CHECK(cmd)                                                  \
   do {                                                                   \
@@ -623,6 +634,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gelu_bwd", &gelu_bwd);
   m.def("attn_softmax_fwd", &attn_softmax_fwd);
   m.def("attn_softmax_bwd", &attn_softmax_bwd);
+  m.def("pack_weights", &pack_weights);
   m.def("rccl_unique_id", &rccl_unique_id);
   m.def("rccl_comm_init", &rccl_comm_init);
   m.def("rccl_allreduce_avg", &rccl_allreduce_avg);

@@ -359,8 +359,11 @@ class LinearFn(torch.autograd.Function):
     """y = x @ W^T + b; x [M,K] compute dtype, W master fp32 [N,K]."""
 
     @staticmethod
-    def forward(ctx, x, weight, w_packed, bias):
-        ctx.save_for_backward(x, weight, w_packed)
+    def forward(ctx, x, weight, w_packed, bias, w_packed_t=None):
+        ctx.save_for_backward(x, weight, w_packed,
+                              w_packed_t if w_packed_t is not None
+                              else w_packed)
+        ctx.has_t = w_packed_t is not None
         ctx.has_bias = bias is not None
         if x.is_cuda:
             return ext().gemm_nt(x, w_packed, bias)
@@ -371,7 +374,7 @@ class LinearFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        x, weight, w_packed = ctx.saved_tensors
+        x, weight, w_packed, w_packed_t = ctx.saved_tensors
         dy = dy.contiguous()
         dx = dw = db = None
         if x.is_cuda:
@@ -379,7 +382,8 @@ class LinearFn(torch.autograd.Function):
             # zero-padded to multiples of 64 (zero rows/cols add nothing)
             if ctx.needs_input_grad[0]:
                 dyp = _pad_cols64(dy)
-                wtp = _pad_cols64(w_packed.t().contiguous())  # [K_in, O]
+                wtp = _pad_cols64(w_packed_t if ctx.has_t
+                                  else w_packed.t().contiguous())
                 dx = ext().gemm_nt(dyp, wtp)                  # [M, K_in]
             if ctx.needs_input_grad[1]:
                 dyt = _pad_cols64(dy.t().contiguous())        # [O, M]
@@ -395,11 +399,11 @@ class LinearFn(torch.autograd.Function):
                 dw = dy32.t() @ x32
             if ctx.has_bias and ctx.needs_input_grad[3]:
                 db = dy32.sum(dim=0)
-        return dx, dw, None, db
+        return dx, dw, None, db, None
 
 
-def linear(x, weight, w_packed, bias=None):
-    return LinearFn.apply(x, weight, w_packed, bias)
+def linear(x, weight, w_packed, bias=None, w_packed_t=None):
+    return LinearFn.apply(x, weight, w_packed, bias, w_packed_t)
 
 
 # ---------------------------------------------------------------------------
