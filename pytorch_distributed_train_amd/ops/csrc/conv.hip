@@ -293,7 +293,11 @@ __global__ __launch_bounds__(TPB) void igemm_wrw_k(
     float* __restrict__ dw, const T* __restrict__ zero_page, int Nn, int H,
     int W, int C, int P, int Q, int Kout, int R, int S, int stride, int pad,
     long rows_per_chunk, int ko_tiles, int c_tiles) {
-  __shared__ __attribute__((aligned(16))) char lds[2][2][64 * 64 * 2];
+  // 3-deep staging ring with counted vmcnt + raw barrier (guide T3/T4:
+  // wrw's compute phase (8 MFMA) is much shorter than the gather's HBM
+  // latency, so two tiles stay in flight across each barrier; hipcc's
+  // __syncthreads would emit vmcnt(0) and drain them — raw s_barrier only)
+  __shared__ __attribute__((aligned(16))) char lds[3][2][64 * 64 * 2];
 
   int bid = blockIdx.x;
   int tap = bid / (ko_tiles * c_tiles);
@@ -377,38 +381,63 @@ __global__ __launch_bounds__(TPB) void igemm_wrw_k(
     }
   };
 
+  // prologue: two tiles in flight (each stage = 4 glds per wave)
   stage(r0, 0);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
-
+  if (r0 + 64 < r1) {
+    advance();
+    stage(r0 + 64, 1);
+  }
+  unsigned ldy_base[3] = {lds_addr_of(lds[0][0]), lds_addr_of(lds[1][0]),
+                          lds_addr_of(lds[2][0])};
+  unsigned lx_base[3] = {lds_addr_of(lds[0][1]), lds_addr_of(lds[1][1]),
+                         lds_addr_of(lds[2][1])};
   int cur = 0;
-  unsigned ldy_base[2] = {lds_addr_of(lds[0][0]), lds_addr_of(lds[1][0])};
-  unsigned lx_base[2] = {lds_addr_of(lds[0][1]), lds_addr_of(lds[1][1])};
   for (long rb = r0; rb < r1; rb += 64) {
-    if (rb + 64 < r1) {
-      advance();
-      stage(rb + 64, cur ^ 1);
-    }
+    // wait for THIS tile's 4 glds (<=4 others stay in flight), then a raw
+    // barrier so every wave's rows of tile `cur` have landed
+    if (rb + 64 < r1)
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    // issue ALL 16 transpose reads (both K=32 steps), wait once, then run
+    // the 8 MFMAs — the reads' latencies overlap instead of chaining
+    unsigned long long alo[2][2], ahi[2][2], blo[2][2], bhi[2][2];
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      // hardware-transposed fragment reads (ds_read_b64_tr_b16)
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+        frag_read_tr_issue(ldy_base[cur], kk * 32, wm * 32 + i * 16, lane,
+                           &alo[kk][i], &ahi[kk][i]);
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+        frag_read_tr_issue(lx_base[cur], kk * 32, wn * 32 + i * 16, lane,
+                           &blo[kk][i], &bhi[kk][i]);
+    }
+    tr_read_wait();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
       s16x8 af[2], bf[2];
 #pragma unroll
-      for (int i = 0; i < 2; ++i)
-        af[i] = frag_read_tr(ldy_base[cur], kk * 32, wm * 32 + i * 16,
-                             lane);
-#pragma unroll
-      for (int i = 0; i < 2; ++i)
-        bf[i] = frag_read_tr(lx_base[cur], kk * 32, wn * 32 + i * 16, lane);
+      for (int i = 0; i < 2; ++i) {
+        af[i] = tr_pair_to_frag(alo[kk][i], ahi[kk][i]);
+        bf[i] = tr_pair_to_frag(blo[kk][i], bhi[kk][i]);
+      }
 #pragma unroll
       for (int i = 0; i < 2; ++i)
 #pragma unroll
         for (int j = 0; j < 2; ++j)
           acc[i][j] = MfmaOp<T>::run(af[i], bf[j], acc[i][j]);
     }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
-    cur ^= 1;
+    // all waves done READING tile cur before its buffer is re-staged two
+    // iterations from now (the next loop's barrier alone would only cover
+    // one iteration) — raw barrier, no vmcnt drain
+    __builtin_amdgcn_s_barrier();
+    if (rb + 128 < r1) {
+      advance();
+      stage(rb + 128, (cur + 2) % 3);
+    }
+    cur = (cur + 1) % 3;
   }
 
 #pragma unroll

@@ -52,7 +52,14 @@ DEV_INLINE void glds16(const void* gsrc, void* lds_dst) {
 // Reads an MFMA A/B fragment (lane l: element j = M[m0 + (l&15)]
 // [k0 + (l>>4)*8 + j]) from a [k][m] LDS image with 128-B rows carrying
 // the chunk^=(k&7) 16-B swizzle (write side pre-swizzles the glds source).
-DEV_INLINE s16x8 frag_read_tr(unsigned lds_base, int k0, int m0, int lane) {
+// Issue form: the two tr reads are ISSUED without a wait so many
+// fragments' reads fly concurrently; the caller places ONE
+// tr_read_wait() before the first consumer (a per-call lgkmcnt(0) was a
+// serial ~50-cycle chain per fragment pair — 47% issue-stall in the wrw
+// kernel's PMC). Outputs are early-clobber and only valid after the wait.
+DEV_INLINE void frag_read_tr_issue(unsigned lds_base, int k0, int m0,
+                                   int lane, unsigned long long* lo,
+                                   unsigned long long* hi) {
   int g = lane >> 4;
   int row = (lane & 15) >> 2;  // which of the 4 k-rows this lane fetches
   int seg = lane & 3;          // which 8-B segment of the 32-B m-span
@@ -62,14 +69,23 @@ DEV_INLINE s16x8 frag_read_tr(unsigned lds_base, int k0, int m0, int lane) {
   int kb = ka + 4;                // rows for elements j=4..7
   unsigned a0 = lds_base + ka * 128 + (((chunk ^ (ka & 7)) << 4) | rem);
   unsigned a1 = lds_base + kb * 128 + (((chunk ^ (kb & 7)) << 4) | rem);
-  unsigned long long lo, hi;
   asm volatile(
       "ds_read_b64_tr_b16 %0, %2\n\t"
-      "ds_read_b64_tr_b16 %1, %3\n\t"
-      "s_waitcnt lgkmcnt(0)"
-      : "=&v"(lo), "=&v"(hi)
+      "ds_read_b64_tr_b16 %1, %3"
+      : "=&v"(*lo), "=&v"(*hi)
       : "v"(a0), "v"(a1));
+}
+
+// Wait for every outstanding LDS read, then fence the scheduler so no
+// consumer (incl. register-only MFMA — guide §5.4 rule 18) is hoisted
+// above the wait.
+DEV_INLINE void tr_read_wait() {
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_sched_barrier(0);
+}
+
+DEV_INLINE s16x8 tr_pair_to_frag(unsigned long long lo,
+                                 unsigned long long hi) {
   union {
     struct { unsigned long long lo, hi; } u;
     s16x8 v;
