@@ -50,9 +50,10 @@ void pdta_conv_fwd(const void*, const void*, void*, const void*, int, int,
 void pdta_conv_bwd_data(const void*, const void*, void*, const void*, int,
                         int, int, int, int, int, int, int, int, int, int,
                         int, hipStream_t);
-void pdta_conv_bwd_weight(const void*, const void*, float*, const void*, int,
-                          int, int, int, int, int, int, int, int, int, int,
-                          int, hipStream_t);
+void pdta_wrw_plan(int, int, int, int, int, int, int, long*, long*);
+void pdta_conv_bwd_weight(const void*, const void*, float*, float*,
+                          const void*, int, int, int, int, int, int, int,
+                          int, int, int, int, int, hipStream_t);
 void pdta_stem_im2col(const void*, void*, int, int, int, int, int, int,
                       int, int, int, int, int, int, hipStream_t);
 void pdta_gemm_nt_batched(const void*, const void*, void*, int, int, int,
@@ -401,9 +402,14 @@ at::Tensor conv2d_bwd_weight(at::Tensor x, at::Tensor dy, int64_t R,
   int P = dy.size(1), Q = dy.size(2), Kout = dy.size(3);
   if (C % 64 == 0) {
     auto dw = at::zeros({Kout, C, R, S}, x.options().dtype(at::kFloat));
+    long tiles, chunks;
+    pdta_wrw_plan(N, P, Q, Kout, C, R, S, &tiles, &chunks);
+    auto part = at::empty({tiles * chunks * 4096},
+                          x.options().dtype(at::kFloat));
     pdta_conv_bwd_weight(x.data_ptr(), dy.data_ptr(), dw.data_ptr<float>(),
-                         zero_page(x), N, H, W, C, P, Q, Kout, R, S, stride,
-                         pad, dt_of(x), cur_stream());
+                         part.data_ptr<float>(), zero_page(x), N, H, W, C,
+                         P, Q, Kout, R, S, stride, pad, dt_of(x),
+                         cur_stream());
     return dw;
   }
   // stem: im2col + 1x1 igemm wrw, then repack [K][r][s][c] -> [K,C,R,S]
@@ -412,10 +418,14 @@ at::Tensor conv2d_bwd_weight(at::Tensor x, at::Tensor dy, int64_t R,
   pdta_stem_im2col(x.data_ptr(), xcol.data_ptr(), N, H, W, C, P, Q, R, S,
                    stride, pad, rsc_pad, dt_of(x), cur_stream());
   auto dwcol = at::zeros({Kout, rsc_pad}, x.options().dtype(at::kFloat));
+  long tiles, chunks;
+  pdta_wrw_plan(N, P, Q, Kout, rsc_pad, 1, 1, &tiles, &chunks);
+  auto part = at::empty({tiles * chunks * 4096},
+                        x.options().dtype(at::kFloat));
   pdta_conv_bwd_weight(xcol.data_ptr(), dy.data_ptr(),
-                       dwcol.data_ptr<float>(), zero_page(x), N, P, Q,
-                       rsc_pad, P, Q, Kout, 1, 1, 1, 0, dt_of(x),
-                       cur_stream());
+                       dwcol.data_ptr<float>(), part.data_ptr<float>(),
+                       zero_page(x), N, P, Q, rsc_pad, P, Q, Kout, 1, 1, 1,
+                       0, dt_of(x), cur_stream());
   return dwcol.slice(1, 0, rsc)
       .view({Kout, R, S, C})
       .permute({0, 3, 1, 2})
