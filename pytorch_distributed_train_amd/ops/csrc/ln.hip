@@ -50,8 +50,12 @@ __global__ void ln_fwd_k(const T* __restrict__ x, T* __restrict__ y,
   }
 }
 
-// dx = is * w .* (dy - mean(dy.*w) ... ) per row; per-chunk partial
-// [chunk][2d] for dw/db (sum dy*xhat, sum dy), finalized by ln_bwd_final.
+// dx = is * w .* (dy - mean(dy.*w) ... ) per row; dw/db partials are
+// accumulated in REGISTERS per thread (a per-element LDS-atomic version
+// was 261 us/call), combined across the block's 4 waves through LDS once
+// per chunk, written to part [chunk][2d]; ln_bwd_final reduces chunks
+// with 64-col x 4-slice blocks (a d-threads-only finalize was one wave
+// chip-wide and chunk-serial).
 template <typename T>
 __global__ void ln_bwd_k(const T* __restrict__ x, const T* __restrict__ dy,
                          T* __restrict__ dx, const float* __restrict__ w,
@@ -59,17 +63,24 @@ __global__ void ln_bwd_k(const T* __restrict__ x, const T* __restrict__ dy,
                          const float* __restrict__ invstd,
                          float* __restrict__ part, long rows, int d,
                          long rows_per_chunk) {
-  // one wave per row; block accumulates dw/db partials in LDS (fp32 [2d])
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* pdw = reinterpret_cast<float*>(smem);      // [d]
-  float* pdb = pdw + d;                             // [d]
-  for (int i = threadIdx.x; i < 2 * d; i += blockDim.x) pdw[i] = 0.f;
-  __syncthreads();
+  float* red = reinterpret_cast<float*>(smem);  // [4 waves][2d]
 
   long r0 = (long)blockIdx.x * rows_per_chunk;
   long r1 = min(rows, r0 + rows_per_chunk);
   int lane = threadIdx.x & 63;
   int wv = threadIdx.x >> 6;
+  constexpr int MAXG = 8;           // supports d up to 4096
+  int ngroups = (d + 511) / 512;
+  float adw[MAXG][8], adb[MAXG][8];
+#pragma unroll
+  for (int g = 0; g < MAXG; ++g)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      adw[g][j] = 0.f;
+      adb[g][j] = 0.f;
+    }
+
   for (long row = r0 + wv; row < r1; row += 4) {
     const T* xr = x + row * d;
     const T* gr = dy + row * d;
@@ -89,39 +100,64 @@ __global__ void ln_bwd_k(const T* __restrict__ x, const T* __restrict__ dy,
     s1 = wave_reduce_sum(s1) / d;
     s2 = wave_reduce_sum(s2) / d;
     T* dr = dx + row * d;
-    for (int c = lane * 8; c < d; c += 512) {
-      Vec8<T> g = load8(gr + c);
+    int g = 0;
+    for (int c = lane * 8; c < d; c += 512, ++g) {
+      Vec8<T> gv = load8(gr + c);
       Vec8<T> v = load8(xr + c);
       Vec8<T> o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float gg = to_f32(g.v[j]);
+        float gg = to_f32(gv.v[j]);
         float xh = (to_f32(v.v[j]) - mu) * is;
         o.v[j] = from_f32<T>(is * (gg * w[c + j] - s1 - xh * s2));
-        // dw/db partials into LDS (atomic: 4 waves share the buffer)
-        atomicAdd(&pdw[c + j], gg * xh);
-        atomicAdd(&pdb[c + j], gg);
+        adw[g][j] += gg * xh;
+        adb[g][j] += gg;
       }
       store8(dr + c, o);
     }
   }
+  // combine the 4 waves through LDS (no atomics), one pass per chunk
+  float* my = red + wv * 2 * d;
+  {
+    int g = 0;
+    for (int c = lane * 8; c < d; c += 512, ++g)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        my[c + j] = adw[g][j];
+        my[d + c + j] = adb[g][j];
+      }
+  }
   __syncthreads();
   float* out = part + (long)blockIdx.x * 2 * d;
-  for (int i = threadIdx.x; i < 2 * d; i += blockDim.x) out[i] = pdw[i];
+  for (int i = threadIdx.x; i < 2 * d; i += blockDim.x)
+    out[i] = red[i] + red[2 * d + i] + red[4 * d + i] + red[6 * d + i];
 }
 
 __global__ void ln_bwd_final_k(const float* __restrict__ part, int chunks,
                                int d, float* __restrict__ dw,
                                float* __restrict__ db) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= d) return;
+  __shared__ float red[2][256];
+  int c = blockIdx.x * 64 + (int)(threadIdx.x % 64);
+  int slice = threadIdx.x / 64;
   float sw = 0.f, sb = 0.f;
-  for (int k = 0; k < chunks; ++k) {
-    sw += part[(long)k * 2 * d + c];
-    sb += part[(long)k * 2 * d + d + c];
+  if (c < d) {
+    for (int k = slice; k < chunks; k += 4) {
+      sw += part[(long)k * 2 * d + c];
+      sb += part[(long)k * 2 * d + d + c];
+    }
   }
-  dw[c] = sw;
-  db[c] = sb;
+  red[0][threadIdx.x] = sw;
+  red[1][threadIdx.x] = sb;
+  __syncthreads();
+  if (slice == 0 && c < d) {
+#pragma unroll
+    for (int i = 1; i < 4; ++i) {
+      sw += red[0][threadIdx.x + i * 64];
+      sb += red[1][threadIdx.x + i * 64];
+    }
+    dw[c] = sw;
+    db[c] = sb;
+  }
 }
 
 // ---- GELU (exact erf form, matching torch GELU default) -----------------
@@ -234,13 +270,13 @@ void pdta_ln_bwd(const void* x, const void* dy, void* dx, const float* w,
                  hipStream_t s) {
   int chunks = pdta_ln_chunks(rows);
   long rpc = (rows + chunks - 1) / chunks;
-  int lds = 2 * d * sizeof(float);
+  int lds = 4 * 2 * d * sizeof(float);
   PDTA_DISPATCH_DTYPE16(dt, T, {
     ln_bwd_k<T><<<chunks, TPB, lds, s>>>((const T*)x, (const T*)dy, (T*)dx,
                                          w, mean, invstd, part, rows, d,
                                          rpc);
   });
-  ln_bwd_final_k<<<(d + 255) / 256, 256, 0, s>>>(part, chunks, d, dw, db);
+  ln_bwd_final_k<<<(d + 63) / 64, 256, 0, s>>>(part, chunks, d, dw, db);
 }
 
 void pdta_gelu_fwd(const void* x, void* y, long n, int dt, hipStream_t s) {

@@ -386,9 +386,19 @@ class LinearFn(torch.autograd.Function):
                                   else w_packed.t().contiguous())
                 dx = ext().gemm_nt(dyp, wtp)                  # [M, K_in]
             if ctx.needs_input_grad[1]:
-                dyt = _pad_cols64(dy.t().contiguous())        # [O, M]
-                xt = _pad_cols64(x.t().contiguous())          # [K_in, M]
-                dw = ext().gemm_nt(dyt, xt, None, True)       # fp32 [O, K_in]
+                o_dim, i_dim = weight.shape
+                rows = x.shape[0]
+                if o_dim % 64 == 0 and i_dim % 64 == 0:
+                    # dW = dy^T @ x is exactly the conv-wrw TN GEMM: reuse
+                    # its split-K slab kernel (a plain-nt path serialized
+                    # K=rows in a handful of blocks — 358us per MLP grad)
+                    dw = ext().conv2d_bwd_weight(
+                        x.view(1, rows, 1, i_dim), dy.view(1, rows, 1, o_dim),
+                        1, 1, 1, 0).view(o_dim, i_dim)
+                else:
+                    dyt = _pad_cols64(dy.t().contiguous())    # [O, M]
+                    xt = _pad_cols64(x.t().contiguous())      # [K_in, M]
+                    dw = ext().gemm_nt(dyt, xt, None, True)   # fp32 [O, K_in]
             if ctx.has_bias and ctx.needs_input_grad[3]:
                 db = ext().colsum_f32(dy)                     # [M,N] -> [N]
         else:
