@@ -60,11 +60,16 @@ class MultiheadSelfAttention(nn.Module):
             packed_t = None
         qkv = OF.linear(x.reshape(b * s, d), self.in_proj_weight, packed,
                         self.in_proj_bias, packed_t)
+        scale = 1.0 / (hd ** 0.5)
+        if x.is_cuda:
+            # zero-copy path: q/k/v read in place, grads scattered in place
+            out = OF.attention_qkv(qkv, b, s, h, hd, nvalid, scale)
+            return self.out_proj(out.reshape(b * s, d)).view(b, s, d)
         qkv = qkv.view(b, s, 3, h, hd).permute(2, 0, 3, 1, 4)
         q = qkv[0].reshape(b * h, s, hd).contiguous()
         k = qkv[1].reshape(b * h, s, hd).contiguous()
         v = qkv[2].reshape(b * h, s, hd).contiguous()
-        out = OF.attention(q, k, v, nvalid, 1.0 / (hd ** 0.5))
+        out = OF.attention(q, k, v, nvalid, scale)
         out = out.view(b, h, s, hd).permute(0, 2, 1, 3).reshape(b * s, d)
         return self.out_proj(out.contiguous()).view(b, s, d)
 

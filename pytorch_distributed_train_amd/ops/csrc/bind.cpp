@@ -68,8 +68,11 @@ void pdta_gelu_fwd(const void*, void*, long, int, hipStream_t);
 void pdta_gelu_bwd(const void*, const void*, void*, long, int, hipStream_t);
 void pdta_softmax_fwd(const float*, void*, long, int, int, float, int,
                       hipStream_t);
-void pdta_softmax_bwd(const void*, const void*, float*, long, int, float,
+void pdta_softmax_bwd(const void*, const void*, void*, long, int, float,
                       int, hipStream_t);
+void pdta_gemm_bs(const void*, const void*, void*, int, int, int, int, int,
+                  int, int, int, long, long, long, long, long, long, int,
+                  int, int, int, hipStream_t);
 void pdta_pack_weights(const float*, void*, const long*, const long*, int,
                        long, int, hipStream_t);
 }
@@ -518,11 +521,73 @@ at::Tensor attn_softmax_bwd(at::Tensor probs, at::Tensor dprobs,
   CHECK_IN(dprobs);
   int n = probs.size(-1);
   long rows = probs.numel() / n;
-  auto ds = at::empty(probs.sizes(), probs.options().dtype(at::kFloat));
-  pdta_softmax_bwd(probs.data_ptr(), dprobs.data_ptr(),
-                   ds.data_ptr<float>(), rows, n, scale, dt_of(probs),
-                   cur_stream());
+  auto ds = at::empty(probs.sizes(), probs.options());
+  pdta_softmax_bwd(probs.data_ptr(), dprobs.data_ptr(), ds.data_ptr(),
+                   rows, n, scale, dt_of(probs), cur_stream());
   return ds;
+}
+
+// ---- zero-copy multi-head attention over the interleaved qkv ------------
+// qkv [b, S, 3*h*hd] contiguous; returns (out [b, S, h*hd], p [b*h, S, S]).
+std::vector<at::Tensor> attn_fwd(at::Tensor qkv, int64_t b, int64_t S,
+                                 int64_t h, int64_t hd, int64_t nvalid,
+                                 double scale) {
+  CHECK_IN(qkv);
+  int dt = dt_of(qkv);
+  long d = h * hd, ld = 3 * d;
+  char* q = (char*)qkv.data_ptr();
+  char* k = q + d * 2;   // 16-bit elements
+  char* v = q + 2 * d * 2;
+  auto scores = at::empty({b * h, S, S}, qkv.options().dtype(at::kFloat));
+  pdta_gemm_bs(q, k, scores.data_ptr(), S, S, hd, ld, ld, S, b * h, h,
+               S * ld, hd, S * ld, hd, (long)h * S * S, S * S, 0, 0, 1, dt,
+               cur_stream());
+  auto p = at::empty({b * h, S, S}, qkv.options());
+  pdta_softmax_fwd(scores.data_ptr<float>(), p.data_ptr(), b * h * S, S,
+                   nvalid, scale, dt, cur_stream());
+  auto out = at::empty({b, S, d}, qkv.options());
+  pdta_gemm_bs(p.data_ptr(), v, out.data_ptr(), S, hd, S, S, ld, d, b * h,
+               h, (long)h * S * S, S * S, S * ld, hd, S * d, hd, 0, 1, 0,
+               dt, cur_stream());
+  return {out, p};
+}
+
+// returns d(qkv) [b, S, 3*h*hd] (written in place per head, zero copies)
+at::Tensor attn_bwd(at::Tensor qkv, at::Tensor p, at::Tensor dout,
+                    int64_t b, int64_t S, int64_t h, int64_t hd,
+                    double scale) {
+  CHECK_IN(qkv);
+  CHECK_IN(p);
+  CHECK_IN(dout);
+  int dt = dt_of(qkv);
+  long d = h * hd, ld = 3 * d;
+  char* q = (char*)qkv.data_ptr();
+  char* k = q + d * 2;
+  char* v = q + 2 * d * 2;
+  auto dp = at::empty({b * h, S, S}, qkv.options());
+  pdta_gemm_bs(dout.data_ptr(), v, dp.data_ptr(), S, S, hd, d, ld, S,
+               b * h, h, S * d, hd, S * ld, hd, (long)h * S * S, S * S, 0,
+               0, 0, dt, cur_stream());
+  auto ds = at::empty({b * h, S, S}, qkv.options());
+  pdta_softmax_bwd(p.data_ptr(), dp.data_ptr(), ds.data_ptr(), b * h * S,
+                   S, scale, dt, cur_stream());
+  auto dqkv = at::empty({b, S, ld}, qkv.options());
+  char* dq = (char*)dqkv.data_ptr();
+  char* dk = dq + d * 2;
+  char* dv = dq + 2 * d * 2;
+  // dq = ds @ K
+  pdta_gemm_bs(ds.data_ptr(), k, dq, S, hd, S, S, ld, ld, b * h, h,
+               (long)h * S * S, S * S, S * ld, hd, S * ld, hd, 0, 1, 0, dt,
+               cur_stream());
+  // dk = ds^T @ Q
+  pdta_gemm_bs(ds.data_ptr(), q, dk, S, hd, S, S, ld, ld, b * h, h,
+               (long)h * S * S, S * S, S * ld, hd, S * ld, hd, 1, 1, 0, dt,
+               cur_stream());
+  // dv = P^T @ dOut
+  pdta_gemm_bs(p.data_ptr(), dout.data_ptr(), dv, S, hd, S, S, d, ld,
+               b * h, h, (long)h * S * S, S * S, S * d, hd, S * ld, hd, 1,
+               1, 0, dt, cur_stream());
+  return dqkv;
 }
 
 void pack_weights(at::Tensor master, at::Tensor pool, at::Tensor desc,
@@ -644,6 +709,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gelu_bwd", &gelu_bwd);
   m.def("attn_softmax_fwd", &attn_softmax_fwd);
   m.def("attn_softmax_bwd", &attn_softmax_bwd);
+  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
   m.def("pack_weights", &pack_weights);
   m.def("rccl_unique_id", &rccl_unique_id);
   m.def("rccl_comm_init", &rccl_comm_init);

@@ -225,10 +225,10 @@ __global__ void softmax_fwd_f32_k(const float* __restrict__ s,
   }
 }
 
-// dscores(f32) = scale * p .* (dp - sum(p .* dp))
+// dscores = scale * p .* (dp - sum(p .* dp)), emitted in compute dtype
 template <typename T>
 __global__ void softmax_bwd_k(const T* __restrict__ p, const T* __restrict__ dp,
-                              float* __restrict__ ds, long rows, int n,
+                              T* __restrict__ ds, long rows, int n,
                               float scale) {
   long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
   if (row >= rows) return;
@@ -238,9 +238,9 @@ __global__ void softmax_bwd_k(const T* __restrict__ p, const T* __restrict__ dp,
   float dot = 0.f;
   for (int c = lane; c < n; c += 64) dot += to_f32(pr[c]) * to_f32(gr[c]);
   dot = wave_reduce_sum(dot);
-  float* dr = ds + row * n;
+  T* dr = ds + row * n;
   for (int c = lane; c < n; c += 64)
-    dr[c] = scale * to_f32(pr[c]) * (to_f32(gr[c]) - dot);
+    dr[c] = from_f32<T>(scale * to_f32(pr[c]) * (to_f32(gr[c]) - dot));
 }
 
 }  // namespace
@@ -303,12 +303,12 @@ void pdta_softmax_fwd(const float* scores, void* probs, long rows, int n,
                                                      nvalid, scale, dt);
 }
 
-void pdta_softmax_bwd(const void* probs, const void* dprobs, float* dscores,
+void pdta_softmax_bwd(const void* probs, const void* dprobs, void* dscores,
                       long rows, int n, float scale, int dt, hipStream_t s) {
   long blocks = (rows + 3) / 4;
   PDTA_DISPATCH_DTYPE16(dt, T, {
     softmax_bwd_k<T><<<(unsigned)blocks, TPB, 0, s>>>(
-        (const T*)probs, (const T*)dprobs, dscores, rows, n, scale);
+        (const T*)probs, (const T*)dprobs, (T*)dscores, rows, n, scale);
   });
 }
 

@@ -124,3 +124,36 @@ def test_vit_train_step_gpu():
         losses.append(float(loss.detach()))
     assert all(l == l for l in losses), losses
     assert min(losses[1:]) < 0.7 * losses[0], losses
+
+
+def test_attention_qkv_zero_copy_matches_permute_path():
+    """attn_fwd/attn_bwd (in-place strided GEMMs) vs the permute-based
+    AttentionFn on the same qkv tensor — fwd and d(qkv)."""
+    from pytorch_distributed_train_amd.ops import functional as OF
+    b, s, h, hd, nv = 2, 128, 4, 64, 100
+    scale = 0.125
+    qkv = _rand(b * s, 3 * h * hd, seed=20)
+    # zero the padded V rows (as the model guarantees via input padding)
+    qkv3 = qkv.view(b, s, 3, h, hd)
+    with torch.no_grad():
+        qkv3[:, nv:, 2] = 0
+    qkv_a = qkv.clone().requires_grad_(True)
+    qkv_b = qkv.clone().requires_grad_(True)
+
+    out_a = OF.attention_qkv(qkv_a, b, s, h, hd, nv, scale)  # [b,s,h*hd]
+
+    qp = qkv_b.view(b, s, 3, h, hd).permute(2, 0, 3, 1, 4)
+    q = qp[0].reshape(b * h, s, hd).contiguous()
+    k = qp[1].reshape(b * h, s, hd).contiguous()
+    v = qp[2].reshape(b * h, s, hd).contiguous()
+    out_ref = OF.attention(q, k, v, nv, scale)
+    out_ref = out_ref.view(b, h, s, hd).permute(0, 2, 1, 3).reshape(
+        b, s, h * hd)
+
+    _close(out_a[:, :nv], out_ref[:, :nv], what="attn_qkv fwd")
+
+    g = torch.zeros_like(out_a)
+    g[:, :nv] = _rand(b, nv, h * hd, seed=21)
+    out_a.backward(g)
+    out_ref.backward(g.clone())
+    _close(qkv_a.grad, qkv_b.grad, rtol=3e-2, atol=3e-2, what="attn_qkv dqkv")
