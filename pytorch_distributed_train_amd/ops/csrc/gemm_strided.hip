@@ -83,6 +83,8 @@ __global__ __launch_bounds__(TPB) void gemm_bs_k(
     if (kt + 1 < K / 64) stage_tiles((kt + 1) * 64, cur ^ 1);
     unsigned la = lds_addr_of(lds[cur][0]);
     unsigned lb = lds_addr_of(lds[cur][1]);
+    const char* la_p = lds[cur][0];
+    const char* lb_p = lds[cur][1];
     // issue all LDS reads for both K=32 steps, wait once, then MFMA
     unsigned long long alo[2][2], ahi[2][2], blo[2][2], bhi[2][2];
     s16x8 afn[2][2], bfn[2][2];
@@ -94,15 +96,13 @@ __global__ __launch_bounds__(TPB) void gemm_bs_k(
           frag_read_tr_issue(la, kk * 32, wm * 32 + i * 16, lane,
                              &alo[kk][i], &ahi[kk][i]);
         else
-          afn[kk][i] = frag_read128((const char*)nullptr + la,
-                                    wm * 32 + i * 16 + (lane & 15),
+          afn[kk][i] = frag_read128(la_p, wm * 32 + i * 16 + (lane & 15),
                                     (lane >> 4) + kk * 4);
         if (TRB)
           frag_read_tr_issue(lb, kk * 32, wn * 32 + i * 16, lane,
                              &blo[kk][i], &bhi[kk][i]);
         else
-          bfn[kk][i] = frag_read128((const char*)nullptr + lb,
-                                    wn * 32 + i * 16 + (lane & 15),
+          bfn[kk][i] = frag_read128(lb_p, wn * 32 + i * 16 + (lane & 15),
                                     (lane >> 4) + kk * 4);
       }
     }
