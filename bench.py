@@ -96,7 +96,7 @@ def main(argv=None) -> int:
     if env.rank == 0:
         ms = t_max / b.steps * 1000.0
         total_images = b.batch_size * n_gpus * b.steps
-        print(json.dumps({
+        line = json.dumps({
             "metric": ("images/sec (whole node) ResNet-50 ImageNet-shape"
                        if b.model == "resnet50" else
                        f"images/sec (whole node) {b.model} ImageNet-shape"),
@@ -115,7 +115,14 @@ def main(argv=None) -> int:
                        "global_batch": b.batch_size * n_gpus,
                        "input": "3x224x224",
                        "parallelism": f"dp{n_gpus}"},
-        }))
+        })
+        print(line)
+        try:
+            os.makedirs("gpurun_out", exist_ok=True)
+            with open(os.path.join("gpurun_out", "bench.jsonl"), "a") as f:
+                f.write(line + "\n")
+        except OSError:
+            pass
     return 0
 
 

@@ -31,6 +31,10 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--steps-per-epoch", type=int, default=0,
                    help="cap steps per epoch (0 = full sampler length)")
     p.add_argument("--lr", type=float, default=0.1)
+    p.add_argument("--lr-schedule", default="none",
+                   choices=("none", "step", "cosine"),
+                   help="step: x0.1 at 30/60/80%% of epochs; cosine anneal")
+    p.add_argument("--warmup-epochs", type=int, default=0)
     p.add_argument("--momentum", type=float, default=0.9)
     p.add_argument("--weight-decay", type=float, default=1e-4)
     p.add_argument("--dtype", default="bf16", choices=("fp32", "bf16", "fp16"),

@@ -110,12 +110,13 @@ class Encoder(nn.Module):
         self.ln = LayerNorm(dim)
 
     def forward(self, x):  # x [B, SEQ, dim]
-        x = x + self.pos_embedding.to(x.dtype)
         b, s, d = x.shape
         if s < SEQ_PAD:  # pad token rows for GEMM-friendly shapes
-            pad = torch.zeros(b, SEQ_PAD - s, d, dtype=x.dtype,
-                              device=x.device)
-            x = torch.cat([x, pad], dim=1)
+            xp = x.new_zeros(b, SEQ_PAD, d)
+            xp[:, :s] = x + self.pos_embedding.to(x.dtype)
+            x = xp
+        else:
+            x = x + self.pos_embedding.to(x.dtype)
         x = self.layers(x)
         return self.ln(x)
 

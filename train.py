@@ -55,6 +55,26 @@ def evaluate(trainer, dataset, env, batch_size, max_batches=0):
     return (top1 / max(total, 1), top5 / max(total, 1))
 
 
+def lr_at(args, epoch: int) -> float:
+    """Step (x0.1 at 30/60/80% of epochs) or cosine schedule with linear
+    warmup — the reference-style training recipes."""
+    import math
+    if args.warmup_epochs and epoch < args.warmup_epochs:
+        return args.lr * (epoch + 1) / args.warmup_epochs
+    if args.lr_schedule == "step":
+        frac = epoch / max(args.epochs, 1)
+        scale = 1.0
+        for boundary in (0.3, 0.6, 0.8):
+            if frac >= boundary:
+                scale *= 0.1
+        return args.lr * scale
+    if args.lr_schedule == "cosine":
+        t = (epoch - args.warmup_epochs) / max(
+            args.epochs - args.warmup_epochs, 1)
+        return args.lr * 0.5 * (1 + math.cos(math.pi * t))
+    return args.lr
+
+
 def main(argv=None):
     args = build_parser().parse_args(argv)
     env = DistEnv.from_env(args.local_rank)
@@ -88,6 +108,7 @@ def main(argv=None):
 
     for epoch in range(start_epoch, args.epochs):
         trainer.epoch = epoch
+        trainer.optimizer.lr = lr_at(args, epoch)
         sampler.set_epoch(epoch)
         trainer.model.train()
         timer.start()
@@ -100,7 +121,8 @@ def main(argv=None):
             ips = timer.tick(args.batch_size * env.world_size)
             if step % args.print_freq == 0:
                 logger.log(epoch=epoch, step=step, loss=float(loss),
-                           images_per_sec=round(ips, 1), lr=args.lr)
+                           images_per_sec=round(ips, 1),
+                           lr=trainer.optimizer.lr)
         if args.eval_freq and (epoch + 1) % args.eval_freq == 0:
             t1, t5 = evaluate(trainer, dataset, env, args.batch_size,
                               max_batches=args.eval_batches)
@@ -116,5 +138,21 @@ def main(argv=None):
     return 0
 
 
+def _recorded_main():
+    """@record analog: serialize worker failures so
+    the launcher/operator can see which rank died and why."""
+    try:
+        return main()
+    except Exception:
+        import traceback
+        rank = os.environ.get("RANK", "0")
+        os.makedirs("gpurun_out", exist_ok=True)
+        path = os.path.join("gpurun_out", f"worker_error_rank{rank}.txt")
+        with open(path, "w") as f:
+            traceback.print_exc(file=f)
+        traceback.print_exc()
+        return 1
+
+
 if __name__ == "__main__":
-    sys.exit(main())
+    sys.exit(_recorded_main())
