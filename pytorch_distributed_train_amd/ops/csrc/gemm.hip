@@ -130,6 +130,139 @@ __global__ __launch_bounds__(TPB) void gemm_nt_k(
   }
 }
 
+
+// ---- large-M GEMM: 256x128 tile, 8 waves, 3-slot K-ring ----------------
+// The pipelined big-tile core (guide T3/T4: the gain of phase-splitting IS
+// the counted vmcnt): three K-tile slots (48 KiB each = A 256x64 + B
+// 128x64), tile T+2's six glds issue during tile T's compute, and each
+// tile boundary costs ONE raw s_barrier + s_waitcnt vmcnt(6) — loads stay
+// in flight across the barrier (hipcc's __syncthreads would drain them).
+// Wave grid 2(M) x 4(N): 128x32 per wave, 8x2 fragments, 32 MFMA/tile.
+template <typename T, typename OutT, bool BIAS>
+__global__ __launch_bounds__(512) void gemm_nt_big_k(
+    const T* __restrict__ A, const T* __restrict__ B, OutT* __restrict__ C,
+    const float* __restrict__ bias, int M, int N, int K, int tiles_n,
+    long batch_sa, long batch_sb, long batch_sc) {
+  constexpr int BOFF = 256 * 64 * 2;  // A image bytes per slot
+  __shared__ __attribute__((aligned(16))) char lds[3][BOFF + 128 * 64 * 2];
+
+  A += (long)blockIdx.y * batch_sa;
+  B += (long)blockIdx.y * batch_sb;
+  C += (long)blockIdx.y * batch_sc;
+
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  int tm = bid / tiles_n, tn = bid % tiles_n;
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  int wm = wave >> 2, wn = wave & 3;  // 2 x 4
+  int m0 = tm * 256, n0 = tn * 128;
+  int ktiles = K / 64;
+
+  int row_in_chunk = lane >> 3;
+  int col16 = lane & 7;
+
+  // stage tile kt into slot: A 4 glds/wave (rows 8*(w*4+i)...), B 2
+  auto stage_a = [&](int kt, int slot, int i) {
+    int row = (wave * 4 + i) * 8 + row_in_chunk;
+    int grow = m0 + row;
+    if (grow > M - 1) grow = M - 1;
+    const char* g = reinterpret_cast<const char*>(A) +
+                    ((long)grow * K + kt * 64) * 2 +
+                    (col16 ^ (row & 7)) * 16;
+    glds16(g, (char*)lds[slot] + (wave * 4 + i) * 8 * 128);
+  };
+  auto stage_b = [&](int kt, int slot, int i) {
+    int row = (wave * 2 + i) * 8 + row_in_chunk;
+    int grow = n0 + row;
+    if (grow > N - 1) grow = N - 1;
+    const char* g = reinterpret_cast<const char*>(B) +
+                    ((long)grow * K + kt * 64) * 2 +
+                    (col16 ^ (row & 7)) * 16;
+    glds16(g, (char*)lds[slot] + BOFF + (wave * 2 + i) * 8 * 128);
+  };
+  auto stage_all = [&](int kt, int slot) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) stage_a(kt, slot, i);
+#pragma unroll
+    for (int i = 0; i < 2; ++i) stage_b(kt, slot, i);
+  };
+
+  f32x4 acc[8][2];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  stage_all(0, 0);
+  if (ktiles > 1) stage_all(1, 1);
+
+  for (int kt = 0; kt < ktiles; ++kt) {
+    int slot = kt % 3;
+    // tile kt's 6 glds were issued >= 2 tiles ago; at most the newest 6
+    // (tile kt+1's) may still be in flight
+    if (kt + 1 < ktiles)
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    const char* la = lds[slot];
+    const char* lb = lds[slot] + BOFF;
+    s16x8 bf[2][2];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+        bf[kk][j] = frag_read128(lb, wn * 32 + j * 16 + (lane & 15),
+                                 (lane >> 4) + kk * 4);
+    int nslot = (kt + 2) % 3;
+    bool pre = kt + 2 < ktiles;
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {  // quadrants of the wave's 128 rows
+      s16x8 af[2][2];
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+          af[kk][i] = frag_read128(
+              la, wm * 128 + q * 32 + i * 16 + (lane & 15),
+              (lane >> 4) + kk * 4);
+      // drip tile kt+2's staging through the quadrant phases
+      if (pre) {
+        if (q < 2) {
+          stage_a(kt + 2, nslot, q * 2);
+          stage_a(kt + 2, nslot, q * 2 + 1);
+        } else {
+          stage_b(kt + 2, nslot, q - 2);
+        }
+      }
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+          for (int j = 0; j < 2; ++j)
+            acc[q * 2 + i][j] =
+                MfmaOp<T>::run(af[kk][i], bf[kk][j], acc[q * 2 + i][j]);
+    }
+  }
+
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    long rbase = m0 + wm * 128 + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int col = n0 + wn * 32 + j * 16 + (lane & 15);
+      if (col >= N) continue;
+      float badd = BIAS ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = rbase + r;
+        if (row < M)
+          C[(long)row * N + col] = from_f32<OutT>(acc[i][j][r] + badd);
+      }
+    }
+  }
+}
+
 // column sum: db[n] = sum_m dy[m][n]  (linear bias grad)
 template <typename T>
 __global__ void colsum_f32_k(const T* __restrict__ dy, float* __restrict__ db,
@@ -149,6 +282,32 @@ extern "C" {
 // C[M,N] = A[M,K] B[N,K]^T (+bias); K % 64 == 0. out_f32: C is fp32.
 void pdta_gemm_nt(const void* A, const void* B, void* C, const float* bias,
                   int M, int N, int K, int dt, int out_f32, hipStream_t s) {
+  if (M >= 1024 && N >= 128 && K % 64 == 0) {
+    int tiles_m = (M + 255) / 256, tiles_n = (N + 127) / 128;
+    dim3 grid(tiles_m * tiles_n);
+    PDTA_DISPATCH_DTYPE16(dt, T, {
+      if (out_f32) {
+        if (bias)
+          gemm_nt_big_k<T, float, true><<<grid, 512, 0, s>>>(
+              (const T*)A, (const T*)B, (float*)C, bias, M, N, K, tiles_n,
+              0, 0, 0);
+        else
+          gemm_nt_big_k<T, float, false><<<grid, 512, 0, s>>>(
+              (const T*)A, (const T*)B, (float*)C, nullptr, M, N, K,
+              tiles_n, 0, 0, 0);
+      } else {
+        if (bias)
+          gemm_nt_big_k<T, T, true><<<grid, 512, 0, s>>>(
+              (const T*)A, (const T*)B, (T*)C, bias, M, N, K, tiles_n, 0, 0,
+              0);
+        else
+          gemm_nt_big_k<T, T, false><<<grid, 512, 0, s>>>(
+              (const T*)A, (const T*)B, (T*)C, nullptr, M, N, K, tiles_n, 0,
+              0, 0);
+      }
+    });
+    return;
+  }
   int tiles_m = (M + BM - 1) / BM, tiles_n = (N + BN - 1) / BN;
   dim3 grid(tiles_m * tiles_n);
   PDTA_DISPATCH_DTYPE16(dt, T, {

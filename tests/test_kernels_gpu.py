@@ -41,7 +41,9 @@ def test_gemm_nt_identity_asymmetric():
 
 
 @pytest.mark.parametrize("m,n,k", [(128, 128, 64), (256, 512, 128),
-                                   (200, 100, 192), (1000, 64, 256)])
+                                   (200, 100, 192), (1000, 64, 256),
+                                   (2048, 768, 768), (4096, 256, 192),
+                                   (1100, 130, 128)])
 def test_gemm_nt_random(m, n, k):
     a = _rand(m, k, seed=1)
     b = _rand(n, k, seed=2)
