@@ -131,20 +131,21 @@ __global__ __launch_bounds__(TPB) void gemm_nt_k(
 }
 
 
-// ---- large-M GEMM: 256x128 tile, 8 waves, 3-slot K-ring ----------------
-// The pipelined big-tile core (guide T3/T4: the gain of phase-splitting IS
-// the counted vmcnt): three K-tile slots (48 KiB each = A 256x64 + B
-// 128x64), tile T+2's six glds issue during tile T's compute, and each
-// tile boundary costs ONE raw s_barrier + s_waitcnt vmcnt(6) — loads stay
-// in flight across the barrier (hipcc's __syncthreads would drain them).
-// Wave grid 2(M) x 4(N): 128x32 per wave, 8x2 fragments, 32 MFMA/tile.
+// ---- large-M GEMM: 256x256 tile, 8 waves, 2-slot K-ring ----------------
+// Big-tile pipelined core: per 512-thread block a 256x256 output tile
+// (wave grid 2x4 -> 128x64 per wave, 64 MFMA/K-tile/wave), two 64-KiB
+// K-slots (A 256x64 + B 256x64), tile T+1's eight glds issued at the top
+// of tile T's compute so the ~L2 latency hides under ~2200 cycles of
+// MFMA; one raw s_barrier + vmcnt drain per K-tile. Halves the staged
+// bytes per FLOP vs the 128-wide tiles (the glds path, not HBM, is the
+// bound at these shapes).
 template <typename T, typename OutT, bool BIAS>
 __global__ __launch_bounds__(512) void gemm_nt_big_k(
     const T* __restrict__ A, const T* __restrict__ B, OutT* __restrict__ C,
     const float* __restrict__ bias, int M, int N, int K, int tiles_n,
     long batch_sa, long batch_sb, long batch_sc) {
   constexpr int BOFF = 256 * 64 * 2;  // A image bytes per slot
-  __shared__ __attribute__((aligned(16))) char lds[3][BOFF + 128 * 64 * 2];
+  __shared__ __attribute__((aligned(16))) char lds[2][2 * BOFF];
 
   A += (long)blockIdx.y * batch_sa;
   B += (long)blockIdx.y * batch_sb;
@@ -154,67 +155,60 @@ __global__ __launch_bounds__(512) void gemm_nt_big_k(
   int tm = bid / tiles_n, tn = bid % tiles_n;
   int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
   int wm = wave >> 2, wn = wave & 3;  // 2 x 4
-  int m0 = tm * 256, n0 = tn * 128;
+  int m0 = tm * 256, n0 = tn * 256;
   int ktiles = K / 64;
 
   int row_in_chunk = lane >> 3;
   int col16 = lane & 7;
 
-  // stage tile kt into slot: A 4 glds/wave (rows 8*(w*4+i)...), B 2
-  auto stage_a = [&](int kt, int slot, int i) {
-    int row = (wave * 4 + i) * 8 + row_in_chunk;
-    int grow = m0 + row;
-    if (grow > M - 1) grow = M - 1;
-    const char* g = reinterpret_cast<const char*>(A) +
-                    ((long)grow * K + kt * 64) * 2 +
-                    (col16 ^ (row & 7)) * 16;
-    glds16(g, (char*)lds[slot] + (wave * 4 + i) * 8 * 128);
-  };
-  auto stage_b = [&](int kt, int slot, int i) {
-    int row = (wave * 2 + i) * 8 + row_in_chunk;
-    int grow = n0 + row;
-    if (grow > N - 1) grow = N - 1;
-    const char* g = reinterpret_cast<const char*>(B) +
-                    ((long)grow * K + kt * 64) * 2 +
-                    (col16 ^ (row & 7)) * 16;
-    glds16(g, (char*)lds[slot] + BOFF + (wave * 2 + i) * 8 * 128);
-  };
   auto stage_all = [&](int kt, int slot) {
 #pragma unroll
-    for (int i = 0; i < 4; ++i) stage_a(kt, slot, i);
+    for (int i = 0; i < 4; ++i) {
+      int row = (wave * 4 + i) * 8 + row_in_chunk;
+      int grow = m0 + row;
+      if (grow > M - 1) grow = M - 1;
+      const char* g = reinterpret_cast<const char*>(A) +
+                      ((long)grow * K + kt * 64) * 2 +
+                      (col16 ^ (row & 7)) * 16;
+      glds16(g, (char*)lds[slot] + (wave * 4 + i) * 8 * 128);
+    }
 #pragma unroll
-    for (int i = 0; i < 2; ++i) stage_b(kt, slot, i);
+    for (int i = 0; i < 4; ++i) {
+      int row = (wave * 4 + i) * 8 + row_in_chunk;
+      int grow = n0 + row;
+      if (grow > N - 1) grow = N - 1;
+      const char* g = reinterpret_cast<const char*>(B) +
+                      ((long)grow * K + kt * 64) * 2 +
+                      (col16 ^ (row & 7)) * 16;
+      glds16(g, (char*)lds[slot] + BOFF + (wave * 4 + i) * 8 * 128);
+    }
   };
 
-  f32x4 acc[8][2];
+  f32x4 acc[8][4];
 #pragma unroll
   for (int i = 0; i < 8; ++i)
 #pragma unroll
-    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   stage_all(0, 0);
-  if (ktiles > 1) stage_all(1, 1);
 
   for (int kt = 0; kt < ktiles; ++kt) {
-    int slot = kt % 3;
-    // tile kt's 6 glds were issued >= 2 tiles ago; at most the newest 6
-    // (tile kt+1's) may still be in flight
-    if (kt + 1 < ktiles)
-      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
-    else
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    int slot = kt & 1;
+    // drain this tile's loads (they had the previous tile's ~2200-cycle
+    // MFMA stream to fly) and rendezvous
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
+    // issue the NEXT tile's staging first so it spans this tile's compute
+    if (kt + 1 < ktiles) stage_all(kt + 1, slot ^ 1);
     const char* la = lds[slot];
     const char* lb = lds[slot] + BOFF;
-    s16x8 bf[2][2];
+    s16x8 bf[2][4];
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
-      for (int j = 0; j < 2; ++j)
-        bf[kk][j] = frag_read128(lb, wn * 32 + j * 16 + (lane & 15),
+      for (int j = 0; j < 4; ++j)
+        bf[kk][j] = frag_read128(lb, wn * 64 + j * 16 + (lane & 15),
                                  (lane >> 4) + kk * 4);
-    int nslot = (kt + 2) % 3;
-    bool pre = kt + 2 < ktiles;
 #pragma unroll
     for (int q = 0; q < 4; ++q) {  // quadrants of the wave's 128 rows
       s16x8 af[2][2];
@@ -225,32 +219,25 @@ __global__ __launch_bounds__(512) void gemm_nt_big_k(
           af[kk][i] = frag_read128(
               la, wm * 128 + q * 32 + i * 16 + (lane & 15),
               (lane >> 4) + kk * 4);
-      // drip tile kt+2's staging through the quadrant phases
-      if (pre) {
-        if (q < 2) {
-          stage_a(kt + 2, nslot, q * 2);
-          stage_a(kt + 2, nslot, q * 2 + 1);
-        } else {
-          stage_b(kt + 2, nslot, q - 2);
-        }
-      }
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
         for (int i = 0; i < 2; ++i)
 #pragma unroll
-          for (int j = 0; j < 2; ++j)
+          for (int j = 0; j < 4; ++j)
             acc[q * 2 + i][j] =
                 MfmaOp<T>::run(af[kk][i], bf[kk][j], acc[q * 2 + i][j]);
     }
+    // all waves done reading this slot before it is re-staged next tile
+    __builtin_amdgcn_s_barrier();
   }
 
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
     long rbase = m0 + wm * 128 + i * 16 + (lane >> 4) * 4;
 #pragma unroll
-    for (int j = 0; j < 2; ++j) {
-      int col = n0 + wn * 32 + j * 16 + (lane & 15);
+    for (int j = 0; j < 4; ++j) {
+      int col = n0 + wn * 64 + j * 16 + (lane & 15);
       if (col >= N) continue;
       float badd = BIAS ? bias[col] : 0.f;
 #pragma unroll
@@ -282,8 +269,8 @@ extern "C" {
 // C[M,N] = A[M,K] B[N,K]^T (+bias); K % 64 == 0. out_f32: C is fp32.
 void pdta_gemm_nt(const void* A, const void* B, void* C, const float* bias,
                   int M, int N, int K, int dt, int out_f32, hipStream_t s) {
-  if (M >= 1024 && N >= 128 && K % 64 == 0) {
-    int tiles_m = (M + 255) / 256, tiles_n = (N + 127) / 128;
+  if (M >= 1024 && N >= 256 && K % 64 == 0) {
+    int tiles_m = (M + 255) / 256, tiles_n = (N + 255) / 256;
     dim3 grid(tiles_m * tiles_n);
     PDTA_DISPATCH_DTYPE16(dt, T, {
       if (out_f32) {
