@@ -147,6 +147,134 @@ __global__ __launch_bounds__(TPB) void igemm_fwd_k(
   }
 }
 
+// ---- big-channel fwd: 256x256 tile, 8 waves, 2-slot K-ring -------------
+// The gemm.hip big-tile structure with the im2col gather in the A-source
+// (layers with Kout % 256 == 0 carry most of the conv FLOPs; the 128-wide
+// tiles ran at ~190 TF from staging re-reads).
+template <typename T>
+__global__ __launch_bounds__(512) void igemm_fwd_big_k(
+    const T* __restrict__ x, const T* __restrict__ wt, T* __restrict__ y,
+    const T* __restrict__ zero_page, int Nn, int H, int W, int C, int P,
+    int Q, int Kout, int R, int S, int stride, int pad, int tiles_n,
+    unsigned mq, int sq, unsigned mpq, int spq) {
+  constexpr int BOFF = 256 * 64 * 2;
+  __shared__ __attribute__((aligned(16))) char lds[2][2 * BOFF];
+
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  int tm = bid / tiles_n, tn = bid % tiles_n;
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  int wm = wave >> 2, wn = wave & 3;
+  long M = (long)Nn * P * Q;
+  int m0 = tm * 256, n0 = tn * 256;
+  int RSC = R * S * C;
+
+  int row_in_chunk = lane >> 3;
+  int col16 = lane & 7;
+  int a_n[4], a_h0[4], a_w0[4], a_swz[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int row = (wave * 4 + i) * 8 + row_in_chunk;
+    long m = m0 + row;
+    if (m > M - 1) m = M - 1;
+    unsigned npq = mdiv((unsigned)m, mpq, spq);
+    unsigned rem = (unsigned)m - npq * (unsigned)(P * Q);
+    unsigned pp = mdiv(rem, mq, sq);
+    unsigned qq = rem - pp * (unsigned)Q;
+    a_n[i] = (int)npq;
+    a_h0[i] = (int)pp * stride - pad;
+    a_w0[i] = (int)qq * stride - pad;
+    a_swz[i] = col16 ^ (row & 7);
+  }
+
+  int cchunks = C / 64;
+  int ktiles = R * S * cchunks;
+
+  auto stage_all = [&](int kt, int slot) {
+    int tap = kt / cchunks, ct = kt % cchunks;
+    int r = tap / S, ss = tap % S;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int h = a_h0[i] + r, w = a_w0[i] + ss;
+      const char* g;
+      if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
+        g = reinterpret_cast<const char*>(x) +
+            ((((long)a_n[i] * H + h) * W + w) * C + ct * 64) * 2 +
+            a_swz[i] * 16;
+      else
+        g = reinterpret_cast<const char*>(zero_page);
+      glds16(g, (char*)lds[slot] + (wave * 4 + i) * 8 * 128);
+    }
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int row = (wave * 4 + i) * 8 + row_in_chunk;
+      int ko = n0 + row;
+      if (ko > Kout - 1) ko = Kout - 1;
+      const char* g = reinterpret_cast<const char*>(wt) +
+                      ((long)ko * RSC + tap * C + ct * 64) * 2 +
+                      (col16 ^ (row & 7)) * 16;
+      glds16(g, (char*)lds[slot] + BOFF + (wave * 4 + i) * 8 * 128);
+    }
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  stage_all(0, 0);
+  for (int kt = 0; kt < ktiles; ++kt) {
+    int slot = kt & 1;
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (kt + 1 < ktiles) stage_all(kt + 1, slot ^ 1);
+    const char* la = lds[slot];
+    const char* lb = lds[slot] + BOFF;
+    s16x8 bf[2][4];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        bf[kk][j] = frag_read128(lb, wn * 64 + j * 16 + (lane & 15),
+                                 (lane >> 4) + kk * 4);
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      s16x8 af[2][2];
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+          af[kk][i] = frag_read128(
+              la, wm * 128 + q * 32 + i * 16 + (lane & 15),
+              (lane >> 4) + kk * 4);
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[q * 2 + i][j] =
+                MfmaOp<T>::run(af[kk][i], bf[kk][j], acc[q * 2 + i][j]);
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    long rbase = m0 + wm * 128 + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = n0 + wn * 64 + j * 16 + (lane & 15);
+      if (col >= Kout) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = rbase + r;
+        if (row < M) y[row * Kout + col] = from_f32<T>(acc[i][j][r]);
+      }
+    }
+  }
+}
+
 // ----------------------------------------------------------- bwd-data --
 template <typename T, int STRIDE, int BNT>  // STRIDE 1/2/0(runtime)
 __global__ __launch_bounds__(TPB) void igemm_bwd_data_k(
@@ -541,6 +669,18 @@ void pdta_conv_fwd(const void* x, const void* wt, void* y,
                    int Q, int Kout, int R, int S, int stride, int pad,
                    int dt, hipStream_t s) {
   long M = (long)Nn * P * Q;
+  if (Kout % 256 == 0 && M >= 4096) {
+    int tiles_m = (int)((M + 255) / 256), tiles_n = Kout / 256;
+    dim3 grid(tiles_m * tiles_n);
+    MagicDiv mq = magic_div(Q), mpq = magic_div(P * Q);
+    PDTA_DISPATCH_DTYPE16(dt, T, {
+      igemm_fwd_big_k<T><<<grid, 512, 0, s>>>(
+          (const T*)x, (const T*)wt, (T*)y, (const T*)zero_page, Nn, H, W,
+          C, P, Q, Kout, R, S, stride, pad, tiles_n, mq.mul, mq.shift,
+          mpq.mul, mpq.shift);
+    });
+    return;
+  }
   int bnt = (Kout % 128 == 0) ? 128 : 64;
   int tiles_m = (int)((M + 127) / 128), tiles_n = (Kout + bnt - 1) / bnt;
   dim3 grid(tiles_m * tiles_n);
