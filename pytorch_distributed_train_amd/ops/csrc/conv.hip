@@ -410,6 +410,141 @@ __global__ __launch_bounds__(TPB) void igemm_bwd_data_k(
   }
 }
 
+// ---- big-channel bwd-data: 256x256 tile, 2-slot K-ring -----------------
+template <typename T, int STRIDE>
+__global__ __launch_bounds__(512) void igemm_bwd_data_big_k(
+    const T* __restrict__ dy, const T* __restrict__ w2, T* __restrict__ dx,
+    const T* __restrict__ zero_page, int Nn, int H, int W, int C, int P,
+    int Q, int Kout, int R, int S, int stride, int pad, int tiles_n,
+    unsigned mw, int sw, unsigned mhw, int shw) {
+  constexpr int BOFF = 256 * 64 * 2;
+  __shared__ __attribute__((aligned(16))) char lds[2][2 * BOFF];
+
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  int tm = bid / tiles_n, tn = bid % tiles_n;
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  int wm = wave >> 2, wn = wave & 3;
+  long M = (long)Nn * H * W;
+  int m0 = tm * 256, n0 = tn * 256;
+
+  int row_in_chunk = lane >> 3;
+  int col16 = lane & 7;
+  int a_n[4], a_h[4], a_w[4], a_swz[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int row = (wave * 4 + i) * 8 + row_in_chunk;
+    long m = m0 + row;
+    if (m > M - 1) m = M - 1;
+    unsigned nhw = mdiv((unsigned)m, mhw, shw);
+    unsigned rem = (unsigned)m - nhw * (unsigned)(H * W);
+    unsigned hh = mdiv(rem, mw, sw);
+    a_w[i] = (int)(rem - hh * (unsigned)W);
+    a_h[i] = (int)hh;
+    a_n[i] = (int)nhw;
+    a_swz[i] = col16 ^ (row & 7);
+  }
+
+  int kchunks = Kout / 64;
+  int ktiles = R * S * kchunks;
+
+  auto stage_all = [&](int kt, int slot) {
+    int tap = kt / kchunks, kot = kt % kchunks;
+    int r = tap / S, ss = tap % S;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int pn = a_h[i] + pad - r;
+      int qn = a_w[i] + pad - ss;
+      int p, q;
+      bool valid;
+      if constexpr (STRIDE == 1) {
+        p = pn;
+        q = qn;
+        valid = pn >= 0 && qn >= 0 && p < P && q < Q;
+      } else {
+        p = pn >> 1;
+        q = qn >> 1;
+        valid = pn >= 0 && qn >= 0 && !((pn | qn) & 1) && p < P && q < Q;
+      }
+      const char* g;
+      if (valid)
+        g = reinterpret_cast<const char*>(dy) +
+            ((((long)a_n[i] * P + p) * Q + q) * Kout + kot * 64) * 2 +
+            a_swz[i] * 16;
+      else
+        g = reinterpret_cast<const char*>(zero_page);
+      glds16(g, (char*)lds[slot] + (wave * 4 + i) * 8 * 128);
+    }
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int row = (wave * 4 + i) * 8 + row_in_chunk;
+      int c = n0 + row;
+      if (c > C - 1) c = C - 1;
+      const char* g = reinterpret_cast<const char*>(w2) +
+                      ((((long)tap) * C + c) * Kout + kot * 64) * 2 +
+                      (col16 ^ (row & 7)) * 16;
+      glds16(g, (char*)lds[slot] + BOFF + (wave * 4 + i) * 8 * 128);
+    }
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  stage_all(0, 0);
+  for (int kt = 0; kt < ktiles; ++kt) {
+    int slot = kt & 1;
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (kt + 1 < ktiles) stage_all(kt + 1, slot ^ 1);
+    const char* la = lds[slot];
+    const char* lb = lds[slot] + BOFF;
+    s16x8 bf[2][4];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        bf[kk][j] = frag_read128(lb, wn * 64 + j * 16 + (lane & 15),
+                                 (lane >> 4) + kk * 4);
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      s16x8 af[2][2];
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+          af[kk][i] = frag_read128(
+              la, wm * 128 + q * 32 + i * 16 + (lane & 15),
+              (lane >> 4) + kk * 4);
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[q * 2 + i][j] =
+                MfmaOp<T>::run(af[kk][i], bf[kk][j], acc[q * 2 + i][j]);
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    long rbase = m0 + wm * 128 + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = n0 + wn * 64 + j * 16 + (lane & 15);
+      if (col >= C) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = rbase + r;
+        if (row < M) dx[row * C + col] = from_f32<T>(acc[i][j][r]);
+      }
+    }
+  }
+}
+
 // ---------------------------------------------------------- bwd-weight --
 // dW[kout][c] at one (r,s): 64x64 tile per block, split-K over npq rows
 // (64-row LDS tiles, transposed fragment reads), fp32 atomicAdd into the
@@ -701,6 +836,24 @@ void pdta_conv_bwd_data(const void* dy, const void* w2, void* dx,
                         int P, int Q, int Kout, int R, int S, int stride,
                         int pad, int dt, hipStream_t s) {
   long M = (long)Nn * H * W;
+  if (C % 256 == 0 && M >= 4096 && (stride == 1 || stride == 2)) {
+    int tiles_m = (int)((M + 255) / 256), tiles_n = C / 256;
+    dim3 grid(tiles_m * tiles_n);
+    MagicDiv mw = magic_div(W), mhw = magic_div(H * W);
+    PDTA_DISPATCH_DTYPE16(dt, T, {
+      if (stride == 1)
+        igemm_bwd_data_big_k<T, 1><<<grid, 512, 0, s>>>(
+            (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H,
+            W, C, P, Q, Kout, R, S, stride, pad, tiles_n, mw.mul, mw.shift,
+            mhw.mul, mhw.shift);
+      else
+        igemm_bwd_data_big_k<T, 2><<<grid, 512, 0, s>>>(
+            (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H,
+            W, C, P, Q, Kout, R, S, stride, pad, tiles_n, mw.mul, mw.shift,
+            mhw.mul, mhw.shift);
+    });
+    return;
+  }
   int bnt = (C % 128 == 0) ? 128 : 64;
   int tiles_m = (int)((M + 127) / 128), tiles_n = (C + bnt - 1) / bnt;
   dim3 grid(tiles_m * tiles_n);
