@@ -90,6 +90,10 @@ class TorchDistComm:
         if not torch_dist.is_initialized():
             os.environ.setdefault("MASTER_ADDR", env.master_addr)
             os.environ.setdefault("MASTER_PORT", str(env.master_port))
+            if env.master_addr in ("127.0.0.1", "localhost"):
+                # single-node: bind gloo to loopback (container hostnames
+                # may not resolve — see the torchrun 127.0.0.1 contract)
+                os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
             torch_dist.init_process_group(
                 backend=backend, init_method="env://",
                 rank=env.rank, world_size=env.world_size)
