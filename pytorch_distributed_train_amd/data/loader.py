@@ -1,28 +1,30 @@
 """Batch loader over a sampler.
 
 The reference uses torch DataLoader with worker processes + pinned memory
-(SURVEY.md §2b row 14). For synthetic data the bottleneck is batch assembly,
-not IO, so the MI355X-native loader is simpler: it materializes batches
-directly on the host as one tensor (optionally pinned) and overlaps H2D copy
-with compute via a dedicated copy stream in the trainer.
-
-For the bench path (`synthetic=True` prefab batches) it pre-builds one
-device-resident batch and reuses it — BASELINE.json's benchmark contract is
-synthetic data, and prefab batches keep the loader off the hot path entirely.
+(SURVEY.md §2b row 14). For synthetic data the cost is batch assembly
+(host randn), not disk IO, so the MI355X-native loader uses a background
+prefetch thread pool (torch ops release the GIL) filling a bounded queue
+of pinned batches; the trainer overlaps H2D copy with compute via
+non_blocking copies. workers=0 falls back to synchronous assembly.
 """
 from __future__ import annotations
+
+import queue
+import threading
 
 import torch
 
 
 class Loader:
     def __init__(self, dataset, sampler, batch_size: int, pin_memory: bool = False,
-                 drop_last: bool = True):
+                 drop_last: bool = True, workers: int = 0, prefetch: int = 4):
         self.dataset = dataset
         self.sampler = sampler
         self.batch_size = batch_size
         self.pin_memory = pin_memory
         self.drop_last = drop_last
+        self.workers = workers
+        self.prefetch = max(prefetch, 2)
 
     def __len__(self) -> int:
         n = len(self.sampler)
@@ -30,15 +32,15 @@ class Loader:
             return n // self.batch_size
         return (n + self.batch_size - 1) // self.batch_size
 
-    def __iter__(self):
+    def _batches(self):
         idxs = []
         for i in self.sampler:
             idxs.append(i)
             if len(idxs) == self.batch_size:
-                yield self._collate(idxs)
+                yield idxs
                 idxs = []
         if idxs and not self.drop_last:
-            yield self._collate(idxs)
+            yield idxs
 
     def _collate(self, idxs):
         imgs = []
@@ -52,3 +54,49 @@ class Loader:
             x = x.pin_memory()
             labels = labels.pin_memory()
         return x, labels
+
+    def __iter__(self):
+        if self.workers <= 0:
+            for idxs in self._batches():
+                yield self._collate(idxs)
+            return
+        yield from self._iter_prefetch()
+
+    def _iter_prefetch(self):
+        batch_q: "queue.Queue" = queue.Queue()
+        out_q: "queue.Queue" = queue.Queue(maxsize=self.prefetch)
+        batches = list(self._batches())
+        for bi, idxs in enumerate(batches):
+            batch_q.put((bi, idxs))
+        stop = threading.Event()
+        results: dict = {}
+        lock = threading.Lock()
+
+        def worker():
+            while not stop.is_set():
+                try:
+                    bi, idxs = batch_q.get_nowait()
+                except queue.Empty:
+                    return
+                data = self._collate(idxs)
+                with lock:
+                    results[bi] = data
+                out_q.put(bi)
+
+        threads = [threading.Thread(target=worker, daemon=True)
+                   for _ in range(self.workers)]
+        for t in threads:
+            t.start()
+        try:
+            # deliver in order (ranks must see identical batch sequence)
+            next_bi = 0
+            ready = set()
+            while next_bi < len(batches):
+                while next_bi not in ready:
+                    ready.add(out_q.get())
+                with lock:
+                    data = results.pop(next_bi)
+                yield data
+                next_bi += 1
+        finally:
+            stop.set()

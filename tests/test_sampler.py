@@ -62,3 +62,19 @@ def test_drop_last():
 def test_rank_out_of_range():
     with pytest.raises(ValueError):
         DistributedSampler(_dataset(8), 2, 2)
+
+
+def test_prefetch_loader_matches_sync():
+    from pytorch_distributed_train_amd.data import (DistributedSampler,
+                                                    Loader, make_dataset)
+    ds = make_dataset("synthetic-cifar10", 64, seed=3)
+    s1 = DistributedSampler(ds, 1, 0, shuffle=True, seed=5)
+    s2 = DistributedSampler(ds, 1, 0, shuffle=True, seed=5)
+    s1.set_epoch(1)
+    s2.set_epoch(1)
+    sync = list(Loader(ds, s1, 8, workers=0))
+    pref = list(Loader(ds, s2, 8, workers=4))
+    assert len(sync) == len(pref)
+    for (x1, y1), (x2, y2) in zip(sync, pref):
+        assert torch.equal(x1, x2)
+        assert torch.equal(y1, y2)

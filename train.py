@@ -84,7 +84,8 @@ def main(argv=None):
     sampler = DistributedSampler(dataset, env.world_size, env.rank,
                                  shuffle=True, seed=args.seed)
     loader = Loader(dataset, sampler, args.batch_size,
-                    pin_memory=(trainer.device.type == "cuda"))
+                    pin_memory=(trainer.device.type == "cuda"),
+                    workers=args.workers)
 
     start_epoch = 0
     if args.resume:
