@@ -758,7 +758,8 @@ template <typename T>
 __global__ void stem_im2col_k(const T* __restrict__ x, T* __restrict__ out,
                               int Nn, int H, int W, int Cin, int P, int Q,
                               int R, int S, int stride, int pad, int rsc_pad,
-                              unsigned mq, int sq, unsigned mpq, int spq) {
+                              unsigned mq, int sq, unsigned mpq, int spq,
+                              unsigned mc, int sc_, unsigned msc, int ssc) {
   // item = (row, 8-element group): consecutive threads write consecutive
   // 16-B chunks of one row (fully coalesced); row decode via magic
   // division (runtime-divisor div was ~40% of this kernel's issue slots)
@@ -782,9 +783,10 @@ __global__ void stem_im2col_k(const T* __restrict__ x, T* __restrict__ out,
       int e = grp * 8 + j;
       float val = 0.f;
       if (e < rsc) {
-        int c = e % Cin;
-        int t = e / Cin;
-        int sc = t % S, rr = t / S;
+        int t = (int)mdiv((unsigned)e, mc, sc_);      // e / Cin
+        int c = e - t * Cin;
+        int rr = (int)mdiv((unsigned)t, msc, ssc);    // t / S
+        int sc = t - rr * S;
         int h = h0 + rr, w = w0 + sc;
         if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
           val = to_f32(x[(((long)n * H + h) * W + w) * Cin + c]);
@@ -936,10 +938,12 @@ void pdta_stem_im2col(const void* x, void* out, int Nn, int H, int W,
                       int pad, int rsc_pad, int dt, hipStream_t s) {
   long total = (long)Nn * P * Q * (rsc_pad / 8);
   MagicDiv mq = magic_div(Q), mpq = magic_div(P * Q);
+  MagicDiv mc = magic_div(Cin), msc = magic_div(S);
   PDTA_DISPATCH_DTYPE16(dt, T, {
     stem_im2col_k<T><<<ew_grid(total, TPB, 1), TPB, 0, s>>>(
         (const T*)x, (T*)out, Nn, H, W, Cin, P, Q, R, S, stride, pad,
-        rsc_pad, mq.mul, mq.shift, mpq.mul, mpq.shift);
+        rsc_pad, mq.mul, mq.shift, mpq.mul, mpq.shift, mc.mul, mc.shift,
+        msc.mul, msc.shift);
   });
 }
 
