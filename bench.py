@@ -39,6 +39,9 @@ def parse(argv=None):
     p.add_argument("--grad-comm-dtype", default="same")
     p.add_argument("--sync-bn", action="store_true")
     p.add_argument("--local_rank", "--local-rank", type=int, default=None)
+    p.add_argument("--hip-graph", action="store_true",
+                   help="capture the step into a HIP graph and replay it "
+                        "(single-GPU)")
     return p.parse_args(argv)
 
 
@@ -74,13 +77,19 @@ def main(argv=None) -> int:
         if on_gpu:
             torch.cuda.synchronize()
 
+    use_graph = False
+    if b.hip_graph and env.world_size == 1 and on_gpu:
+        use_graph = trainer.enable_hip_graph(x, y)
+        if env.rank == 0 and not use_graph:
+            print("warning: hip-graph capture unavailable; eager steps",
+                  file=sys.stderr)
     for _ in range(b.warmup):
-        trainer.train_step(x, y)
+        trainer.graph_step() if use_graph else trainer.train_step(x, y)
     barrier_sync()
 
     t0 = time.perf_counter()
     for _ in range(b.steps):
-        trainer.train_step(x, y)
+        trainer.graph_step() if use_graph else trainer.train_step(x, y)
     if on_gpu:
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0

@@ -117,6 +117,50 @@ class Trainer:
         self.watchdog.heartbeat()
         return loss.detach()
 
+    # ---- hipGraph-captured step (single-GPU) ---------------------------
+    def enable_hip_graph(self, example_x, example_y) -> bool:
+        """Capture one whole training step (weight repack, forward, loss,
+        backward, fused SGD) into a HIP graph; graph_step() replays it.
+        Launch-bound overhead (200+ kernel launches/step) collapses into
+        one graph launch. Single-GPU only for now (the DDP bucket hooks
+        enqueue RCCL work from the autograd thread mid-capture)."""
+        if not isinstance(self.comm, NullComm) or self.scaler.enabled \
+                or self.device.type != "cuda":
+            return False
+        self._gx = example_x.clone()
+        self._gy = example_y.clone()
+
+        def body():
+            self.optimizer.zero_grad()
+            if self.packs is not None:
+                self.packs.refresh()
+            logits = self.model(self._gx)
+            loss = OF.softmax_cross_entropy(logits, self._gy,
+                                            self.args.label_smoothing)
+            loss.backward()
+            self.optimizer.step()
+            return loss.detach()
+
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                body()
+        torch.cuda.current_stream().wait_stream(s)
+        self._graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self._graph):
+            self._gloss = body()
+        return True
+
+    def graph_step(self, images=None, labels=None) -> torch.Tensor:
+        if images is not None:
+            self._gx.copy_(images, non_blocking=True)
+            self._gy.copy_(labels, non_blocking=True)
+        self._graph.replay()
+        self.global_step += 1
+        self.watchdog.heartbeat()
+        return self._gloss
+
     def state(self):
         return dict(model=self.model, optimizer=self.optimizer,
                     scaler=self.scaler, epoch=self.epoch,
