@@ -94,9 +94,12 @@ static inline MagicDiv magic_div(unsigned d) {
     m.shift = 31 + k;
     return m;
   }
+  // non-power-of-two: shift = 31 + ceil(log2 d) keeps the multiplier in
+  // 32 bits (a 32+L shift makes it 33 bits, which silently truncated);
+  // valid for n < 2^31 — all index spaces in this framework.
   int s = 0;
   while ((1u << s) < d) ++s;
-  m.shift = 32 + s;
+  m.shift = 31 + s;
   m.mul = (unsigned)(((1ull << m.shift) + d - 1) / d);
   return m;
 }

@@ -80,7 +80,9 @@ def _conv_ref(x, w, stride, pad):
 
 @pytest.mark.parametrize("n,h,c,k,r,stride", [
     (2, 14, 64, 64, 1, 1), (2, 14, 64, 128, 1, 2), (2, 14, 64, 64, 3, 1),
-    (2, 14, 128, 256, 3, 2), (1, 7, 2048, 512, 1, 1)])
+    (2, 14, 128, 256, 3, 2), (1, 7, 2048, 512, 1, 1),
+    # big-tile path (M >= 4096) with NON-power-of-two spatial dims
+    (2, 46, 64, 256, 3, 1), (3, 46, 64, 512, 1, 2), (2, 56, 128, 256, 1, 1)])
 def test_conv_fwd(n, h, c, k, r, stride):
     pad = r // 2
     x = _rand(n, h, h, c, seed=7)
@@ -91,7 +93,9 @@ def test_conv_fwd(n, h, c, k, r, stride):
 
 
 @pytest.mark.parametrize("n,h,c,k,r,stride", [
-    (2, 14, 64, 64, 3, 1), (2, 14, 64, 128, 1, 2), (2, 14, 128, 64, 3, 2)])
+    (2, 14, 64, 64, 3, 1), (2, 14, 64, 128, 1, 2), (2, 14, 128, 64, 3, 2),
+    # big-tile path with non-power-of-two spatial dims
+    (2, 46, 256, 64, 3, 1), (2, 46, 256, 128, 1, 2)])
 def test_conv_bwd_data(n, h, c, k, r, stride):
     import torch.nn.functional as F
     pad = r // 2
