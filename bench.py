@@ -39,9 +39,11 @@ def parse(argv=None):
     p.add_argument("--grad-comm-dtype", default="same")
     p.add_argument("--sync-bn", action="store_true")
     p.add_argument("--local_rank", "--local-rank", type=int, default=None)
-    p.add_argument("--hip-graph", action="store_true",
+    p.add_argument("--hip-graph", dest="hip_graph", action="store_true",
+                   default=True,
                    help="capture the step into a HIP graph and replay it "
-                        "(single-GPU)")
+                        "(single-GPU; default on, falls back to eager)")
+    p.add_argument("--no-hip-graph", dest="hip_graph", action="store_false")
     return p.parse_args(argv)
 
 
@@ -79,7 +81,12 @@ def main(argv=None) -> int:
 
     use_graph = False
     if b.hip_graph and env.world_size == 1 and on_gpu:
-        use_graph = trainer.enable_hip_graph(x, y)
+        try:
+            use_graph = trainer.enable_hip_graph(x, y)
+        except Exception as e:  # never let capture break the bench contract
+            print(f"warning: hip-graph capture failed ({e}); eager steps",
+                  file=sys.stderr)
+            use_graph = False
         if env.rank == 0 and not use_graph:
             print("warning: hip-graph capture unavailable; eager steps",
                   file=sys.stderr)
