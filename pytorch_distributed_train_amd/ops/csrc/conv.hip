@@ -747,6 +747,140 @@ __global__ void wrw_reduce_k(const float* __restrict__ part,
   }
 }
 
+// ---- big bwd-weight: 256x256 (ko x c) tile, 8 waves, 2-slot ring -------
+// For Kout % 256 == 0 && C % 256 == 0 (including the linear-dW TN GEMMs
+// routed through this path): the 64x64-tile kernel re-stages every row
+// once per output tile — 16x the staged bytes of this one. Tiles are
+// [64 rows][256 features] (512-B rows), both fragments hardware-transposed
+// (ds_read_b64_tr_b16). Partials land as 64x64 granules in the SAME
+// layout as the small kernel, so wrw_reduce_k is shared.
+template <typename T, bool GATHER>
+__global__ __launch_bounds__(512) void igemm_wrw_big_k(
+    const T* __restrict__ x, const T* __restrict__ dy,
+    float* __restrict__ part, const T* __restrict__ zero_page, int Nn,
+    int H, int W, int C, int P, int Q, int Kout, int R, int S, int stride,
+    int pad, long rows_per_chunk, int ko_tiles, int c_tiles,
+    unsigned mq, int sq, unsigned mpq, int spq) {
+  constexpr int BOFF = 64 * 256 * 2;  // one operand image per slot
+  __shared__ __attribute__((aligned(16))) char lds[2][2 * BOFF];
+
+  int nbk = ko_tiles / 4, nbc = c_tiles / 4;  // 256-wide big tiles
+  int tap = blockIdx.x / (nbk * nbc);
+  int rem = blockIdx.x % (nbk * nbc);
+  int bko = rem / nbc, bc = rem % nbc;
+  int r = tap / S, ss = tap % S;
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  int wm = wave >> 2, wn = wave & 3;  // 2 x 4 over (256 ko x 256 c)
+
+  long NPQ = (long)Nn * P * Q;
+  long r0 = (long)blockIdx.y * rows_per_chunk;
+  long r1 = min(NPQ, r0 + rows_per_chunk);
+  if (r0 >= r1) return;
+
+  auto stage_all = [&](long rowbase, int slot) {
+    // each glds: 64 lanes x 16B = 2 rows of 512 B; 4 glds/wave x 8 waves
+    // = the full 64-row image per operand
+    int lrow_in = lane >> 5;   // 0..1 within the glds' 2 rows
+    int c32 = lane & 31;       // 16-B chunk within the 512-B row
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int lrow = (wave * 4 + i) * 2 + lrow_in;
+      long row = rowbase + lrow;
+      int swz = (c32 ^ (lrow & 7)) * 16;
+      const char* g;
+      if (row < r1)
+        g = reinterpret_cast<const char*>(dy) +
+            (row * Kout + bko * 256) * 2 + swz;
+      else
+        g = reinterpret_cast<const char*>(zero_page);
+      glds16(g, (char*)lds[slot] + (wave * 4 + i) * 1024);
+    }
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int lrow = (wave * 4 + i) * 2 + lrow_in;
+      long row = rowbase + lrow;
+      int swz = (c32 ^ (lrow & 7)) * 16;
+      const char* g = reinterpret_cast<const char*>(zero_page);
+      if (row < r1) {
+        if (GATHER) {
+          unsigned npq = mdiv((unsigned)row, mpq, spq);
+          unsigned rm = (unsigned)row - npq * (unsigned)(P * Q);
+          unsigned pp = mdiv(rm, mq, sq);
+          unsigned qq = rm - pp * (unsigned)Q;
+          int h = (int)pp * stride - pad + r;
+          int w = (int)qq * stride - pad + ss;
+          if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
+            g = reinterpret_cast<const char*>(x) +
+                ((((long)npq * H + h) * W + w) * C + bc * 256) * 2 + swz;
+        } else {
+          g = reinterpret_cast<const char*>(x) +
+              (row * C + bc * 256) * 2 + swz;
+        }
+      }
+      glds16(g, (char*)lds[slot] + BOFF + (wave * 4 + i) * 1024);
+    }
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  stage_all(r0, 0);
+  unsigned dy_base[2] = {lds_addr_of(lds[0]), lds_addr_of(lds[1])};
+  for (long rb = r0; rb < r1; rb += 64) {
+    int slot = (int)((rb - r0) >> 6) & 1;
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (rb + 64 < r1) stage_all(rb + 64, slot ^ 1);
+    unsigned la = dy_base[slot];
+    unsigned lb = dy_base[slot] + BOFF;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      unsigned long long alo[8], ahi[8], blo[4], bhi[4];
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        frag_read_tr_issue_s<512>(la, kk * 32, wm * 128 + i * 16, lane,
+                                  &alo[i], &ahi[i]);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        frag_read_tr_issue_s<512>(lb, kk * 32, wn * 64 + j * 16, lane,
+                                  &blo[j], &bhi[j]);
+      tr_read_wait();
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        s16x8 af = tr_pair_to_frag(alo[i], ahi[i]);
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = MfmaOp<T>::run(af, tr_pair_to_frag(blo[j], bhi[j]),
+                                     acc[i][j]);
+      }
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // write 4x4 granules of 64x64 fp32 in the shared wrw partial layout
+  int tiles_total = ko_tiles * c_tiles * R * S;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    int ko_local = wm * 128 + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int c_local = wn * 64 + j * 16 + (lane & 15);
+#pragma unroll
+      for (int rg = 0; rg < 4; ++rg) {
+        int kol = ko_local + rg;
+        int t64 = tap * ko_tiles * c_tiles +
+                  (bko * 4 + (kol >> 6)) * c_tiles + (bc * 4 + (c_local >> 6));
+        float* out =
+            part + ((long)blockIdx.y * tiles_total + t64) * 4096;
+        out[(kol & 63) * 64 + (c_local & 63)] = acc[i][j][rg];
+      }
+    }
+  }
+}
+
 // ------------------------------------------------------------- stem ----
 // The C=3 stem (7x7/s2) goes through explicit im2col into the SAME MFMA
 // igemm path as every other conv: xcol[npq][rsc_pad] with rsc = R*S*Cin
@@ -896,7 +1030,10 @@ void pdta_wrw_plan(int Nn, int P, int Q, int Kout, int C, int R, int S,
   long NPQ = (long)Nn * P * Q;
   int ko_tiles = (Kout + 63) / 64, c_tiles = (C + 63) / 64;
   int tiles = ko_tiles * c_tiles * R * S;
-  long splitk = 4096 / tiles;
+  long launch_tiles = tiles;
+  if (Kout % 256 == 0 && C % 256 == 0)  // big-tile path
+    launch_tiles = (long)(ko_tiles / 4) * (c_tiles / 4) * R * S;
+  long splitk = 4096 / launch_tiles;
   if (splitk < 1) splitk = 1;
   if (splitk > 256) splitk = 256;
   long rows_per_chunk = ((NPQ + splitk - 1) / splitk + 63) / 64 * 64;
@@ -916,18 +1053,38 @@ void pdta_conv_bwd_weight(const void* x, const void* dy, float* dw,
   pdta_wrw_plan(Nn, P, Q, Kout, C, R, S, &tiles, &chunks);
   long rows_per_chunk = ((NPQ + chunks - 1) / chunks + 63) / 64 * 64;
   chunks = (NPQ + rows_per_chunk - 1) / rows_per_chunk;
-  dim3 grid((unsigned)tiles, (unsigned)chunks);
   bool identity = (R == 1 && S == 1 && stride == 1 && pad == 0);
-  PDTA_DISPATCH_DTYPE16(dt, T, {
-    if (identity)
-      igemm_wrw_k<T, false><<<grid, TPB, 0, s>>>(
-          (const T*)x, (const T*)dy, part, (const T*)zero_page, Nn, H, W, C,
-          P, Q, Kout, R, S, stride, pad, rows_per_chunk, ko_tiles, c_tiles);
-    else
-      igemm_wrw_k<T, true><<<grid, TPB, 0, s>>>(
-          (const T*)x, (const T*)dy, part, (const T*)zero_page, Nn, H, W, C,
-          P, Q, Kout, R, S, stride, pad, rows_per_chunk, ko_tiles, c_tiles);
-  });
+  if (Kout % 256 == 0 && C % 256 == 0) {
+    long big_tiles = (long)(ko_tiles / 4) * (c_tiles / 4) * R * S;
+    dim3 grid((unsigned)big_tiles, (unsigned)chunks);
+    MagicDiv mq = magic_div(Q), mpq = magic_div(P * Q);
+    PDTA_DISPATCH_DTYPE16(dt, T, {
+      if (identity)
+        igemm_wrw_big_k<T, false><<<grid, 512, 0, s>>>(
+            (const T*)x, (const T*)dy, part, (const T*)zero_page, Nn, H, W,
+            C, P, Q, Kout, R, S, stride, pad, rows_per_chunk, ko_tiles,
+            c_tiles, mq.mul, mq.shift, mpq.mul, mpq.shift);
+      else
+        igemm_wrw_big_k<T, true><<<grid, 512, 0, s>>>(
+            (const T*)x, (const T*)dy, part, (const T*)zero_page, Nn, H, W,
+            C, P, Q, Kout, R, S, stride, pad, rows_per_chunk, ko_tiles,
+            c_tiles, mq.mul, mq.shift, mpq.mul, mpq.shift);
+    });
+  } else {
+    dim3 grid((unsigned)tiles, (unsigned)chunks);
+    PDTA_DISPATCH_DTYPE16(dt, T, {
+      if (identity)
+        igemm_wrw_k<T, false><<<grid, TPB, 0, s>>>(
+            (const T*)x, (const T*)dy, part, (const T*)zero_page, Nn, H, W,
+            C, P, Q, Kout, R, S, stride, pad, rows_per_chunk, ko_tiles,
+            c_tiles);
+      else
+        igemm_wrw_k<T, true><<<grid, TPB, 0, s>>>(
+            (const T*)x, (const T*)dy, part, (const T*)zero_page, Nn, H, W,
+            C, P, Q, Kout, R, S, stride, pad, rows_per_chunk, ko_tiles,
+            c_tiles);
+    });
+  }
   long total = tiles * 4096;
   wrw_reduce_k<<<ew_grid(total, 256, 1), 256, 0, s>>>(
       part, dw, (int)chunks, (int)tiles, ko_tiles, c_tiles, C, Kout, R, S);

@@ -76,6 +76,31 @@ DEV_INLINE void frag_read_tr_issue(unsigned lds_base, int k0, int m0,
       : "v"(a0), "v"(a1));
 }
 
+// Stride-parameterized variant: [k][m] LDS image with ROW_BYTES-wide rows
+// (e.g. 512 B for 256-feature tiles) and the chunk^(k&7) swizzle on 16-B
+// chunks within the row.
+template <int ROW_BYTES>
+DEV_INLINE void frag_read_tr_issue_s(unsigned lds_base, int k0, int m0,
+                                     int lane, unsigned long long* lo,
+                                     unsigned long long* hi) {
+  int g = lane >> 4;
+  int row = (lane & 15) >> 2;
+  int seg = lane & 3;
+  int byte0 = m0 * 2 + seg * 8;
+  int chunk = byte0 >> 4, rem = byte0 & 15;
+  int ka = k0 + g * 8 + row;
+  int kb = ka + 4;
+  unsigned a0 =
+      lds_base + ka * ROW_BYTES + (((chunk ^ (ka & 7)) << 4) | rem);
+  unsigned a1 =
+      lds_base + kb * ROW_BYTES + (((chunk ^ (kb & 7)) << 4) | rem);
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3"
+      : "=&v"(*lo), "=&v"(*hi)
+      : "v"(a0), "v"(a1));
+}
+
 // Wait for every outstanding LDS read, then fence the scheduler so no
 // consumer (incl. register-only MFMA — guide §5.4 rule 18) is hoisted
 // above the wait.

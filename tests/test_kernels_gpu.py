@@ -111,7 +111,9 @@ def test_conv_bwd_data(n, h, c, k, r, stride):
 
 
 @pytest.mark.parametrize("n,h,c,k,r,stride", [
-    (2, 14, 64, 64, 3, 1), (2, 14, 64, 128, 1, 2), (2, 8, 128, 128, 3, 2)])
+    (2, 14, 64, 64, 3, 1), (2, 14, 64, 128, 1, 2), (2, 8, 128, 128, 3, 2),
+    # big-tile wrw (Kout,C % 256) incl. gather + non-pow2 spatial
+    (2, 20, 256, 256, 3, 1), (2, 23, 256, 512, 1, 2), (1, 30, 512, 256, 1, 1)])
 def test_conv_bwd_weight(n, h, c, k, r, stride):
     pad = r // 2
     p = (h + 2 * pad - r) // stride + 1
