@@ -1024,16 +1024,29 @@ void pdta_conv_bwd_data(const void* dy, const void* w2, void* dx,
   });
 }
 
+// big-tile wrw pays off only with long reductions and enough big tiles
+int pdta_wrw_use_big(int Nn, int P, int Q, int Kout, int C, int R, int S) {
+  if (Kout % 256 != 0 || C % 256 != 0) return 0;
+  long NPQ = (long)Nn * P * Q;
+  long big_tiles = (long)(Kout / 256) * (C / 256) * R * S;
+  return (NPQ >= 16384 && big_tiles >= 16) ? 1 : 0;
+}
+
 // planning: number of split-K chunks and tiles for the workspace
 void pdta_wrw_plan(int Nn, int P, int Q, int Kout, int C, int R, int S,
                    long* tiles_out, long* chunks_out) {
   long NPQ = (long)Nn * P * Q;
   int ko_tiles = (Kout + 63) / 64, c_tiles = (C + 63) / 64;
   int tiles = ko_tiles * c_tiles * R * S;
-  long launch_tiles = tiles;
-  if (Kout % 256 == 0 && C % 256 == 0)  // big-tile path
-    launch_tiles = (long)(ko_tiles / 4) * (c_tiles / 4) * R * S;
-  long splitk = 4096 / launch_tiles;
+  long splitk;
+  if (pdta_wrw_use_big(Nn, P, Q, Kout, C, R, S)) {
+    // 512-thread big-tile blocks: fewer of them, each must carry a long
+    // reduction (tiny chunks made the epilogue slab dominate)
+    long big_tiles = (long)(ko_tiles / 4) * (c_tiles / 4) * R * S;
+    splitk = 1024 / big_tiles;
+  } else {
+    splitk = 4096 / tiles;
+  }
   if (splitk < 1) splitk = 1;
   if (splitk > 256) splitk = 256;
   long rows_per_chunk = ((NPQ + splitk - 1) / splitk + 63) / 64 * 64;
@@ -1054,7 +1067,7 @@ void pdta_conv_bwd_weight(const void* x, const void* dy, float* dw,
   long rows_per_chunk = ((NPQ + chunks - 1) / chunks + 63) / 64 * 64;
   chunks = (NPQ + rows_per_chunk - 1) / rows_per_chunk;
   bool identity = (R == 1 && S == 1 && stride == 1 && pad == 0);
-  if (Kout % 256 == 0 && C % 256 == 0) {
+  if (pdta_wrw_use_big(Nn, P, Q, Kout, C, R, S)) {
     long big_tiles = (long)(ko_tiles / 4) * (c_tiles / 4) * R * S;
     dim3 grid((unsigned)big_tiles, (unsigned)chunks);
     MagicDiv mq = magic_div(Q), mpq = magic_div(P * Q);
