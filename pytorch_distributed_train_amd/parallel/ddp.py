@@ -33,6 +33,20 @@ from .flat import FlatParams
 from .comm import NullComm
 
 
+class _Work_then:
+    def __init__(self, work, post):
+        self._work, self._post = work, post
+
+    def wait(self):
+        self._work.wait()
+        if self._post is not None:
+            self._post()
+            self._post = None
+
+    def sync_stream(self, stream=None):
+        self.wait()
+
+
 class _Bucket:
     __slots__ = ("index", "start", "end", "param_ids", "pending", "work")
 
@@ -76,9 +90,10 @@ class GradReducer:
         self._shadow = None
         if self.enabled and flat.master.is_cuda:
             self._comm_stream = torch.cuda.Stream(device=flat.device)
-            if comm_dtype is not None and comm_dtype != torch.float32:
-                self._shadow = torch.zeros(flat.total, dtype=comm_dtype,
-                                           device=flat.device)
+        if self.enabled and comm_dtype is not None \
+                and comm_dtype != torch.float32:
+            self._shadow = torch.zeros(flat.total, dtype=comm_dtype,
+                                       device=flat.device)
 
         self._hooks = []
         if self.enabled:
@@ -127,6 +142,11 @@ class GradReducer:
             else:
                 b.work = self.comm.allreduce_avg(view,
                                                  stream=self._comm_stream)
+        elif self._shadow is not None:  # CPU compressed path (tests)
+            shadow = self._shadow[b.start:b.end]
+            shadow.copy_(view)
+            w = self.comm.allreduce_avg(shadow)
+            b.work = _Work_then(w, lambda: view.copy_(shadow))
         else:
             b.work = self.comm.allreduce_avg(view)
 
