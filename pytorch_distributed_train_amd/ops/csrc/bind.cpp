@@ -476,7 +476,8 @@ std::vector<at::Tensor> ln_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
   auto dx = at::empty_like(x);
   auto opts = x.options().dtype(at::kFloat);
   int chunks = pdta_ln_chunks(rows);
-  auto part = at::empty({(long)chunks * 2 * d}, opts);
+  // +16 rows: scratch for the kernel-side two-level finalize
+  auto part = at::empty({(long)(chunks + 16) * 2 * d}, opts);
   auto dw = at::empty({d}, opts), db = at::empty({d}, opts);
   auto wf = w.to(at::kFloat).contiguous();
   pdta_ln_bwd(x.data_ptr(), dy.data_ptr(), dx.data_ptr(),

@@ -10,7 +10,11 @@ namespace {
 constexpr int TPB = 256;  // 4 waves -> 4 rows per block
 
 // ---- LayerNorm ----------------------------------------------------------
-template <typename T>
+// NG = number of 512-column groups covering d (register arrays are sized
+// exactly: a fixed max-size version cost 128 VGPRs of dead partials and
+// single-wave occupancy — 6x off the HBM traffic floor in rocprof).
+// The row is loaded ONCE into registers and reused for the apply pass.
+template <typename T, int NG>
 __global__ void ln_fwd_k(const T* __restrict__ x, T* __restrict__ y,
                          const float* __restrict__ w,
                          const float* __restrict__ b,
@@ -21,14 +25,19 @@ __global__ void ln_fwd_k(const T* __restrict__ x, T* __restrict__ y,
   if (row >= rows) return;
   int lane = threadIdx.x & 63;
   const T* xr = x + row * d;
+  Vec8<T> xv[NG];
   float s = 0.f, ss = 0.f;
-  for (int c = lane * 8; c < d; c += 64 * 8) {
-    Vec8<T> v = load8(xr + c);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      float f = to_f32(v.v[j]);
-      s += f;
-      ss += f * f;
+  for (int g = 0; g < NG; ++g) {
+    int c = lane * 8 + g * 512;
+    if (c < d) {
+      xv[g] = load8(xr + c);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = to_f32(xv[g].v[j]);
+        s += f;
+        ss += f * f;
+      }
     }
   }
   s = wave_reduce_sum(s);
@@ -41,12 +50,25 @@ __global__ void ln_fwd_k(const T* __restrict__ x, T* __restrict__ y,
     invstd_out[row] = is;
   }
   T* yr = y + row * d;
-  for (int c = lane * 8; c < d; c += 64 * 8) {
-    Vec8<T> v = load8(xr + c);
 #pragma unroll
-    for (int j = 0; j < 8; ++j)
-      v.v[j] = from_f32<T>((to_f32(v.v[j]) - mu) * is * w[c + j] + b[c + j]);
-    store8(yr + c, v);
+  for (int g = 0; g < NG; ++g) {
+    int c = lane * 8 + g * 512;
+    if (c < d) {
+      float4 w0 = *reinterpret_cast<const float4*>(w + c);
+      float4 w1 = *reinterpret_cast<const float4*>(w + c + 4);
+      float4 b0 = *reinterpret_cast<const float4*>(b + c);
+      float4 b1 = *reinterpret_cast<const float4*>(b + c + 4);
+      const float* wp = reinterpret_cast<const float*>(&w0);
+      const float* bp = reinterpret_cast<const float*>(&b0);
+      Vec8<T> o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float wj = j < 4 ? wp[j] : reinterpret_cast<const float*>(&w1)[j - 4];
+        float bj = j < 4 ? bp[j] : reinterpret_cast<const float*>(&b1)[j - 4];
+        o.v[j] = from_f32<T>((to_f32(xv[g].v[j]) - mu) * is * wj + bj);
+      }
+      store8(yr + c, o);
+    }
   }
 }
 
@@ -56,7 +78,7 @@ __global__ void ln_fwd_k(const T* __restrict__ x, T* __restrict__ y,
 // per chunk, written to part [chunk][2d]; ln_bwd_final reduces chunks
 // with 64-col x 4-slice blocks (a d-threads-only finalize was one wave
 // chip-wide and chunk-serial).
-template <typename T>
+template <typename T, int NG>
 __global__ void ln_bwd_k(const T* __restrict__ x, const T* __restrict__ dy,
                          T* __restrict__ dx, const float* __restrict__ w,
                          const float* __restrict__ mean,
@@ -70,57 +92,75 @@ __global__ void ln_bwd_k(const T* __restrict__ x, const T* __restrict__ dy,
   long r1 = min(rows, r0 + rows_per_chunk);
   int lane = threadIdx.x & 63;
   int wv = threadIdx.x >> 6;
-  constexpr int MAXG = 8;           // supports d up to 4096
-  int ngroups = (d + 511) / 512;
-  float adw[MAXG][8], adb[MAXG][8];
+  float adw[NG][8], adb[NG][8];
+  float wreg[NG][8];  // w is row-invariant: load once per kernel
 #pragma unroll
-  for (int g = 0; g < MAXG; ++g)
+  for (int g = 0; g < NG; ++g) {
+    int c = lane * 8 + g * 512;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       adw[g][j] = 0.f;
       adb[g][j] = 0.f;
+      wreg[g][j] = 0.f;
     }
+    if (c < d) {
+      float4 w0 = *reinterpret_cast<const float4*>(w + c);
+      float4 w1 = *reinterpret_cast<const float4*>(w + c + 4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        wreg[g][j] = reinterpret_cast<const float*>(&w0)[j];
+        wreg[g][j + 4] = reinterpret_cast<const float*>(&w1)[j];
+      }
+    }
+  }
 
   for (long row = r0 + wv; row < r1; row += 4) {
     const T* xr = x + row * d;
     const T* gr = dy + row * d;
     float mu = mean[row], is = invstd[row];
+    Vec8<T> gv[NG], xv[NG];  // row cached in registers: one HBM read
     float s1 = 0.f, s2 = 0.f;  // sum(dy*w), sum(dy*w*xhat)
-    for (int c = lane * 8; c < d; c += 512) {
-      Vec8<T> g = load8(gr + c);
-      Vec8<T> v = load8(xr + c);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float gw = to_f32(g.v[j]) * w[c + j];
-        float xh = (to_f32(v.v[j]) - mu) * is;
-        s1 += gw;
-        s2 += gw * xh;
+    for (int g = 0; g < NG; ++g) {
+      int c = lane * 8 + g * 512;
+      if (c < d) {
+        gv[g] = load8(gr + c);
+        xv[g] = load8(xr + c);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float gw = to_f32(gv[g].v[j]) * wreg[g][j];
+          float xh = (to_f32(xv[g].v[j]) - mu) * is;
+          s1 += gw;
+          s2 += gw * xh;
+        }
       }
     }
     s1 = wave_reduce_sum(s1) / d;
     s2 = wave_reduce_sum(s2) / d;
     T* dr = dx + row * d;
-    int g = 0;
-    for (int c = lane * 8; c < d; c += 512, ++g) {
-      Vec8<T> gv = load8(gr + c);
-      Vec8<T> v = load8(xr + c);
-      Vec8<T> o;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float gg = to_f32(gv.v[j]);
-        float xh = (to_f32(v.v[j]) - mu) * is;
-        o.v[j] = from_f32<T>(is * (gg * w[c + j] - s1 - xh * s2));
-        adw[g][j] += gg * xh;
-        adb[g][j] += gg;
+    for (int g = 0; g < NG; ++g) {
+      int c = lane * 8 + g * 512;
+      if (c < d) {
+        Vec8<T> o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float gg = to_f32(gv[g].v[j]);
+          float xh = (to_f32(xv[g].v[j]) - mu) * is;
+          o.v[j] = from_f32<T>(is * (gg * wreg[g][j] - s1 - xh * s2));
+          adw[g][j] += gg * xh;
+          adb[g][j] += gg;
+        }
+        store8(dr + c, o);
       }
-      store8(dr + c, o);
     }
   }
   // combine the 4 waves through LDS (no atomics), one pass per chunk
   float* my = red + wv * 2 * d;
-  {
-    int g = 0;
-    for (int c = lane * 8; c < d; c += 512, ++g)
+#pragma unroll
+  for (int g = 0; g < NG; ++g) {
+    int c = lane * 8 + g * 512;
+    if (c < d)
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         my[c + j] = adw[g][j];
@@ -131,6 +171,36 @@ __global__ void ln_bwd_k(const T* __restrict__ x, const T* __restrict__ dy,
   float* out = part + (long)blockIdx.x * 2 * d;
   for (int i = threadIdx.x; i < 2 * d; i += blockDim.x)
     out[i] = red[i] + red[2 * d + i] + red[4 * d + i] + red[6 * d + i];
+}
+
+// Sums chunk rows s, s+nslab, s+2*nslab, ... of part into out[s][2d]
+// (first level of the two-level finalize: with ~1000 chunks a single
+// d/64-block finalize was 12 blocks chip-wide and latency-serial, 77 us).
+__global__ void ln_bwd_psum_k(const float* __restrict__ part, int chunks,
+                              int d, int nslab, float* __restrict__ out) {
+  __shared__ float red[2][256];
+  int c = blockIdx.x * 64 + (int)(threadIdx.x % 64);
+  int slice = threadIdx.x / 64;
+  int s = blockIdx.y;
+  float sw = 0.f, sb = 0.f;
+  if (c < d) {
+    for (int k = s + slice * nslab; k < chunks; k += 4 * nslab) {
+      sw += part[(long)k * 2 * d + c];
+      sb += part[(long)k * 2 * d + d + c];
+    }
+  }
+  red[0][threadIdx.x] = sw;
+  red[1][threadIdx.x] = sb;
+  __syncthreads();
+  if (slice == 0 && c < d) {
+#pragma unroll
+    for (int i = 1; i < 4; ++i) {
+      sw += red[0][threadIdx.x + i * 64];
+      sb += red[1][threadIdx.x + i * 64];
+    }
+    out[(long)s * 2 * d + c] = sw;
+    out[(long)s * 2 * d + d + c] = sb;
+  }
 }
 
 __global__ void ln_bwd_final_k(const float* __restrict__ part, int chunks,
@@ -299,13 +369,31 @@ __global__ void softmax_bwd_k(const T* __restrict__ p, const T* __restrict__ dp,
 
 extern "C" {
 
+// register-array groups covering d (columns handled per thread = NG*8)
+#define PDTA_LN_DISPATCH_NG(d, NG, ...)        \
+  if ((d) <= 512) {                            \
+    constexpr int NG = 1;                      \
+    __VA_ARGS__                                \
+  } else if ((d) <= 1024) {                    \
+    constexpr int NG = 2;                      \
+    __VA_ARGS__                                \
+  } else if ((d) <= 2048) {                    \
+    constexpr int NG = 4;                      \
+    __VA_ARGS__                                \
+  } else {                                     \
+    constexpr int NG = 8;                      \
+    __VA_ARGS__                                \
+  }
+
 void pdta_ln_fwd(const void* x, void* y, const float* w, const float* b,
                  float* mean, float* invstd, long rows, int d, float eps,
                  int dt, hipStream_t s) {
   long blocks = (rows + 3) / 4;
   PDTA_DISPATCH_DTYPE16(dt, T, {
-    ln_fwd_k<T><<<(unsigned)blocks, TPB, 0, s>>>(
-        (const T*)x, (T*)y, w, b, mean, invstd, rows, d, eps);
+    PDTA_LN_DISPATCH_NG(d, NG, {
+      ln_fwd_k<T, NG><<<(unsigned)blocks, TPB, 0, s>>>(
+          (const T*)x, (T*)y, w, b, mean, invstd, rows, d, eps);
+    });
   });
 }
 
@@ -324,11 +412,23 @@ void pdta_ln_bwd(const void* x, const void* dy, void* dx, const float* w,
   long rpc = (rows + chunks - 1) / chunks;
   int lds = 4 * 2 * d * sizeof(float);
   PDTA_DISPATCH_DTYPE16(dt, T, {
-    ln_bwd_k<T><<<chunks, TPB, lds, s>>>((const T*)x, (const T*)dy, (T*)dx,
-                                         w, mean, invstd, part, rows, d,
-                                         rpc);
+    PDTA_LN_DISPATCH_NG(d, NG, {
+      ln_bwd_k<T, NG><<<chunks, TPB, lds, s>>>(
+          (const T*)x, (const T*)dy, (T*)dx, w, mean, invstd, part, rows, d,
+          rpc);
+    });
   });
-  ln_bwd_final_k<<<(d + 63) / 64, 256, 0, s>>>(part, chunks, d, dw, db);
+  // two-level finalize for long chunk lists (part carries 16 extra rows —
+  // bind.cpp ln_bwd allocates chunks+16)
+  constexpr int NSLAB = 16;
+  if (chunks > 64) {
+    float* p2 = part + (long)chunks * 2 * d;
+    dim3 g1((d + 63) / 64, NSLAB);
+    ln_bwd_psum_k<<<g1, 256, 0, s>>>(part, chunks, d, NSLAB, p2);
+    ln_bwd_final_k<<<(d + 63) / 64, 256, 0, s>>>(p2, NSLAB, d, dw, db);
+  } else {
+    ln_bwd_final_k<<<(d + 63) / 64, 256, 0, s>>>(part, chunks, d, dw, db);
+  }
 }
 
 void pdta_gelu_fwd(const void* x, void* y, long n, int dt, hipStream_t s) {
