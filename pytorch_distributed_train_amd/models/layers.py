@@ -47,6 +47,16 @@ class Conv2dNHWC(nn.Module):
             w2 = w.permute(2, 3, 1, 0).to(x.dtype).contiguous()
         else:
             w2 = self._pack_w2
+        # emit_bn_stats (set by the model builder on conv->BN pairs): the
+        # fwd kernel's epilogue writes per-tile channel (sum, sumsq)
+        # partials so the BN that follows never re-reads y for statistics
+        if (getattr(self, "emit_bn_stats", False) and self.training
+                and x.is_cuda and self.bias is None
+                and torch.is_grad_enabled()):
+            y, part = OF.conv2d_nhwc(x, w, wt, w2, self.stride,
+                                     self.padding, want_bn_stats=True)
+            y._pdta_bn_part = part
+            return y
         y = OF.conv2d_nhwc(x, w, wt, w2, self.stride, self.padding)
         if self.bias is not None:
             y = y + self.bias.to(y.dtype)  # broadcast over trailing C
@@ -92,10 +102,18 @@ class BatchNorm2dNHWC(nn.Module):
         # residual: fused ResNet join y = relu(bn(x) + residual)
         if self.training:
             if x.is_cuda and self.comm is None:
-                # fast path: stats + running update in one kernel pair
-                mean, invstd = OF.bn_stats_train(
-                    x, self.running_mean, self.running_var, self.momentum,
-                    self.eps)
+                part = getattr(x, "_pdta_bn_part", None)
+                if part is not None:
+                    # stats pre-reduced by the producing conv's epilogue
+                    rows = x.shape[0] * x.shape[1] * x.shape[2]
+                    mean, invstd = OF.bn_stats_train_from_part(
+                        part, rows, self.running_mean, self.running_var,
+                        self.momentum, self.eps)
+                else:
+                    # fast path: stats + running update in one kernel pair
+                    mean, invstd = OF.bn_stats_train(
+                        x, self.running_mean, self.running_var,
+                        self.momentum, self.eps)
                 with torch.no_grad():
                     self.num_batches_tracked += 1
                 return OF.bn_train(x, self.weight, self.bias, mean, invstd,

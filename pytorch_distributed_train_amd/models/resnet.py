@@ -96,6 +96,18 @@ class ResNet(nn.Module):
         self.layer4 = self._make_layer(block, 512, layers[3], stride=2)
         self.avgpool = GlobalAvgPoolNHWC()
         self.fc = Linear(512 * block.expansion, num_classes)
+        # every conv feeds a BN: its fwd epilogue pre-reduces BN statistics
+        for m in self.modules():
+            for cname, bname in (("conv1", "bn1"), ("conv2", "bn2"),
+                                 ("conv3", "bn3")):
+                conv = getattr(m, cname, None)
+                if isinstance(conv, Conv2dNHWC) and \
+                        isinstance(getattr(m, bname, None), BatchNorm2dNHWC):
+                    conv.emit_bn_stats = True
+            if isinstance(m, nn.Sequential) and len(m) == 2 and \
+                    isinstance(m[0], Conv2dNHWC) and \
+                    isinstance(m[1], BatchNorm2dNHWC):
+                m[0].emit_bn_stats = True  # downsample pairs
 
     def _make_layer(self, block, planes, blocks, stride=1):
         downsample = None

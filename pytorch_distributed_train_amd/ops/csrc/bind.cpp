@@ -44,9 +44,12 @@ void pdta_ce_bwd(const void*, const long*, const float*, const float*, void*,
 void pdta_gemm_nt(const void*, const void*, void*, const float*, int, int,
                   int, int, int, hipStream_t);
 void pdta_colsum_f32(const void*, float*, long, int, int, hipStream_t);
+long pdta_conv_fwd_chunks(long, int);
+void pdta_bn_stats_from_part(float*, int, float*, float*, float*, float*,
+                             float, float, int, long, int, hipStream_t);
 void pdta_conv_fwd(const void*, const void*, void*, const void*, int, int,
                    int, int, int, int, int, int, int, int, int, int,
-                   hipStream_t);
+                   hipStream_t, float*);
 void pdta_conv_bwd_data(const void*, const void*, void*, const void*, int,
                         int, int, int, int, int, int, int, int, int, int,
                         int, hipStream_t);
@@ -351,9 +354,13 @@ at::Tensor colsum_f32(at::Tensor dy) {
 }
 
 // ---- conv ----------------------------------------------------------------
-at::Tensor conv2d_fwd(at::Tensor x, at::Tensor wt, int64_t stride,
-                      int64_t pad) {
-  // x [N,H,W,C]; wt pack [Kout,R,S,C]
+std::vector<at::Tensor> conv2d_fwd(at::Tensor x, at::Tensor wt,
+                                   int64_t stride, int64_t pad,
+                                   bool want_bn_stats) {
+  // x [N,H,W,C]; wt pack [Kout,R,S,C]. want_bn_stats: the fwd kernel's
+  // epilogue also emits per-m-tile (sum, sumsq) channel partials of y so
+  // the following BatchNorm skips its stage-1 read of y
+  // (bn_stats_train_from_part finalizes them).
   CHECK_IN(x);
   CHECK_IN(wt);
   int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
@@ -361,10 +368,18 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor wt, int64_t stride,
   TORCH_CHECK(wt.size(3) == C, "conv: channel mismatch");
   int P = conv_out(H, R, stride, pad), Q = conv_out(W, S, stride, pad);
   auto y = at::empty({N, P, Q, Kout}, x.options());
+  at::Tensor part;
+  float* part_ptr = nullptr;
+  if (want_bn_stats) {
+    long chunks = pdta_conv_fwd_chunks((long)N * P * Q, Kout);
+    part = at::empty({(chunks + 64) * 2 * Kout},
+                     x.options().dtype(at::kFloat));
+    part_ptr = part.data_ptr<float>();
+  }
   if (C % 64 == 0) {
     pdta_conv_fwd(x.data_ptr(), wt.data_ptr(), y.data_ptr(), zero_page(x), N,
                   H, W, C, P, Q, Kout, R, S, stride, pad, dt_of(x),
-                  cur_stream());
+                  cur_stream(), part_ptr);
   } else {
     // stem (small C): im2col into the SAME igemm path (1x1 over xcol)
     int rsc = R * S * C, rsc_pad = (rsc + 63) / 64 * 64;
@@ -375,9 +390,30 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor wt, int64_t stride,
                                       {0, rsc_pad - rsc}).contiguous();
     pdta_conv_fwd(xcol.data_ptr(), wt_pad.data_ptr(), y.data_ptr(),
                   zero_page(x), N, P, Q, rsc_pad, P, Q, Kout, 1, 1, 1, 0,
-                  dt_of(x), cur_stream());
+                  dt_of(x), cur_stream(), part_ptr);
   }
-  return y;
+  if (want_bn_stats) return {y, part};
+  return {y};
+}
+
+std::vector<at::Tensor> bn_stats_train_from_part(at::Tensor part, long rows,
+                                                 at::Tensor running_mean,
+                                                 at::Tensor running_var,
+                                                 double momentum,
+                                                 double eps) {
+  // (mean, invstd) from a conv2d_fwd(want_bn_stats=True) partial slab;
+  // updates running stats in-kernel (mirrors bn_stats_train)
+  CHECK_IN(part);
+  int C = running_mean.size(0);
+  int chunks = (int)(part.numel() / (2 * C)) - 64;
+  auto opts = part.options();
+  auto mean = at::empty({C}, opts), invstd = at::empty({C}, opts);
+  pdta_bn_stats_from_part(part.data_ptr<float>(), chunks,
+                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                          running_mean.data_ptr<float>(),
+                          running_var.data_ptr<float>(), momentum, eps, 1,
+                          rows, C, cur_stream());
+  return {mean, invstd};
 }
 
 at::Tensor conv2d_bwd_data(at::Tensor dy, at::Tensor w2, int64_t H,
@@ -699,7 +735,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_nt", &gemm_nt, py::arg("a"), py::arg("b"),
         py::arg("bias") = c10::nullopt, py::arg("out_f32") = false);
   m.def("colsum_f32", &colsum_f32);
-  m.def("conv2d_fwd", &conv2d_fwd);
+  m.def("conv2d_fwd", &conv2d_fwd, py::arg("x"), py::arg("wt"),
+        py::arg("stride"), py::arg("pad"),
+        py::arg("want_bn_stats") = false);
+  m.def("bn_stats_train_from_part", &bn_stats_train_from_part);
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
   m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
   m.def("gemm_nt_batched", &gemm_nt_batched, py::arg("a"), py::arg("b"),

@@ -328,6 +328,28 @@ void pdta_bn_stats(const void* x, float* part, float* mean, float* out2,
       running_var, momentum, eps, mode);
 }
 
+// Finalize-only path: `part` already holds [chunks][2C] partials written by
+// a producing kernel's epilogue (conv.hip igemm_fwd STATS variants) — the
+// stage-1 read of x never happens. part must carry 64 extra scratch rows.
+void pdta_bn_stats_from_part(float* part, int chunks, float* mean,
+                             float* out2, float* running_mean,
+                             float* running_var, float momentum, float eps,
+                             int mode, long rows, int C, hipStream_t s) {
+  const float* fin = part;
+  int fchunks = chunks;
+  if (chunks > 64) {
+    float* part2 = part + (long)chunks * 2 * C;
+    bn_mid_k<<<dim3((2 * C + 255) / 256, 64), 256, 0, s>>>(part, part2,
+                                                           chunks, 2 * C);
+    fin = part2;
+    fchunks = 64;
+  }
+  float unbias = rows > 1 ? (float)rows / (float)(rows - 1) : 1.f;
+  bn_stats_final_k<<<(C + 63) / 64, 256, 0, s>>>(
+      fin, fchunks, C, 1.f / (float)rows, unbias, mean, out2, running_mean,
+      running_var, momentum, eps, mode);
+}
+
 // chunks needed by the caller to size `part`:
 int pdta_bn_chunks(long rows, int C) {
   int rpb, chunks;

@@ -24,11 +24,16 @@ namespace {
 constexpr int TPB = 256;
 
 // ---------------------------------------------------------------- fwd --
-template <typename T, int BNT>  // BNT: output-channel tile (64 or 128)
+// STATS: also emit per-m-tile partial (sum, sumsq) of the f32 output tile
+// into bn_part (bn_part_k's [chunk][2C] layout, chunk = tm) so the BN that
+// consumes y never re-reads it for statistics (bn.hip pdta_bn_stats stage 1
+// eliminated; at B=256 that pass was ~7% of the ResNet-50 step).
+template <typename T, int BNT, bool STATS = false>
 __global__ __launch_bounds__(TPB) void igemm_fwd_k(
     const T* __restrict__ x, const T* __restrict__ wt, T* __restrict__ y,
     const T* __restrict__ zero_page, int Nn, int H, int W, int C, int P,
-    int Q, int Kout, int R, int S, int stride, int pad, int tiles_n) {
+    int Q, int Kout, int R, int S, int stride, int pad, int tiles_n,
+    float* __restrict__ bn_part = nullptr) {
   // single LDS object (guide §5 trap (a)): A tile 16 KiB + B tile
   constexpr int BOFF = 128 * 64 * 2;
   constexpr int NW_N = (BNT == 128) ? 2 : 1;  // waves along n
@@ -145,18 +150,66 @@ __global__ __launch_bounds__(TPB) void igemm_fwd_k(
       }
     }
   }
+
+  if (STATS) {
+    constexpr int FR = BNT / NW_N;  // columns covered by one wn group
+    float* sred = reinterpret_cast<float*>(lds[0]);  // [4 waves][2*FR]
+#pragma unroll
+    for (int j = 0; j < NJ; ++j) {
+      float sj = 0.f, qj = 0.f;
+#pragma unroll
+      for (int i = 0; i < MI; ++i) {
+        long rbase = m0 + wm * (128 / NW_M) + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          if (rbase + r < M) {
+            float v = acc[i][j][r];
+            sj += v;
+            qj += v * v;
+          }
+      }
+      // sum the 4 row-subgroups (lanes l, l+16, l+32, l+48 share a column)
+      sj += __shfl_xor(sj, 16);
+      sj += __shfl_xor(sj, 32);
+      qj += __shfl_xor(qj, 16);
+      qj += __shfl_xor(qj, 32);
+      if (lane < 16) {
+        sred[wave * 2 * FR + j * 16 + lane] = sj;
+        sred[wave * 2 * FR + FR + j * 16 + lane] = qj;
+      }
+    }
+    __syncthreads();
+    int cpb = Kout < 256 ? Kout : 256;
+    for (int t = threadIdx.x; t < BNT; t += TPB) {
+      int wn_c = t / FR, off = t % FR;
+      float s = 0.f, q = 0.f;
+#pragma unroll
+      for (int wmi = 0; wmi < NW_M; ++wmi) {
+        int wv = wmi * NW_N + wn_c;
+        s += sred[wv * 2 * FR + off];
+        q += sred[wv * 2 * FR + FR + off];
+      }
+      int c = n0 + t;
+      if (c < Kout) {
+        long base = ((long)tm * Kout + (c / cpb) * (long)cpb) * 2;
+        bn_part[base + c % cpb] = s;
+        bn_part[base + cpb + c % cpb] = q;
+      }
+    }
+  }
 }
 
 // ---- big-channel fwd: 256x256 tile, 8 waves, 2-slot K-ring -------------
 // The gemm.hip big-tile structure with the im2col gather in the A-source
 // (layers with Kout % 256 == 0 carry most of the conv FLOPs; the 128-wide
 // tiles ran at ~190 TF from staging re-reads).
-template <typename T>
+template <typename T, bool STATS = false>
 __global__ __launch_bounds__(512) void igemm_fwd_big_k(
     const T* __restrict__ x, const T* __restrict__ wt, T* __restrict__ y,
     const T* __restrict__ zero_page, int Nn, int H, int W, int C, int P,
     int Q, int Kout, int R, int S, int stride, int pad, int tiles_n,
-    unsigned mq, int sq, unsigned mpq, int spq) {
+    unsigned mq, int sq, unsigned mpq, int spq,
+    float* __restrict__ bn_part = nullptr) {
   constexpr int BOFF = 256 * 64 * 2;
   __shared__ __attribute__((aligned(16))) char lds[2][2 * BOFF];
 
@@ -270,6 +323,46 @@ __global__ __launch_bounds__(512) void igemm_fwd_big_k(
       for (int r = 0; r < 4; ++r) {
         long row = rbase + r;
         if (row < M) y[row * Kout + col] = from_f32<T>(acc[i][j][r]);
+      }
+    }
+  }
+
+  if (STATS) {  // per-m-tile (sum, sumsq) partials, see igemm_fwd_k
+    float* sred = reinterpret_cast<float*>(lds[0]);  // [8 waves][2*64]
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float sj = 0.f, qj = 0.f;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        long rbase = m0 + wm * 128 + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          if (rbase + r < M) {
+            float v = acc[i][j][r];
+            sj += v;
+            qj += v * v;
+          }
+      }
+      sj += __shfl_xor(sj, 16);
+      sj += __shfl_xor(sj, 32);
+      qj += __shfl_xor(qj, 16);
+      qj += __shfl_xor(qj, 32);
+      if (lane < 16) {
+        sred[wave * 128 + j * 16 + lane] = sj;
+        sred[wave * 128 + 64 + j * 16 + lane] = qj;
+      }
+    }
+    __syncthreads();
+    int cpb = Kout < 256 ? Kout : 256;
+    for (int t = threadIdx.x; t < 256; t += 512) {
+      int wn_c = t / 64, off = t % 64;
+      float s = sred[wn_c * 128 + off] + sred[(4 + wn_c) * 128 + off];
+      float q = sred[wn_c * 128 + 64 + off] + sred[(4 + wn_c) * 128 + 64 + off];
+      int c = n0 + t;
+      if (c < Kout) {
+        long base = ((long)tm * Kout + (c / cpb) * (long)cpb) * 2;
+        bn_part[base + c % cpb] = s;
+        bn_part[base + cpb + c % cpb] = q;
       }
     }
   }
@@ -948,20 +1041,33 @@ __global__ void stem_im2col_k(const T* __restrict__ x, T* __restrict__ out,
 
 extern "C" {
 
+// m-tile count of the kernel pdta_conv_fwd will pick = the chunk count of
+// the bn_part slab it writes when bn_part != nullptr
+long pdta_conv_fwd_chunks(long M, int Kout) {
+  if (Kout % 256 == 0 && M >= 4096) return (M + 255) / 256;
+  return (M + 127) / 128;
+}
+
 void pdta_conv_fwd(const void* x, const void* wt, void* y,
                    const void* zero_page, int Nn, int H, int W, int C, int P,
                    int Q, int Kout, int R, int S, int stride, int pad,
-                   int dt, hipStream_t s) {
+                   int dt, hipStream_t s, float* bn_part) {
   long M = (long)Nn * P * Q;
   if (Kout % 256 == 0 && M >= 4096) {
     int tiles_m = (int)((M + 255) / 256), tiles_n = Kout / 256;
     dim3 grid(tiles_m * tiles_n);
     MagicDiv mq = magic_div(Q), mpq = magic_div(P * Q);
     PDTA_DISPATCH_DTYPE16(dt, T, {
-      igemm_fwd_big_k<T><<<grid, 512, 0, s>>>(
-          (const T*)x, (const T*)wt, (T*)y, (const T*)zero_page, Nn, H, W,
-          C, P, Q, Kout, R, S, stride, pad, tiles_n, mq.mul, mq.shift,
-          mpq.mul, mpq.shift);
+      if (bn_part)
+        igemm_fwd_big_k<T, true><<<grid, 512, 0, s>>>(
+            (const T*)x, (const T*)wt, (T*)y, (const T*)zero_page, Nn, H, W,
+            C, P, Q, Kout, R, S, stride, pad, tiles_n, mq.mul, mq.shift,
+            mpq.mul, mpq.shift, bn_part);
+      else
+        igemm_fwd_big_k<T><<<grid, 512, 0, s>>>(
+            (const T*)x, (const T*)wt, (T*)y, (const T*)zero_page, Nn, H, W,
+            C, P, Q, Kout, R, S, stride, pad, tiles_n, mq.mul, mq.shift,
+            mpq.mul, mpq.shift);
     });
     return;
   }
@@ -969,14 +1075,25 @@ void pdta_conv_fwd(const void* x, const void* wt, void* y,
   int tiles_m = (int)((M + 127) / 128), tiles_n = (Kout + bnt - 1) / bnt;
   dim3 grid(tiles_m * tiles_n);
   PDTA_DISPATCH_DTYPE16(dt, T, {
-    if (bnt == 128)
-      igemm_fwd_k<T, 128><<<grid, TPB, 0, s>>>(
-          (const T*)x, (const T*)wt, (T*)y, (const T*)zero_page, Nn, H, W,
-          C, P, Q, Kout, R, S, stride, pad, tiles_n);
-    else
-      igemm_fwd_k<T, 64><<<grid, TPB, 0, s>>>(
-          (const T*)x, (const T*)wt, (T*)y, (const T*)zero_page, Nn, H, W,
-          C, P, Q, Kout, R, S, stride, pad, tiles_n);
+    if (bnt == 128) {
+      if (bn_part)
+        igemm_fwd_k<T, 128, true><<<grid, TPB, 0, s>>>(
+            (const T*)x, (const T*)wt, (T*)y, (const T*)zero_page, Nn, H, W,
+            C, P, Q, Kout, R, S, stride, pad, tiles_n, bn_part);
+      else
+        igemm_fwd_k<T, 128><<<grid, TPB, 0, s>>>(
+            (const T*)x, (const T*)wt, (T*)y, (const T*)zero_page, Nn, H, W,
+            C, P, Q, Kout, R, S, stride, pad, tiles_n);
+    } else {
+      if (bn_part)
+        igemm_fwd_k<T, 64, true><<<grid, TPB, 0, s>>>(
+            (const T*)x, (const T*)wt, (T*)y, (const T*)zero_page, Nn, H, W,
+            C, P, Q, Kout, R, S, stride, pad, tiles_n, bn_part);
+      else
+        igemm_fwd_k<T, 64><<<grid, TPB, 0, s>>>(
+            (const T*)x, (const T*)wt, (T*)y, (const T*)zero_page, Nn, H, W,
+            C, P, Q, Kout, R, S, stride, pad, tiles_n);
+    }
   });
 }
 

@@ -53,18 +53,24 @@ class Conv2dNHWC(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, x, weight, wt, w2, stride: int, pad: int):
+    def forward(ctx, x, weight, wt, w2, stride: int, pad: int,
+                want_bn_stats: bool = False):
         ctx.stride, ctx.pad = stride, pad
         ctx.save_for_backward(x, weight, w2)
         if x.is_cuda:
-            return ext().conv2d_fwd(x, wt, stride, pad)
+            out = ext().conv2d_fwd(x, wt, stride, pad, want_bn_stats)
+            if want_bn_stats:
+                y, part = out
+                ctx.mark_non_differentiable(part)
+                return y, part
+            return out[0]
         # CPU reference (fp32)
         w = weight.float()
         y = F.conv2d(_nhwc_to_nchw(x).float(), w, stride=stride, padding=pad)
         return _nchw_to_nhwc(y).to(x.dtype)
 
     @staticmethod
-    def backward(ctx, dy):
+    def backward(ctx, dy, _dpart=None):
         x, weight, w2 = ctx.saved_tensors
         stride, pad = ctx.stride, ctx.pad
         dy = dy.contiguous()
@@ -89,10 +95,13 @@ class Conv2dNHWC(torch.autograd.Function):
             if need_dw:
                 dw = torch.nn.grad.conv2d_weight(x_nchw, w.shape, dy_nchw,
                                                  stride=stride, padding=pad)
-        return dx, dw, None, None, None, None
+        return dx, dw, None, None, None, None, None
 
 
-def conv2d_nhwc(x, weight, wt, w2, stride: int = 1, pad: int = 0):
+def conv2d_nhwc(x, weight, wt, w2, stride: int = 1, pad: int = 0,
+                want_bn_stats: bool = False):
+    if want_bn_stats:
+        return Conv2dNHWC.apply(x, weight, wt, w2, stride, pad, True)
     return Conv2dNHWC.apply(x, weight, wt, w2, stride, pad)
 
 
@@ -206,6 +215,14 @@ def bn_stats_train(x, running_mean, running_var, momentum, eps):
     """(mean, invstd) + in-kernel running-stat update — GPU training fast
     path (one kernel pair instead of stats + 4 python-side buffer ops)."""
     return ext().bn_stats_train(x, running_mean, running_var, momentum, eps)
+
+
+def bn_stats_train_from_part(part, rows, running_mean, running_var,
+                             momentum, eps):
+    """(mean, invstd) finalized from a conv2d_nhwc(want_bn_stats=True)
+    partial slab — the stage-1 pass over the activation never runs."""
+    return ext().bn_stats_train_from_part(part, rows, running_mean,
+                                          running_var, momentum, eps)
 
 
 def bn_train(x, weight, bias, mean, invstd, fuse_relu=False, comm=None,

@@ -88,7 +88,7 @@ def test_conv_fwd(n, h, c, k, r, stride):
     x = _rand(n, h, h, c, seed=7)
     w = (_rand(k, c, r, r, seed=8).float() / (c ** 0.5)).to(torch.bfloat16)
     wt = w.permute(0, 2, 3, 1).contiguous()
-    y = ext().conv2d_fwd(x, wt, stride, pad)
+    y = ext().conv2d_fwd(x, wt, stride, pad)[0]
     _close(y, _conv_ref(x, w, stride, pad), what=f"conv {c}->{k} r{r}s{stride}")
 
 
@@ -131,7 +131,7 @@ def test_stem_conv_fwd_and_wrw():
     x = _rand(2, 32, 32, 3, seed=13)
     w = (_rand(64, 3, 7, 7, seed=14).float() / 3.0).to(torch.bfloat16)
     wt = w.permute(0, 2, 3, 1).contiguous()
-    y = ext().conv2d_fwd(x, wt, 2, 3)
+    y = ext().conv2d_fwd(x, wt, 2, 3)[0]
     _close(y, _conv_ref(x, w, 2, 3), what="stem fwd")
     p = y.shape[1]
     dy = _rand(2, p, p, 64, seed=15)
@@ -304,3 +304,39 @@ def test_unscale_check():
     g2[100] = float("inf")
     found2 = ext().unscale_check(g2, 1.0)
     assert float(found2) == 1.0
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [
+    # (N, HW, C, K, ksize, stride) — small-tile, big-tile, and stem-ish
+    (2, 30, 64, 128, 3, 1),
+    (2, 46, 64, 256, 1, 1),
+    (3, 23, 128, 512, 3, 2),
+])
+def test_conv_fwd_fused_bn_stats(shape):
+    """conv2d_fwd(want_bn_stats=True) epilogue partials finalize to the
+    same mean/invstd as the standalone bn_stats_train pass over y."""
+    N, HW, C, K, ks, stride = shape
+    torch.manual_seed(5)
+    dev = "cuda"
+    x = torch.randn(N, HW, HW, C, device=dev, dtype=torch.bfloat16)
+    w = torch.randn(K, C, ks, ks, device=dev) * 0.05
+    wt = w.permute(0, 2, 3, 1).to(torch.bfloat16).contiguous()
+    pad = ks // 2
+    y, part = ext().conv2d_fwd(x, wt, stride, pad, True)
+    rows = y.numel() // K
+    rm1 = torch.zeros(K, device=dev)
+    rv1 = torch.ones(K, device=dev)
+    mean1, invstd1 = ext().bn_stats_train_from_part(
+        part, rows, rm1, rv1, 0.1, 1e-5)
+    rm2 = torch.zeros(K, device=dev)
+    rv2 = torch.ones(K, device=dev)
+    mean2, invstd2 = ext().bn_stats_train(y, rm2, rv2, 0.1, 1e-5)
+    # the epilogue sums f32 accumulators pre-cast; bn_stats_train reads the
+    # bf16-stored y — equal to bf16 rounding
+    assert torch.allclose(mean1, mean2, atol=5e-3, rtol=1e-2), \
+        (mean1 - mean2).abs().max()
+    assert torch.allclose(invstd1, invstd2, atol=5e-3, rtol=2e-2), \
+        (invstd1 - invstd2).abs().max()
+    assert torch.allclose(rm1, rm2, atol=5e-3, rtol=1e-2)
+    assert torch.allclose(rv1, rv2, atol=5e-3, rtol=2e-2)
