@@ -22,12 +22,13 @@ void pdta_bn_apply(const void*, void*, const void*, const float*,
                    const float*, const float*, const float*, long, int, int,
                    int, hipStream_t);
 void pdta_bn_bwd_reduce(const void*, const void*, const void*,
-                        const float*, float*, float*, float*, long, int,
-                        int, hipStream_t);
+                        const float*, const float*, const float*,
+                        const float*, int, float*, float*, float*, long,
+                        int, int, hipStream_t);
 void pdta_bn_bwd_elemt(const void*, const void*, const void*, void*,
-                       void*, const float*, const float*, const float*,
-                       const float*, const float*, float, long, int, int,
-                       hipStream_t);
+                       void*, const float*, const float*, int,
+                       const float*, const float*, const float*,
+                       const float*, float, long, int, int, hipStream_t);
 void pdta_maxpool_fwd(const void*, void*, uint8_t*, int, int, int, int, int,
                       int, int, int, int, int, hipStream_t);
 void pdta_maxpool_bwd(const void*, const uint8_t*, void*, int, int, int, int,
@@ -193,7 +194,13 @@ at::Tensor bn_apply(at::Tensor x, at::Tensor weight, at::Tensor bias,
 
 std::vector<at::Tensor> bn_bwd_reduce(at::Tensor x, at::Tensor dy,
                                       c10::optional<at::Tensor> relu_y,
-                                      at::Tensor mean) {
+                                      at::Tensor mean,
+                                      c10::optional<at::Tensor> invstd,
+                                      c10::optional<at::Tensor> weight,
+                                      c10::optional<at::Tensor> bias) {
+  // mask modes: relu_y given -> read it; else invstd/weight/bias given ->
+  // recompute mask (gamma*xhat+beta > 0), skipping the relu_y pass;
+  // else no mask
   CHECK_IN(x);
   CHECK_IN(dy);
   int C = x.size(3);
@@ -202,9 +209,19 @@ std::vector<at::Tensor> bn_bwd_reduce(at::Tensor x, at::Tensor dy,
   int chunks = pdta_bn_chunks(rows, C);
   auto part = at::empty({(long)chunks * 2 * C}, opts);
   auto sum_dy = at::empty({C}, opts), sum_dy_xmu = at::empty({C}, opts);
+  int mask = relu_y.has_value() ? 1 : (bias.has_value() ? 2 : 0);
+  at::Tensor isf, wf, bf;
+  if (mask == 2) {
+    isf = invstd->to(at::kFloat).contiguous();
+    wf = weight->to(at::kFloat).contiguous();
+    bf = bias->to(at::kFloat).contiguous();
+  }
   pdta_bn_bwd_reduce(x.data_ptr(), dy.data_ptr(),
                      relu_y.has_value() ? relu_y->data_ptr() : nullptr,
                      mean.contiguous().data_ptr<float>(),
+                     mask == 2 ? isf.data_ptr<float>() : nullptr,
+                     mask == 2 ? wf.data_ptr<float>() : nullptr,
+                     mask == 2 ? bf.data_ptr<float>() : nullptr, mask,
                      part.data_ptr<float>(), sum_dy.data_ptr<float>(),
                      sum_dy_xmu.data_ptr<float>(), rows, C, dt_of(x),
                      cur_stream());
@@ -216,7 +233,8 @@ std::vector<at::Tensor> bn_bwd_elemt(at::Tensor x, at::Tensor dy,
                                      at::Tensor mean, at::Tensor invstd,
                                      at::Tensor weight, at::Tensor sum_dy,
                                      at::Tensor sum_dy_xmu, double count,
-                                     bool want_dres) {
+                                     bool want_dres,
+                                     c10::optional<at::Tensor> bias) {
   CHECK_IN(x);
   CHECK_IN(dy);
   int C = x.size(3);
@@ -225,10 +243,16 @@ std::vector<at::Tensor> bn_bwd_elemt(at::Tensor x, at::Tensor dy,
   at::Tensor dres;
   if (want_dres) dres = at::empty_like(x);
   auto w = weight.to(at::kFloat).contiguous();
+  int mask = relu_y.has_value() ? 1 : (bias.has_value() ? 2 : 0);
+  at::Tensor bf;
+  if (mask == 2) bf = bias->to(at::kFloat).contiguous();
+  TORCH_CHECK(!(want_dres && mask != 1),
+              "bn_bwd_elemt: dres requires the stored-relu mask");
   pdta_bn_bwd_elemt(x.data_ptr(), dy.data_ptr(),
                     relu_y.has_value() ? relu_y->data_ptr() : nullptr,
                     dx.data_ptr(), want_dres ? dres.data_ptr() : nullptr,
                     w.data_ptr<float>(),
+                    mask == 2 ? bf.data_ptr<float>() : nullptr, mask,
                     mean.contiguous().data_ptr<float>(),
                     invstd.contiguous().data_ptr<float>(),
                     sum_dy.contiguous().data_ptr<float>(),
@@ -719,11 +743,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_apply", &bn_apply, py::arg("x"), py::arg("weight"),
         py::arg("bias"), py::arg("mean"), py::arg("invstd"),
         py::arg("relu"), py::arg("res") = c10::nullopt);
-  m.def("bn_bwd_reduce", &bn_bwd_reduce);
+  m.def("bn_bwd_reduce", &bn_bwd_reduce, py::arg("x"), py::arg("dy"),
+        py::arg("relu_y"), py::arg("mean"),
+        py::arg("invstd") = c10::nullopt, py::arg("weight") = c10::nullopt,
+        py::arg("bias") = c10::nullopt);
   m.def("bn_bwd_elemt", &bn_bwd_elemt, py::arg("x"), py::arg("dy"),
         py::arg("relu_y"), py::arg("mean"), py::arg("invstd"),
         py::arg("weight"), py::arg("sum_dy"), py::arg("sum_dy_xmu"),
-        py::arg("count"), py::arg("want_dres") = false);
+        py::arg("count"), py::arg("want_dres") = false,
+        py::arg("bias") = c10::nullopt);
   m.def("maxpool_fwd", &maxpool_fwd);
   m.def("maxpool_bwd", &maxpool_bwd);
   m.def("avgpool_global_fwd", &avgpool_global_fwd);

@@ -141,12 +141,19 @@ __global__ void bn_apply_k(const T* __restrict__ x, T* __restrict__ y,
 }
 
 // -- backward stage 1: per-chunk partial (sum_dy, sum_dy_xmu), with the
-//    preceding-ReLU mask fused when RELU (dy := y>0 ? dy : 0) ------------
-template <typename T, bool RELU>
+//    preceding-ReLU mask fused. MASK 0: none. MASK 1: dy := y>0 ? dy : 0
+//    (read the stored relu output — required for residual joins, where the
+//    pre-activation includes the identity branch). MASK 2: recompute the
+//    mask as gamma*xhat+beta > 0 from x — one whole HBM pass over relu_y
+//    never happens (exact up to bf16 boundary rounding of y) -------------
+template <typename T, int MASK>
 __global__ void bn_bwd_part_k(const T* __restrict__ x,
                               const T* __restrict__ dy,
                               const T* __restrict__ relu_y,
                               const float* __restrict__ mean,
+                              const float* __restrict__ invstd,
+                              const float* __restrict__ weight,
+                              const float* __restrict__ bias,
                               float* __restrict__ part, long rows, int C,
                               int rows_per_block) {
   __shared__ float red[2][TPB];
@@ -157,11 +164,17 @@ __global__ void bn_bwd_part_k(const T* __restrict__ x,
   long r0 = (long)blockIdx.y * rows_per_block + threadIdx.x / cpb;
   long r_end = min(rows, (long)(blockIdx.y + 1) * rows_per_block);
   float m = mean[c];
+  float k1 = 0.f, b0 = 0.f;
+  if (MASK == 2) {
+    k1 = weight[c] * invstd[c];
+    b0 = bias[c];
+  }
   float s = 0.f, sx = 0.f;
   for (long r = r0; r < r_end; r += rpar) {
     float g = to_f32(dy[r * C + c]);
-    if (RELU) g = to_f32(relu_y[r * C + c]) > 0.f ? g : 0.f;
     float v = to_f32(x[r * C + c]);
+    if (MASK == 1) g = to_f32(relu_y[r * C + c]) > 0.f ? g : 0.f;
+    if (MASK == 2) g = k1 * (v - m) + b0 > 0.f ? g : 0.f;
     s += g;
     sx += g * (v - m);
   }
@@ -211,12 +224,13 @@ __global__ void bn_bwd_final_k(const float* __restrict__ part, int chunks,
 
 // -- backward elementwise (ReLU mask fused when RELU):
 //    dx = w*invstd * (dy - sum_dy/n - (x-mean)*invstd^2*sum_dy_xmu/n)
-template <typename T, bool RELU, bool DRES>
+template <typename T, int MASK, bool DRES>
 __global__ void bn_bwd_elemt_k(const T* __restrict__ x,
                                const T* __restrict__ dy,
                                const T* __restrict__ relu_y,
                                T* __restrict__ dx, T* __restrict__ dres,
                                const float* __restrict__ weight,
+                               const float* __restrict__ bias,
                                const float* __restrict__ mean,
                                const float* __restrict__ invstd,
                                const float* __restrict__ sum_dy,
@@ -227,7 +241,7 @@ __global__ void bn_bwd_elemt_k(const T* __restrict__ x,
   int cg = blockIdx.x * cpg + (int)(threadIdx.x % cpg);
   if (cg >= c8) return;
   int c0 = cg * 8;
-  float k1[8], k2[8], k3[8], mu[8];
+  float k1[8], k2[8], k3[8], mu[8], b0[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     float is = invstd[c0 + j];
@@ -235,6 +249,7 @@ __global__ void bn_bwd_elemt_k(const T* __restrict__ x,
     k2[j] = sum_dy[c0 + j] * inv_n;
     k3[j] = is * is * sum_dy_xmu[c0 + j] * inv_n;
     mu[j] = mean[c0 + j];
+    b0[j] = (MASK == 2) ? bias[c0 + j] : 0.f;
   }
   const int C = c8 * 8;
   long rstride = (long)gridDim.y * rpar;
@@ -243,13 +258,15 @@ __global__ void bn_bwd_elemt_k(const T* __restrict__ x,
     Vec8<T> g = load8(dy + r * C + c0);
     Vec8<T> v = load8(x + r * C + c0);
     Vec8<T> yv, dr;
-    if (RELU) yv = load8(relu_y + r * C + c0);
+    if (MASK == 1) yv = load8(relu_y + r * C + c0);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float gg = to_f32(g.v[j]);
-      if (RELU) gg = to_f32(yv.v[j]) > 0.f ? gg : 0.f;
+      float xm = to_f32(v.v[j]) - mu[j];
+      if (MASK == 1) gg = to_f32(yv.v[j]) > 0.f ? gg : 0.f;
+      if (MASK == 2) gg = k1[j] * xm + b0[j] > 0.f ? gg : 0.f;
       if (DRES) dr.v[j] = from_f32<T>(gg);
-      float o = k1[j] * (gg - k2[j] - (to_f32(v.v[j]) - mu[j]) * k3[j]);
+      float o = k1[j] * (gg - k2[j] - xm * k3[j]);
       g.v[j] = from_f32<T>(o);
     }
     if (DRES) store8(dres + r * C + c0, dr);
@@ -385,21 +402,29 @@ void pdta_bn_apply(const void* x, void* y, const void* res,
   });
 }
 
-// relu_y: non-null => fuse dy := (relu_y > 0) ? dy : 0.
+// mask: 0 none; 1 read relu_y; 2 recompute gamma*xhat+beta > 0 (needs
+// invstd/weight/bias, skips the relu_y HBM pass)
 void pdta_bn_bwd_reduce(const void* x, const void* dy, const void* relu_y,
-                        const float* mean, float* part, float* sum_dy,
+                        const float* mean, const float* invstd,
+                        const float* weight, const float* bias, int mask,
+                        float* part, float* sum_dy,
                         float* sum_dy_xmu, long rows, int C, int dt,
                         hipStream_t s) {
   int rpb, chunks;
   dim3 g = reduce_grid(rows, C, &rpb, &chunks);
   PDTA_DISPATCH_DTYPE(dt, T, {
-    if (relu_y)
-      bn_bwd_part_k<T, true><<<g, TPB, 0, s>>>(
-          (const T*)x, (const T*)dy, (const T*)relu_y, mean, part, rows, C,
-          rpb);
+    if (mask == 1)
+      bn_bwd_part_k<T, 1><<<g, TPB, 0, s>>>(
+          (const T*)x, (const T*)dy, (const T*)relu_y, mean, nullptr,
+          nullptr, nullptr, part, rows, C, rpb);
+    else if (mask == 2)
+      bn_bwd_part_k<T, 2><<<g, TPB, 0, s>>>(
+          (const T*)x, (const T*)dy, nullptr, mean, invstd, weight, bias,
+          part, rows, C, rpb);
     else
-      bn_bwd_part_k<T, false><<<g, TPB, 0, s>>>(
-          (const T*)x, (const T*)dy, nullptr, mean, part, rows, C, rpb);
+      bn_bwd_part_k<T, 0><<<g, TPB, 0, s>>>(
+          (const T*)x, (const T*)dy, nullptr, mean, nullptr, nullptr,
+          nullptr, part, rows, C, rpb);
   });
   const float* fin = part;
   int fchunks = chunks;
@@ -416,27 +441,33 @@ void pdta_bn_bwd_reduce(const void* x, const void* dy, const void* relu_y,
 
 void pdta_bn_bwd_elemt(const void* x, const void* dy, const void* relu_y,
                        void* dx, void* dres, const float* weight,
+                       const float* bias, int mask,
                        const float* mean, const float* invstd,
                        const float* sum_dy, const float* sum_dy_xmu,
                        float count, long rows, int C, int dt,
                        hipStream_t s) {
   dim3 g = ew_grid2(rows, C);
   PDTA_DISPATCH_DTYPE(dt, T, {
-    if (relu_y) {
+    if (mask == 1) {
       if (dres)
-        bn_bwd_elemt_k<T, true, true><<<g, TPB, 0, s>>>(
+        bn_bwd_elemt_k<T, 1, true><<<g, TPB, 0, s>>>(
             (const T*)x, (const T*)dy, (const T*)relu_y, (T*)dx, (T*)dres,
-            weight, mean, invstd, sum_dy, sum_dy_xmu, 1.f / count, rows,
-            C / 8);
+            weight, nullptr, mean, invstd, sum_dy, sum_dy_xmu, 1.f / count,
+            rows, C / 8);
       else
-        bn_bwd_elemt_k<T, true, false><<<g, TPB, 0, s>>>(
+        bn_bwd_elemt_k<T, 1, false><<<g, TPB, 0, s>>>(
             (const T*)x, (const T*)dy, (const T*)relu_y, (T*)dx, nullptr,
-            weight, mean, invstd, sum_dy, sum_dy_xmu, 1.f / count, rows,
-            C / 8);
+            weight, nullptr, mean, invstd, sum_dy, sum_dy_xmu, 1.f / count,
+            rows, C / 8);
+    } else if (mask == 2) {
+      bn_bwd_elemt_k<T, 2, false><<<g, TPB, 0, s>>>(
+          (const T*)x, (const T*)dy, nullptr, (T*)dx, nullptr, weight, bias,
+          mean, invstd, sum_dy, sum_dy_xmu, 1.f / count, rows, C / 8);
     } else {
-      bn_bwd_elemt_k<T, false, false><<<g, TPB, 0, s>>>(
-          (const T*)x, (const T*)dy, nullptr, (T*)dx, nullptr, weight, mean,
-          invstd, sum_dy, sum_dy_xmu, 1.f / count, rows, C / 8);
+      bn_bwd_elemt_k<T, 0, false><<<g, TPB, 0, s>>>(
+          (const T*)x, (const T*)dy, nullptr, (T*)dx, nullptr, weight,
+          nullptr, mean, invstd, sum_dy, sum_dy_xmu, 1.f / count, rows,
+          C / 8);
     }
   });
 }

@@ -150,12 +150,12 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
         ctx.comm = comm
         ctx.count_override = count_override
         ctx.has_residual = residual is not None
-        ctx.save_for_backward(x, weight, mean, invstd, y)
+        ctx.save_for_backward(x, weight, bias, mean, invstd, y)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, weight, mean, invstd, y = ctx.saved_tensors
+        x, weight, bias, mean, invstd, y = ctx.saved_tensors
         fuse_relu = ctx.fuse_relu
         dy = dy.contiguous()
         n = x.shape[0] * x.shape[1] * x.shape[2]
@@ -163,8 +163,17 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
             n = ctx.count_override
         dres = None
         if x.is_cuda:
-            relu_y = y if fuse_relu else None  # mask fused in the kernels
-            sum_dy, sum_dy_xmu = ext().bn_bwd_reduce(x, dy, relu_y, mean)
+            # ReLU mask: residual joins must read the stored relu output
+            # (pre-activation includes the identity branch); plain conv->
+            # bn->relu recomputes the mask from xhat in-kernel and never
+            # re-reads y (one whole HBM pass saved per kernel)
+            relu_y = y if (fuse_relu and ctx.has_residual) else None
+            premask = fuse_relu and not ctx.has_residual
+            sum_dy, sum_dy_xmu = ext().bn_bwd_reduce(
+                x, dy, relu_y, mean,
+                invstd=invstd if premask else None,
+                weight=weight if premask else None,
+                bias=bias if premask else None)
             # param grads from LOCAL sums (DDP averages them afterwards,
             # matching torch SyncBN's split of responsibilities)
             dweight = sum_dy_xmu * invstd
@@ -177,7 +186,8 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
             out = ext().bn_bwd_elemt(x, dy, relu_y, mean, invstd, weight,
                                      sum_dy.contiguous(),
                                      sum_dy_xmu.contiguous(), float(n),
-                                     ctx.has_residual)
+                                     ctx.has_residual,
+                                     bias=bias if premask else None)
             dx = out[0]
             if ctx.has_residual:
                 dres = out[1]

@@ -340,3 +340,32 @@ def test_conv_fwd_fused_bn_stats(shape):
         (invstd1 - invstd2).abs().max()
     assert torch.allclose(rm1, rm2, atol=5e-3, rtol=1e-2)
     assert torch.allclose(rv1, rv2, atol=5e-3, rtol=2e-2)
+
+
+@pytest.mark.gpu
+def test_bn_bwd_premask_matches_stored_relu_mask():
+    """MASK=2 (gamma*xhat+beta > 0 recomputed in-kernel) must equal MASK=1
+    (reading the stored relu output) away from the bf16 rounding boundary."""
+    torch.manual_seed(7)
+    dev = "cuda"
+    N, HW, C = 3, 17, 128
+    x = torch.randn(N, HW, HW, C, device=dev, dtype=torch.bfloat16)
+    dy = torch.randn_like(x)
+    mean, var = ext().bn_stats(x)
+    invstd = (var + 1e-5).rsqrt()
+    w = torch.randn(C, device=dev) * 0.5 + 1.0
+    b = torch.randn(C, device=dev) * 0.1
+    y = ext().bn_apply(x, w, b, mean, invstd, True, None)
+    s1, sx1 = ext().bn_bwd_reduce(x, dy, y, mean)
+    s2, sx2 = ext().bn_bwd_reduce(x, dy, None, mean, invstd=invstd,
+                                  weight=w, bias=b)
+    assert torch.allclose(s1, s2, atol=1e-1, rtol=1e-3), \
+        (s1 - s2).abs().max()
+    assert torch.allclose(sx1, sx2, atol=1e-1, rtol=1e-3)
+    n = float(N * HW * HW)
+    dx1 = ext().bn_bwd_elemt(x, dy, y, mean, invstd, w, s1, sx1, n,
+                             False)[0]
+    dx2 = ext().bn_bwd_elemt(x, dy, None, mean, invstd, w, s1, sx1, n,
+                             False, bias=b)[0]
+    diff = (dx1.float() - dx2.float()).abs()
+    assert (diff > 0).float().mean() < 1e-3  # only boundary elements differ
