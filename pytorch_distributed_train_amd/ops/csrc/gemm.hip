@@ -360,7 +360,12 @@ void pdta_colsum_f32(const void* dy, float* db, long M, int N, int dt,
                      hipStream_t s) {
   dim3 block(256);
   if (N % 8 == 0) {
-    dim3 grid((N / 8 + 255) / 256, (unsigned)min((long)128, M));
+    // grid.y sized for ~64K active lanes (N/8 lanes per row-slab); a fixed
+    // 128 starved the chip at N=768 (125 us vs 62 for the scalar version)
+    long gy = 524288 / N;
+    if (gy < 64) gy = 64;
+    if (gy > M) gy = M;
+    dim3 grid((N / 8 + 255) / 256, (unsigned)gy);
     PDTA_DISPATCH_DTYPE16(dt, T, {
       colsum8_f32_k<T><<<grid, block, 0, s>>>((const T*)dy, db, M, N);
     });

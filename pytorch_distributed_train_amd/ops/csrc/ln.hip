@@ -277,6 +277,46 @@ __global__ void softmax_fwd_f32_k(const float* __restrict__ s,
   int lane = threadIdx.x & 63;
   const float4* sr = reinterpret_cast<const float4*>(s + row * n);
   int n4 = n >> 2;
+  if (n4 <= 64) {
+    // row-in-registers fast path (n <= 256 — every ViT attention shape):
+    // ONE read of the scores instead of three (the generic path was
+    // measured exactly at its 3-read traffic floor)
+    bool act = lane < n4;
+    int c0 = lane * 4;
+    float4 v = act ? sr[lane] : float4{0.f, 0.f, 0.f, 0.f};
+    float e[4];
+    float mx = -3.4e38f;
+    if (act) {
+      if (c0 + 0 < nvalid) mx = fmaxf(mx, v.x);
+      if (c0 + 1 < nvalid) mx = fmaxf(mx, v.y);
+      if (c0 + 2 < nvalid) mx = fmaxf(mx, v.z);
+      if (c0 + 3 < nvalid) mx = fmaxf(mx, v.w);
+    }
+    mx = wave_reduce_max(mx) * scale;
+    e[0] = (act && c0 + 0 < nvalid) ? __expf(v.x * scale - mx) : 0.f;
+    e[1] = (act && c0 + 1 < nvalid) ? __expf(v.y * scale - mx) : 0.f;
+    e[2] = (act && c0 + 2 < nvalid) ? __expf(v.z * scale - mx) : 0.f;
+    e[3] = (act && c0 + 3 < nvalid) ? __expf(v.w * scale - mx) : 0.f;
+    float sum = wave_reduce_sum(e[0] + e[1] + e[2] + e[3]);
+    float inv = 1.f / sum;
+    if (act) {
+      if (dt == PDTA_DT_BF16) {
+        __hip_bfloat16 o[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          o[j] = from_f32<__hip_bfloat16>(e[j] * inv);
+        *reinterpret_cast<uint2*>((__hip_bfloat16*)p_out + row * n + c0) =
+            *reinterpret_cast<uint2*>(o);
+      } else {
+        __half o[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) o[j] = from_f32<__half>(e[j] * inv);
+        *reinterpret_cast<uint2*>((__half*)p_out + row * n + c0) =
+            *reinterpret_cast<uint2*>(o);
+      }
+    }
+    return;
+  }
   float mx = -3.4e38f;
   for (int c = lane; c < n4; c += 64) {
     float4 v = sr[c];
@@ -344,6 +384,27 @@ __global__ void softmax_bwd_k(const T* __restrict__ p, const T* __restrict__ dp,
   const T* pr = p + row * n;
   const T* gr = dp + row * n;
   int n8 = n >> 3;
+  if (n8 <= 64) {  // row cached in registers: one read of p and dp
+    bool act = lane < n8;
+    Vec8<T> pv, gv;
+    float dot = 0.f;
+    if (act) {
+      pv = load8(pr + lane * 8);
+      gv = load8(gr + lane * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dot += to_f32(pv.v[j]) * to_f32(gv.v[j]);
+    }
+    dot = wave_reduce_sum(dot);
+    if (act) {
+      Vec8<T> o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o.v[j] = from_f32<T>(scale * to_f32(pv.v[j]) *
+                             (to_f32(gv.v[j]) - dot));
+      store8(ds + row * n + lane * 8, o);
+    }
+    return;
+  }
   float dot = 0.f;
   for (int c = lane; c < n8; c += 64) {
     Vec8<T> pv = load8(pr + c * 8);
