@@ -44,7 +44,9 @@ void pdta_ce_bwd(const void*, const long*, const float*, const float*, void*,
                  int, int, float, float, int, hipStream_t);
 void pdta_gemm_nt(const void*, const void*, void*, const float*, int, int,
                   int, int, int, hipStream_t);
-void pdta_colsum_f32(const void*, float*, long, int, int, hipStream_t);
+int pdta_colsum_ws(int);
+void pdta_colsum_f32(const void*, float*, float*, long, int, int,
+                     hipStream_t);
 long pdta_conv_fwd_chunks(long, int);
 void pdta_bn_stats_from_part(float*, int, float*, float*, float*, float*,
                              float, float, int, long, int, hipStream_t);
@@ -371,9 +373,19 @@ at::Tensor gemm_nt(at::Tensor a, at::Tensor b,
 
 at::Tensor colsum_f32(at::Tensor dy) {
   CHECK_IN(dy);
-  auto db = at::zeros({dy.size(1)}, dy.options().dtype(at::kFloat));
-  pdta_colsum_f32(dy.data_ptr(), db.data_ptr<float>(), dy.size(0),
-                  dy.size(1), dt_of(dy), cur_stream());
+  int N = dy.size(1);
+  int ws = pdta_colsum_ws(N);
+  auto opts = dy.options().dtype(at::kFloat);
+  // scalar fallback (ws == 0) atomics into db and needs it zeroed
+  auto db = ws ? at::empty({N}, opts) : at::zeros({N}, opts);
+  at::Tensor part;
+  float* pp = nullptr;
+  if (ws) {
+    part = at::empty({ws}, opts);
+    pp = part.data_ptr<float>();
+  }
+  pdta_colsum_f32(dy.data_ptr(), db.data_ptr<float>(), pp, dy.size(0), N,
+                  dt_of(dy), cur_stream());
   return db;
 }
 

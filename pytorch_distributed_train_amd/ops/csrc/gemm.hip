@@ -262,10 +262,13 @@ __global__ void colsum_f32_k(const T* __restrict__ dy, float* __restrict__ db,
   atomicAdd(&db[c], acc);
 }
 
-// vectorized variant (N % 8 == 0): 16 B loads, 8 columns per thread
+// vectorized two-stage variant (N % 8 == 0): 16 B loads, 8 columns per
+// thread, per-row-slab partials in plain stores + a finalize kernel (an
+// atomicAdd version serialized ~700 RMWs per column at the L2 and ran 2x
+// SLOWER than the scalar original)
 template <typename T>
-__global__ void colsum8_f32_k(const T* __restrict__ dy,
-                              float* __restrict__ db, long M, int N) {
+__global__ void colsum8_part_k(const T* __restrict__ dy,
+                               float* __restrict__ part, long M, int N) {
   int c8 = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
   if (c8 >= N) return;
   long r0 = blockIdx.y, rs = gridDim.y;
@@ -275,8 +278,26 @@ __global__ void colsum8_f32_k(const T* __restrict__ dy,
 #pragma unroll
     for (int j = 0; j < 8; ++j) acc[j] += to_f32(v.v[j]);
   }
+  float* o = part + (long)blockIdx.y * N + c8;
+  *reinterpret_cast<float4*>(o) = *reinterpret_cast<float4*>(acc);
+  *reinterpret_cast<float4*>(o + 4) = *reinterpret_cast<float4*>(acc + 4);
+}
+
+__global__ void colsum_final_k(const float* __restrict__ part,
+                               float* __restrict__ db, int gy, int N) {
+  __shared__ float red[256];
+  int c = blockIdx.x * 64 + (int)(threadIdx.x % 64);
+  int slice = (int)(threadIdx.x / 64);
+  float s = 0.f;
+  if (c < N)
+    for (int k = slice; k < gy; k += 4) s += part[(long)k * N + c];
+  red[threadIdx.x] = s;
+  __syncthreads();
+  if (slice == 0 && c < N) {
 #pragma unroll
-  for (int j = 0; j < 8; ++j) atomicAdd(&db[c8 + j], acc[j]);
+    for (int i = 1; i < 4; ++i) s += red[threadIdx.x + i * 64];
+    db[c] = s;
+  }
 }
 
 }  // namespace
@@ -355,20 +376,24 @@ void pdta_gemm_nt_batched(const void* A, const void* B, void* C, int M,
   });
 }
 
-// db (fp32 [N], zeroed by caller) += column sums of dy [M,N]
-void pdta_colsum_f32(const void* dy, float* db, long M, int N, int dt,
-                     hipStream_t s) {
+// db (fp32 [N]) = column sums of dy [M,N]. part: fp32 workspace of
+// pdta_colsum_ws(N) elements (vectorized path); db must be pre-zeroed
+// only for the scalar fallback (N % 8 != 0), which atomics into it.
+int pdta_colsum_ws(int N) { return N % 8 == 0 ? 512 * N : 0; }
+
+void pdta_colsum_f32(const void* dy, float* db, float* part, long M, int N,
+                     int dt, hipStream_t s) {
   dim3 block(256);
   if (N % 8 == 0) {
-    // grid.y sized for ~64K active lanes (N/8 lanes per row-slab); a fixed
-    // 128 starved the chip at N=768 (125 us vs 62 for the scalar version)
-    long gy = 524288 / N;
+    long gy = 524288 / N;  // ~64K active lanes in stage 1
     if (gy < 64) gy = 64;
+    if (gy > 512) gy = 512;
     if (gy > M) gy = M;
     dim3 grid((N / 8 + 255) / 256, (unsigned)gy);
     PDTA_DISPATCH_DTYPE16(dt, T, {
-      colsum8_f32_k<T><<<grid, block, 0, s>>>((const T*)dy, db, M, N);
+      colsum8_part_k<T><<<grid, block, 0, s>>>((const T*)dy, part, M, N);
     });
+    colsum_final_k<<<(N + 63) / 64, 256, 0, s>>>(part, db, (int)gy, N);
   } else {
     dim3 grid((N + 255) / 256, (unsigned)min((long)512, M));
     PDTA_DISPATCH_DTYPE16(dt, T, {
