@@ -479,7 +479,7 @@ at::Tensor conv2d_bwd_weight(at::Tensor x, at::Tensor dy, int64_t R,
     auto dw = at::zeros({Kout, C, R, S}, x.options().dtype(at::kFloat));
     long tiles, chunks;
     pdta_wrw_plan(N, P, Q, Kout, C, R, S, &tiles, &chunks);
-    auto part = at::empty({tiles * chunks * 4096},
+    auto part = at::empty({tiles * (chunks + 16) * 4096},
                           x.options().dtype(at::kFloat));
     pdta_conv_bwd_weight(x.data_ptr(), dy.data_ptr(), dw.data_ptr<float>(),
                          part.data_ptr<float>(), zero_page(x), N, H, W, C,
@@ -495,7 +495,7 @@ at::Tensor conv2d_bwd_weight(at::Tensor x, at::Tensor dy, int64_t R,
   auto dwcol = at::zeros({Kout, rsc_pad}, x.options().dtype(at::kFloat));
   long tiles, chunks;
   pdta_wrw_plan(N, P, Q, Kout, rsc_pad, 1, 1, &tiles, &chunks);
-  auto part = at::empty({tiles * chunks * 4096},
+  auto part = at::empty({tiles * (chunks + 16) * 4096},
                         x.options().dtype(at::kFloat));
   pdta_conv_bwd_weight(xcol.data_ptr(), dy.data_ptr(),
                        dwcol.data_ptr<float>(), part.data_ptr<float>(),

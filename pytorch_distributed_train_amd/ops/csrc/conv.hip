@@ -815,6 +815,29 @@ __global__ __launch_bounds__(TPB) void igemm_wrw_k(
 }
 
 // sum the per-chunk partials into dW's torch layout [Kout,C,R,S]
+// first-level slab reduce for long chunk lists: [chunks][tiles*4096] ->
+// [16][tiles*4096] (small-tile layers run splitk up to 256; a single
+// reduce pass there had only tiles*16 waves, each chunk-serial)
+__global__ void wrw_mid_k(const float* __restrict__ part,
+                          float* __restrict__ out, int chunks, long total4) {
+  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride_ = (long)gridDim.x * blockDim.x;
+  int slab = blockIdx.y;  // 16 slabs
+  const float4* p4 = reinterpret_cast<const float4*>(part);
+  float4* o4 = reinterpret_cast<float4*>(out);
+  for (; idx < total4; idx += stride_) {
+    float4 sv = {0.f, 0.f, 0.f, 0.f};
+    for (int k = slab; k < chunks; k += 16) {
+      float4 v = p4[(long)k * total4 + idx];
+      sv.x += v.x;
+      sv.y += v.y;
+      sv.z += v.z;
+      sv.w += v.w;
+    }
+    o4[(long)slab * total4 + idx] = sv;
+  }
+}
+
 __global__ void wrw_reduce_k(const float* __restrict__ part,
                              float* __restrict__ dw, int chunks, int tiles,
                              int ko_tiles, int c_tiles, int C, int Kout,
@@ -1185,7 +1208,8 @@ void pdta_wrw_plan(int Nn, int P, int Q, int Kout, int C, int R, int S,
   *chunks_out = chunks;
 }
 
-// dw fp32 [Kout,C,R,S] ZEROED by caller; part fp32 [chunks*tiles*4096].
+// dw fp32 [Kout,C,R,S] ZEROED by caller; part fp32 [(chunks+16)*tiles*4096]
+// (16 extra rows: wrw_mid_k scratch).
 void pdta_conv_bwd_weight(const void* x, const void* dy, float* dw,
                           float* part, const void* zero_page, int Nn, int H,
                           int W, int C, int P, int Q, int Kout, int R, int S,
@@ -1229,8 +1253,18 @@ void pdta_conv_bwd_weight(const void* x, const void* dy, float* dw,
     });
   }
   long total4 = tiles * 1024;  // float4 elements
+  const float* red_src = part;
+  int red_chunks = (int)chunks;
+  if (chunks > 32) {  // two-level: parallel slab pass, then 16-deep final
+    float* mid = part + (long)chunks * tiles * 4096;
+    dim3 gmid(ew_grid(total4, 256, 1), 16);
+    wrw_mid_k<<<gmid, 256, 0, s>>>(part, mid, (int)chunks, total4);
+    red_src = mid;
+    red_chunks = 16;
+  }
   wrw_reduce_k<<<ew_grid(total4, 256, 1), 256, 0, s>>>(
-      part, dw, (int)chunks, (int)tiles, ko_tiles, c_tiles, C, Kout, R, S);
+      red_src, dw, red_chunks, (int)tiles, ko_tiles, c_tiles, C, Kout, R,
+      S);
 }
 
 void pdta_stem_im2col(const void* x, void* out, int Nn, int H, int W,
