@@ -432,6 +432,21 @@ std::vector<at::Tensor> conv2d_fwd(at::Tensor x, at::Tensor wt,
   return {y};
 }
 
+std::vector<at::Tensor> bn_stats_from_part(at::Tensor part, long rows,
+                                            int64_t C) {
+  // (mean, var) from a conv2d_fwd(want_bn_stats=True) slab — SyncBN local
+  // stats without re-reading the activation
+  CHECK_IN(part);
+  int chunks = (int)(part.numel() / (2 * C)) - 64;
+  auto opts = part.options();
+  auto mean = at::empty({C}, opts), var = at::empty({C}, opts);
+  pdta_bn_stats_from_part(part.data_ptr<float>(), chunks,
+                          mean.data_ptr<float>(), var.data_ptr<float>(),
+                          nullptr, nullptr, 0.f, 0.f, 0, rows, C,
+                          cur_stream());
+  return {mean, var};
+}
+
 std::vector<at::Tensor> bn_stats_train_from_part(at::Tensor part, long rows,
                                                  at::Tensor running_mean,
                                                  at::Tensor running_var,
@@ -779,6 +794,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("stride"), py::arg("pad"),
         py::arg("want_bn_stats") = false);
   m.def("bn_stats_train_from_part", &bn_stats_train_from_part);
+  m.def("bn_stats_from_part", &bn_stats_from_part);
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
   m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
   m.def("gemm_nt_batched", &gemm_nt_batched, py::arg("a"), py::arg("b"),

@@ -215,8 +215,13 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
 
 
 def bn_stats(x):
-    """Per-channel (mean, var) over N*H*W in fp32. GPU: reduction kernel."""
+    """Per-channel (mean, var) over N*H*W in fp32. GPU: reduction kernel,
+    or finalize-only when the producing conv pre-reduced the partials."""
     if x.is_cuda:
+        part = getattr(x, "_pdta_bn_part", None)
+        if part is not None:
+            rows = x.shape[0] * x.shape[1] * x.shape[2]
+            return ext().bn_stats_from_part(part, rows, x.shape[3])
         return ext().bn_stats(x)
     return _bn_cpu_stats(x)
 
