@@ -20,8 +20,11 @@ def main(path, out=None):
     ctab = [t for t in tables if "counter" in t.lower()]
     dtab = [t for t in tables if "kernel_dispatch" in t]
     if not ctab:
-        print("tables:", "\n".join(tables))
-        sys.exit("no counter table")
+        text = "no counter table; tables:\n" + "\n".join(tables)
+        print(text)
+        if out:
+            open(out, "w").write(text + "\n")
+        sys.exit(1)
     lines = []
     for t in ctab:
         cols = [r[1] for r in cur.execute(f"PRAGMA table_info({t})")]
