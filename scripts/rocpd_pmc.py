@@ -44,11 +44,27 @@ def main(path, out=None):
     except Exception as e:
         lines.append(f"primary query failed: {e}")
         rows = []
-        # fallback: dump any counters_collection-like table schema
-        for t in tables:
-            if "collection" in t or "counter" in t:
-                cols = [r[1] for r in cur.execute(f"PRAGMA table_info({t})")]
-                lines.append(f"{t}: {cols}")
+        # fallback: try the rocpd_pmc_event layout (ROCm 7.x rocpd)
+        try:
+            q = f"""
+            SELECT s.display_name, i.name, SUM(p.value), COUNT(*)
+            FROM rocpd_pmc_event_{uuid} p
+            JOIN rocpd_info_pmc_{uuid} i ON i.id = p.pmc_id
+            JOIN rocpd_event_{uuid} e ON e.id = p.event_id
+            JOIN rocpd_kernel_dispatch_{uuid} d ON d.event_id = e.id
+            JOIN rocpd_info_kernel_symbol_{uuid} s ON s.id = d.kernel_id
+            GROUP BY s.display_name, i.name"""
+            rows = list(cur.execute(q))
+        except Exception as e2:
+            lines.append(f"fallback query failed: {e2}")
+            for t in tables:
+                if any(k in t for k in ("pmc", "kernel_dispatch", "event",
+                                        "kernel_symbol")):
+                    cols = [r[1] for r in
+                            cur.execute(f"PRAGMA table_info({t})")]
+                    lines.append(f"{t}:\n  cols={cols}")
+                    for r in cur.execute(f"SELECT * FROM {t} LIMIT 2"):
+                        lines.append(f"  row={r!r:.300s}")
     agg = {}
     for name, cname, val, n in rows:
         short = name.split("(")[0][:70]
