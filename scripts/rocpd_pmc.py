@@ -17,7 +17,8 @@ def main(path, out=None):
     cur = db.cursor()
     tables = [n for (n,) in cur.execute(
         "SELECT name FROM sqlite_master WHERE type='table'")]
-    ctab = [t for t in tables if "counter" in t.lower()]
+    ctab = [t for t in tables
+            if "counter" in t.lower() or "pmc_event" in t.lower()]
     dtab = [t for t in tables if "kernel_dispatch" in t]
     if not ctab:
         text = "no counter table; tables:\n" + "\n".join(tables)
