@@ -68,7 +68,8 @@ def main(path, out=None):
                         lines.append(f"  row={r!r:.300s}")
     agg = {}
     for name, cname, val, n in rows:
-        short = name.split("(")[0][:70]
+        short = name.replace("(anonymous namespace)::", "")
+        short = short.split("(")[0][:70]
         agg.setdefault(short, {})[cname] = (val, n)
     for k, counters in sorted(
             agg.items(),
