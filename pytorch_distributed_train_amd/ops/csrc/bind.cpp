@@ -392,7 +392,7 @@ at::Tensor colsum_f32(at::Tensor dy) {
 // ---- conv ----------------------------------------------------------------
 std::vector<at::Tensor> conv2d_fwd(at::Tensor x, at::Tensor wt,
                                    int64_t stride, int64_t pad,
-                                   bool want_bn_stats) {
+                                   bool want_bn_stats, bool want_xcol) {
   // x [N,H,W,C]; wt pack [Kout,R,S,C]. want_bn_stats: the fwd kernel's
   // epilogue also emits per-m-tile (sum, sumsq) channel partials of y so
   // the following BatchNorm skips its stage-1 read of y
@@ -427,6 +427,10 @@ std::vector<at::Tensor> conv2d_fwd(at::Tensor x, at::Tensor wt,
     pdta_conv_fwd(xcol.data_ptr(), wt_pad.data_ptr(), y.data_ptr(),
                   zero_page(x), N, P, Q, rsc_pad, P, Q, Kout, 1, 1, 1, 0,
                   dt_of(x), cur_stream(), part_ptr);
+    std::vector<at::Tensor> ret = {y};
+    if (want_bn_stats) ret.push_back(part);
+    if (want_xcol) ret.push_back(xcol);  // reused by conv2d_bwd_weight
+    return ret;
   }
   if (want_bn_stats) return {y, part};
   return {y};
@@ -485,7 +489,8 @@ at::Tensor conv2d_bwd_data(at::Tensor dy, at::Tensor w2, int64_t H,
 }
 
 at::Tensor conv2d_bwd_weight(at::Tensor x, at::Tensor dy, int64_t R,
-                             int64_t S, int64_t stride, int64_t pad) {
+                             int64_t S, int64_t stride, int64_t pad,
+                             c10::optional<at::Tensor> xcol_cached) {
   CHECK_IN(x);
   CHECK_IN(dy);
   int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
@@ -504,9 +509,16 @@ at::Tensor conv2d_bwd_weight(at::Tensor x, at::Tensor dy, int64_t R,
   }
   // stem: im2col + 1x1 igemm wrw, then repack [K][r][s][c] -> [K,C,R,S]
   int rsc = R * S * C, rsc_pad = (rsc + 63) / 64 * 64;
-  auto xcol = at::empty({(long)N * P * Q, rsc_pad}, x.options());
-  pdta_stem_im2col(x.data_ptr(), xcol.data_ptr(), N, H, W, C, P, Q, R, S,
-                   stride, pad, rsc_pad, dt_of(x), cur_stream());
+  at::Tensor xcol;
+  if (xcol_cached.has_value()) {  // conv2d_fwd(want_xcol=True) output
+    xcol = *xcol_cached;
+    TORCH_CHECK(xcol.size(0) == (long)N * P * Q &&
+                xcol.size(1) == rsc_pad, "cached xcol shape mismatch");
+  } else {
+    xcol = at::empty({(long)N * P * Q, rsc_pad}, x.options());
+    pdta_stem_im2col(x.data_ptr(), xcol.data_ptr(), N, H, W, C, P, Q, R, S,
+                     stride, pad, rsc_pad, dt_of(x), cur_stream());
+  }
   auto dwcol = at::zeros({Kout, rsc_pad}, x.options().dtype(at::kFloat));
   long tiles, chunks;
   pdta_wrw_plan(N, P, Q, Kout, rsc_pad, 1, 1, &tiles, &chunks);
@@ -792,11 +804,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsum_f32", &colsum_f32);
   m.def("conv2d_fwd", &conv2d_fwd, py::arg("x"), py::arg("wt"),
         py::arg("stride"), py::arg("pad"),
-        py::arg("want_bn_stats") = false);
+        py::arg("want_bn_stats") = false, py::arg("want_xcol") = false);
   m.def("bn_stats_train_from_part", &bn_stats_train_from_part);
   m.def("bn_stats_from_part", &bn_stats_from_part);
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
-  m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
+  m.def("conv2d_bwd_weight", &conv2d_bwd_weight, py::arg("x"),
+        py::arg("dy"), py::arg("R"), py::arg("S"),
+        py::arg("stride"), py::arg("pad"),
+        py::arg("xcol_cached") = c10::nullopt);
   m.def("gemm_nt_batched", &gemm_nt_batched, py::arg("a"), py::arg("b"),
         py::arg("out_f32") = false);
   m.def("ln_fwd", &ln_fwd);

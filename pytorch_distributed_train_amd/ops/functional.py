@@ -56,14 +56,19 @@ class Conv2dNHWC(torch.autograd.Function):
     def forward(ctx, x, weight, wt, w2, stride: int, pad: int,
                 want_bn_stats: bool = False):
         ctx.stride, ctx.pad = stride, pad
-        ctx.save_for_backward(x, weight, w2)
         if x.is_cuda:
-            out = ext().conv2d_fwd(x, wt, stride, pad, want_bn_stats)
+            # stem (C % 64 != 0): cache the im2col image for bwd-weight
+            want_xcol = x.shape[3] % 64 != 0 and torch.is_grad_enabled()
+            out = ext().conv2d_fwd(x, wt, stride, pad, want_bn_stats,
+                                   want_xcol)
+            xcol = out[-1] if want_xcol else None
+            ctx.save_for_backward(x, weight, w2, xcol)
             if want_bn_stats:
-                y, part = out
+                y, part = out[0], out[1]
                 ctx.mark_non_differentiable(part)
                 return y, part
             return out[0]
+        ctx.save_for_backward(x, weight, w2, None)
         # CPU reference (fp32)
         w = weight.float()
         y = F.conv2d(_nhwc_to_nchw(x).float(), w, stride=stride, padding=pad)
@@ -71,7 +76,7 @@ class Conv2dNHWC(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy, _dpart=None):
-        x, weight, w2 = ctx.saved_tensors
+        x, weight, w2, xcol = ctx.saved_tensors
         stride, pad = ctx.stride, ctx.pad
         dy = dy.contiguous()
         need_dx = ctx.needs_input_grad[0]
@@ -83,7 +88,8 @@ class Conv2dNHWC(torch.autograd.Function):
                                            stride, pad)
             if need_dw:
                 dw = ext().conv2d_bwd_weight(x, dy, weight.shape[2],
-                                             weight.shape[3], stride, pad)
+                                             weight.shape[3], stride, pad,
+                                             xcol_cached=xcol)
         else:
             x_nchw = _nhwc_to_nchw(x).float()
             dy_nchw = _nhwc_to_nchw(dy).float()
