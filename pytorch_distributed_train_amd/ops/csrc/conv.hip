@@ -915,7 +915,8 @@ __global__ __launch_bounds__(512) void igemm_wrw_big_k(
     for (int i = 0; i < 4; ++i) {
       int lrow = (wave * 4 + i) * 2 + lrow_in;
       long row = rowbase + lrow;
-      int swz = (c32 ^ (lrow & 7)) * 16;
+      int swz = (c32 ^ (lrow & 31)) * 16;  // full-range XOR (see
+      // frag_read_tr_issue_s)
       const char* g;
       if (row < r1)
         g = reinterpret_cast<const char*>(dy) +
@@ -928,7 +929,8 @@ __global__ __launch_bounds__(512) void igemm_wrw_big_k(
     for (int i = 0; i < 4; ++i) {
       int lrow = (wave * 4 + i) * 2 + lrow_in;
       long row = rowbase + lrow;
-      int swz = (c32 ^ (lrow & 7)) * 16;
+      int swz = (c32 ^ (lrow & 31)) * 16;  // full-range XOR (see
+      // frag_read_tr_issue_s)
       const char* g = reinterpret_cast<const char*>(zero_page);
       if (row < r1) {
         if (GATHER) {

@@ -77,12 +77,17 @@ DEV_INLINE void frag_read_tr_issue(unsigned lds_base, int k0, int m0,
 }
 
 // Stride-parameterized variant: [k][m] LDS image with ROW_BYTES-wide rows
-// (e.g. 512 B for 256-feature tiles) and the chunk^(k&7) swizzle on 16-B
-// chunks within the row.
+// (e.g. 512 B for 256-feature tiles). The XOR swizzle uses the FULL chunk
+// range of the row (k & (ROW_BYTES/16 - 1)): one ds_read_b64_tr_b16
+// touches rows up to 24 apart across its four 16-lane groups, and a 3-bit
+// mask left rows 8/16/24 apart on identical banks (PMC: 18% of wrw-big
+// wave cycles in LDS bank-conflict stalls; 6 conflict cycles per access
+// vs the 2-cycle floor of an 8-B/lane read).
 template <int ROW_BYTES>
 DEV_INLINE void frag_read_tr_issue_s(unsigned lds_base, int k0, int m0,
                                      int lane, unsigned long long* lo,
                                      unsigned long long* hi) {
+  constexpr int KMASK = ROW_BYTES / 16 - 1;
   int g = lane >> 4;
   int row = (lane & 15) >> 2;
   int seg = lane & 3;
@@ -91,9 +96,9 @@ DEV_INLINE void frag_read_tr_issue_s(unsigned lds_base, int k0, int m0,
   int ka = k0 + g * 8 + row;
   int kb = ka + 4;
   unsigned a0 =
-      lds_base + ka * ROW_BYTES + (((chunk ^ (ka & 7)) << 4) | rem);
+      lds_base + ka * ROW_BYTES + (((chunk ^ (ka & KMASK)) << 4) | rem);
   unsigned a1 =
-      lds_base + kb * ROW_BYTES + (((chunk ^ (kb & 7)) << 4) | rem);
+      lds_base + kb * ROW_BYTES + (((chunk ^ (kb & KMASK)) << 4) | rem);
   asm volatile(
       "ds_read_b64_tr_b16 %0, %2\n\t"
       "ds_read_b64_tr_b16 %1, %3"
