@@ -39,12 +39,13 @@ def _global_batch(n=8):
     return x, y
 
 
-def _worker(rank, world, port, sync_bn, identical_batches, out_dir):
+def _worker(rank, world, port, sync_bn, identical_batches, out_dir,
+            extra_flags=()):
     os.environ.update(RANK=str(rank), LOCAL_RANK=str(rank),
                       WORLD_SIZE=str(world), MASTER_ADDR="127.0.0.1",
                       MASTER_PORT=str(port))
     from pytorch_distributed_train_amd.engine import Trainer
-    extra = ["--sync-bn"] if sync_bn else []
+    extra = (["--sync-bn"] if sync_bn else []) + list(extra_flags)
     trainer = Trainer(_make_args(extra))
     x, y = _global_batch()
     if identical_batches:
@@ -70,14 +71,14 @@ def _single_process_grads(x, y, sync_semantics):
     return trainer.flat.grad.clone(), trainer.flat.master.clone()
 
 
-def _run_workers(sync_bn, identical_batches, tmp_path):
+def _run_workers(sync_bn, identical_batches, tmp_path, extra_flags=()):
     port = _free_port()
     ctx = mp.get_context("spawn")
     procs = []
     for r in range(2):
         p = ctx.Process(target=_worker,
                         args=(r, 2, port, sync_bn, identical_batches,
-                              str(tmp_path)))
+                              str(tmp_path), tuple(extra_flags)))
         p.start()
         procs.append(p)
     for p in procs:
@@ -152,3 +153,19 @@ def test_grad_compression_bf16(tmp_path):
     # round-trip through bf16: grads equal to bf16 precision
     assert T.allclose(flat.grad, ref.to(T.bfloat16).float(),
                       atol=0, rtol=0)
+
+
+def test_ddp_bf16_compressed_comm_end_to_end(tmp_path):
+    """--grad-comm-dtype bf16 through the full Trainer: ranks stay
+    identical and match the uncompressed gradients to bf16 precision."""
+    r0, r1 = _run_workers(sync_bn=False, identical_batches=True,
+                          tmp_path=tmp_path,
+                          extra_flags=("--grad-comm-dtype", "bf16"))
+    assert torch.allclose(r0["grad"], r1["grad"], atol=1e-6)
+    assert torch.allclose(r0["master"], r1["master"], atol=1e-6)
+    x, y = _global_batch()
+    ref_grad, _ = _single_process_grads(x[:4], y[:4], False)
+    # identical ranks -> AVG is identity up to the bf16 round-trip
+    err = (r0["grad"] - ref_grad).abs()
+    tol = ref_grad.abs() * 2 ** -8 + 1e-5
+    assert (err <= tol).float().mean() > 0.999, err.max()
