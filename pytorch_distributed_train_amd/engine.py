@@ -91,6 +91,17 @@ class Trainer:
         parts = spec.split(":")
         return int(parts[1]), int(parts[3])
 
+    def _check_grads_finite(self):
+        """NanCheck analog: one reduction over the flat gradient; raises
+        with rank info so the launcher's gang restart (or the operator)
+        can attribute the failure."""
+        if not bool(torch.isfinite(self.flat.grad).all()):
+            n_bad = int((~torch.isfinite(self.flat.grad)).sum())
+            raise RuntimeError(
+                f"non-finite gradient after all-reduce: rank "
+                f"{self.env.rank} step {self.global_step} "
+                f"({n_bad} bad elements)")
+
     def train_step(self, images, labels) -> torch.Tensor:
         """One optimization step; returns the (device) scalar loss."""
         if self._fault and self.global_step == self._fault[0] \
@@ -109,6 +120,8 @@ class Trainer:
             self.scaler.scale(loss).backward()
         with record_function("pdta.reduce_finalize"):
             self.reducer.finalize()
+            if getattr(self.args, 'nan_check', False):
+                self._check_grads_finite()
         skip = self.scaler.unscale_and_check(self.flat.grad, self.comm)
         if not skip:
             with record_function("pdta.optimizer"):

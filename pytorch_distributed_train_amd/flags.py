@@ -78,6 +78,11 @@ def build_parser() -> argparse.ArgumentParser:
                         "progress for this long (0 = off)")
     p.add_argument("--fault-inject", default="",
                    help="step:N:rank:R -> raise at step N on rank R (testing)")
+    p.add_argument("--nan-check", action="store_true",
+                   help="after each backward+all-reduce, verify the flat "
+                        "gradient is finite and abort with rank info if "
+                        "not (ProcessGroupNCCL NanCheck analog; adds one "
+                        "host sync per step)")
     return p
 
 

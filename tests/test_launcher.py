@@ -60,3 +60,44 @@ def test_launcher_exhausts_restarts(tmp_path):
     r = _launch(script, tmp_path, extra=("--max-restarts=1",))
     assert r.returncode == 3
     assert "exhausted" in r.stderr
+
+
+def test_fault_injection_raises_at_step(tmp_path):
+    """--fault-inject step:N:rank:R trips exactly at step N on rank R
+    (single-process: rank 0)."""
+    import torch
+    import pytest as _pytest
+    from pytorch_distributed_train_amd.flags import build_parser
+    from pytorch_distributed_train_amd.engine import Trainer
+
+    args = build_parser().parse_args([
+        "--model", "resnet18", "--data", "synthetic-cifar10",
+        "--dtype", "fp32", "--device", "cpu", "--batch-size", "2",
+        "--fault-inject", "step:1:rank:0"])
+    trainer = Trainer(args)
+    trainer.model.train()
+    x = torch.randn(2, 3, 32, 32)
+    y = torch.randint(0, 10, (2,))
+    trainer.train_step(x, y)  # step 0 fine
+    with _pytest.raises(RuntimeError, match="fault injection: step 1"):
+        trainer.train_step(x, y)
+
+
+def test_nan_check_trips_on_poisoned_grad():
+    import torch
+    import pytest as _pytest
+    from pytorch_distributed_train_amd.flags import build_parser
+    from pytorch_distributed_train_amd.engine import Trainer
+
+    args = build_parser().parse_args([
+        "--model", "resnet18", "--data", "synthetic-cifar10",
+        "--dtype", "fp32", "--device", "cpu", "--batch-size", "2",
+        "--nan-check"])
+    trainer = Trainer(args)
+    trainer.model.train()
+    x = torch.randn(2, 3, 32, 32)
+    y = torch.randint(0, 10, (2,))
+    trainer.train_step(x, y)  # clean step passes the check
+    trainer.flat.grad[3] = float("nan")
+    with _pytest.raises(RuntimeError, match="non-finite gradient"):
+        trainer._check_grads_finite()
