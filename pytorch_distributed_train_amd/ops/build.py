@@ -22,7 +22,7 @@ ROCM = os.environ.get("ROCM_HOME", "/opt/rocm")
 TORCH_DIR = os.path.dirname(os.path.abspath(torch.__file__))
 
 HIP_SOURCES = ["elementwise.hip", "bn.hip", "pool.hip", "sgd.hip", "ce.hip",
-               "gemm.hip", "conv.hip", "ln.hip", "packs.hip", "gemm_strided.hip"]
+               "gemm.hip", "conv.hip", "ln.hip", "packs.hip", "gemm_strided.hip", "attn.hip"]
 CPP_SOURCES = ["bind.cpp"]
 
 SO_NAME = "_hip_ops" + sysconfig.get_config_var("EXT_SUFFIX")

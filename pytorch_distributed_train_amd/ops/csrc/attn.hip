@@ -1,0 +1,197 @@
+// Fused attention forward for the ViT path (SURVEY.md §2d softmax row;
+// BASELINE config #5): one kernel computes P = softmax(scale · Q·K^T)
+// (masked past nvalid — sequence padding to 256 stays exact, proven in
+// tests/test_vit_cpu.py) and O = P·V for one (batch·head, 64-query) block.
+//
+// Replaces the scores-GEMM (fp32 [S,S] round trip through HBM), the
+// standalone softmax, and the P·V GEMM of the decomposed path
+// (bind.cpp attn_fwd): the scores never leave registers, P makes exactly
+// one HBM write (kept for the decomposed backward), and K/V share one LDS
+// slab (V is staged while softmax computes, hiding its latency).
+//
+// Zero-copy operands: q/k/v are element offsets into the interleaved
+// [b, S, 3·h·hd] qkv projection (row stride 3·h·hd) — no permute/copy
+// kernels anywhere, as in the decomposed path.
+//
+// Shape-specialized: S(padded) == 256, head_dim == 64 (ViT-B/16 and
+// anything padded to it); other shapes take the decomposed path.
+#include "common.h"
+#include "mfma_tile.h"
+
+namespace {
+
+constexpr int SP = 256;  // padded sequence length
+constexpr int HD = 64;   // head dim
+
+template <typename T>
+__global__ __launch_bounds__(256) void attn_fused_fwd_k(
+    const T* __restrict__ qkv, T* __restrict__ probs, T* __restrict__ out,
+    int h, int nvalid, float scale) {
+  // grid: (SP/64 query blocks, b*h). 4 waves; each owns 16 query rows.
+  __shared__ __attribute__((aligned(16))) char qtile[64 * 128];
+  __shared__ __attribute__((aligned(16))) char kv[SP * 128];  // K, then V
+  __shared__ __attribute__((aligned(16))) char pimg[64 * 512];
+
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  int bh = blockIdx.y;
+  int batch = bh / h, head = bh % h;
+  long d = (long)h * HD, ld3 = 3 * d;
+  const T* qbase = qkv + (long)batch * SP * ld3 + (long)head * HD;
+  const T* kbase = qbase + d;
+  const T* vbase = qbase + 2 * d;
+  int q0 = blockIdx.x * 64;
+
+  // ---- stage Q [64][128B] and K [256][128B] (glds: 1 KiB = 8 rows each)
+  int row_in = lane >> 3, c16 = lane & 7;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    int row = (wave * 2 + i) * 8 + row_in;
+    const char* g =
+        reinterpret_cast<const char*>(qbase + (long)(q0 + row) * ld3) +
+        (c16 ^ (row & 7)) * 16;
+    glds16(g, qtile + (wave * 2 + i) * 1024);
+  }
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    int row = (wave * 8 + i) * 8 + row_in;
+    const char* g =
+        reinterpret_cast<const char*>(kbase + (long)row * ld3) +
+        (c16 ^ (row & 7)) * 16;
+    glds16(g, kv + (wave * 8 + i) * 1024);
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  // ---- S = Q·K^T for this wave's 16 query rows (all 256 key cols)
+  f32x4 acc[16];
+#pragma unroll
+  for (int jn = 0; jn < 16; ++jn) acc[jn] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+  for (int kk = 0; kk < 2; ++kk) {
+    s16x8 af =
+        frag_read128(qtile, wave * 16 + (lane & 15), (lane >> 4) + kk * 4);
+#pragma unroll
+    for (int jn = 0; jn < 16; ++jn) {
+      s16x8 bf =
+          frag_read128(kv, jn * 16 + (lane & 15), (lane >> 4) + kk * 4);
+      acc[jn] = MfmaOp<T>::run(af, bf, acc[jn]);
+    }
+  }
+  __syncthreads();  // all waves done reading K: the slab becomes V
+
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {  // stage V; its latency hides under softmax
+    int row = (wave * 8 + i) * 8 + row_in;
+    const char* g =
+        reinterpret_cast<const char*>(vbase + (long)row * ld3) +
+        (c16 ^ (row & 7)) * 16;
+    glds16(g, kv + (wave * 8 + i) * 1024);
+  }
+
+  // ---- masked softmax in registers (C layout: row (lane>>4)*4+r,
+  //      col jn*16 + (lane&15); a row lives in the 16 lanes of one group)
+  int colbase = lane & 15;
+  float mx[4] = {-3.4e38f, -3.4e38f, -3.4e38f, -3.4e38f};
+#pragma unroll
+  for (int jn = 0; jn < 16; ++jn) {
+    int col = jn * 16 + colbase;
+    if (col < nvalid)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) mx[r] = fmaxf(mx[r], acc[jn][r]);
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+#pragma unroll
+    for (int m = 1; m < 16; m <<= 1)
+      mx[r] = fmaxf(mx[r], __shfl_xor(mx[r], m));
+    mx[r] *= scale;
+  }
+  float sum[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+  for (int jn = 0; jn < 16; ++jn) {
+    int col = jn * 16 + colbase;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float e = (col < nvalid) ? __expf(acc[jn][r] * scale - mx[r]) : 0.f;
+      acc[jn][r] = e;
+      sum[r] += e;
+    }
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+#pragma unroll
+    for (int m = 1; m < 16; m <<= 1) sum[r] += __shfl_xor(sum[r], m);
+    sum[r] = 1.f / sum[r];
+  }
+
+  // ---- write P: LDS A-image [64][512B] for the P·V pass + ONE global
+  //      store (bf16/fp16) for the decomposed backward
+  T* pg = probs + (long)bh * SP * SP + (long)q0 * SP;
+#pragma unroll
+  for (int jn = 0; jn < 16; ++jn) {
+    int col = jn * 16 + colbase;
+    int byte = col * 2;
+    int chunk = byte >> 4, rem = byte & 15;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = wave * 16 + (lane >> 4) * 4 + r;
+      T pv = from_f32<T>(acc[jn][r] * sum[r]);
+      *reinterpret_cast<T*>(pimg + row * 512 +
+                            (((chunk ^ (row & 7)) << 4) | rem)) = pv;
+      pg[(long)row * SP + col] = pv;
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // V staged
+  __syncthreads();                                  // P image visible
+
+  // ---- O = P·V (K-dim = 256 keys, 8 MFMA steps; V^T fragments via
+  //      ds_read_b64_tr_b16 from the [k][m] V image)
+  f32x4 acc2[4];
+#pragma unroll
+  for (int jn = 0; jn < 4; ++jn) acc2[jn] = {0.f, 0.f, 0.f, 0.f};
+  unsigned vb = lds_addr_of(kv);
+#pragma unroll
+  for (int kt = 0; kt < 8; ++kt) {
+    s16x8 af = frag_read_s<512>(pimg, wave * 16 + (lane & 15),
+                                kt * 4 + (lane >> 4));
+    unsigned long long lo[4], hi[4];
+#pragma unroll
+    for (int jn = 0; jn < 4; ++jn)
+      frag_read_tr_issue(vb, kt * 32, jn * 16, lane, &lo[jn], &hi[jn]);
+    tr_read_wait();
+#pragma unroll
+    for (int jn = 0; jn < 4; ++jn)
+      acc2[jn] = MfmaOp<T>::run(af, tr_pair_to_frag(lo[jn], hi[jn]),
+                                acc2[jn]);
+  }
+
+  T* obase = out + (long)batch * SP * d + (long)head * HD;
+#pragma unroll
+  for (int jn = 0; jn < 4; ++jn) {
+    int col = jn * 16 + colbase;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = q0 + wave * 16 + (lane >> 4) * 4 + r;
+      obase[(long)row * d + col] = from_f32<T>(acc2[jn][r]);
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+// 1 if the fused kernel handles this shape (bind.cpp attn_fwd dispatch)
+int pdta_attn_fused_ok(int S, int hd) { return S == SP && hd == HD; }
+
+void pdta_attn_fused_fwd(const void* qkv, void* probs, void* out, int b,
+                         int h, int nvalid, float scale, int dt,
+                         hipStream_t s) {
+  dim3 grid(SP / 64, b * h);
+  PDTA_DISPATCH_DTYPE16(dt, T, {
+    attn_fused_fwd_k<T><<<grid, 256, 0, s>>>(
+        (const T*)qkv, (T*)probs, (T*)out, h, nvalid, scale);
+  });
+}
+
+}  // extern "C"

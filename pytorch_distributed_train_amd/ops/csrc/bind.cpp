@@ -81,6 +81,9 @@ void pdta_gemm_bs(const void*, const void*, void*, int, int, int, int, int,
                   int, int, int, hipStream_t);
 void pdta_pack_weights(const float*, void*, const long*, const long*, int,
                        long, int, hipStream_t);
+int pdta_attn_fused_ok(int, int);
+void pdta_attn_fused_fwd(const void*, void*, void*, int, int, int, float,
+                         int, hipStream_t);
 }
 
 namespace {
@@ -638,6 +641,14 @@ std::vector<at::Tensor> attn_fwd(at::Tensor qkv, int64_t b, int64_t S,
   char* q = (char*)qkv.data_ptr();
   char* k = q + d * 2;   // 16-bit elements
   char* v = q + 2 * d * 2;
+  if (pdta_attn_fused_ok(S, hd)) {
+    // one kernel: scores never touch HBM, P written once for backward
+    auto p = at::empty({b * h, S, S}, qkv.options());
+    auto out = at::empty({b, S, d}, qkv.options());
+    pdta_attn_fused_fwd(qkv.data_ptr(), p.data_ptr(), out.data_ptr(), b, h,
+                        nvalid, (float)scale, dt, cur_stream());
+    return {out, p};
+  }
   auto scores = at::empty({b * h, S, S}, qkv.options().dtype(at::kFloat));
   pdta_gemm_bs(q, k, scores.data_ptr(), S, S, hd, ld, ld, S, b * h, h,
                S * ld, hd, S * ld, hd, (long)h * S * S, S * S, 0, 0, 1, dt,

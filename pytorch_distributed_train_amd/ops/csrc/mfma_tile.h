@@ -106,6 +106,14 @@ DEV_INLINE void frag_read_tr_issue_s(unsigned lds_base, int k0, int m0,
       : "v"(a0), "v"(a1));
 }
 
+// Row-pitch-parameterized plain fragment read (frag_read128 for images
+// whose rows are ROW_BYTES wide; same 16-B-chunk XOR swizzle).
+template <int ROW_BYTES>
+DEV_INLINE s16x8 frag_read_s(const char* lds, int row, int chunk) {
+  return *reinterpret_cast<const s16x8*>(
+      lds + row * ROW_BYTES + ((chunk ^ (row & 7)) * 16));
+}
+
 // Wait for every outstanding LDS read, then fence the scheduler so no
 // consumer (incl. register-only MFMA — guide §5.4 rule 18) is hoisted
 // above the wait.
