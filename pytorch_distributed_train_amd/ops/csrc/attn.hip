@@ -177,6 +177,96 @@ __global__ __launch_bounds__(256) void attn_fused_fwd_k(
   }
 }
 
+// Fused attention backward, stage 1: dP = dO·V^T and
+// dS = scale · P ∘ (dP − rowsum(P ∘ dP)) in one kernel — the [S,S] dP
+// tensor never touches HBM (the decomposed path wrote and re-read it).
+// dS is materialized (bf16/fp16) because both dQ = dS·K and
+// dK = dS^T·Q consume it.
+template <typename T>
+__global__ __launch_bounds__(256) void attn_fused_bwd_ds_k(
+    const T* __restrict__ qkv, const T* __restrict__ p,
+    const T* __restrict__ dout, T* __restrict__ ds, int h, float scale) {
+  __shared__ __attribute__((aligned(16))) char dotile[64 * 128];
+  __shared__ __attribute__((aligned(16))) char vimg[SP * 128];
+
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  int bh = blockIdx.y;
+  int batch = bh / h, head = bh % h;
+  long d = (long)h * HD, ld3 = 3 * d;
+  const T* vbase =
+      qkv + (long)batch * SP * ld3 + (long)head * HD + 2 * d;
+  const T* dobase = dout + (long)batch * SP * d + (long)head * HD;
+  int q0 = blockIdx.x * 64;
+
+  int row_in = lane >> 3, c16 = lane & 7;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    int row = (wave * 2 + i) * 8 + row_in;
+    const char* g =
+        reinterpret_cast<const char*>(dobase + (long)(q0 + row) * d) +
+        (c16 ^ (row & 7)) * 16;
+    glds16(g, dotile + (wave * 2 + i) * 1024);
+  }
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    int row = (wave * 8 + i) * 8 + row_in;
+    const char* g =
+        reinterpret_cast<const char*>(vbase + (long)row * ld3) +
+        (c16 ^ (row & 7)) * 16;
+    glds16(g, vimg + (wave * 8 + i) * 1024);
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  // dP[m, n] = sum_f dO[m, f] · V[n, f]  (nt shape)
+  f32x4 acc[16];
+#pragma unroll
+  for (int jn = 0; jn < 16; ++jn) acc[jn] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+  for (int kk = 0; kk < 2; ++kk) {
+    s16x8 af =
+        frag_read128(dotile, wave * 16 + (lane & 15), (lane >> 4) + kk * 4);
+#pragma unroll
+    for (int jn = 0; jn < 16; ++jn) {
+      s16x8 bf =
+          frag_read128(vimg, jn * 16 + (lane & 15), (lane >> 4) + kk * 4);
+      acc[jn] = MfmaOp<T>::run(af, bf, acc[jn]);
+    }
+  }
+
+  // read this block's P tile (C layout), rowdot, dS
+  int colbase = lane & 15;
+  const T* pg = p + (long)bh * SP * SP + (long)q0 * SP;
+  T* dg = ds + (long)bh * SP * SP + (long)q0 * SP;
+  float pv[16][4];
+  float dot[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+  for (int jn = 0; jn < 16; ++jn) {
+    int col = jn * 16 + colbase;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = wave * 16 + (lane >> 4) * 4 + r;
+      pv[jn][r] = to_f32(pg[(long)row * SP + col]);
+      dot[r] += pv[jn][r] * acc[jn][r];
+    }
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+#pragma unroll
+    for (int m = 1; m < 16; m <<= 1) dot[r] += __shfl_xor(dot[r], m);
+  }
+#pragma unroll
+  for (int jn = 0; jn < 16; ++jn) {
+    int col = jn * 16 + colbase;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = wave * 16 + (lane >> 4) * 4 + r;
+      dg[(long)row * SP + col] =
+          from_f32<T>(scale * pv[jn][r] * (acc[jn][r] - dot[r]));
+    }
+  }
+}
+
 }  // namespace
 
 extern "C" {
@@ -191,6 +281,16 @@ void pdta_attn_fused_fwd(const void* qkv, void* probs, void* out, int b,
   PDTA_DISPATCH_DTYPE16(dt, T, {
     attn_fused_fwd_k<T><<<grid, 256, 0, s>>>(
         (const T*)qkv, (T*)probs, (T*)out, h, nvalid, scale);
+  });
+}
+
+void pdta_attn_fused_bwd_ds(const void* qkv, const void* p,
+                            const void* dout, void* ds, int b, int h,
+                            float scale, int dt, hipStream_t s) {
+  dim3 grid(SP / 64, b * h);
+  PDTA_DISPATCH_DTYPE16(dt, T, {
+    attn_fused_bwd_ds_k<T><<<grid, 256, 0, s>>>(
+        (const T*)qkv, (const T*)p, (const T*)dout, (T*)ds, h, scale);
   });
 }
 

@@ -84,6 +84,8 @@ void pdta_pack_weights(const float*, void*, const long*, const long*, int,
 int pdta_attn_fused_ok(int, int);
 void pdta_attn_fused_fwd(const void*, void*, void*, int, int, int, float,
                          int, hipStream_t);
+void pdta_attn_fused_bwd_ds(const void*, const void*, const void*, void*,
+                            int, int, float, int, hipStream_t);
 }
 
 namespace {
@@ -675,13 +677,20 @@ at::Tensor attn_bwd(at::Tensor qkv, at::Tensor p, at::Tensor dout,
   char* q = (char*)qkv.data_ptr();
   char* k = q + d * 2;
   char* v = q + 2 * d * 2;
-  auto dp = at::empty({b * h, S, S}, qkv.options());
-  pdta_gemm_bs(dout.data_ptr(), v, dp.data_ptr(), S, S, hd, d, ld, S,
-               b * h, h, S * d, hd, S * ld, hd, (long)h * S * S, S * S, 0,
-               0, 0, dt, cur_stream());
   auto ds = at::empty({b * h, S, S}, qkv.options());
-  pdta_softmax_bwd(p.data_ptr(), dp.data_ptr(), ds.data_ptr(), b * h * S,
-                   S, scale, dt, cur_stream());
+  if (pdta_attn_fused_ok(S, hd)) {
+    // dP never touches HBM (fused dO·V^T + softmax backward)
+    pdta_attn_fused_bwd_ds(qkv.data_ptr(), p.data_ptr(), dout.data_ptr(),
+                           ds.data_ptr(), b, h, (float)scale,
+                           dt, cur_stream());
+  } else {
+    auto dp = at::empty({b * h, S, S}, qkv.options());
+    pdta_gemm_bs(dout.data_ptr(), v, dp.data_ptr(), S, S, hd, d, ld, S,
+                 b * h, h, S * d, hd, S * ld, hd, (long)h * S * S, S * S,
+                 0, 0, 0, dt, cur_stream());
+    pdta_softmax_bwd(p.data_ptr(), dp.data_ptr(), ds.data_ptr(), b * h * S,
+                     S, scale, dt, cur_stream());
+  }
   auto dqkv = at::empty({b, S, ld}, qkv.options());
   char* dq = (char*)dqkv.data_ptr();
   char* dk = dq + d * 2;
