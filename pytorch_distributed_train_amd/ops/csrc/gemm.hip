@@ -285,17 +285,20 @@ __global__ void colsum8_part_k(const T* __restrict__ dy,
 
 __global__ void colsum_final_k(const float* __restrict__ part,
                                float* __restrict__ db, int gy, int N) {
-  __shared__ float red[256];
+  // 16 slices x 64 cols (1024 threads): N=768 is only 12 column-blocks
+  // chip-wide, so depth-parallelism inside the block is what keeps this
+  // off the latency floor
+  __shared__ float red[1024];
   int c = blockIdx.x * 64 + (int)(threadIdx.x % 64);
   int slice = (int)(threadIdx.x / 64);
   float s = 0.f;
   if (c < N)
-    for (int k = slice; k < gy; k += 4) s += part[(long)k * N + c];
+    for (int k = slice; k < gy; k += 16) s += part[(long)k * N + c];
   red[threadIdx.x] = s;
   __syncthreads();
   if (slice == 0 && c < N) {
 #pragma unroll
-    for (int i = 1; i < 4; ++i) s += red[threadIdx.x + i * 64];
+    for (int i = 1; i < 16; ++i) s += red[threadIdx.x + i * 64];
     db[c] = s;
   }
 }
@@ -393,7 +396,8 @@ void pdta_colsum_f32(const void* dy, float* db, float* part, long M, int N,
     PDTA_DISPATCH_DTYPE16(dt, T, {
       colsum8_part_k<T><<<grid, block, 0, s>>>((const T*)dy, part, M, N);
     });
-    colsum_final_k<<<(N + 63) / 64, 256, 0, s>>>(part, db, (int)gy, N);
+    colsum_final_k<<<(N + 63) / 64, 1024, 0, s>>>(part, db, (int)gy,
+                                                  N);
   } else {
     dim3 grid((N + 255) / 256, (unsigned)min((long)512, M));
     PDTA_DISPATCH_DTYPE16(dt, T, {
