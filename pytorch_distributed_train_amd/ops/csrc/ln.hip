@@ -114,41 +114,18 @@ __global__ void ln_bwd_k(const T* __restrict__ x, const T* __restrict__ dy,
     }
   }
 
-  // software-pipelined over rows: row i+1's loads are issued before row
-  // i's two serial wave reductions, so the ~row of HBM latency hides
-  // under the DPP chains (unpipelined form measured ~7x the traffic
-  // floor at d=768)
-  Vec8<T> gv[NG], xv[NG];
-  long row = r0 + wv;
-  if (row < r1) {
-#pragma unroll
-    for (int g = 0; g < NG; ++g) {
-      int c = lane * 8 + g * 512;
-      if (c < d) {
-        gv[g] = load8(dy + row * d + c);
-        xv[g] = load8(x + row * d + c);
-      }
-    }
-  }
-  for (; row < r1; ) {
-    long nrow = row + 4;
-    Vec8<T> gn[NG], xn[NG];
-    if (nrow < r1) {
-#pragma unroll
-      for (int g = 0; g < NG; ++g) {
-        int c = lane * 8 + g * 512;
-        if (c < d) {
-          gn[g] = load8(dy + nrow * d + c);
-          xn[g] = load8(x + nrow * d + c);
-        }
-      }
-    }
+  for (long row = r0 + wv; row < r1; row += 4) {
+    const T* xr = x + row * d;
+    const T* gr = dy + row * d;
     float mu = mean[row], is = invstd[row];
+    Vec8<T> gv[NG], xv[NG];  // row cached in registers: one HBM read
     float s1 = 0.f, s2 = 0.f;  // sum(dy*w), sum(dy*w*xhat)
 #pragma unroll
     for (int g = 0; g < NG; ++g) {
       int c = lane * 8 + g * 512;
       if (c < d) {
+        gv[g] = load8(gr + c);
+        xv[g] = load8(xr + c);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float gw = to_f32(gv[g].v[j]) * wreg[g][j];
@@ -176,12 +153,6 @@ __global__ void ln_bwd_k(const T* __restrict__ x, const T* __restrict__ dy,
         }
         store8(dr + c, o);
       }
-    }
-    row = nrow;
-#pragma unroll
-    for (int g = 0; g < NG; ++g) {
-      gv[g] = gn[g];
-      xv[g] = xn[g];
     }
   }
   // combine the 4 waves through LDS (no atomics), one pass per chunk
