@@ -273,7 +273,20 @@ __global__ void colsum8_part_k(const T* __restrict__ dy,
   if (c8 >= N) return;
   long r0 = blockIdx.y, rs = gridDim.y;
   float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-  for (long r = r0; r < M; r += rs) {
+  // 4 independent loads in flight per thread (the rolled loop kept one
+  // load outstanding and ran ~5x off the traffic floor)
+  long r = r0;
+  for (; r + 3 * rs < M; r += 4 * rs) {
+    Vec8<T> v0 = load8(dy + r * N + c8);
+    Vec8<T> v1 = load8(dy + (r + rs) * N + c8);
+    Vec8<T> v2 = load8(dy + (r + 2 * rs) * N + c8);
+    Vec8<T> v3 = load8(dy + (r + 3 * rs) * N + c8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      acc[j] += to_f32(v0.v[j]) + to_f32(v1.v[j]) + to_f32(v2.v[j]) +
+                to_f32(v3.v[j]);
+  }
+  for (; r < M; r += rs) {
     Vec8<T> v = load8(dy + r * N + c8);
 #pragma unroll
     for (int j = 0; j < 8; ++j) acc[j] += to_f32(v.v[j]);
