@@ -280,8 +280,47 @@ __global__ void bn_bwd_elemt_k(const T* __restrict__ x,
   }
   const int C = c8 * 8;
   long rstride = (long)gridDim.y * rpar;
-  for (long r = (long)blockIdx.y * rpar + threadIdx.x / cpg; r < rows;
-       r += rstride) {
+  long r = (long)blockIdx.y * rpar + threadIdx.x / cpg;
+  // 2 rows per iteration: 4-6 loads in flight (latency hiding)
+  for (; r + rstride < rows; r += 2 * rstride) {
+    long r2 = r + rstride;
+    Vec8<T> ga = load8(dy + r * C + c0);
+    Vec8<T> gb = load8(dy + r2 * C + c0);
+    Vec8<T> va = load8(x + r * C + c0);
+    Vec8<T> vb = load8(x + r2 * C + c0);
+    Vec8<T> ya, yb, da, db;
+    if (MASK == 1) {
+      ya = load8(relu_y + r * C + c0);
+      yb = load8(relu_y + r2 * C + c0);
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float g0 = to_f32(ga.v[j]), g1 = to_f32(gb.v[j]);
+      float xm0 = to_f32(va.v[j]) - mu[j];
+      float xm1 = to_f32(vb.v[j]) - mu[j];
+      if (MASK == 1) {
+        g0 = to_f32(ya.v[j]) > 0.f ? g0 : 0.f;
+        g1 = to_f32(yb.v[j]) > 0.f ? g1 : 0.f;
+      }
+      if (MASK == 2) {
+        g0 = k1[j] * xm0 + b0[j] > 0.f ? g0 : 0.f;
+        g1 = k1[j] * xm1 + b0[j] > 0.f ? g1 : 0.f;
+      }
+      if (DRES) {
+        da.v[j] = from_f32<T>(g0);
+        db.v[j] = from_f32<T>(g1);
+      }
+      ga.v[j] = from_f32<T>(k1[j] * (g0 - k2[j] - xm0 * k3[j]));
+      gb.v[j] = from_f32<T>(k1[j] * (g1 - k2[j] - xm1 * k3[j]));
+    }
+    if (DRES) {
+      store8(dres + r * C + c0, da);
+      store8(dres + r2 * C + c0, db);
+    }
+    store8(dx + r * C + c0, ga);
+    store8(dx + r2 * C + c0, gb);
+  }
+  for (; r < rows; r += rstride) {
     Vec8<T> g = load8(dy + r * C + c0);
     Vec8<T> v = load8(x + r * C + c0);
     Vec8<T> yv, dr;
