@@ -124,8 +124,35 @@ __global__ void bn_apply_k(const T* __restrict__ x, T* __restrict__ y,
   }
   const int C = c8 * 8;
   long rstride = (long)gridDim.y * rpar;
-  for (long r = (long)blockIdx.y * rpar + threadIdx.x / cpg; r < rows;
-       r += rstride) {
+  long r = (long)blockIdx.y * rpar + threadIdx.x / cpg;
+  for (; r + rstride < rows; r += 2 * rstride) {  // 2-4 loads in flight
+    long r2 = r + rstride;
+    Vec8<T> va = load8(x + r * C + c0);
+    Vec8<T> vb = load8(x + r2 * C + c0);
+    Vec8<T> ra, rb;
+    if (RES) {
+      ra = load8(res + r * C + c0);
+      rb = load8(res + r2 * C + c0);
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float o0 = to_f32(va.v[j]) * scale[j] + shift[j];
+      float o1 = to_f32(vb.v[j]) * scale[j] + shift[j];
+      if (RES) {
+        o0 += to_f32(ra.v[j]);
+        o1 += to_f32(rb.v[j]);
+      }
+      if (RELU) {
+        o0 = fmaxf(o0, 0.f);
+        o1 = fmaxf(o1, 0.f);
+      }
+      va.v[j] = from_f32<T>(o0);
+      vb.v[j] = from_f32<T>(o1);
+    }
+    store8(y + r * C + c0, va);
+    store8(y + r2 * C + c0, vb);
+  }
+  for (; r < rows; r += rstride) {
     Vec8<T> v = load8(x + r * C + c0);
     Vec8<T> rv;
     if (RES) rv = load8(res + r * C + c0);
