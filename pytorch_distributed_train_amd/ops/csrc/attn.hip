@@ -177,17 +177,19 @@ __global__ __launch_bounds__(256) void attn_fused_fwd_k(
   }
 }
 
-// Fused attention backward, stage 1: dP = dO·V^T and
-// dS = scale · P ∘ (dP − rowsum(P ∘ dP)) in one kernel — the [S,S] dP
-// tensor never touches HBM (the decomposed path wrote and re-read it).
-// dS is materialized (bf16/fp16) because both dQ = dS·K and
-// dK = dS^T·Q consume it.
+// Fused attention backward, stage 1: dP = dO·V^T,
+// dS = scale · P ∘ (dP − rowsum(P ∘ dP)), and dQ = dS·K in one kernel —
+// the [S,S] dP never touches HBM, dS goes straight from registers into
+// the dQ MFMA pass (K reuses the V LDS slab), and dS is materialized
+// once (bf16/fp16) for dK = dS^T·Q only.
 template <typename T>
 __global__ __launch_bounds__(256) void attn_fused_bwd_ds_k(
     const T* __restrict__ qkv, const T* __restrict__ p,
-    const T* __restrict__ dout, T* __restrict__ ds, int h, float scale) {
+    const T* __restrict__ dout, T* __restrict__ ds, T* __restrict__ dqkv,
+    int h, float scale) {
   __shared__ __attribute__((aligned(16))) char dotile[64 * 128];
   __shared__ __attribute__((aligned(16))) char vimg[SP * 128];
+  __shared__ __attribute__((aligned(16))) char dsimg[64 * 512];
 
   int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
   int bh = blockIdx.y;
@@ -195,7 +197,9 @@ __global__ __launch_bounds__(256) void attn_fused_bwd_ds_k(
   long d = (long)h * HD, ld3 = 3 * d;
   const T* vbase =
       qkv + (long)batch * SP * ld3 + (long)head * HD + 2 * d;
+  const T* kbase = vbase - d;  // k plane of the interleaved qkv
   const T* dobase = dout + (long)batch * SP * d + (long)head * HD;
+  T* dqbase = dqkv + (long)batch * SP * ld3 + (long)head * HD;
   int q0 = blockIdx.x * 64;
 
   int row_in = lane >> 3, c16 = lane & 7;
@@ -255,14 +259,59 @@ __global__ __launch_bounds__(256) void attn_fused_bwd_ds_k(
 #pragma unroll
     for (int m = 1; m < 16; m <<= 1) dot[r] += __shfl_xor(dot[r], m);
   }
+  __syncthreads();  // all waves done with V fragments: the slab becomes K
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {  // stage K; hides under the dS writes
+    int krow = (wave * 8 + i) * 8 + row_in;
+    const char* g =
+        reinterpret_cast<const char*>(kbase + (long)krow * ld3) +
+        (c16 ^ (krow & 7)) * 16;
+    glds16(g, vimg + (wave * 8 + i) * 1024);
+  }
+  // dS -> global (consumed by dK = dS^T·Q) and -> LDS A-image for dQ
 #pragma unroll
   for (int jn = 0; jn < 16; ++jn) {
     int col = jn * 16 + colbase;
+    int byte = col * 2;
+    int chunk = byte >> 4, rem = byte & 15;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int row = wave * 16 + (lane >> 4) * 4 + r;
-      dg[(long)row * SP + col] =
-          from_f32<T>(scale * pv[jn][r] * (acc[jn][r] - dot[r]));
+      T dsv = from_f32<T>(scale * pv[jn][r] * (acc[jn][r] - dot[r]));
+      *reinterpret_cast<T*>(dsimg + row * 512 +
+                            (((chunk ^ (row & 7)) << 4) | rem)) = dsv;
+      dg[(long)row * SP + col] = dsv;
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // K staged
+  __syncthreads();                                  // dS image visible
+
+  // dQ = dS·K (K-dim = 256 keys; K^T fragments via ds_read_b64_tr_b16)
+  f32x4 acc2[4];
+#pragma unroll
+  for (int jn = 0; jn < 4; ++jn) acc2[jn] = {0.f, 0.f, 0.f, 0.f};
+  unsigned kb = lds_addr_of(vimg);
+#pragma unroll
+  for (int kt = 0; kt < 8; ++kt) {
+    s16x8 af = frag_read_s<512>(dsimg, wave * 16 + (lane & 15),
+                                kt * 4 + (lane >> 4));
+    unsigned long long lo[4], hi[4];
+#pragma unroll
+    for (int jn = 0; jn < 4; ++jn)
+      frag_read_tr_issue(kb, kt * 32, jn * 16, lane, &lo[jn], &hi[jn]);
+    tr_read_wait();
+#pragma unroll
+    for (int jn = 0; jn < 4; ++jn)
+      acc2[jn] = MfmaOp<T>::run(af, tr_pair_to_frag(lo[jn], hi[jn]),
+                                acc2[jn]);
+  }
+#pragma unroll
+  for (int jn = 0; jn < 4; ++jn) {
+    int col = jn * 16 + colbase;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = q0 + wave * 16 + (lane >> 4) * 4 + r;
+      dqbase[(long)row * ld3 + col] = from_f32<T>(acc2[jn][r]);
     }
   }
 }
@@ -285,12 +334,13 @@ void pdta_attn_fused_fwd(const void* qkv, void* probs, void* out, int b,
 }
 
 void pdta_attn_fused_bwd_ds(const void* qkv, const void* p,
-                            const void* dout, void* ds, int b, int h,
-                            float scale, int dt, hipStream_t s) {
+                            const void* dout, void* ds, void* dqkv, int b,
+                            int h, float scale, int dt, hipStream_t s) {
   dim3 grid(SP / 64, b * h);
   PDTA_DISPATCH_DTYPE16(dt, T, {
     attn_fused_bwd_ds_k<T><<<grid, 256, 0, s>>>(
-        (const T*)qkv, (const T*)p, (const T*)dout, (T*)ds, h, scale);
+        (const T*)qkv, (const T*)p, (const T*)dout, (T*)ds, (T*)dqkv, h,
+        scale);
   });
 }
 

@@ -85,7 +85,7 @@ int pdta_attn_fused_ok(int, int);
 void pdta_attn_fused_fwd(const void*, void*, void*, int, int, int, float,
                          int, hipStream_t);
 void pdta_attn_fused_bwd_ds(const void*, const void*, const void*, void*,
-                            int, int, float, int, hipStream_t);
+                            void*, int, int, float, int, hipStream_t);
 }
 
 namespace {
@@ -678,11 +678,16 @@ at::Tensor attn_bwd(at::Tensor qkv, at::Tensor p, at::Tensor dout,
   char* k = q + d * 2;
   char* v = q + 2 * d * 2;
   auto ds = at::empty({b * h, S, S}, qkv.options());
+  auto dqkv = at::empty({b, S, ld}, qkv.options());
+  char* dq = (char*)dqkv.data_ptr();
+  char* dk = dq + d * 2;
+  char* dv = dq + 2 * d * 2;
   if (pdta_attn_fused_ok(S, hd)) {
-    // dP never touches HBM (fused dO·V^T + softmax backward)
+    // one kernel: dP never touches HBM, dQ comes straight from the
+    // in-register dS (K reuses the V LDS slab)
     pdta_attn_fused_bwd_ds(qkv.data_ptr(), p.data_ptr(), dout.data_ptr(),
-                           ds.data_ptr(), b, h, (float)scale,
-                           dt, cur_stream());
+                           ds.data_ptr(), dqkv.data_ptr(), b, h,
+                           (float)scale, dt, cur_stream());
   } else {
     auto dp = at::empty({b * h, S, S}, qkv.options());
     pdta_gemm_bs(dout.data_ptr(), v, dp.data_ptr(), S, S, hd, d, ld, S,
@@ -690,15 +695,11 @@ at::Tensor attn_bwd(at::Tensor qkv, at::Tensor p, at::Tensor dout,
                  0, 0, 0, dt, cur_stream());
     pdta_softmax_bwd(p.data_ptr(), dp.data_ptr(), ds.data_ptr(), b * h * S,
                      S, scale, dt, cur_stream());
+    // dq = ds @ K
+    pdta_gemm_bs(ds.data_ptr(), k, dq, S, hd, S, S, ld, ld, b * h, h,
+                 (long)h * S * S, S * S, S * ld, hd, S * ld, hd, 0, 1, 0,
+                 dt, cur_stream());
   }
-  auto dqkv = at::empty({b, S, ld}, qkv.options());
-  char* dq = (char*)dqkv.data_ptr();
-  char* dk = dq + d * 2;
-  char* dv = dq + 2 * d * 2;
-  // dq = ds @ K
-  pdta_gemm_bs(ds.data_ptr(), k, dq, S, hd, S, S, ld, ld, b * h, h,
-               (long)h * S * S, S * S, S * ld, hd, S * ld, hd, 0, 1, 0, dt,
-               cur_stream());
   // dk = ds^T @ Q
   pdta_gemm_bs(ds.data_ptr(), q, dk, S, hd, S, S, ld, ld, b * h, h,
                (long)h * S * S, S * S, S * ld, hd, S * ld, hd, 1, 1, 0, dt,
