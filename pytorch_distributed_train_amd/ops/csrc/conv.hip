@@ -278,16 +278,9 @@ __global__ __launch_bounds__(512) void igemm_fwd_big_k(
   stage_all(0, 0);
   for (int kt = 0; kt < ktiles; ++kt) {
     int slot = kt & 1;
-    // issue tile kt+1's loads BEFORE stalling on tile kt's: the counted
-    // wait (8 glds per wave per stage) lets the prefetch fly during the
-    // stall instead of after it
-    if (kt + 1 < ktiles) {
-      stage_all(kt + 1, slot ^ 1);
-      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-    } else {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
+    if (kt + 1 < ktiles) stage_all(kt + 1, slot ^ 1);
     const char* la = lds[slot];
     const char* lb = lds[slot] + BOFF;
     s16x8 bf[2][4];
@@ -595,16 +588,9 @@ __global__ __launch_bounds__(512) void igemm_bwd_data_big_k(
   stage_all(0, 0);
   for (int kt = 0; kt < ktiles; ++kt) {
     int slot = kt & 1;
-    // issue tile kt+1's loads BEFORE stalling on tile kt's: the counted
-    // wait (8 glds per wave per stage) lets the prefetch fly during the
-    // stall instead of after it
-    if (kt + 1 < ktiles) {
-      stage_all(kt + 1, slot ^ 1);
-      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-    } else {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
+    if (kt + 1 < ktiles) stage_all(kt + 1, slot ^ 1);
     const char* la = lds[slot];
     const char* lb = lds[slot] + BOFF;
     s16x8 bf[2][4];
@@ -976,15 +962,9 @@ __global__ __launch_bounds__(512) void igemm_wrw_big_k(
   unsigned dy_base[2] = {lds_addr_of(lds[0]), lds_addr_of(lds[1])};
   for (long rb = r0; rb < r1; rb += 64) {
     int slot = (int)((rb - r0) >> 6) & 1;
-    // stage-first + counted wait (8 glds/wave/stage): prefetch flies
-    // during the stall on this slab's loads
-    if (rb + 64 < r1) {
-      stage_all(rb + 64, slot ^ 1);
-      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-    } else {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
+    if (rb + 64 < r1) stage_all(rb + 64, slot ^ 1);
     unsigned la = dy_base[slot];
     unsigned lb = dy_base[slot] + BOFF;
 #pragma unroll

@@ -194,16 +194,12 @@ __global__ __launch_bounds__(512) void gemm_nt_big_k(
 
   for (int kt = 0; kt < ktiles; ++kt) {
     int slot = kt & 1;
-    // issue tile kt+1's loads BEFORE stalling on tile kt's: the counted
-    // wait (8 glds per wave per stage) lets the prefetch fly during the
-    // stall instead of after it
-    if (kt + 1 < ktiles) {
-      stage_all(kt + 1, slot ^ 1);
-      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-    } else {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    }
+    // drain this tile's loads (they had the previous tile's ~2200-cycle
+    // MFMA stream to fly) and rendezvous
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
+    // issue the NEXT tile's staging first so it spans this tile's compute
+    if (kt + 1 < ktiles) stage_all(kt + 1, slot ^ 1);
     const char* la = lds[slot];
     const char* lb = lds[slot] + BOFF;
     s16x8 bf[2][4];
