@@ -12,7 +12,8 @@ import os
 from dataclasses import dataclass, field
 
 
-MODEL_CHOICES = ("resnet18", "resnet50", "resnet152", "vit_b_16")
+MODEL_CHOICES = ("resnet18", "resnet34", "resnet50", "resnet101",
+                 "resnet152", "vit_b_16")
 
 
 def build_parser() -> argparse.ArgumentParser:

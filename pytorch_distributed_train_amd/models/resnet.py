@@ -1,4 +1,4 @@
-"""ResNet-18/50/152 from scratch (torchvision is not installed here).
+"""ResNet-18/34/50/101/152 from scratch (torchvision is not installed here).
 
 Module names follow the torchvision resnet naming (conv1, bn1, layer1..4,
 <block>.convN/bnN/downsample.0/.1, fc) so state_dicts keep the reference
@@ -139,8 +139,16 @@ def resnet18(num_classes=1000, compute_dtype=None):
     return ResNet(BasicBlock, [2, 2, 2, 2], num_classes, compute_dtype)
 
 
+def resnet34(num_classes=1000, compute_dtype=None):
+    return ResNet(BasicBlock, [3, 4, 6, 3], num_classes, compute_dtype)
+
+
 def resnet50(num_classes=1000, compute_dtype=None):
     return ResNet(Bottleneck, [3, 4, 6, 3], num_classes, compute_dtype)
+
+
+def resnet101(num_classes=1000, compute_dtype=None):
+    return ResNet(Bottleneck, [3, 4, 23, 3], num_classes, compute_dtype)
 
 
 def resnet152(num_classes=1000, compute_dtype=None):

@@ -99,3 +99,18 @@ def test_label_smoothing_matches_torch():
                                            label_smoothing=0.1)
     g_torch = torch.autograd.grad(l2, logits2)[0]
     assert torch.allclose(g_ours, g_torch, atol=1e-6)
+
+
+def test_resnet34_101_structural():
+    """New depth variants: torchvision parameter counts and a fwd/bwd step."""
+    import torch
+    from pytorch_distributed_train_amd.models import build_model
+    from pytorch_distributed_train_amd.ops import functional as OF
+
+    for name, nparams in (("resnet34", 21797672), ("resnet101", 44549160)):
+        m = build_model(name, 1000, torch.float32)
+        assert sum(p.numel() for p in m.parameters()) == nparams, name
+        x = torch.randn(2, 3, 64, 64)
+        y = torch.randint(0, 1000, (2,))
+        OF.softmax_cross_entropy(m(x), y).backward()
+        assert all(p.grad is not None for p in m.parameters())
