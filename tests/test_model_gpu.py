@@ -197,3 +197,24 @@ def test_hip_graph_step_matches_eager():
     diff = (ta.flat.master - tb.flat.master).abs().max()
     assert torch.allclose(ta.flat.master, tb.flat.master,
                           atol=1e-6, rtol=1e-5), diff
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("name", ["resnet34", "resnet101"])
+def test_depth_variants_step(name):
+    """New depth variants take a full fwd+bwd+step on the HIP path."""
+    import torch
+    from pytorch_distributed_train_amd.flags import build_parser
+    from pytorch_distributed_train_amd.engine import Trainer
+
+    args = build_parser().parse_args([
+        "--model", name, "--data", "synthetic-imagenet",
+        "--dtype", "bf16", "--batch-size", "4", "--seed", "2"])
+    t = Trainer(args)
+    t.model.train()
+    x = torch.randn(4, 3, 224, 224).cuda()
+    y = torch.randint(0, 1000, (4,)).cuda()
+    l0 = float(t.train_step(x, y))
+    for _ in range(3):
+        loss = t.train_step(x, y)
+    assert float(loss) < l0  # memorizing one batch
