@@ -1,0 +1,54 @@
+"""End-to-end train.py main() on CPU: epoch loop, lr schedule, eval,
+checkpoint write, resume (single process — the gloo ws=2 configs are in
+test_ddp_gloo.py)."""
+import json
+import os
+import sys
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import train as train_mod  # repo-root entrypoint
+
+
+def _argv(tmp_path, extra=()):
+    return ["--model", "resnet18", "--data", "synthetic-cifar10",
+            "--dtype", "fp32", "--device", "cpu", "--batch-size", "4",
+            "--num-samples", "16", "--epochs", "2", "--lr", "0.01",
+            "--eval-freq", "1", "--eval-batches", "2",
+            "--checkpoint-dir", str(tmp_path), "--checkpoint-freq", "1",
+            "--seed", "7", *extra]
+
+
+def test_main_trains_evals_checkpoints(tmp_path, monkeypatch):
+    monkeypatch.delenv("RANK", raising=False)
+    monkeypatch.delenv("WORLD_SIZE", raising=False)
+    assert train_mod.main(_argv(tmp_path)) == 0
+    ckpt = tmp_path / "resnet18_checkpoint.pt"
+    assert ckpt.exists()
+    state = torch.load(str(ckpt), weights_only=False)
+    assert state["epoch"] == 2
+
+    # resume continues from the saved epoch without error
+    assert train_mod.main(_argv(tmp_path,
+                                ("--resume", str(ckpt),
+                                 "--epochs", "3"))) == 0
+    state = torch.load(str(ckpt), weights_only=False)
+    assert state["epoch"] == 3
+
+
+def test_lr_schedules():
+    from pytorch_distributed_train_amd.flags import build_parser
+    a = build_parser().parse_args([
+        "--lr", "1.0", "--epochs", "10", "--warmup-epochs", "2",
+        "--lr-schedule", "step"])
+    assert train_mod.lr_at(a, 0) == 0.5        # linear warmup 1/2
+    assert train_mod.lr_at(a, 1) == 1.0        # warmup done
+    assert train_mod.lr_at(a, 2) == 1.0        # before 30%
+    assert abs(train_mod.lr_at(a, 3) - 0.1) < 1e-9   # past 30%
+    assert abs(train_mod.lr_at(a, 6) - 0.01) < 1e-9  # past 60%
+    assert abs(train_mod.lr_at(a, 8) - 0.001) < 1e-9  # past 80%
+    a = build_parser().parse_args([
+        "--lr", "1.0", "--epochs", "10", "--lr-schedule", "cosine"])
+    assert abs(train_mod.lr_at(a, 0) - 1.0) < 1e-9
+    assert train_mod.lr_at(a, 9) < 0.05
