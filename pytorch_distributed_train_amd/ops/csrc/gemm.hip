@@ -15,6 +15,8 @@
 //   - LDS XOR swizzle col16 ^= (row&7) applied on the *source* address
 //     (glds writes lane-linear — guide §5.4 rule 21) and on fragment reads
 //   - XCD-aware bijective blockIdx swizzle (T1) for L2 locality
+#include <cstdlib>
+
 #include "common.h"
 #include "mfma_tile.h"
 
@@ -262,6 +264,106 @@ __global__ void colsum_f32_k(const T* __restrict__ dy, float* __restrict__ db,
   atomicAdd(&db[c], acc);
 }
 
+// ---- v2 experimental: 256x128 tile, BK=32, 8 waves (4x2) ---------------
+// Occupancy experiment (docs/DESIGN.md §6): acc 64 VGPR/thread and
+// 2x24 KiB ring slots -> 3 blocks/CU by LDS, ~4 waves/SIMD by VGPR
+// (the 256x256/BK=64 kernel is pinned at 2 waves/SIMD and 52% WAIT).
+// Rows are 64 B (32 K-elems); swizzle XOR uses the row's 4 chunks.
+DEV_INLINE s16x8 frag_read64(const char* lds, int row, int chunk) {
+  return *reinterpret_cast<const s16x8*>(
+      lds + row * 64 + ((chunk ^ (row & 3)) * 16));
+}
+
+template <typename T, typename OutT, bool BIAS>
+__global__ __launch_bounds__(512) void gemm_nt_v2_k(
+    const T* __restrict__ A, const T* __restrict__ B, OutT* __restrict__ C,
+    const float* __restrict__ bias, int M, int N, int K, int tiles_n,
+    long batch_sa, long batch_sb, long batch_sc) {
+  constexpr int BOFF = 256 * 64;  // A slot bytes
+  __shared__ __attribute__((aligned(16))) char lds[2][BOFF + 128 * 64];
+
+  A += (long)blockIdx.y * batch_sa;
+  B += (long)blockIdx.y * batch_sb;
+  C += (long)blockIdx.y * batch_sc;
+
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  int tm = bid / tiles_n, tn = bid % tiles_n;
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  int wm = wave >> 1, wn = wave & 1;  // 4 x 2 wave grid of 64x64
+  int m0 = tm * 256, n0 = tn * 128;
+  int ktiles = K / 32;
+
+  int row_in = lane >> 2, c4 = lane & 3;
+
+  auto stage_all = [&](int kt, int slot) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {  // A: 16 KiB = 16 glds, 2 per wave
+      int row = (wave * 2 + i) * 16 + row_in;
+      int grow = m0 + row;
+      if (grow > M - 1) grow = M - 1;
+      const char* g = reinterpret_cast<const char*>(A) +
+                      ((long)grow * K + kt * 32) * 2 +
+                      ((c4 ^ (row_in & 3)) * 16);
+      glds16(g, (char*)lds[slot] + (wave * 2 + i) * 1024);
+    }
+    {  // B: 8 KiB = 8 glds, 1 per wave
+      int row = wave * 16 + row_in;
+      int grow = n0 + row;
+      if (grow > N - 1) grow = N - 1;
+      const char* g = reinterpret_cast<const char*>(B) +
+                      ((long)grow * K + kt * 32) * 2 +
+                      ((c4 ^ (row_in & 3)) * 16);
+      glds16(g, (char*)lds[slot] + BOFF + wave * 1024);
+    }
+  };
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  stage_all(0, 0);
+  for (int kt = 0; kt < ktiles; ++kt) {
+    int slot = kt & 1;
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (kt + 1 < ktiles) stage_all(kt + 1, slot ^ 1);
+    const char* la = lds[slot];
+    const char* lb = lds[slot] + BOFF;
+    s16x8 af[4], bf[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+      af[i] = frag_read64(la, wm * 64 + i * 16 + (lane & 15), lane >> 4);
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      bf[j] = frag_read64(lb, wn * 64 + j * 16 + (lane & 15), lane >> 4);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        acc[i][j] = MfmaOp<T>::run(af[i], bf[j], acc[i][j]);
+    __builtin_amdgcn_s_barrier();
+  }
+
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    long rbase = m0 + wm * 64 + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = n0 + wn * 64 + j * 16 + (lane & 15);
+      if (col >= N) continue;
+      float badd = BIAS ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = rbase + r;
+        if (row < M)
+          C[(long)row * N + col] = from_f32<OutT>(acc[i][j][r] + badd);
+      }
+    }
+  }
+}
+
 // vectorized two-stage variant (N % 8 == 0): 16 B loads, 8 columns per
 // thread, per-row-slab partials in plain stores + a finalize kernel (an
 // atomicAdd version serialized ~700 RMWs per column at the L2 and ran 2x
@@ -323,6 +425,24 @@ extern "C" {
 // C[M,N] = A[M,K] B[N,K]^T (+bias); K % 64 == 0. out_f32: C is fp32.
 void pdta_gemm_nt(const void* A, const void* B, void* C, const float* bias,
                   int M, int N, int K, int dt, int out_f32, hipStream_t s) {
+  // occupancy experiment (docs/DESIGN.md §6), opt-in via PDTA_GEMM_V2=1
+  static int use_v2 = -1;
+  if (use_v2 < 0) use_v2 = getenv("PDTA_GEMM_V2") != nullptr;
+  if (use_v2 && M >= 1024 && N >= 128 && K % 32 == 0 && !out_f32) {
+    int tiles_m = (M + 255) / 256, tiles_n = (N + 127) / 128;
+    dim3 grid(tiles_m * tiles_n);
+    PDTA_DISPATCH_DTYPE16(dt, T, {
+      if (bias)
+        gemm_nt_v2_k<T, T, true><<<grid, 512, 0, s>>>(
+            (const T*)A, (const T*)B, (T*)C, bias, M, N, K, tiles_n, 0, 0,
+            0);
+      else
+        gemm_nt_v2_k<T, T, false><<<grid, 512, 0, s>>>(
+            (const T*)A, (const T*)B, (T*)C, nullptr, M, N, K, tiles_n, 0,
+            0, 0);
+    });
+    return;
+  }
   if (M >= 1024 && N >= 256 && K % 64 == 0) {
     int tiles_m = (M + 255) / 256, tiles_n = (N + 255) / 256;
     dim3 grid(tiles_m * tiles_n);
