@@ -47,8 +47,33 @@ def parse(argv=None):
     return p.parse_args(argv)
 
 
+def _free_port() -> int:
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def _self_launch(argv) -> int:
+    """`python bench.py --gpus N` with no rendezvous env: spawn N local
+    ranks of this script via the gang launcher (driver contract — the
+    bare invocation must produce an n_gpus=N record)."""
+    from pytorch_distributed_train_amd.launch import main as launch_main
+    return launch_main([
+        "--nproc_per_node", str(parse(argv).gpus),
+        "--use-env",
+        "--master_addr", "127.0.0.1",
+        "--master_port", str(_free_port()),
+        "--max-restarts", "0",
+        os.path.abspath(__file__),
+        *(argv if argv is not None else sys.argv[1:]),
+    ])
+
+
 def main(argv=None) -> int:
     b = parse(argv)
+    if b.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        return _self_launch(argv)
     env = DistEnv.from_env(b.local_rank)
     if env.world_size > 1 and env.world_size != b.gpus and env.rank == 0:
         print(f"warning: WORLD_SIZE={env.world_size} != --gpus {b.gpus}; "

@@ -38,8 +38,11 @@ class Trainer:
         if dev == "auto":
             dev = "cuda" if torch.cuda.is_available() else "cpu"
         if dev == "cuda":
-            torch.cuda.set_device(self.env.local_rank)
-            self.device = torch.device("cuda", self.env.local_rank)
+            # modulo: more ranks than GPUs is legal (2-ranks-on-1-GPU RCCL
+            # validation on the single-GPU box; 8-GPU node maps 1:1)
+            idx = self.env.local_rank % max(torch.cuda.device_count(), 1)
+            torch.cuda.set_device(idx)
+            self.device = torch.device("cuda", idx)
         else:
             self.device = torch.device("cpu")
         self.compute_dtype = _DTYPES[args.dtype]
