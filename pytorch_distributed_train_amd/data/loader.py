@@ -81,7 +81,16 @@ class Loader:
                 data = self._collate(idxs)
                 with lock:
                     results[bi] = data
-                out_q.put(bi)
+                # interruptible put: a consumer that abandons the iterator
+                # early (steps-per-epoch break, exception) sets stop; an
+                # unconditional blocking put would leak this thread and its
+                # pinned batch forever
+                while not stop.is_set():
+                    try:
+                        out_q.put(bi, timeout=0.1)
+                        break
+                    except queue.Full:
+                        continue
 
         threads = [threading.Thread(target=worker, daemon=True)
                    for _ in range(self.workers)]
@@ -100,3 +109,13 @@ class Loader:
                 next_bi += 1
         finally:
             stop.set()
+            # unblock any worker already parked in a pre-stop put()
+            try:
+                while True:
+                    out_q.get_nowait()
+            except queue.Empty:
+                pass
+            for t in threads:
+                t.join(timeout=5.0)
+            with lock:
+                results.clear()

@@ -123,9 +123,12 @@ class Trainer:
             self.scaler.scale(loss).backward()
         with record_function("pdta.reduce_finalize"):
             self.reducer.finalize()
-            if getattr(self.args, 'nan_check', False):
-                self._check_grads_finite()
         skip = self.scaler.unscale_and_check(self.flat.grad, self.comm)
+        # nan-check AFTER unscale and only on non-skipped steps: under fp16
+        # AMP a loss-scale overflow is a recoverable event the scaler
+        # handles by skipping + backoff, not a reason to kill the gang
+        if not skip and getattr(self.args, 'nan_check', False):
+            self._check_grads_finite()
         if not skip:
             with record_function("pdta.optimizer"):
                 self.optimizer.step()

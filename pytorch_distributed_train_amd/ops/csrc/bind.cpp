@@ -36,7 +36,7 @@ void pdta_maxpool_bwd(const void*, const uint8_t*, void*, int, int, int, int,
 void pdta_gap_fwd(const void*, void*, int, long, int, int, hipStream_t);
 void pdta_gap_bwd(const void*, void*, int, long, int, int, hipStream_t);
 void pdta_fused_sgd(float*, const float*, float*, long, float, float, float,
-                    float, int, float, hipStream_t);
+                    float, int, float, const float*, hipStream_t);
 void pdta_unscale_check(float*, long, float, float*, hipStream_t);
 void pdta_ce_fwd(const void*, const long*, float*, float*, int, int, float,
                  int, hipStream_t);
@@ -313,12 +313,15 @@ at::Tensor avgpool_global_bwd(at::Tensor dy, int64_t H, int64_t W) {
 
 // ---- optimizer / amp -----------------------------------------------------
 void fused_sgd(at::Tensor p, at::Tensor g, at::Tensor m, double lr,
-               double mu, double wd, double damp, bool nesterov, double gs) {
+               double mu, double wd, double damp, bool nesterov, double gs,
+               c10::optional<at::Tensor> lr_dev) {
   CHECK_IN(p);
   CHECK_IN(g);
   pdta_fused_sgd(p.data_ptr<float>(), g.data_ptr<float>(),
                  m.numel() ? m.data_ptr<float>() : nullptr, p.numel(), lr,
-                 mu, wd, damp, nesterov ? 1 : 0, gs, cur_stream());
+                 mu, wd, damp, nesterov ? 1 : 0, gs,
+                 lr_dev.has_value() ? lr_dev->data_ptr<float>() : nullptr,
+                 cur_stream());
 }
 
 at::Tensor unscale_check(at::Tensor g, double inv) {
@@ -816,7 +819,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool_bwd", &maxpool_bwd);
   m.def("avgpool_global_fwd", &avgpool_global_fwd);
   m.def("avgpool_global_bwd", &avgpool_global_bwd);
-  m.def("fused_sgd", &fused_sgd);
+  m.def("fused_sgd", &fused_sgd, py::arg("p"), py::arg("g"),
+        py::arg("m"), py::arg("lr"), py::arg("mu"), py::arg("wd"),
+        py::arg("damp"), py::arg("nesterov"), py::arg("gs"),
+        py::arg("lr_dev") = py::none());
   m.def("unscale_check", &unscale_check);
   m.def("softmax_ce_fwd", &softmax_ce_fwd);
   m.def("softmax_ce_bwd", &softmax_ce_bwd);

@@ -14,7 +14,11 @@ constexpr int TPB = 256;
 template <bool MOMENTUM, bool NESTEROV>
 __global__ void fused_sgd_k(float* __restrict__ p, const float* __restrict__ g,
                             float* __restrict__ m, long n4, float lr,
-                            float mu, float wd, float damp, float gs) {
+                            float mu, float wd, float damp, float gs,
+                            const float* __restrict__ lr_dev) {
+  // lr from a device scalar when given: a hipGraph-captured launch then
+  // honors later schedule updates (host lr would be baked into the graph)
+  if (lr_dev) lr = *lr_dev;
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
   for (; i < n4; i += stride) {
@@ -66,19 +70,19 @@ extern "C" {
 // n must be a multiple of 4 (the flat buffer is 64-element aligned).
 void pdta_fused_sgd(float* p, const float* g, float* m, long n, float lr,
                     float mu, float wd, float damp, int nesterov, float gs,
-                    hipStream_t s) {
+                    const float* lr_dev, hipStream_t s) {
   long n4 = n / 4;
   int grid = ew_grid(n4, TPB, 1);
   if (mu != 0.f) {
     if (nesterov)
       fused_sgd_k<true, true><<<grid, TPB, 0, s>>>(p, g, m, n4, lr, mu, wd,
-                                                   damp, gs);
+                                                   damp, gs, lr_dev);
     else
       fused_sgd_k<true, false><<<grid, TPB, 0, s>>>(p, g, m, n4, lr, mu, wd,
-                                                    damp, gs);
+                                                    damp, gs, lr_dev);
   } else {
     fused_sgd_k<false, false><<<grid, TPB, 0, s>>>(p, g, nullptr, n4, lr, mu,
-                                                   wd, damp, gs);
+                                                   wd, damp, gs, lr_dev);
   }
 }
 

@@ -48,7 +48,8 @@ class _Work_then:
 
 
 class _Bucket:
-    __slots__ = ("index", "start", "end", "param_ids", "pending", "work")
+    __slots__ = ("index", "start", "end", "param_ids", "pending", "ready",
+                 "work")
 
     def __init__(self, index, start, end, param_ids):
         self.index = index
@@ -56,6 +57,7 @@ class _Bucket:
         self.end = end
         self.param_ids = param_ids
         self.pending = 0
+        self.ready = False
         self.work = None
 
 
@@ -101,6 +103,7 @@ class GradReducer:
                 self._hooks.append(p.register_post_accumulate_grad_hook(
                     self._make_hook(i)))
         self._iteration_open = False
+        self._next_launch = 0
 
     # ------------------------------------------------------------------
     def prepare(self) -> None:
@@ -109,7 +112,9 @@ class GradReducer:
             return
         for b in self.buckets:
             b.pending = len(b.param_ids)
+            b.ready = False
             b.work = None
+        self._next_launch = 0
         self._iteration_open = True
 
     def _make_hook(self, param_id: int):
@@ -119,8 +124,19 @@ class GradReducer:
             b = self._param_bucket[param_id]
             b.pending -= 1
             if b.pending == 0 and self.overlap:
-                self._launch(b)
+                b.ready = True
+                self._launch_ready()
         return hook
+
+    def _launch_ready(self) -> None:
+        """Launch ready buckets strictly in ascending bucket index: RCCL
+        requires every rank to enqueue collectives in the same order, so an
+        early-ready out-of-order bucket is deferred until its predecessors
+        have launched (torch Reducer next_bucket_ semantics)."""
+        while (self._next_launch < len(self.buckets)
+               and self.buckets[self._next_launch].ready):
+            self._launch(self.buckets[self._next_launch])
+            self._next_launch += 1
 
     def _launch(self, b: _Bucket) -> None:
         view = self.flat.grad[b.start:b.end]

@@ -26,6 +26,11 @@ class FusedSGD:
         if nesterov and (momentum <= 0 or dampening != 0):
             raise ValueError("nesterov needs momentum > 0, dampening = 0")
         self.flat = flat
+        # on GPU lr lives in a device scalar the kernel dereferences, so a
+        # hipGraph-captured step sees schedule updates (lr setter below)
+        self._lr_dev = (torch.zeros(1, dtype=torch.float32,
+                                    device=flat.master.device)
+                        if flat.master.is_cuda else None)
         self.lr = lr
         self.momentum = momentum
         self.weight_decay = weight_decay
@@ -36,22 +41,37 @@ class FusedSGD:
             self.momentum_buf = torch.zeros_like(flat.master)
         self._step_count = 0
 
+    @property
+    def lr(self) -> float:
+        return self._lr
+
+    @lr.setter
+    def lr(self, value: float) -> None:
+        self._lr = float(value)
+        if self._lr_dev is not None:
+            self._lr_dev.fill_(self._lr)
+
     @torch.no_grad()
     def step(self, grad_scale: float = 1.0) -> None:
         p, g = self.flat.master, self.flat.grad
+        # torch.optim.SGD initializes the momentum buffer to a CLONE of d_p
+        # on its first use (no (1-dampening) factor); with zero-initialized
+        # flat storage that is equivalent to dampening=0 on the first step
+        damp = self.dampening if self._step_count > 0 else 0.0
         if p.is_cuda:
             mom = self.momentum_buf if self.momentum_buf is not None \
                 else torch.empty(0, device=p.device)
             ext().fused_sgd(p, g, mom, self.lr, self.momentum,
-                            self.weight_decay, self.dampening,
-                            self.nesterov, grad_scale)
+                            self.weight_decay, damp,
+                            self.nesterov, grad_scale,
+                            lr_dev=self._lr_dev)
         else:
             d_p = g if grad_scale == 1.0 else g * grad_scale
             if self.weight_decay != 0.0:
                 d_p = d_p.add(p, alpha=self.weight_decay)
             if self.momentum != 0.0:
                 buf = self.momentum_buf
-                buf.mul_(self.momentum).add_(d_p, alpha=1 - self.dampening)
+                buf.mul_(self.momentum).add_(d_p, alpha=1 - damp)
                 d_p = d_p.add(buf, alpha=self.momentum) if self.nesterov else buf
             p.add_(d_p, alpha=-self.lr)
         self._step_count += 1
@@ -97,3 +117,6 @@ class FusedSGD:
                 self.momentum_buf[off:off + n].copy_(
                     buf.reshape(-1).to(self.momentum_buf.device,
                                        torch.float32))
+                # restored buffers are past their first use: the next step
+                # must apply the real dampening factor
+                self._step_count = max(self._step_count, 1)
