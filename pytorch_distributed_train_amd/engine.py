@@ -54,9 +54,10 @@ class Trainer:
         self.model.to(self.device)
 
         self.comm = make_communicator(self.env, self.device, args.backend)
+        self._bn_comm = None
         if args.sync_bn and self.env.world_size > 1:
-            bn_comm = self.comm.dup()
-            self.model = convert_sync_batchnorm(self.model, bn_comm)
+            self._bn_comm = self.comm.dup()
+            self.model = convert_sync_batchnorm(self.model, self._bn_comm)
 
         self.flat = FlatParams(self.model.named_parameters(),
                                device=self.device)
@@ -82,17 +83,26 @@ class Trainer:
         self.epoch = 0
         self.global_step = 0
         self._fault = self._parse_fault(args.fault_inject)
-        self.watchdog = Watchdog(getattr(args, "watchdog_sec", 0.0),
-                                 rank=self.env.rank)
+        wd_sec = float(getattr(args, "watchdog_sec", -1.0))
+        if wd_sec < 0:  # auto: on for distributed runs (PGNCCL is always-on)
+            wd_sec = 120.0 if self.env.world_size > 1 else 0.0
+        self.watchdog = Watchdog(wd_sec, rank=self.env.rank)
+        for c in (self.comm, getattr(self, "_bn_comm", None)):
+            if c is not None and hasattr(c, "abort"):
+                self.watchdog.add_stall_hook(c.abort)
         self.watchdog.start()
 
     @staticmethod
     def _parse_fault(spec: str):
-        # "step:N:rank:R" -> (N, R)
+        # "step:N:rank:R" -> (N, R, "raise"); "hang:step:N:rank:R" hangs
         if not spec:
             return None
+        mode = "raise"
+        if spec.startswith("hang:"):
+            mode = "hang"
+            spec = spec[len("hang:"):]
         parts = spec.split(":")
-        return int(parts[1]), int(parts[3])
+        return int(parts[1]), int(parts[3]), mode
 
     def _check_grads_finite(self):
         """NanCheck analog: one reduction over the flat gradient; raises
@@ -109,6 +119,11 @@ class Trainer:
         """One optimization step; returns the (device) scalar loss."""
         if self._fault and self.global_step == self._fault[0] \
                 and self.env.rank == self._fault[1]:
+            if self._fault[2] == "hang":
+                # desync scenario: this rank never reaches the collective;
+                # peers block in all-reduce until THEIR watchdog fires
+                while True:
+                    time.sleep(3600)
             raise RuntimeError(
                 f"fault injection: step {self.global_step} rank {self.env.rank}")
         self.optimizer.zero_grad()

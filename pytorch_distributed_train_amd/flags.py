@@ -74,11 +74,16 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--eval-batches", type=int, default=10,
                    help="cap eval batches (synthetic data)")
     p.add_argument("--watchdog-sec", type=float,
-                   default=float(os.environ.get("PDTA_WATCHDOG_SEC", "0")),
-                   help="abort + dump flight recorder if no training "
-                        "progress for this long (0 = off)")
+                   default=float(os.environ.get("PDTA_WATCHDOG_SEC", "-1")),
+                   help="abort + dump flight recorder + ncclCommAbort if no "
+                        "training progress for this long; -1 = auto (120 s "
+                        "when distributed, off single-process — the "
+                        "always-on ProcessGroupNCCL watchdog analog); 0 = "
+                        "off")
     p.add_argument("--fault-inject", default="",
-                   help="step:N:rank:R -> raise at step N on rank R (testing)")
+                   help="step:N:rank:R -> raise at step N on rank R; "
+                        "hang:step:N:rank:R -> block forever instead "
+                        "(watchdog/desync testing)")
     p.add_argument("--nan-check", action="store_true",
                    help="after each backward+all-reduce, verify the flat "
                         "gradient is finite and abort with rank info if "

@@ -795,6 +795,13 @@ void rccl_comm_destroy(int64_t comm) {
   ncclCommDestroy((ncclComm_t)(uintptr_t)comm);
 }
 
+void rccl_comm_abort(int64_t comm) {
+  // watchdog stall path: tears down in-flight collectives so a
+  // desynchronized gang member cannot wedge the GPU (ncclCommAbort
+  // semantics — safe to call from a non-enqueueing thread)
+  ncclCommAbort((ncclComm_t)(uintptr_t)comm);
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -857,4 +864,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rccl_broadcast", &rccl_broadcast);
   m.def("rccl_allgather", &rccl_allgather);
   m.def("rccl_comm_destroy", &rccl_comm_destroy);
+  m.def("rccl_comm_abort", &rccl_comm_abort);
 }

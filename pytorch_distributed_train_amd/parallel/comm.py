@@ -78,6 +78,9 @@ class NullComm:
     def barrier(self) -> None:
         pass
 
+    def abort(self) -> None:
+        pass
+
     def dup(self) -> "NullComm":
         return self
 
@@ -124,6 +127,9 @@ class TorchDistComm:
 
     def barrier(self) -> None:
         torch_dist.barrier()
+
+    def abort(self) -> None:
+        pass  # gloo collectives are host-blocking; os.abort reaps them
 
     def dup(self) -> "TorchDistComm":
         return self  # gloo enqueue is host-ordered; one PG suffices on CPU
@@ -185,6 +191,14 @@ class RcclComm:
     def barrier(self) -> None:
         self._control.barrier()
         torch.cuda.synchronize()
+
+    def abort(self) -> None:
+        """Tear down in-flight RCCL work (watchdog stall path) so a hung
+        collective cannot wedge the device past process death."""
+        try:
+            self._ext.rccl_comm_abort(self._comm)
+        except Exception:
+            pass
 
     def dup(self) -> "RcclComm":
         """A second communicator (own ncclComm_t) for collectives that must

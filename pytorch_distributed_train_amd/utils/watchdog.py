@@ -52,6 +52,13 @@ class Watchdog:
         self._armed = False
         self._stop = threading.Event()
         self._thread = None
+        self._on_stall = []
+
+    def add_stall_hook(self, fn) -> None:
+        """Called (best-effort) on stall before the process aborts —
+        e.g. RcclComm.abort so in-flight collectives are torn down
+        (ncclCommAbort analog of ProcessGroupNCCL's watchdog)."""
+        self._on_stall.append(fn)
 
     def start(self) -> None:
         if self.timeout <= 0 or self._thread is not None:
@@ -78,5 +85,10 @@ class Watchdog:
                       "dumping state and aborting", file=sys.stderr)
                 dump_flight_recorder()
                 faulthandler.dump_traceback(file=sys.stderr)
+                for fn in self._on_stall:
+                    try:
+                        fn()
+                    except Exception:
+                        pass
                 sys.stderr.flush()
                 os.abort()  # SIGABRT -> launcher restarts the gang

@@ -101,3 +101,24 @@ def test_nan_check_trips_on_poisoned_grad():
     trainer.flat.grad[3] = float("nan")
     with _pytest.raises(RuntimeError, match="non-finite gradient"):
         trainer._check_grads_finite()
+
+
+def test_watchdog_exits_gang_on_hang(tmp_path):
+    """A rank that hangs pre-collective (hang:step:1:rank:1) leaves its
+    peer blocked in all-reduce; the peer's watchdog must fire, dump the
+    flight recorder, and abort so the launcher reaps the gang instead of
+    hanging forever (PGNCCL watchdog behavior)."""
+    cmd = [sys.executable, "-m", "pytorch_distributed_train_amd.launch",
+           "--nproc_per_node=2", "--master_port=29731",
+           "--monitor-interval=0.05", "--max-restarts=0",
+           os.path.join(REPO, "train.py"),
+           "--model", "resnet18", "--data", "synthetic-cifar10",
+           "--batch-size", "8", "--num-samples", "64",
+           "--epochs", "1", "--device", "cpu",
+           "--watchdog-sec", "5",
+           "--fault-inject", "hang:step:1:rank:1"]
+    r = subprocess.run(cmd, cwd=REPO, capture_output=True, text=True,
+                       timeout=180)
+    assert r.returncode != 0
+    assert "[watchdog]" in r.stderr, r.stderr[-2000:]
+    assert "flight recorder" in r.stderr
