@@ -61,6 +61,14 @@ class Trainer:
 
         self.flat = FlatParams(self.model.named_parameters(),
                                device=self.device)
+        # buffer sync (torch DDP broadcast_buffers analog,
+        # nn/parallel/distributed.py:1557 behavior): float buffers are
+        # re-viewed into ONE flat tensor at init, so the per-step sync is a
+        # single small broadcast with zero pack/unpack copies
+        self._buffer_flat = None
+        if getattr(args, "broadcast_buffers", False) \
+                and self.env.world_size > 1:
+            self._buffer_flat = self._flatten_buffers()
         comm_dtype = None
         if args.grad_comm_dtype == "bf16":
             comm_dtype = torch.bfloat16
@@ -91,6 +99,28 @@ class Trainer:
             if c is not None and hasattr(c, "abort"):
                 self.watchdog.add_stall_hook(c.abort)
         self.watchdog.start()
+
+    def _flatten_buffers(self):
+        mods = [(m, n, b) for m in self.model.modules()
+                for n, b in m.named_buffers(recurse=False)
+                if b.dtype == torch.float32]
+        if not mods:
+            return None
+        total = sum(b.numel() for _, _, b in mods)
+        flat = torch.empty(total, dtype=torch.float32, device=self.device)
+        off = 0
+        for m, name, b in mods:
+            n = b.numel()
+            flat[off:off + n].copy_(b.reshape(-1))
+            m._buffers[name] = flat[off:off + n].view_as(b)
+            off += n
+        return flat
+
+    def _sync_buffers(self) -> None:
+        """Rank-0 buffer broadcast before forward; enqueued on the compute
+        stream so no host sync is added on GPU."""
+        if self._buffer_flat is not None and self.model.training:
+            self.comm.broadcast(self._buffer_flat, src=0).sync_stream()
 
     @staticmethod
     def _parse_fault(spec: str):
@@ -128,6 +158,7 @@ class Trainer:
                 f"fault injection: step {self.global_step} rank {self.env.rank}")
         self.optimizer.zero_grad()
         self.reducer.prepare()
+        self._sync_buffers()
         if self.packs is not None:
             self.packs.refresh()  # one kernel repacks every layer's weights
         with record_function("pdta.forward"):

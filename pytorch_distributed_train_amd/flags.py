@@ -57,6 +57,13 @@ def build_parser() -> argparse.ArgumentParser:
                    help="compress gradient all-reduce payload")
     p.add_argument("--no-overlap", action="store_true",
                    help="disable backward/all-reduce overlap (debugging)")
+    p.add_argument("--broadcast-buffers", dest="broadcast_buffers",
+                   action="store_true", default=True,
+                   help="broadcast module buffers (BN running stats) from "
+                        "rank 0 each training step (torch DDP "
+                        "broadcast_buffers=True parity)")
+    p.add_argument("--no-broadcast-buffers", dest="broadcast_buffers",
+                   action="store_false")
     p.add_argument("--backend", default="auto", choices=("auto", "rccl", "gloo"),
                    help="collective backend; auto = rccl on GPU, gloo on CPU")
     p.add_argument("--device", default="auto", choices=("auto", "cuda", "cpu"))

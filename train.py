@@ -31,8 +31,9 @@ from pytorch_distributed_train_amd.utils.metrics import RankLogger, StepTimer
 
 
 def evaluate(trainer, dataset, env, batch_size, max_batches=0):
-    """Top-1/top-5 on a synthetic validation split (rank-local; averaged
-    across ranks by the caller if distributed)."""
+    """Top-1/top-5 on a synthetic validation split. Counts are summed
+    across ranks (one tiny all-reduce), so every rank returns the same
+    global accuracy."""
     from pytorch_distributed_train_amd.data import DistributedSampler, Loader
     sampler = DistributedSampler(dataset, env.world_size, env.rank,
                                  shuffle=False)
@@ -52,6 +53,13 @@ def evaluate(trainer, dataset, env, batch_size, max_batches=0):
             top5 += int(correct.any(dim=1).sum())
             total += labels.numel()
     trainer.model.train()
+    if env.world_size > 1:
+        counts = torch.tensor([top1, top5, total], dtype=torch.float32,
+                              device=trainer.device)
+        trainer.comm.allreduce_sum(counts).wait()
+        if counts.is_cuda:
+            torch.cuda.synchronize()
+        top1, top5, total = (int(v) for v in counts.tolist())
     return (top1 / max(total, 1), top5 / max(total, 1))
 
 
