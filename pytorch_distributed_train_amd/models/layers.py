@@ -123,7 +123,11 @@ class BatchNorm2dNHWC(nn.Module):
                 self.num_batches_tracked += 1
                 m = self.momentum
                 # running_var uses the unbiased estimator, as torch BN does
-                unbiased = var * (count / max(count - 1.0, 1.0))
+                # (count may be a device scalar on the SyncBN path)
+                if torch.is_tensor(count):
+                    unbiased = var * (count / (count - 1.0).clamp(min=1.0))
+                else:
+                    unbiased = var * (count / max(count - 1.0, 1.0))
                 self.running_mean.mul_(1 - m).add_(mean, alpha=m)
                 self.running_var.mul_(1 - m).add_(unbiased, alpha=m)
             invstd = (var + self.eps).rsqrt()

@@ -185,10 +185,19 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
             dweight = sum_dy_xmu * invstd
             dbias = sum_dy.clone() if ctx.comm is not None else sum_dy
             if ctx.comm is not None:
-                packed = torch.cat([sum_dy, sum_dy_xmu])
-                ctx.comm.allreduce_sum(packed).sync_stream()
+                # [sum_dy ‖ sum_dy_xmu ‖ local_count] in ONE all-reduce;
+                # normalizing by the reduced count on-device keeps the
+                # path host-sync-free AND exact for ragged rank batches
                 c = sum_dy.numel()
-                sum_dy, sum_dy_xmu = packed[:c], packed[c:]
+                local_n = float(x.shape[0] * x.shape[1] * x.shape[2])
+                packed = torch.cat([sum_dy, sum_dy_xmu,
+                                    torch.full((1,), local_n,
+                                               dtype=torch.float32,
+                                               device=sum_dy.device)])
+                ctx.comm.allreduce_sum(packed).sync_stream()
+                packed[:2 * c] /= packed[2 * c]
+                sum_dy, sum_dy_xmu = packed[:c], packed[c:2 * c]
+                n = 1.0
             out = ext().bn_bwd_elemt(x, dy, relu_y, mean, invstd, weight,
                                      sum_dy.contiguous(),
                                      sum_dy_xmu.contiguous(), float(n),
@@ -208,10 +217,15 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
             dweight = sum_dy_xmu * invstd
             dbias = sum_dy.clone() if ctx.comm is not None else sum_dy
             if ctx.comm is not None:
-                packed = torch.cat([sum_dy, sum_dy_xmu])
-                ctx.comm.allreduce_sum(packed).wait()
                 c = sum_dy.numel()
-                sum_dy, sum_dy_xmu = packed[:c], packed[c:]
+                local_n = float(x.shape[0] * x.shape[1] * x.shape[2])
+                packed = torch.cat([sum_dy, sum_dy_xmu,
+                                    torch.full((1,), local_n,
+                                               dtype=torch.float32)])
+                ctx.comm.allreduce_sum(packed).wait()
+                packed[:2 * c] /= packed[2 * c]
+                sum_dy, sum_dy_xmu = packed[:c], packed[c:2 * c]
+                n = 1.0
             w32 = weight.float()
             if ctx.has_residual:
                 dres = dy32.to(x.dtype)

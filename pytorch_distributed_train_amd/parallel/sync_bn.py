@@ -41,10 +41,10 @@ class SyncBatchNorm2dNHWC(BatchNorm2dNHWC):
             gmean = g[:, :c].sum(dim=0) / total
             gvar = g[:, c:2 * c].sum(dim=0) / total - gmean * gmean
             gvar.clamp_(min=0.0)
-            return gmean, gvar, float(total.item()) if not x.is_cuda \
-                else local_count * comm.world_size
-        # note: on GPU the global count equals local*world (equal shapes per
-        # rank by the sampler's padding contract), avoiding a host sync.
+            # the REDUCED count is returned as a device scalar (no host
+            # sync, and exact even for ragged last batches — the
+            # equal-shapes shortcut it replaces silently broke there)
+            return gmean, gvar, total
 
 
 def convert_sync_batchnorm(module: torch.nn.Module, comm):
