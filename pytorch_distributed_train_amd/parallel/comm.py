@@ -135,6 +135,66 @@ class TorchDistComm:
         return self  # gloo enqueue is host-ordered; one PG suffices on CPU
 
 
+class StagedGlooComm(TorchDistComm):
+    """GPU tensors staged through host memory over gloo.
+
+    RCCL rejects two ranks of one communicator on the same device
+    ("Duplicate GPU detected", hard error in transport init), so
+    multi-rank-per-GPU validation runs — on the single-GPU box only —
+    with collectives staged D2H → gloo → H2D. Every other piece of the
+    distributed path (reducer buckets, SyncBN packed stats, flat
+    broadcast) is exercised unchanged on the HIP kernels. Never selected
+    when ranks map 1:1 onto devices (the 8-GPU node runs RcclComm)."""
+
+    def _staged(self, t, run):
+        if not t.is_cuda:
+            return run(t)
+        host = t.cpu()  # sync copy: grads on the caller's stream are done
+        w = run(host)
+        return _StagedWork(w, lambda: t.copy_(host))
+
+    def allreduce_sum(self, tensor, stream=None) -> Work:
+        return self._staged(tensor,
+                            lambda h: super(StagedGlooComm, self)
+                            .allreduce_sum(h))
+
+    def allreduce_avg(self, tensor, stream=None) -> Work:
+        return self._staged(tensor,
+                            lambda h: super(StagedGlooComm, self)
+                            .allreduce_avg(h))
+
+    def broadcast(self, tensor, src: int = 0, stream=None) -> Work:
+        return self._staged(tensor,
+                            lambda h: super(StagedGlooComm, self)
+                            .broadcast(h, src=src))
+
+    def allgather_into(self, out, inp, stream=None) -> Work:
+        if not out.is_cuda:
+            return super().allgather_into(out, inp)
+        hout = out.cpu()
+        hinp = inp.cpu()
+        w = super().allgather_into(hout, hinp)
+        return _StagedWork(w, lambda: out.copy_(hout))
+
+    def dup(self) -> "StagedGlooComm":
+        return self
+
+
+class _StagedWork(Work):
+    def __init__(self, inner: Work, post):
+        self._inner = inner
+        self._post = post
+
+    def wait(self) -> None:
+        if self._inner is not None:
+            self._inner.wait()
+            self._inner = None
+            self._post()
+
+    def sync_stream(self, stream=None) -> None:
+        self.wait()
+
+
 class RcclComm:
     """RCCL over xGMI; the extension owns the ncclComm_t.
 
@@ -225,6 +285,17 @@ def make_communicator(env: DistEnv, device, backend: str = "auto"):
     dev = torch.device(device)
     if backend == "auto":
         backend = "rccl" if dev.type == "cuda" else "gloo"
+    if backend == "rccl" and dev.type == "cuda" \
+            and env.world_size > torch.cuda.device_count():
+        # more ranks than GPUs: RCCL hard-errors on duplicate devices in
+        # one communicator — stage through gloo (single-GPU validation
+        # only; never taken when ranks map 1:1 to devices)
+        import sys
+        print(f"[comm] rank {env.rank}: {env.world_size} ranks > "
+              f"{torch.cuda.device_count()} GPUs — RCCL cannot share a "
+              "device; using host-staged gloo collectives",
+              file=sys.stderr)
+        return StagedGlooComm(env, backend="gloo")
     control = TorchDistComm(env, backend="gloo")
     if backend == "gloo":
         return control
