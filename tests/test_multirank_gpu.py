@@ -140,7 +140,7 @@ def test_rccl_ddp_2ranks_syncbn_split_batch(tmp_path):
 def test_bench_self_launch_2ranks_on_1gpu():
     """`python bench.py --gpus 2` with no rendezvous env must produce an
     n_gpus=2 JSON record (driver contract) — here with both ranks on the
-    single GPU over real RCCL."""
+    single GPU (host-staged collectives; RcclComm on real multi-GPU)."""
     env = dict(os.environ)
     for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE", "MASTER_ADDR",
               "MASTER_PORT"):
