@@ -105,7 +105,7 @@ def main(argv=None) -> int:
             torch.cuda.synchronize()
 
     use_graph = False
-    if b.hip_graph and env.world_size == 1 and on_gpu:
+    if b.hip_graph and on_gpu:
         try:
             use_graph = trainer.enable_hip_graph(x, y)
         except Exception as e:  # never let capture break the bench contract

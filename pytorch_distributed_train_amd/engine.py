@@ -10,6 +10,7 @@ step:
 """
 from __future__ import annotations
 
+import os
 import time
 
 import torch
@@ -182,27 +183,42 @@ class Trainer:
         self.watchdog.heartbeat()
         return loss.detach()
 
-    # ---- hipGraph-captured step (single-GPU) ---------------------------
+    # ---- hipGraph-captured step ----------------------------------------
     def enable_hip_graph(self, example_x, example_y) -> bool:
         """Capture one whole training step (weight repack, forward, loss,
-        backward, fused SGD) into a HIP graph; graph_step() replays it.
-        Launch-bound overhead (200+ kernel launches/step) collapses into
-        one graph launch. Single-GPU only for now (the DDP bucket hooks
-        enqueue RCCL work from the autograd thread mid-capture)."""
-        if not isinstance(self.comm, NullComm) or self.scaler.enabled \
-                or self.device.type != "cuda":
+        backward — including the reducer's RCCL bucket all-reduces on the
+        side stream — fused SGD) into a HIP graph; graph_step() replays
+        it. Launch-bound overhead (200+ kernel launches/step) collapses
+        into one graph launch.
+
+        RCCL collectives are graph-capturable (their enqueue lands on the
+        side stream, which joins the capture through the bucket events);
+        validated on hardware at world-1 RCCL (PDTA_RCCL_WS1, see
+        tests/test_multirank_gpu.py). Multi-rank capture is opt-in via
+        PDTA_DIST_GRAPH=1 until it has run on an 8-GPU node; default at
+        ws>1 is eager (guaranteed-progress choice for the driver bench).
+        Host-staged gloo cannot be captured."""
+        from .parallel.comm import RcclComm
+        if not isinstance(self.comm, (NullComm, RcclComm)) \
+                or self.scaler.enabled or self.device.type != "cuda":
+            return False
+        if self.env.world_size > 1 \
+                and os.environ.get("PDTA_DIST_GRAPH") != "1":
             return False
         self._gx = example_x.clone()
         self._gy = example_y.clone()
 
         def body():
             self.optimizer.zero_grad()
+            self.reducer.prepare()
+            self._sync_buffers()
             if self.packs is not None:
                 self.packs.refresh()
             logits = self.model(self._gx)
             loss = OF.softmax_cross_entropy(logits, self._gy,
                                             self.args.label_smoothing)
             loss.backward()
+            self.reducer.finalize()
             self.optimizer.step()
             return loss.detach()
 

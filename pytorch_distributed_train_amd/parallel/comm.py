@@ -281,6 +281,13 @@ class RcclComm:
 def make_communicator(env: DistEnv, device, backend: str = "auto"):
     """backend: auto|rccl|gloo. auto = rccl on cuda devices, gloo on cpu."""
     if env.world_size <= 1:
+        if os.environ.get("PDTA_RCCL_WS1") == "1" \
+                and torch.device(device).type == "cuda":
+            # validation mode: a real world-1 RCCL communicator so the
+            # reducer's bucket collectives (and their hipGraph capture)
+            # run the genuine enqueue path on a single-GPU box
+            control = TorchDistComm(env, backend="gloo")
+            return RcclComm(env, control, torch.device(device))
         return NullComm()
     dev = torch.device(device)
     if backend == "auto":
