@@ -50,14 +50,19 @@ class Conv2dNHWC(nn.Module):
         # emit_bn_stats (set by the model builder on conv->BN pairs): the
         # fwd kernel's epilogue writes per-tile channel (sum, sumsq)
         # partials so the BN that follows never re-reads y for statistics
+        fuse_grads = x.is_cuda and self.training and torch.is_grad_enabled()
+        sink = getattr(self, "_dx_sink", None) if fuse_grads else None
+        accum = getattr(self, "_dx_accum", None) if fuse_grads else None
         if (getattr(self, "emit_bn_stats", False) and self.training
                 and x.is_cuda and self.bias is None
                 and torch.is_grad_enabled()):
             y, part = OF.conv2d_nhwc(x, w, wt, w2, self.stride,
-                                     self.padding, want_bn_stats=True)
+                                     self.padding, want_bn_stats=True,
+                                     dx_sink=sink, dx_accum=accum)
             y._pdta_bn_part = part
             return y
-        y = OF.conv2d_nhwc(x, w, wt, w2, self.stride, self.padding)
+        y = OF.conv2d_nhwc(x, w, wt, w2, self.stride, self.padding,
+                           dx_sink=sink, dx_accum=accum)
         if self.bias is not None:
             y = y + self.bias.to(y.dtype)  # broadcast over trailing C
         return y
@@ -117,7 +122,9 @@ class BatchNorm2dNHWC(nn.Module):
                 with torch.no_grad():
                     self.num_batches_tracked += 1
                 return OF.bn_train(x, self.weight, self.bias, mean, invstd,
-                                   self.fuse_relu, residual=residual)
+                                   self.fuse_relu, residual=residual,
+                                   dres_sink=getattr(self, "_dres_sink",
+                                                     None))
             mean, var, count = self._compute_stats(x)
             with torch.no_grad():
                 self.num_batches_tracked += 1
@@ -134,7 +141,9 @@ class BatchNorm2dNHWC(nn.Module):
             return OF.bn_train(x, self.weight, self.bias, mean, invstd,
                                self.fuse_relu, self.comm,
                                count if self.comm is not None else None,
-                               residual)
+                               residual,
+                               dres_sink=(getattr(self, "_dres_sink", None)
+                                          if x.is_cuda else None))
         return OF.bn_infer(x, self.weight, self.bias, self.running_mean,
                            self.running_var, self.eps, self.fuse_relu,
                            residual)

@@ -42,12 +42,24 @@ class BasicBlock(nn.Module):
         self.bn2.fuse_relu = True  # fused with the residual join
         self.downsample = downsample
         self.stride = stride
+        # residual-fork gradient fusion (GradJoinFn protocol): the
+        # identity-path gradient (bn2 dres, or the downsample conv's dx)
+        # is folded into conv1's bwd-data epilogue — no autograd add
+        self._fork = OF.GradHolder()
+        self.conv1._dx_accum = self._fork
+        if downsample is not None:
+            downsample[0]._dx_sink = self._fork
+        else:
+            self.bn2._dres_sink = self._fork
 
     def forward(self, x):
-        identity = x
+        if x.is_cuda and self.training and torch.is_grad_enabled():
+            x, identity = OF.grad_join(x, self._fork)
+        else:
+            identity = x
         out = self.bn1(self.conv1(x))  # ReLU fused into bn1
         if self.downsample is not None:
-            identity = self.downsample(x)
+            identity = self.downsample(identity)
         # bn2 + residual add + ReLU in one kernel
         return self.bn2(self.conv2(out), identity)
 
@@ -69,13 +81,22 @@ class Bottleneck(nn.Module):
         self.relu = ReLUNHWC()
         self.downsample = downsample
         self.stride = stride
+        self._fork = OF.GradHolder()  # see BasicBlock
+        self.conv1._dx_accum = self._fork
+        if downsample is not None:
+            downsample[0]._dx_sink = self._fork
+        else:
+            self.bn3._dres_sink = self._fork
 
     def forward(self, x):
-        identity = x
+        if x.is_cuda and self.training and torch.is_grad_enabled():
+            x, identity = OF.grad_join(x, self._fork)
+        else:
+            identity = x
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
         if self.downsample is not None:
-            identity = self.downsample(x)
+            identity = self.downsample(identity)
         # bn3 + residual add + ReLU in one kernel
         return self.bn3(self.conv3(out), identity)
 

@@ -53,9 +53,9 @@ void pdta_bn_stats_from_part(float*, int, float*, float*, float*, float*,
 void pdta_conv_fwd(const void*, const void*, void*, const void*, int, int,
                    int, int, int, int, int, int, int, int, int, int,
                    hipStream_t, float*);
-void pdta_conv_bwd_data(const void*, const void*, void*, const void*, int,
-                        int, int, int, int, int, int, int, int, int, int,
-                        int, hipStream_t);
+void pdta_conv_bwd_data(const void*, const void*, void*, const void*,
+                        const void*, int, int, int, int, int, int, int,
+                        int, int, int, int, int, hipStream_t);
 void pdta_wrw_plan(int, int, int, int, int, int, int, long*, long*);
 void pdta_conv_bwd_weight(const void*, const void*, float*, float*,
                           const void*, int, int, int, int, int, int, int,
@@ -480,8 +480,10 @@ std::vector<at::Tensor> bn_stats_train_from_part(at::Tensor part, long rows,
 }
 
 at::Tensor conv2d_bwd_data(at::Tensor dy, at::Tensor w2, int64_t H,
-                           int64_t W, int64_t stride, int64_t pad) {
-  // dy [N,P,Q,Kout]; w2 pack [R,S,C,Kout]
+                           int64_t W, int64_t stride, int64_t pad,
+                           c10::optional<at::Tensor> accum) {
+  // dy [N,P,Q,Kout]; w2 pack [R,S,C,Kout]; accum (optional, same shape as
+  // dx): residual-fork gradient folded into the store epilogue
   CHECK_IN(dy);
   CHECK_IN(w2);
   int N = dy.size(0), P = dy.size(1), Q = dy.size(2), Kout = dy.size(3);
@@ -490,9 +492,17 @@ at::Tensor conv2d_bwd_data(at::Tensor dy, at::Tensor w2, int64_t H,
                            "(the stem conv never needs dx)");
   TORCH_CHECK(Kout % 64 == 0, "bwd_data: Kout must be a multiple of 64");
   auto dx = at::empty({N, H, W, (long)C}, dy.options());
+  const void* acc_ptr = nullptr;
+  if (accum.has_value()) {
+    TORCH_CHECK(accum->is_contiguous() &&
+                accum->sizes() == dx.sizes() &&
+                accum->scalar_type() == dy.scalar_type(),
+                "bwd_data accum: shape/dtype mismatch");
+    acc_ptr = accum->data_ptr();
+  }
   pdta_conv_bwd_data(dy.data_ptr(), w2.data_ptr(), dx.data_ptr(),
-                     zero_page(dy), N, H, W, C, P, Q, Kout, R, S, stride,
-                     pad, dt_of(dy), cur_stream());
+                     zero_page(dy), acc_ptr, N, H, W, C, P, Q, Kout, R, S,
+                     stride, pad, dt_of(dy), cur_stream());
   return dx;
 }
 
@@ -504,7 +514,7 @@ at::Tensor conv2d_bwd_weight(at::Tensor x, at::Tensor dy, int64_t R,
   int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
   int P = dy.size(1), Q = dy.size(2), Kout = dy.size(3);
   if (C % 64 == 0) {
-    auto dw = at::zeros({Kout, C, R, S}, x.options().dtype(at::kFloat));
+    auto dw = at::empty({Kout, C, R, S}, x.options().dtype(at::kFloat));
     long tiles, chunks;
     pdta_wrw_plan(N, P, Q, Kout, C, R, S, &tiles, &chunks);
     auto part = at::empty({tiles * (chunks + 16) * 4096},
@@ -527,7 +537,7 @@ at::Tensor conv2d_bwd_weight(at::Tensor x, at::Tensor dy, int64_t R,
     pdta_stem_im2col(x.data_ptr(), xcol.data_ptr(), N, H, W, C, P, Q, R, S,
                      stride, pad, rsc_pad, dt_of(x), cur_stream());
   }
-  auto dwcol = at::zeros({Kout, rsc_pad}, x.options().dtype(at::kFloat));
+  auto dwcol = at::empty({Kout, rsc_pad}, x.options().dtype(at::kFloat));
   long tiles, chunks;
   pdta_wrw_plan(N, P, Q, Kout, rsc_pad, 1, 1, &tiles, &chunks);
   auto part = at::empty({tiles * (chunks + 16) * 4096},
@@ -841,7 +851,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("want_bn_stats") = false, py::arg("want_xcol") = false);
   m.def("bn_stats_train_from_part", &bn_stats_train_from_part);
   m.def("bn_stats_from_part", &bn_stats_from_part);
-  m.def("conv2d_bwd_data", &conv2d_bwd_data);
+  m.def("conv2d_bwd_data", &conv2d_bwd_data, py::arg("dy"),
+        py::arg("w2"), py::arg("H"), py::arg("W"),
+        py::arg("stride"), py::arg("pad"),
+        py::arg("accum") = py::none());
   m.def("conv2d_bwd_weight", &conv2d_bwd_weight, py::arg("x"),
         py::arg("dy"), py::arg("R"), py::arg("S"),
         py::arg("stride"), py::arg("pad"),

@@ -373,7 +373,8 @@ template <typename T, int STRIDE, int BNT>  // STRIDE 1/2/0(runtime)
 __global__ __launch_bounds__(TPB) void igemm_bwd_data_k(
     const T* __restrict__ dy, const T* __restrict__ w2, T* __restrict__ dx,
     const T* __restrict__ zero_page, int Nn, int H, int W, int C, int P,
-    int Q, int Kout, int R, int S, int stride, int pad, int tiles_n) {
+    int Q, int Kout, int R, int S, int stride, int pad, int tiles_n,
+    const T* __restrict__ accum) {
   constexpr int BOFF = 128 * 64 * 2;
   constexpr int NW_N = (BNT == 128) ? 2 : 1;
   constexpr int NW_M = 4 / NW_N;
@@ -497,7 +498,13 @@ __global__ __launch_bounds__(TPB) void igemm_bwd_data_k(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         long row = rbase + r;
-        if (row < M) dx[row * C + col] = from_f32<T>(acc[i][j][r]);
+        if (row < M) {
+          // accum: residual-fork gradient folded into the store (kills
+          // the autograd CUDAFunctor_add pass at every block boundary)
+          float v = acc[i][j][r];
+          if (accum) v += to_f32(accum[row * C + col]);
+          dx[row * C + col] = from_f32<T>(v);
+        }
       }
     }
   }
@@ -509,7 +516,8 @@ __global__ __launch_bounds__(512) void igemm_bwd_data_big_k(
     const T* __restrict__ dy, const T* __restrict__ w2, T* __restrict__ dx,
     const T* __restrict__ zero_page, int Nn, int H, int W, int C, int P,
     int Q, int Kout, int R, int S, int stride, int pad, int tiles_n,
-    unsigned mw, int sw, unsigned mhw, int shw) {
+    unsigned mw, int sw, unsigned mhw, int shw,
+    const T* __restrict__ accum) {
   constexpr int BOFF = 256 * 64 * 2;
   __shared__ __attribute__((aligned(16))) char lds[2][2 * BOFF];
 
@@ -632,7 +640,11 @@ __global__ __launch_bounds__(512) void igemm_bwd_data_big_k(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         long row = rbase + r;
-        if (row < M) dx[row * C + col] = from_f32<T>(acc[i][j][r]);
+        if (row < M) {
+          float v = acc[i][j][r];
+          if (accum) v += to_f32(accum[row * C + col]);
+          dx[row * C + col] = from_f32<T>(v);
+        }
       }
     }
   }
@@ -868,10 +880,12 @@ __global__ void wrw_reduce_k(const float* __restrict__ part,
     int c = c_t * 64 + (local & 63);
     const float* sp = reinterpret_cast<const float*>(&sv);
     if (ko < Kout) {
+      // each (ko,c,r,s) is produced by exactly one (tile,lane): plain
+      // store, so dw needs no zero-fill before the reduce
 #pragma unroll
       for (int j = 0; j < 4; ++j)
         if (c + j < C)
-          dw[(((long)ko * C + c + j) * R + r) * S + ss] += sp[j];
+          dw[(((long)ko * C + c + j) * R + r) * S + ss] = sp[j];
     }
   }
 }
@@ -1123,9 +1137,10 @@ void pdta_conv_fwd(const void* x, const void* wt, void* y,
 }
 
 void pdta_conv_bwd_data(const void* dy, const void* w2, void* dx,
-                        const void* zero_page, int Nn, int H, int W, int C,
-                        int P, int Q, int Kout, int R, int S, int stride,
-                        int pad, int dt, hipStream_t s) {
+                        const void* zero_page, const void* accum, int Nn,
+                        int H, int W, int C, int P, int Q, int Kout, int R,
+                        int S, int stride, int pad, int dt,
+                        hipStream_t s) {
   long M = (long)Nn * H * W;
   if (C % 256 == 0 && M >= 4096 && (stride == 1 || stride == 2)) {
     int tiles_m = (int)((M + 255) / 256), tiles_n = C / 256;
@@ -1136,12 +1151,12 @@ void pdta_conv_bwd_data(const void* dy, const void* w2, void* dx,
         igemm_bwd_data_big_k<T, 1><<<grid, 512, 0, s>>>(
             (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H,
             W, C, P, Q, Kout, R, S, stride, pad, tiles_n, mw.mul, mw.shift,
-            mhw.mul, mhw.shift);
+            mhw.mul, mhw.shift, (const T*)accum);
       else
         igemm_bwd_data_big_k<T, 2><<<grid, 512, 0, s>>>(
             (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H,
             W, C, P, Q, Kout, R, S, stride, pad, tiles_n, mw.mul, mw.shift,
-            mhw.mul, mhw.shift);
+            mhw.mul, mhw.shift, (const T*)accum);
     });
     return;
   }
@@ -1153,28 +1168,34 @@ void pdta_conv_bwd_data(const void* dy, const void* w2, void* dx,
       if (stride == 1)
         igemm_bwd_data_k<T, 1, 128><<<grid, TPB, 0, s>>>(
             (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H,
-            W, C, P, Q, Kout, R, S, stride, pad, tiles_n);
+            W, C, P, Q, Kout, R, S, stride, pad, tiles_n,
+            (const T*)accum);
       else if (stride == 2)
         igemm_bwd_data_k<T, 2, 128><<<grid, TPB, 0, s>>>(
             (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H,
-            W, C, P, Q, Kout, R, S, stride, pad, tiles_n);
+            W, C, P, Q, Kout, R, S, stride, pad, tiles_n,
+            (const T*)accum);
       else
         igemm_bwd_data_k<T, 0, 128><<<grid, TPB, 0, s>>>(
             (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H,
-            W, C, P, Q, Kout, R, S, stride, pad, tiles_n);
+            W, C, P, Q, Kout, R, S, stride, pad, tiles_n,
+            (const T*)accum);
     } else {
       if (stride == 1)
         igemm_bwd_data_k<T, 1, 64><<<grid, TPB, 0, s>>>(
             (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H,
-            W, C, P, Q, Kout, R, S, stride, pad, tiles_n);
+            W, C, P, Q, Kout, R, S, stride, pad, tiles_n,
+            (const T*)accum);
       else if (stride == 2)
         igemm_bwd_data_k<T, 2, 64><<<grid, TPB, 0, s>>>(
             (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H,
-            W, C, P, Q, Kout, R, S, stride, pad, tiles_n);
+            W, C, P, Q, Kout, R, S, stride, pad, tiles_n,
+            (const T*)accum);
       else
         igemm_bwd_data_k<T, 0, 64><<<grid, TPB, 0, s>>>(
             (const T*)dy, (const T*)w2, (T*)dx, (const T*)zero_page, Nn, H,
-            W, C, P, Q, Kout, R, S, stride, pad, tiles_n);
+            W, C, P, Q, Kout, R, S, stride, pad, tiles_n,
+            (const T*)accum);
     }
   });
 }

@@ -58,6 +58,8 @@ def convert_sync_batchnorm(module: torch.nn.Module, comm):
                                   module.momentum)
         out.fuse_relu = module.fuse_relu
         out.comm = comm
+        if hasattr(module, "_dres_sink"):
+            out._dres_sink = module._dres_sink
         out.weight = module.weight
         out.bias = module.bias
         out.running_mean = module.running_mean

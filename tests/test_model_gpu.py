@@ -220,3 +220,41 @@ def test_depth_variants_step(name):
         loss = t.train_step(x, y)
     assert torch.isfinite(loss), float(loss)
     assert not torch.equal(p0, t.flat.master)  # optimizer stepped
+
+
+@pytest.mark.gpu
+def test_residual_fork_gradients_fused_not_added():
+    """Every residual fork's identity gradient must be folded into the
+    conv1 bwd-data epilogue (GradJoinFn protocol) — zero autograd-add
+    fallbacks — and the grads must match a CPU fp32 reference as closely
+    as bf16 allows (CUDAFunctor_add purge)."""
+    import torch
+    from pytorch_distributed_train_amd.flags import build_parser
+    from pytorch_distributed_train_amd.engine import Trainer
+    from pytorch_distributed_train_amd.ops import functional as OF
+
+    def make(dev, dtype):
+        args = build_parser().parse_args([
+            "--model", "resnet50", "--data", "synthetic-imagenet",
+            "--dtype", dtype, "--device", dev, "--batch-size", "8",
+            "--seed", "41"])
+        t = Trainer(args)
+        t.model.train()
+        return t
+
+    g = torch.Generator().manual_seed(3)
+    x = torch.randn(8, 3, 224, 224, generator=g)
+    y = torch.randint(0, 1000, (8,), generator=g)
+
+    OF.GRAD_JOIN_FALLBACKS[0] = 0
+    tg = make("cuda", "bf16")
+    tg.train_step(x.cuda(), y.cuda())
+    torch.cuda.synchronize()
+    assert OF.GRAD_JOIN_FALLBACKS[0] == 0, \
+        f"{OF.GRAD_JOIN_FALLBACKS[0]} forks fell back to autograd add"
+
+    tc = make("cpu", "fp32")
+    tc.train_step(x, y)
+    gg, gc = tg.flat.grad.cpu(), tc.flat.grad
+    rel = float((gg - gc).norm() / gc.norm().clamp(min=1e-12))
+    assert rel < 0.08, rel
