@@ -50,7 +50,8 @@ class Conv2dNHWC(nn.Module):
         # emit_bn_stats (set by the model builder on conv->BN pairs): the
         # fwd kernel's epilogue writes per-tile channel (sum, sumsq)
         # partials so the BN that follows never re-reads y for statistics
-        fuse_grads = x.is_cuda and self.training and torch.is_grad_enabled()
+        fuse_grads = x.is_cuda and self.training \
+            and torch.is_grad_enabled() and OF.fork_fuse_enabled()
         sink = getattr(self, "_dx_sink", None) if fuse_grads else None
         accum = getattr(self, "_dx_accum", None) if fuse_grads else None
         if (getattr(self, "emit_bn_stats", False) and self.training
@@ -123,8 +124,10 @@ class BatchNorm2dNHWC(nn.Module):
                     self.num_batches_tracked += 1
                 return OF.bn_train(x, self.weight, self.bias, mean, invstd,
                                    self.fuse_relu, residual=residual,
-                                   dres_sink=getattr(self, "_dres_sink",
-                                                     None))
+                                   dres_sink=(getattr(self, "_dres_sink",
+                                                      None)
+                                              if OF.fork_fuse_enabled()
+                                              else None))
             mean, var, count = self._compute_stats(x)
             with torch.no_grad():
                 self.num_batches_tracked += 1
@@ -143,7 +146,8 @@ class BatchNorm2dNHWC(nn.Module):
                                count if self.comm is not None else None,
                                residual,
                                dres_sink=(getattr(self, "_dres_sink", None)
-                                          if x.is_cuda else None))
+                                          if x.is_cuda and
+                                          OF.fork_fuse_enabled() else None))
         return OF.bn_infer(x, self.weight, self.bias, self.running_mean,
                            self.running_var, self.eps, self.fuse_relu,
                            residual)

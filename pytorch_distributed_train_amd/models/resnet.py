@@ -42,9 +42,10 @@ class BasicBlock(nn.Module):
         self.bn2.fuse_relu = True  # fused with the residual join
         self.downsample = downsample
         self.stride = stride
-        # residual-fork gradient fusion (GradJoinFn protocol): the
+        # residual-fork gradient fusion (GradHolder protocol): the
         # identity-path gradient (bn2 dres, or the downsample conv's dx)
-        # is folded into conv1's bwd-data epilogue — no autograd add
+        # is folded into conv1's bwd-data epilogue; the producer returns
+        # None to autograd, so no fork add exists
         self._fork = OF.GradHolder()
         self.conv1._dx_accum = self._fork
         if downsample is not None:
@@ -53,10 +54,11 @@ class BasicBlock(nn.Module):
             self.bn2._dres_sink = self._fork
 
     def forward(self, x):
-        if x.is_cuda and self.training and torch.is_grad_enabled():
-            x, identity = OF.grad_join(x, self._fork)
-        else:
-            identity = x
+        if self._fork.t is not None:
+            raise RuntimeError(
+                "residual-fork gradient was never folded (GradHolder "
+                "protocol violation — conv1 backward did not consume it)")
+        identity = x
         out = self.bn1(self.conv1(x))  # ReLU fused into bn1
         if self.downsample is not None:
             identity = self.downsample(identity)
@@ -89,10 +91,11 @@ class Bottleneck(nn.Module):
             self.bn3._dres_sink = self._fork
 
     def forward(self, x):
-        if x.is_cuda and self.training and torch.is_grad_enabled():
-            x, identity = OF.grad_join(x, self._fork)
-        else:
-            identity = x
+        if self._fork.t is not None:
+            raise RuntimeError(
+                "residual-fork gradient was never folded (GradHolder "
+                "protocol violation — conv1 backward did not consume it)")
+        identity = x
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
         if self.downsample is not None:

@@ -85,6 +85,11 @@ class GradJoinFn(torch.autograd.Function):
 GRAD_JOIN_FALLBACKS = [0]  # diagnosed by tests: 0 == every fork fused
 
 
+def fork_fuse_enabled() -> bool:
+    import os
+    return not os.environ.get("PDTA_NO_FORK_FUSE")
+
+
 def grad_join(x, holder):
     return GradJoinFn.apply(x, holder)
 
@@ -135,12 +140,15 @@ class Conv2dNHWC(torch.autograd.Function):
                 accum = None
                 if ctx.dx_accum is not None and ctx.dx_accum.t is not None:
                     accum = ctx.dx_accum.t
+                    ctx.dx_accum.t = None  # consumed (folded into dx)
                 dx = ext().conv2d_bwd_data(dy, w2, x.shape[1], x.shape[2],
                                            stride, pad, accum=accum)
-                if accum is not None:
-                    ctx.dx_accum.consumed = True
                 if ctx.dx_sink is not None:
+                    # residual-path producer: hand dx to the consumer's
+                    # epilogue and contribute NOTHING via autograd (the
+                    # fold is the only path — no fork add exists)
                     ctx.dx_sink.t = dx
+                    dx = None
             if need_dw:
                 dw = ext().conv2d_bwd_weight(x, dy, weight.shape[2],
                                              weight.shape[3], stride, pad,
@@ -265,9 +273,10 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
             if ctx.has_residual:
                 dres = out[1]
                 if ctx.dres_sink is not None:
-                    # hand the residual grad to the NEXT conv1's bwd-data
-                    # epilogue (GradJoinFn fork protocol)
+                    # residual-path producer (GradHolder protocol): the
+                    # consumer conv1 folds this; autograd sees None
                     ctx.dres_sink.t = dres
+                    dres = None
         else:
             dy32 = dy.float()
             if fuse_relu:
