@@ -53,17 +53,16 @@ class Conv2dNHWC(nn.Module):
         fuse_grads = x.is_cuda and self.training \
             and torch.is_grad_enabled() and OF.fork_fuse_enabled()
         sink = getattr(self, "_dx_sink", None) if fuse_grads else None
-        accum = getattr(self, "_dx_accum", None) if fuse_grads else None
         if (getattr(self, "emit_bn_stats", False) and self.training
                 and x.is_cuda and self.bias is None
                 and torch.is_grad_enabled()):
             y, part = OF.conv2d_nhwc(x, w, wt, w2, self.stride,
                                      self.padding, want_bn_stats=True,
-                                     dx_sink=sink, dx_accum=accum)
+                                     dx_sink=sink)
             y._pdta_bn_part = part
             return y
         y = OF.conv2d_nhwc(x, w, wt, w2, self.stride, self.padding,
-                           dx_sink=sink, dx_accum=accum)
+                           dx_sink=sink)
         if self.bias is not None:
             y = y + self.bias.to(y.dtype)  # broadcast over trailing C
         return y
@@ -122,12 +121,15 @@ class BatchNorm2dNHWC(nn.Module):
                         self.momentum, self.eps)
                 with torch.no_grad():
                     self.num_batches_tracked += 1
+                fuse = OF.fork_fuse_enabled()
                 return OF.bn_train(x, self.weight, self.bias, mean, invstd,
                                    self.fuse_relu, residual=residual,
                                    dres_sink=(getattr(self, "_dres_sink",
                                                       None)
-                                              if OF.fork_fuse_enabled()
-                                              else None))
+                                              if fuse else None),
+                                   dy2_source=(getattr(self, "_dy2_source",
+                                                       None)
+                                               if fuse else None))
             mean, var, count = self._compute_stats(x)
             with torch.no_grad():
                 self.num_batches_tracked += 1
@@ -147,7 +149,12 @@ class BatchNorm2dNHWC(nn.Module):
                                residual,
                                dres_sink=(getattr(self, "_dres_sink", None)
                                           if x.is_cuda and
-                                          OF.fork_fuse_enabled() else None))
+                                          OF.fork_fuse_enabled() else None),
+                               dy2_source=(getattr(self, "_dy2_source",
+                                                   None)
+                                           if x.is_cuda and
+                                           OF.fork_fuse_enabled()
+                                           else None))
         return OF.bn_infer(x, self.weight, self.bias, self.running_mean,
                            self.running_var, self.eps, self.fuse_relu,
                            residual)

@@ -47,7 +47,6 @@ class BasicBlock(nn.Module):
         # is folded into conv1's bwd-data epilogue; the producer returns
         # None to autograd, so no fork add exists
         self._fork = OF.GradHolder()
-        self.conv1._dx_accum = self._fork
         if downsample is not None:
             downsample[0]._dx_sink = self._fork
         else:
@@ -84,7 +83,6 @@ class Bottleneck(nn.Module):
         self.downsample = downsample
         self.stride = stride
         self._fork = OF.GradHolder()  # see BasicBlock
-        self.conv1._dx_accum = self._fork
         if downsample is not None:
             downsample[0]._dx_sink = self._fork
         else:
@@ -132,6 +130,23 @@ class ResNet(nn.Module):
                     isinstance(m[0], Conv2dNHWC) and \
                     isinstance(m[1], BatchNorm2dNHWC):
                 m[0].emit_bn_stats = True  # downsample pairs
+        # two-dy backward wiring (GradHolder protocol): block k's residual
+        # producer fills blocks[k]._fork; the CONSUMER is the op that
+        # produced block k's input — the last BN of block k-1 — which sums
+        # it as a second dy inside its backward kernels. Block 0's input
+        # comes from maxpool: its fork stays unwired (one autograd add on
+        # a 56x56x64 tensor — negligible).
+        blocks = [b for layer in (self.layer1, self.layer2, self.layer3,
+                                  self.layer4) for b in layer]
+        first = blocks[0]
+        if first.downsample is not None:
+            first.downsample[0]._dx_sink = None
+        else:
+            last_bn0 = getattr(first, "bn3", None) or first.bn2
+            last_bn0._dres_sink = None
+        for prev, blk in zip(blocks, blocks[1:]):
+            last_bn = getattr(prev, "bn3", None) or prev.bn2
+            last_bn._dy2_source = blk._fork
 
     def _make_layer(self, block, planes, blocks, stride=1):
         downsample = None

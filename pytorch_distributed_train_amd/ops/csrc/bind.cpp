@@ -24,11 +24,12 @@ void pdta_bn_apply(const void*, void*, const void*, const float*,
 void pdta_bn_bwd_reduce(const void*, const void*, const void*,
                         const float*, const float*, const float*,
                         const float*, int, float*, float*, float*, long,
-                        int, int, hipStream_t);
+                        int, int, const void*, hipStream_t);
 void pdta_bn_bwd_elemt(const void*, const void*, const void*, void*,
                        void*, const float*, const float*, int,
                        const float*, const float*, const float*,
-                       const float*, float, long, int, int, hipStream_t);
+                       const float*, float, long, int, int, const void*,
+                       hipStream_t);
 void pdta_maxpool_fwd(const void*, void*, uint8_t*, int, int, int, int, int,
                       int, int, int, int, int, hipStream_t);
 void pdta_maxpool_bwd(const void*, const uint8_t*, void*, int, int, int, int,
@@ -204,7 +205,8 @@ std::vector<at::Tensor> bn_bwd_reduce(at::Tensor x, at::Tensor dy,
                                       at::Tensor mean,
                                       c10::optional<at::Tensor> invstd,
                                       c10::optional<at::Tensor> weight,
-                                      c10::optional<at::Tensor> bias) {
+                                      c10::optional<at::Tensor> bias,
+                                      c10::optional<at::Tensor> dy2) {
   // mask modes: relu_y given -> read it; else invstd/weight/bias given ->
   // recompute mask (gamma*xhat+beta > 0), skipping the relu_y pass;
   // else no mask
@@ -223,6 +225,12 @@ std::vector<at::Tensor> bn_bwd_reduce(at::Tensor x, at::Tensor dy,
     wf = weight->to(at::kFloat).contiguous();
     bf = bias->to(at::kFloat).contiguous();
   }
+  TORCH_CHECK(!dy2.has_value() || mask == 1,
+              "bn_bwd_reduce: dy2 requires the stored-relu mask");
+  if (dy2.has_value())
+    TORCH_CHECK(dy2->is_contiguous() && dy2->sizes() == dy.sizes() &&
+                dy2->scalar_type() == dy.scalar_type(),
+                "bn_bwd_reduce: dy2 shape/dtype mismatch");
   pdta_bn_bwd_reduce(x.data_ptr(), dy.data_ptr(),
                      relu_y.has_value() ? relu_y->data_ptr() : nullptr,
                      mean.contiguous().data_ptr<float>(),
@@ -231,6 +239,7 @@ std::vector<at::Tensor> bn_bwd_reduce(at::Tensor x, at::Tensor dy,
                      mask == 2 ? bf.data_ptr<float>() : nullptr, mask,
                      part.data_ptr<float>(), sum_dy.data_ptr<float>(),
                      sum_dy_xmu.data_ptr<float>(), rows, C, dt_of(x),
+                     dy2.has_value() ? dy2->data_ptr() : nullptr,
                      cur_stream());
   return {sum_dy, sum_dy_xmu};
 }
@@ -255,6 +264,12 @@ std::vector<at::Tensor> bn_bwd_elemt(at::Tensor x, at::Tensor dy,
   if (mask == 2) bf = bias->to(at::kFloat).contiguous();
   TORCH_CHECK(!(want_dres && mask != 1),
               "bn_bwd_elemt: dres requires the stored-relu mask");
+  TORCH_CHECK(!dy2.has_value() || mask == 1,
+              "bn_bwd_elemt: dy2 requires the stored-relu mask");
+  if (dy2.has_value())
+    TORCH_CHECK(dy2->is_contiguous() && dy2->sizes() == dy.sizes() &&
+                dy2->scalar_type() == dy.scalar_type(),
+                "bn_bwd_elemt: dy2 shape/dtype mismatch");
   pdta_bn_bwd_elemt(x.data_ptr(), dy.data_ptr(),
                     relu_y.has_value() ? relu_y->data_ptr() : nullptr,
                     dx.data_ptr(), want_dres ? dres.data_ptr() : nullptr,
@@ -264,7 +279,9 @@ std::vector<at::Tensor> bn_bwd_elemt(at::Tensor x, at::Tensor dy,
                     invstd.contiguous().data_ptr<float>(),
                     sum_dy.contiguous().data_ptr<float>(),
                     sum_dy_xmu.contiguous().data_ptr<float>(), (float)count,
-                    rows, C, dt_of(x), cur_stream());
+                    rows, C, dt_of(x),
+                    dy2.has_value() ? dy2->data_ptr() : nullptr,
+                    cur_stream());
   if (want_dres) return {dx, dres};
   return {dx};
 }
@@ -826,12 +843,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_bwd_reduce", &bn_bwd_reduce, py::arg("x"), py::arg("dy"),
         py::arg("relu_y"), py::arg("mean"),
         py::arg("invstd") = c10::nullopt, py::arg("weight") = c10::nullopt,
-        py::arg("bias") = c10::nullopt);
+        py::arg("bias") = c10::nullopt, py::arg("dy2") = c10::nullopt);
   m.def("bn_bwd_elemt", &bn_bwd_elemt, py::arg("x"), py::arg("dy"),
         py::arg("relu_y"), py::arg("mean"), py::arg("invstd"),
         py::arg("weight"), py::arg("sum_dy"), py::arg("sum_dy_xmu"),
         py::arg("count"), py::arg("want_dres") = false,
-        py::arg("bias") = c10::nullopt);
+        py::arg("bias") = c10::nullopt, py::arg("dy2") = c10::nullopt);
   m.def("maxpool_fwd", &maxpool_fwd);
   m.def("maxpool_bwd", &maxpool_bwd);
   m.def("avgpool_global_fwd", &avgpool_global_fwd);

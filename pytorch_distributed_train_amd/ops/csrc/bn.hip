@@ -173,7 +173,7 @@ __global__ void bn_apply_k(const T* __restrict__ x, T* __restrict__ y,
 //    pre-activation includes the identity branch). MASK 2: recompute the
 //    mask as gamma*xhat+beta > 0 from x — one whole HBM pass over relu_y
 //    never happens (exact up to bf16 boundary rounding of y) -------------
-template <typename T, int MASK>
+template <typename T, int MASK, bool DY2 = false>
 __global__ void bn_bwd_part_k(const T* __restrict__ x,
                               const T* __restrict__ dy,
                               const T* __restrict__ relu_y,
@@ -182,7 +182,11 @@ __global__ void bn_bwd_part_k(const T* __restrict__ x,
                               const float* __restrict__ weight,
                               const float* __restrict__ bias,
                               float* __restrict__ part, long rows, int C,
-                              int rows_per_block) {
+                              int rows_per_block,
+                              const T* __restrict__ dy2 = nullptr) {
+  // DY2: second incoming gradient (residual fork) summed on load — the
+  // whole-tensor autograd add this replaces was 3 HBM passes; reading
+  // dy2 here and in the elemt kernel is 2
   __shared__ float red[2][TPB];
   int cpb = min(C, TPB);
   int rpar = TPB / cpb;
@@ -205,6 +209,12 @@ __global__ void bn_bwd_part_k(const T* __restrict__ x,
     float g1 = to_f32(dy[(r + rpar) * C + c]);
     float g2 = to_f32(dy[(r + 2 * rpar) * C + c]);
     float g3 = to_f32(dy[(r + 3 * rpar) * C + c]);
+    if constexpr (DY2) {
+      g0 += to_f32(dy2[r * C + c]);
+      g1 += to_f32(dy2[(r + rpar) * C + c]);
+      g2 += to_f32(dy2[(r + 2 * rpar) * C + c]);
+      g3 += to_f32(dy2[(r + 3 * rpar) * C + c]);
+    }
     float v0 = to_f32(x[r * C + c]);
     float v1 = to_f32(x[(r + rpar) * C + c]);
     float v2 = to_f32(x[(r + 2 * rpar) * C + c]);
@@ -226,6 +236,7 @@ __global__ void bn_bwd_part_k(const T* __restrict__ x,
   }
   for (; r < r_end; r += rpar) {
     float g = to_f32(dy[r * C + c]);
+    if constexpr (DY2) g += to_f32(dy2[r * C + c]);
     float v = to_f32(x[r * C + c]);
     if (MASK == 1) g = to_f32(relu_y[r * C + c]) > 0.f ? g : 0.f;
     if (MASK == 2) g = k1 * (v - m) + b0 > 0.f ? g : 0.f;
@@ -278,7 +289,7 @@ __global__ void bn_bwd_final_k(const float* __restrict__ part, int chunks,
 
 // -- backward elementwise (ReLU mask fused when RELU):
 //    dx = w*invstd * (dy - sum_dy/n - (x-mean)*invstd^2*sum_dy_xmu/n)
-template <typename T, int MASK, bool DRES>
+template <typename T, int MASK, bool DRES, bool DY2 = false>
 __global__ void bn_bwd_elemt_k(const T* __restrict__ x,
                                const T* __restrict__ dy,
                                const T* __restrict__ relu_y,
@@ -289,7 +300,8 @@ __global__ void bn_bwd_elemt_k(const T* __restrict__ x,
                                const float* __restrict__ invstd,
                                const float* __restrict__ sum_dy,
                                const float* __restrict__ sum_dy_xmu,
-                               float inv_n, long rows, int c8) {
+                               float inv_n, long rows, int c8,
+                               const T* __restrict__ dy2 = nullptr) {
   int cpg = min(c8, TPB);
   int rpar = TPB / cpg;
   int cg = blockIdx.x * cpg + (int)(threadIdx.x % cpg);
@@ -313,6 +325,11 @@ __global__ void bn_bwd_elemt_k(const T* __restrict__ x,
     long r2 = r + rstride;
     Vec8<T> ga = load8(dy + r * C + c0);
     Vec8<T> gb = load8(dy + r2 * C + c0);
+    Vec8<T> g2a, g2b;
+    if constexpr (DY2) {
+      g2a = load8(dy2 + r * C + c0);
+      g2b = load8(dy2 + r2 * C + c0);
+    }
     Vec8<T> va = load8(x + r * C + c0);
     Vec8<T> vb = load8(x + r2 * C + c0);
     Vec8<T> ya, yb, da, db;
@@ -323,6 +340,10 @@ __global__ void bn_bwd_elemt_k(const T* __restrict__ x,
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float g0 = to_f32(ga.v[j]), g1 = to_f32(gb.v[j]);
+      if constexpr (DY2) {
+        g0 += to_f32(g2a.v[j]);
+        g1 += to_f32(g2b.v[j]);
+      }
       float xm0 = to_f32(va.v[j]) - mu[j];
       float xm1 = to_f32(vb.v[j]) - mu[j];
       if (MASK == 1) {
@@ -349,12 +370,15 @@ __global__ void bn_bwd_elemt_k(const T* __restrict__ x,
   }
   for (; r < rows; r += rstride) {
     Vec8<T> g = load8(dy + r * C + c0);
+    Vec8<T> g2;
+    if constexpr (DY2) g2 = load8(dy2 + r * C + c0);
     Vec8<T> v = load8(x + r * C + c0);
     Vec8<T> yv, dr;
     if (MASK == 1) yv = load8(relu_y + r * C + c0);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float gg = to_f32(g.v[j]);
+      if constexpr (DY2) gg += to_f32(g2.v[j]);
       float xm = to_f32(v.v[j]) - mu[j];
       if (MASK == 1) gg = to_f32(yv.v[j]) > 0.f ? gg : 0.f;
       if (MASK == 2) gg = k1[j] * xm + b0[j] > 0.f ? gg : 0.f;
@@ -502,11 +526,16 @@ void pdta_bn_bwd_reduce(const void* x, const void* dy, const void* relu_y,
                         const float* weight, const float* bias, int mask,
                         float* part, float* sum_dy,
                         float* sum_dy_xmu, long rows, int C, int dt,
-                        hipStream_t s) {
+                        const void* dy2, hipStream_t s) {
   int rpb, chunks;
   dim3 g = reduce_grid(rows, C, &rpb, &chunks);
   PDTA_DISPATCH_DTYPE(dt, T, {
-    if (mask == 1)
+    if (mask == 1 && dy2)
+      // two-dy: residual-fork gradient summed on load
+      bn_bwd_part_k<T, 1, true><<<g, TPB, 0, s>>>(
+          (const T*)x, (const T*)dy, (const T*)relu_y, mean, nullptr,
+          nullptr, nullptr, part, rows, C, rpb, (const T*)dy2);
+    else if (mask == 1)
       bn_bwd_part_k<T, 1><<<g, TPB, 0, s>>>(
           (const T*)x, (const T*)dy, (const T*)relu_y, mean, nullptr,
           nullptr, nullptr, part, rows, C, rpb);
@@ -538,15 +567,25 @@ void pdta_bn_bwd_elemt(const void* x, const void* dy, const void* relu_y,
                        const float* mean, const float* invstd,
                        const float* sum_dy, const float* sum_dy_xmu,
                        float count, long rows, int C, int dt,
-                       hipStream_t s) {
+                       const void* dy2, hipStream_t s) {
   dim3 g = ew_grid2(rows, C);
   PDTA_DISPATCH_DTYPE(dt, T, {
     if (mask == 1) {
-      if (dres)
+      if (dres && dy2)
+        bn_bwd_elemt_k<T, 1, true, true><<<g, TPB, 0, s>>>(
+            (const T*)x, (const T*)dy, (const T*)relu_y, (T*)dx, (T*)dres,
+            weight, nullptr, mean, invstd, sum_dy, sum_dy_xmu, 1.f / count,
+            rows, C / 8, (const T*)dy2);
+      else if (dres)
         bn_bwd_elemt_k<T, 1, true><<<g, TPB, 0, s>>>(
             (const T*)x, (const T*)dy, (const T*)relu_y, (T*)dx, (T*)dres,
             weight, nullptr, mean, invstd, sum_dy, sum_dy_xmu, 1.f / count,
             rows, C / 8);
+      else if (dy2)
+        bn_bwd_elemt_k<T, 1, false, true><<<g, TPB, 0, s>>>(
+            (const T*)x, (const T*)dy, (const T*)relu_y, (T*)dx, nullptr,
+            weight, nullptr, mean, invstd, sum_dy, sum_dy_xmu, 1.f / count,
+            rows, C / 8, (const T*)dy2);
       else
         bn_bwd_elemt_k<T, 1, false><<<g, TPB, 0, s>>>(
             (const T*)x, (const T*)dy, (const T*)relu_y, (T*)dx, nullptr,

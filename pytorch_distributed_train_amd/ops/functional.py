@@ -106,9 +106,9 @@ class Conv2dNHWC(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, wt, w2, stride: int, pad: int,
-                want_bn_stats: bool = False, dx_sink=None, dx_accum=None):
+                want_bn_stats: bool = False, dx_sink=None):
         ctx.stride, ctx.pad = stride, pad
-        ctx.dx_sink, ctx.dx_accum = dx_sink, dx_accum
+        ctx.dx_sink = dx_sink
         if x.is_cuda:
             # stem (C % 64 != 0): cache the im2col image for bwd-weight
             want_xcol = x.shape[3] % 64 != 0 and torch.is_grad_enabled()
@@ -137,16 +137,12 @@ class Conv2dNHWC(torch.autograd.Function):
         dx = dw = None
         if x.is_cuda:
             if need_dx:
-                accum = None
-                if ctx.dx_accum is not None and ctx.dx_accum.t is not None:
-                    accum = ctx.dx_accum.t
-                    ctx.dx_accum.t = None  # consumed (folded into dx)
                 dx = ext().conv2d_bwd_data(dy, w2, x.shape[1], x.shape[2],
-                                           stride, pad, accum=accum)
+                                           stride, pad)
                 if ctx.dx_sink is not None:
-                    # residual-path producer: hand dx to the consumer's
-                    # epilogue and contribute NOTHING via autograd (the
-                    # fold is the only path — no fork add exists)
+                    # residual-path producer (GradHolder protocol): the
+                    # previous block's BN backward reads this as its
+                    # second dy; autograd sees None — no fork add exists
                     ctx.dx_sink.t = dx
                     dx = None
             if need_dw:
@@ -164,14 +160,14 @@ class Conv2dNHWC(torch.autograd.Function):
             if need_dw:
                 dw = torch.nn.grad.conv2d_weight(x_nchw, w.shape, dy_nchw,
                                                  stride=stride, padding=pad)
-        return dx, dw, None, None, None, None, None, None, None
+        return dx, dw, None, None, None, None, None, None
 
 
 def conv2d_nhwc(x, weight, wt, w2, stride: int = 1, pad: int = 0,
-                want_bn_stats: bool = False, dx_sink=None, dx_accum=None):
-    if want_bn_stats or dx_sink is not None or dx_accum is not None:
+                want_bn_stats: bool = False, dx_sink=None):
+    if want_bn_stats or dx_sink is not None:
         return Conv2dNHWC.apply(x, weight, wt, w2, stride, pad,
-                                want_bn_stats, dx_sink, dx_accum)
+                                want_bn_stats, dx_sink)
     return Conv2dNHWC.apply(x, weight, wt, w2, stride, pad)
 
 
@@ -198,7 +194,7 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, mean, invstd, fuse_relu: bool,
                 comm=None, count_override=None, residual=None,
-                dres_sink=None):
+                dres_sink=None, dy2_source=None):
         # mean/invstd are per-channel fp32, computed by the module (local
         # stats for BN, cross-rank combined for SyncBN). For SyncBN, `comm`
         # all-reduces the backward sums and `count_override` is the GLOBAL
@@ -222,6 +218,7 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
         ctx.count_override = count_override
         ctx.has_residual = residual is not None
         ctx.dres_sink = dres_sink
+        ctx.dy2_source = dy2_source
         ctx.save_for_backward(x, weight, bias, mean, invstd, y)
         return y
 
@@ -230,6 +227,12 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
         x, weight, bias, mean, invstd, y = ctx.saved_tensors
         fuse_relu = ctx.fuse_relu
         dy = dy.contiguous()
+        dy2 = None
+        if ctx.dy2_source is not None and ctx.dy2_source.t is not None:
+            # second incoming gradient (next block's residual path),
+            # summed in-kernel (two-dy backward)
+            dy2 = ctx.dy2_source.t
+            ctx.dy2_source.t = None
         n = x.shape[0] * x.shape[1] * x.shape[2]
         if ctx.count_override is not None:
             n = ctx.count_override
@@ -245,7 +248,7 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
                 x, dy, relu_y, mean,
                 invstd=invstd if premask else None,
                 weight=weight if premask else None,
-                bias=bias if premask else None)
+                bias=bias if premask else None, dy2=dy2)
             # param grads from LOCAL sums (DDP averages them afterwards,
             # matching torch SyncBN's split of responsibilities)
             dweight = sum_dy_xmu * invstd
@@ -268,7 +271,8 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
                                      sum_dy.contiguous(),
                                      sum_dy_xmu.contiguous(), float(n),
                                      ctx.has_residual,
-                                     bias=bias if premask else None)
+                                     bias=bias if premask else None,
+                                     dy2=dy2)
             dx = out[0]
             if ctx.has_residual:
                 dres = out[1]
@@ -279,6 +283,8 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
                     dres = None
         else:
             dy32 = dy.float()
+            if dy2 is not None:
+                dy32 = dy32 + dy2.float()
             if fuse_relu:
                 dy32 = dy32 * (y.float() > 0)
             x32 = x.float()
@@ -303,7 +309,7 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
             dx = (dy32 - sum_dy / n - xmu * invstd * invstd * sum_dy_xmu / n)
             dx = (dx * w32 * invstd).to(x.dtype)
         return (dx, dweight, dbias, None, None, None, None, None,
-                dres, None)
+                dres, None, None)
 
 
 def bn_stats(x):
@@ -333,9 +339,10 @@ def bn_stats_train_from_part(part, rows, running_mean, running_var,
 
 
 def bn_train(x, weight, bias, mean, invstd, fuse_relu=False, comm=None,
-             count=None, residual=None, dres_sink=None):
+             count=None, residual=None, dres_sink=None, dy2_source=None):
     return BatchNorm2dNHWCFn.apply(x, weight, bias, mean, invstd, fuse_relu,
-                                   comm, count, residual, dres_sink)
+                                   comm, count, residual, dres_sink,
+                                   dy2_source)
 
 
 def bn_infer(x, weight, bias, running_mean, running_var, eps, fuse_relu=False,

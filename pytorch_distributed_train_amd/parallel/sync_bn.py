@@ -60,6 +60,8 @@ def convert_sync_batchnorm(module: torch.nn.Module, comm):
         out.comm = comm
         if hasattr(module, "_dres_sink"):
             out._dres_sink = module._dres_sink
+        if hasattr(module, "_dy2_source"):
+            out._dy2_source = module._dy2_source
         out.weight = module.weight
         out.bias = module.bias
         out.running_mean = module.running_mean
