@@ -250,7 +250,8 @@ std::vector<at::Tensor> bn_bwd_elemt(at::Tensor x, at::Tensor dy,
                                      at::Tensor weight, at::Tensor sum_dy,
                                      at::Tensor sum_dy_xmu, double count,
                                      bool want_dres,
-                                     c10::optional<at::Tensor> bias) {
+                                     c10::optional<at::Tensor> bias,
+                                     c10::optional<at::Tensor> dy2) {
   CHECK_IN(x);
   CHECK_IN(dy);
   int C = x.size(3);
