@@ -209,16 +209,24 @@ __global__ void bn_bwd_part_k(const T* __restrict__ x,
     float g1 = to_f32(dy[(r + rpar) * C + c]);
     float g2 = to_f32(dy[(r + 2 * rpar) * C + c]);
     float g3 = to_f32(dy[(r + 3 * rpar) * C + c]);
+    float h0, h1, h2, h3;
     if constexpr (DY2) {
-      g0 += to_f32(dy2[r * C + c]);
-      g1 += to_f32(dy2[(r + rpar) * C + c]);
-      g2 += to_f32(dy2[(r + 2 * rpar) * C + c]);
-      g3 += to_f32(dy2[(r + 3 * rpar) * C + c]);
+      // issue all four dy2 loads before any arithmetic consumes them
+      h0 = to_f32(dy2[r * C + c]);
+      h1 = to_f32(dy2[(r + rpar) * C + c]);
+      h2 = to_f32(dy2[(r + 2 * rpar) * C + c]);
+      h3 = to_f32(dy2[(r + 3 * rpar) * C + c]);
     }
     float v0 = to_f32(x[r * C + c]);
     float v1 = to_f32(x[(r + rpar) * C + c]);
     float v2 = to_f32(x[(r + 2 * rpar) * C + c]);
     float v3 = to_f32(x[(r + 3 * rpar) * C + c]);
+    if constexpr (DY2) {
+      g0 += h0;
+      g1 += h1;
+      g2 += h2;
+      g3 += h3;
+    }
     if (MASK == 1) {
       g0 = to_f32(relu_y[r * C + c]) > 0.f ? g0 : 0.f;
       g1 = to_f32(relu_y[(r + rpar) * C + c]) > 0.f ? g1 : 0.f;
