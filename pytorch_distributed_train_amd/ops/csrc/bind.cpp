@@ -24,7 +24,7 @@ void pdta_bn_apply(const void*, void*, const void*, const float*,
 void pdta_bn_bwd_reduce(const void*, const void*, const void*,
                         const float*, const float*, const float*,
                         const float*, int, float*, float*, float*, long,
-                        int, int, const void*, hipStream_t);
+                        int, int, const void*, void*, hipStream_t);
 void pdta_bn_bwd_elemt(const void*, const void*, const void*, void*,
                        void*, const float*, const float*, int,
                        const float*, const float*, const float*,
@@ -219,6 +219,8 @@ std::vector<at::Tensor> bn_bwd_reduce(at::Tensor x, at::Tensor dy,
   auto part = at::empty({(long)chunks * 2 * C}, opts);
   auto sum_dy = at::empty({C}, opts), sum_dy_xmu = at::empty({C}, opts);
   int mask = relu_y.has_value() ? 1 : (bias.has_value() ? 2 : 0);
+  at::Tensor dres;  // two-dy: masked combined dy, written by the kernel
+  if (dy2.has_value()) dres = at::empty_like(dy);
   at::Tensor isf, wf, bf;
   if (mask == 2) {
     isf = invstd->to(at::kFloat).contiguous();
@@ -240,7 +242,9 @@ std::vector<at::Tensor> bn_bwd_reduce(at::Tensor x, at::Tensor dy,
                      part.data_ptr<float>(), sum_dy.data_ptr<float>(),
                      sum_dy_xmu.data_ptr<float>(), rows, C, dt_of(x),
                      dy2.has_value() ? dy2->data_ptr() : nullptr,
+                     dy2.has_value() ? dres.data_ptr() : nullptr,
                      cur_stream());
+  if (dy2.has_value()) return {sum_dy, sum_dy_xmu, dres};
   return {sum_dy, sum_dy_xmu};
 }
 

@@ -244,11 +244,18 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
             # re-reads y (one whole HBM pass saved per kernel)
             relu_y = y if (fuse_relu and ctx.has_residual) else None
             premask = fuse_relu and not ctx.has_residual
-            sum_dy, sum_dy_xmu = ext().bn_bwd_reduce(
+            red = ext().bn_bwd_reduce(
                 x, dy, relu_y, mean,
                 invstd=invstd if premask else None,
                 weight=weight if premask else None,
                 bias=bias if premask else None, dy2=dy2)
+            sum_dy, sum_dy_xmu = red[0], red[1]
+            dy_comb = None
+            if dy2 is not None:
+                # two-dy: the reduce pass wrote the masked combined dy;
+                # the elemt pass now runs mask-free on that one tensor,
+                # and it IS the residual-path gradient (dres)
+                dy_comb = red[2]
             # param grads from LOCAL sums (DDP averages them afterwards,
             # matching torch SyncBN's split of responsibilities)
             dweight = sum_dy_xmu * invstd
@@ -267,20 +274,28 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
                 packed[:2 * c] /= packed[2 * c]
                 sum_dy, sum_dy_xmu = packed[:c], packed[c:2 * c]
                 n = 1.0
-            out = ext().bn_bwd_elemt(x, dy, relu_y, mean, invstd, weight,
-                                     sum_dy.contiguous(),
-                                     sum_dy_xmu.contiguous(), float(n),
-                                     ctx.has_residual,
-                                     bias=bias if premask else None,
-                                     dy2=dy2)
-            dx = out[0]
-            if ctx.has_residual:
-                dres = out[1]
-                if ctx.dres_sink is not None:
-                    # residual-path producer (GradHolder protocol): the
-                    # consumer conv1 folds this; autograd sees None
-                    ctx.dres_sink.t = dres
-                    dres = None
+            if dy_comb is not None:
+                out = ext().bn_bwd_elemt(x, dy_comb, None, mean, invstd,
+                                         weight, sum_dy.contiguous(),
+                                         sum_dy_xmu.contiguous(), float(n),
+                                         False)
+                dx = out[0]
+                dres = dy_comb if ctx.has_residual else None
+            else:
+                out = ext().bn_bwd_elemt(x, dy, relu_y, mean, invstd,
+                                         weight, sum_dy.contiguous(),
+                                         sum_dy_xmu.contiguous(), float(n),
+                                         ctx.has_residual,
+                                         bias=bias if premask else None)
+                dx = out[0]
+                if ctx.has_residual:
+                    dres = out[1]
+            if dres is not None and ctx.dres_sink is not None:
+                # residual-path producer (GradHolder protocol): the
+                # consumer BN sums this as its second dy; autograd
+                # sees None
+                ctx.dres_sink.t = dres
+                dres = None
         else:
             dy32 = dy.float()
             if dy2 is not None:
