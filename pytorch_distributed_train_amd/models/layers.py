@@ -53,16 +53,18 @@ class Conv2dNHWC(nn.Module):
         fuse_grads = x.is_cuda and self.training \
             and torch.is_grad_enabled() and OF.fork_fuse_enabled()
         sink = getattr(self, "_dx_sink", None) if fuse_grads else None
+        gslot = (getattr(self, "_gslot_w", None)
+                 if x.is_cuda and self.training else None)
         if (getattr(self, "emit_bn_stats", False) and self.training
                 and x.is_cuda and self.bias is None
                 and torch.is_grad_enabled()):
             y, part = OF.conv2d_nhwc(x, w, wt, w2, self.stride,
                                      self.padding, want_bn_stats=True,
-                                     dx_sink=sink)
+                                     dx_sink=sink, gslot=gslot)
             y._pdta_bn_part = part
             return y
         y = OF.conv2d_nhwc(x, w, wt, w2, self.stride, self.padding,
-                           dx_sink=sink)
+                           dx_sink=sink, gslot=gslot)
         if self.bias is not None:
             y = y + self.bias.to(y.dtype)  # broadcast over trailing C
         return y
@@ -119,8 +121,9 @@ class BatchNorm2dNHWC(nn.Module):
                     mean, invstd = OF.bn_stats_train(
                         x, self.running_mean, self.running_var,
                         self.momentum, self.eps)
-                with torch.no_grad():
-                    self.num_batches_tracked += 1
+                if not getattr(self, "_nbt_batched", False):
+                    with torch.no_grad():
+                        self.num_batches_tracked += 1
                 fuse = OF.fork_fuse_enabled()
                 return OF.bn_train(x, self.weight, self.bias, mean, invstd,
                                    self.fuse_relu, residual=residual,
@@ -129,7 +132,9 @@ class BatchNorm2dNHWC(nn.Module):
                                               if fuse else None),
                                    dy2_source=(getattr(self, "_dy2_source",
                                                        None)
-                                               if fuse else None))
+                                               if fuse else None),
+                                   gslot_w=getattr(self, "_gslot_w", None),
+                                   gslot_b=getattr(self, "_gslot_b", None))
             mean, var, count = self._compute_stats(x)
             with torch.no_grad():
                 self.num_batches_tracked += 1
@@ -154,7 +159,11 @@ class BatchNorm2dNHWC(nn.Module):
                                                    None)
                                            if x.is_cuda and
                                            OF.fork_fuse_enabled()
-                                           else None))
+                                           else None),
+                               gslot_w=(getattr(self, "_gslot_w", None)
+                                        if x.is_cuda else None),
+                               gslot_b=(getattr(self, "_gslot_b", None)
+                                        if x.is_cuda else None))
         return OF.bn_infer(x, self.weight, self.bias, self.running_mean,
                            self.running_var, self.eps, self.fuse_relu,
                            residual)
@@ -184,7 +193,12 @@ class Linear(nn.Module):
         if packed is None or packed.dtype != x.dtype:
             packed = self.weight.to(x.dtype)  # [N,K] = B^T (MFMA layout)
             packed_t = None
-        return OF.linear(x, self.weight, packed, self.bias, packed_t)
+        on_direct = x.is_cuda and self.training
+        return OF.linear(x, self.weight, packed, self.bias, packed_t,
+                         gslot_w=(getattr(self, "_gslot_w", None)
+                                  if on_direct else None),
+                         gslot_b=(getattr(self, "_gslot_b", None)
+                                  if on_direct else None))
 
     def extra_repr(self):
         return (f"in_features={self.in_features}, "

@@ -34,7 +34,12 @@ class LayerNorm(nn.Module):
         self.bias = nn.Parameter(torch.zeros(dim))
 
     def forward(self, x):
-        return OF.layer_norm(x, self.weight, self.bias, self.eps)
+        on_direct = x.is_cuda and self.training
+        return OF.layer_norm(x, self.weight, self.bias, self.eps,
+                             gslot_w=(getattr(self, "_gslot_w", None)
+                                      if on_direct else None),
+                             gslot_b=(getattr(self, "_gslot_b", None)
+                                      if on_direct else None))
 
 
 class MultiheadSelfAttention(nn.Module):

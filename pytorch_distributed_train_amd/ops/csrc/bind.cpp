@@ -530,13 +530,24 @@ at::Tensor conv2d_bwd_data(at::Tensor dy, at::Tensor w2, int64_t H,
 
 at::Tensor conv2d_bwd_weight(at::Tensor x, at::Tensor dy, int64_t R,
                              int64_t S, int64_t stride, int64_t pad,
-                             c10::optional<at::Tensor> xcol_cached) {
+                             c10::optional<at::Tensor> xcol_cached,
+                             c10::optional<at::Tensor> out) {
   CHECK_IN(x);
   CHECK_IN(dy);
   int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
   int P = dy.size(1), Q = dy.size(2), Kout = dy.size(3);
+  if (out.has_value())
+    TORCH_CHECK(out->is_contiguous() &&
+                out->scalar_type() == at::kFloat &&
+                out->numel() == (long)Kout * C * R * S,
+                "conv2d_bwd_weight: bad out (direct-grad view)");
   if (C % 64 == 0) {
-    auto dw = at::empty({Kout, C, R, S}, x.options().dtype(at::kFloat));
+    // direct-grad: the reduce writes every element exactly once, so the
+    // flat-gradient slice can be the destination (no AccumulateGrad add)
+    auto dw = out.has_value()
+                  ? out->view({Kout, C, R, S})
+                  : at::empty({Kout, C, R, S},
+                              x.options().dtype(at::kFloat));
     long tiles, chunks;
     pdta_wrw_plan(N, P, Q, Kout, C, R, S, &tiles, &chunks);
     auto part = at::empty({tiles * (chunks + 16) * 4096},
@@ -568,10 +579,13 @@ at::Tensor conv2d_bwd_weight(at::Tensor x, at::Tensor dy, int64_t R,
                        dwcol.data_ptr<float>(), part.data_ptr<float>(),
                        zero_page(x), N, P, Q, rsc_pad, P, Q, Kout, 1, 1, 1,
                        0, dt_of(x), cur_stream());
-  return dwcol.slice(1, 0, rsc)
-      .view({Kout, R, S, C})
-      .permute({0, 3, 1, 2})
-      .contiguous();
+  auto packed = dwcol.slice(1, 0, rsc).view({Kout, R, S, C});
+  if (out.has_value()) {
+    out->view({Kout, C, R, S})
+        .copy_(packed.permute({0, 3, 1, 2}));
+    return *out;
+  }
+  return packed.permute({0, 3, 1, 2}).contiguous();
 }
 
 // ---- transformer ops (ViT path) -----------------------------------------
@@ -880,7 +894,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_bwd_weight", &conv2d_bwd_weight, py::arg("x"),
         py::arg("dy"), py::arg("R"), py::arg("S"),
         py::arg("stride"), py::arg("pad"),
-        py::arg("xcol_cached") = c10::nullopt);
+        py::arg("xcol_cached") = c10::nullopt,
+        py::arg("out") = c10::nullopt);
   m.def("gemm_nt_batched", &gemm_nt_batched, py::arg("a"), py::arg("b"),
         py::arg("out_f32") = false);
   m.def("ln_fwd", &ln_fwd);

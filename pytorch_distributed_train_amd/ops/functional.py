@@ -106,9 +106,10 @@ class Conv2dNHWC(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, wt, w2, stride: int, pad: int,
-                want_bn_stats: bool = False, dx_sink=None):
+                want_bn_stats: bool = False, dx_sink=None, gslot=None):
         ctx.stride, ctx.pad = stride, pad
         ctx.dx_sink = dx_sink
+        ctx.gslot = gslot  # (grad_view, notify) direct-grad destination
         if x.is_cuda:
             # stem (C % 64 != 0): cache the im2col image for bwd-weight
             want_xcol = x.shape[3] % 64 != 0 and torch.is_grad_enabled()
@@ -146,9 +147,21 @@ class Conv2dNHWC(torch.autograd.Function):
                     ctx.dx_sink.t = dx
                     dx = None
             if need_dw:
-                dw = ext().conv2d_bwd_weight(x, dy, weight.shape[2],
-                                             weight.shape[3], stride, pad,
-                                             xcol_cached=xcol)
+                if ctx.gslot is not None:
+                    # direct-grad: write straight into the flat slice and
+                    # return None (no AccumulateGrad add); tell the
+                    # reducer the param arrived
+                    view, notify = ctx.gslot
+                    ext().conv2d_bwd_weight(x, dy, weight.shape[2],
+                                            weight.shape[3], stride, pad,
+                                            xcol_cached=xcol, out=view)
+                    if notify is not None:
+                        notify()
+                    dw = None
+                else:
+                    dw = ext().conv2d_bwd_weight(x, dy, weight.shape[2],
+                                                 weight.shape[3], stride,
+                                                 pad, xcol_cached=xcol)
         else:
             x_nchw = _nhwc_to_nchw(x).float()
             dy_nchw = _nhwc_to_nchw(dy).float()
@@ -160,14 +173,14 @@ class Conv2dNHWC(torch.autograd.Function):
             if need_dw:
                 dw = torch.nn.grad.conv2d_weight(x_nchw, w.shape, dy_nchw,
                                                  stride=stride, padding=pad)
-        return dx, dw, None, None, None, None, None, None
+        return dx, dw, None, None, None, None, None, None, None
 
 
 def conv2d_nhwc(x, weight, wt, w2, stride: int = 1, pad: int = 0,
-                want_bn_stats: bool = False, dx_sink=None):
-    if want_bn_stats or dx_sink is not None:
+                want_bn_stats: bool = False, dx_sink=None, gslot=None):
+    if want_bn_stats or dx_sink is not None or gslot is not None:
         return Conv2dNHWC.apply(x, weight, wt, w2, stride, pad,
-                                want_bn_stats, dx_sink)
+                                want_bn_stats, dx_sink, gslot)
     return Conv2dNHWC.apply(x, weight, wt, w2, stride, pad)
 
 
@@ -194,7 +207,8 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, mean, invstd, fuse_relu: bool,
                 comm=None, count_override=None, residual=None,
-                dres_sink=None, dy2_source=None):
+                dres_sink=None, dy2_source=None, gslot_w=None,
+                gslot_b=None):
         # mean/invstd are per-channel fp32, computed by the module (local
         # stats for BN, cross-rank combined for SyncBN). For SyncBN, `comm`
         # all-reduces the backward sums and `count_override` is the GLOBAL
@@ -219,6 +233,7 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
         ctx.has_residual = residual is not None
         ctx.dres_sink = dres_sink
         ctx.dy2_source = dy2_source
+        ctx.gslot_w, ctx.gslot_b = gslot_w, gslot_b
         ctx.save_for_backward(x, weight, bias, mean, invstd, y)
         return y
 
@@ -258,8 +273,20 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
                 dy_comb = red[2]
             # param grads from LOCAL sums (DDP averages them afterwards,
             # matching torch SyncBN's split of responsibilities)
-            dweight = sum_dy_xmu * invstd
-            dbias = sum_dy.clone() if ctx.comm is not None else sum_dy
+            if ctx.gslot_w is not None and x.is_cuda:
+                torch.mul(sum_dy_xmu, invstd, out=ctx.gslot_w[0])
+                if ctx.gslot_w[1] is not None:
+                    ctx.gslot_w[1]()
+                dweight = None
+            else:
+                dweight = sum_dy_xmu * invstd
+            if ctx.gslot_b is not None and x.is_cuda:
+                ctx.gslot_b[0].copy_(sum_dy)
+                if ctx.gslot_b[1] is not None:
+                    ctx.gslot_b[1]()
+                dbias = None
+            else:
+                dbias = sum_dy.clone() if ctx.comm is not None else sum_dy
             if ctx.comm is not None:
                 # [sum_dy ‖ sum_dy_xmu ‖ local_count] in ONE all-reduce;
                 # normalizing by the reduced count on-device keeps the
@@ -324,7 +351,7 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
             dx = (dy32 - sum_dy / n - xmu * invstd * invstd * sum_dy_xmu / n)
             dx = (dx * w32 * invstd).to(x.dtype)
         return (dx, dweight, dbias, None, None, None, None, None,
-                dres, None, None)
+                dres, None, None, None, None)
 
 
 def bn_stats(x):
@@ -354,10 +381,11 @@ def bn_stats_train_from_part(part, rows, running_mean, running_var,
 
 
 def bn_train(x, weight, bias, mean, invstd, fuse_relu=False, comm=None,
-             count=None, residual=None, dres_sink=None, dy2_source=None):
+             count=None, residual=None, dres_sink=None, dy2_source=None,
+             gslot_w=None, gslot_b=None):
     return BatchNorm2dNHWCFn.apply(x, weight, bias, mean, invstd, fuse_relu,
                                    comm, count, residual, dres_sink,
-                                   dy2_source)
+                                   dy2_source, gslot_w, gslot_b)
 
 
 def bn_infer(x, weight, bias, running_mean, running_var, eps, fuse_relu=False,
@@ -505,12 +533,14 @@ class LinearFn(torch.autograd.Function):
     """y = x @ W^T + b; x [M,K] compute dtype, W master fp32 [N,K]."""
 
     @staticmethod
-    def forward(ctx, x, weight, w_packed, bias, w_packed_t=None):
+    def forward(ctx, x, weight, w_packed, bias, w_packed_t=None,
+                gslot_w=None, gslot_b=None):
         ctx.save_for_backward(x, weight, w_packed,
                               w_packed_t if w_packed_t is not None
                               else w_packed)
         ctx.has_t = w_packed_t is not None
         ctx.has_bias = bias is not None
+        ctx.gslot_w, ctx.gslot_b = gslot_w, gslot_b
         if x.is_cuda:
             return ext().gemm_nt(x, w_packed, bias)
         y = x.float() @ weight.float().t()
@@ -534,19 +564,34 @@ class LinearFn(torch.autograd.Function):
             if ctx.needs_input_grad[1]:
                 o_dim, i_dim = weight.shape
                 rows = x.shape[0]
+                w_out = ctx.gslot_w[0] if ctx.gslot_w is not None else None
                 if o_dim % 64 == 0 and i_dim % 64 == 0:
                     # dW = dy^T @ x is exactly the conv-wrw TN GEMM: reuse
                     # its split-K slab kernel (a plain-nt path serialized
                     # K=rows in a handful of blocks — 358us per MLP grad)
                     dw = ext().conv2d_bwd_weight(
                         x.view(1, rows, 1, i_dim), dy.view(1, rows, 1, o_dim),
-                        1, 1, 1, 0).view(o_dim, i_dim)
+                        1, 1, 1, 0,
+                        out=(w_out.view(o_dim, i_dim, 1, 1)
+                             if w_out is not None else None)
+                    ).view(o_dim, i_dim)
                 else:
                     dyt = _pad_cols64(dy.t().contiguous())    # [O, M]
                     xt = _pad_cols64(x.t().contiguous())      # [K_in, M]
                     dw = ext().gemm_nt(dyt, xt, None, True)   # fp32 [O, K_in]
+                    if w_out is not None:
+                        w_out.copy_(dw)
+                if w_out is not None:
+                    if ctx.gslot_w[1] is not None:
+                        ctx.gslot_w[1]()
+                    dw = None
             if ctx.has_bias and ctx.needs_input_grad[3]:
                 db = ext().colsum_f32(dy)                     # [M,N] -> [N]
+                if ctx.gslot_b is not None:
+                    ctx.gslot_b[0].copy_(db)
+                    if ctx.gslot_b[1] is not None:
+                        ctx.gslot_b[1]()
+                    db = None
         else:
             dy32, x32, w32 = dy.float(), x.float(), weight.float()
             if ctx.needs_input_grad[0]:
@@ -555,11 +600,13 @@ class LinearFn(torch.autograd.Function):
                 dw = dy32.t() @ x32
             if ctx.has_bias and ctx.needs_input_grad[3]:
                 db = dy32.sum(dim=0)
-        return dx, dw, None, db, None
+        return dx, dw, None, db, None, None, None
 
 
-def linear(x, weight, w_packed, bias=None, w_packed_t=None):
-    return LinearFn.apply(x, weight, w_packed, bias, w_packed_t)
+def linear(x, weight, w_packed, bias=None, w_packed_t=None, gslot_w=None,
+           gslot_b=None):
+    return LinearFn.apply(x, weight, w_packed, bias, w_packed_t, gslot_w,
+                          gslot_b)
 
 
 # ---------------------------------------------------------------------------
@@ -619,7 +666,9 @@ def softmax_cross_entropy(logits, labels, label_smoothing: float = 0.0):
 
 class LayerNormFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, bias, eps: float):
+    def forward(ctx, x, weight, bias, eps: float, gslot_w=None,
+                gslot_b=None):
+        ctx.gslot_w, ctx.gslot_b = gslot_w, gslot_b
         if x.is_cuda:
             y, mean, invstd = ext().ln_fwd(x, weight, bias, eps)
         else:
@@ -638,6 +687,16 @@ class LayerNormFn(torch.autograd.Function):
         dy = dy.contiguous()
         if x.is_cuda:
             dx, dw, db = ext().ln_bwd(x, dy, weight, mean, invstd)
+            if ctx.gslot_w is not None:
+                ctx.gslot_w[0].copy_(dw)
+                if ctx.gslot_w[1] is not None:
+                    ctx.gslot_w[1]()
+                dw = None
+            if ctx.gslot_b is not None:
+                ctx.gslot_b[0].copy_(db)
+                if ctx.gslot_b[1] is not None:
+                    ctx.gslot_b[1]()
+                db = None
         else:
             d = x.shape[-1]
             x32, g32, w32 = x.float(), dy.float(), weight.float()
@@ -649,11 +708,12 @@ class LayerNormFn(torch.autograd.Function):
             dims = tuple(range(x.dim() - 1))
             dw = (g32 * xhat).sum(dim=dims)
             db = g32.sum(dim=dims)
-        return dx, dw, db, None
+        return dx, dw, db, None, None, None
 
 
-def layer_norm(x, weight, bias, eps: float = 1e-6):
-    return LayerNormFn.apply(x, weight, bias, eps)
+def layer_norm(x, weight, bias, eps: float = 1e-6, gslot_w=None,
+               gslot_b=None):
+    return LayerNormFn.apply(x, weight, bias, eps, gslot_w, gslot_b)
 
 
 class GELUFn(torch.autograd.Function):

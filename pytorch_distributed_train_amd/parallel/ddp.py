@@ -117,15 +117,22 @@ class GradReducer:
         self._next_launch = 0
         self._iteration_open = True
 
+    def param_ready(self, param_id: int) -> None:
+        """Mark one param's gradient as produced this iteration. Called
+        by the post-accumulate hook, or DIRECTLY by a Function backward
+        that wrote its gradient straight into the flat slice (direct-grad
+        path — no AccumulateGrad node ever runs for those params)."""
+        if not self._iteration_open:
+            return
+        b = self._param_bucket[param_id]
+        b.pending -= 1
+        if b.pending == 0 and self.overlap:
+            b.ready = True
+            self._launch_ready()
+
     def _make_hook(self, param_id: int):
         def hook(_param):
-            if not self._iteration_open:
-                return
-            b = self._param_bucket[param_id]
-            b.pending -= 1
-            if b.pending == 0 and self.overlap:
-                b.ready = True
-                self._launch_ready()
+            self.param_ready(param_id)
         return hook
 
     def _launch_ready(self) -> None:

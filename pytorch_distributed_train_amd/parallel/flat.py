@@ -63,6 +63,7 @@ class FlatParams:
             p.grad = self.grad[off:off + n].view(shape)
 
         self.num_params = len(params)
+        self.index_by_param = {id(p): i for i, p in enumerate(params)}
 
     def grad_view(self, i: int) -> torch.Tensor:
         off, n = self.offsets[i], self.sizes[i]
