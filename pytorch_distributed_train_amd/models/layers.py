@@ -137,7 +137,8 @@ class BatchNorm2dNHWC(nn.Module):
                                    gslot_b=getattr(self, "_gslot_b", None))
             mean, var, count = self._compute_stats(x)
             with torch.no_grad():
-                self.num_batches_tracked += 1
+                if not getattr(self, "_nbt_batched", False):
+                    self.num_batches_tracked += 1
                 m = self.momentum
                 # running_var uses the unbiased estimator, as torch BN does
                 # (count may be a device scalar on the SyncBN path)
