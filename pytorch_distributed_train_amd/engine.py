@@ -84,7 +84,8 @@ class Trainer:
                                    overlap=not args.no_overlap,
                                    comm_dtype=comm_dtype)
         self.reducer.broadcast_params(src=0)
-        if self.device.type == "cuda":
+        if self.device.type == "cuda" \
+                and not os.environ.get("PDTA_NO_DIRECT_GRAD"):
             self._wire_direct_grads()
         self.optimizer = FusedSGD(self.flat, lr=args.lr,
                                   momentum=args.momentum,

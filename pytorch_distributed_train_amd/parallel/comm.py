@@ -149,6 +149,8 @@ class StagedGlooComm(TorchDistComm):
     def _staged(self, t, run):
         if not t.is_cuda:
             return run(t)
+        if os.environ.get("PDTA_STAGED_SYNC"):
+            torch.cuda.synchronize()
         host = t.cpu()  # sync copy: grads on the caller's stream are done
         w = run(host)
         return _StagedWork(w, lambda: t.copy_(host))
