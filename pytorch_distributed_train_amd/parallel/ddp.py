@@ -115,6 +115,7 @@ class GradReducer:
             b.ready = False
             b.work = None
         self._next_launch = 0
+        self._seen = set()
         self._iteration_open = True
 
     def param_ready(self, param_id: int) -> None:
@@ -124,6 +125,13 @@ class GradReducer:
         path — no AccumulateGrad node ever runs for those params)."""
         if not self._iteration_open:
             return
+        if param_id in self._seen:
+            raise RuntimeError(
+                f"gradient for param {param_id} "
+                f"({self.flat.names[param_id]}) arrived twice in one "
+                "iteration — a bucket would all-reduce before its grads "
+                "are complete (direct-grad double notify?)")
+        self._seen.add(param_id)
         b = self._param_bucket[param_id]
         b.pending -= 1
         if b.pending == 0 and self.overlap:
