@@ -125,6 +125,7 @@ class Trainer:
             notify = None
             if self.reducer.enabled:
                 notify = (lambda idx=i: self.reducer.param_ready(idx))
+                self.reducer.take_over_arrival(i)
             return (view, notify)
 
         for m in self.model.modules():

@@ -97,11 +97,11 @@ class GradReducer:
             self._shadow = torch.zeros(flat.total, dtype=comm_dtype,
                                        device=flat.device)
 
-        self._hooks = []
+        self._hooks = {}
         if self.enabled:
             for i, p in enumerate(flat.params):
-                self._hooks.append(p.register_post_accumulate_grad_hook(
-                    self._make_hook(i)))
+                self._hooks[i] = p.register_post_accumulate_grad_hook(
+                    self._make_hook(i))
         self._iteration_open = False
         self._next_launch = 0
 
@@ -142,6 +142,15 @@ class GradReducer:
         def hook(_param):
             self.param_ready(param_id)
         return hook
+
+    def take_over_arrival(self, param_id: int) -> None:
+        """Direct-grad params signal arrival via param_ready from their
+        Function backward; their post-accumulate hook must go away —
+        torch fires it even when the Function returns None for the param
+        (verified), which would double-count the arrival."""
+        h = self._hooks.pop(param_id, None)
+        if h is not None:
+            h.remove()
 
     def _launch_ready(self) -> None:
         """Launch ready buckets strictly in ascending bucket index: RCCL
