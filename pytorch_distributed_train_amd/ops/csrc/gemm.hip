@@ -252,6 +252,132 @@ __global__ __launch_bounds__(512) void gemm_nt_big_k(
   }
 }
 
+
+// ---- 8-phase 256x256 schedule (guide "8-phase template" geometry):
+// same tile/wave/fragment layout as gemm_nt_big_k, but the K-loop is
+// split into 4 phases per K-tile (one C-quadrant each), with per-phase
+// raw-barrier pairs, s_setprio(1) around the MFMA burst, and the next
+// tile's staging spread one 64-row batch per phase. The whole-tile
+// rendezvous of the 2-phase structure (52% SQ_WAIT at 2 waves/SIMD) is
+// replaced by 8 short lockstep phases that keep ds_read/glds/MFMA
+// interleaved — the documented lever past the ~900 TF plain-HIP ceiling.
+template <typename T, typename OutT, bool BIAS>
+__global__ __launch_bounds__(512) void gemm_nt_8p_k(
+    const T* __restrict__ A, const T* __restrict__ B, OutT* __restrict__ C,
+    const float* __restrict__ bias, int M, int N, int K, int tiles_n,
+    long batch_sa, long batch_sb, long batch_sc) {
+  constexpr int BOFF = 256 * 64 * 2;  // 32 KiB per operand per slot
+  __shared__ __attribute__((aligned(16))) char lds[2][2 * BOFF];
+
+  A += (long)blockIdx.y * batch_sa;
+  B += (long)blockIdx.y * batch_sb;
+  C += (long)blockIdx.y * batch_sc;
+
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  int tm = bid / tiles_n, tn = bid % tiles_n;
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  int wm = wave >> 2, wn = wave & 3;  // 2 x 4
+  int m0 = tm * 256, n0 = tn * 256;
+  int ktiles = K / 64;
+
+  int row_in_chunk = lane >> 3;
+  int col16 = lane & 7;
+
+  // one 64-row batch (8 KiB; one glds16 per thread) of operand A (b<4)
+  // or B (b>=4) for K-tile kt into `slot`
+  auto stage_batch = [&](int kt, int slot, int b) {
+    int i = b & 3;
+    int row = (wave * 4 + i) * 8 + row_in_chunk;
+    if (b < 4) {
+      int grow = m0 + row;
+      if (grow > M - 1) grow = M - 1;
+      const char* g = reinterpret_cast<const char*>(A) +
+                      ((long)grow * K + kt * 64) * 2 +
+                      (col16 ^ (row & 7)) * 16;
+      glds16(g, (char*)lds[slot] + (wave * 4 + i) * 8 * 128);
+    } else {
+      int grow = n0 + row;
+      if (grow > N - 1) grow = N - 1;
+      const char* g = reinterpret_cast<const char*>(B) +
+                      ((long)grow * K + kt * 64) * 2 +
+                      (col16 ^ (row & 7)) * 16;
+      glds16(g, (char*)lds[slot] + BOFF + (wave * 4 + i) * 8 * 128);
+    }
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+#pragma unroll
+  for (int b = 0; b < 8; ++b) stage_batch(0, 0, b);
+
+  for (int kt = 0; kt < ktiles; ++kt) {
+    int slot = kt & 1;
+    // publish this slot: drain own glds, then whole-block rendezvous
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    const char* la = lds[slot];
+    const char* lb = lds[slot] + BOFF;
+    s16x8 bf[2][4];
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {  // one phase per C-quadrant
+      if (q == 0) {
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            bf[kk][j] = frag_read128(lb, wn * 64 + j * 16 + (lane & 15),
+                                     (lane >> 4) + kk * 4);
+      }
+      s16x8 af[2][2];
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+          af[kk][i] = frag_read128(
+              la, wm * 128 + q * 32 + i * 16 + (lane & 15),
+              (lane >> 4) + kk * 4);
+      if (kt + 1 < ktiles) {  // spread next tile's staging over phases
+        stage_batch(kt + 1, slot ^ 1, q);
+        stage_batch(kt + 1, slot ^ 1, q + 4);
+      }
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[q * 2 + i][j] =
+                MfmaOp<T>::run(af[kk][i], bf[kk][j], acc[q * 2 + i][j]);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    long rbase = m0 + wm * 128 + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = n0 + wn * 64 + j * 16 + (lane & 15);
+      if (col >= N) continue;
+      float badd = BIAS ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = rbase + r;
+        if (row < M)
+          C[(long)row * N + col] = from_f32<OutT>(acc[i][j][r] + badd);
+      }
+    }
+  }
+}
+
 // column sum: db[n] = sum_m dy[m][n]  (linear bias grad)
 template <typename T>
 __global__ void colsum_f32_k(const T* __restrict__ dy, float* __restrict__ db,
@@ -428,6 +554,23 @@ void pdta_gemm_nt(const void* A, const void* B, void* C, const float* bias,
   // occupancy experiment (docs/DESIGN.md §6), opt-in via PDTA_GEMM_V2=1
   static int use_v2 = -1;
   if (use_v2 < 0) use_v2 = getenv("PDTA_GEMM_V2") != nullptr;
+  static int use_8p = -1;
+  if (use_8p < 0) use_8p = getenv("PDTA_GEMM_8P") != nullptr;
+  if (use_8p && M >= 1024 && N >= 256 && K % 64 == 0 && !out_f32) {
+    int tiles_m = (M + 255) / 256, tiles_n = (N + 255) / 256;
+    dim3 grid(tiles_m * tiles_n);
+    PDTA_DISPATCH_DTYPE16(dt, T, {
+      if (bias)
+        gemm_nt_8p_k<T, T, true><<<grid, 512, 0, s>>>(
+            (const T*)A, (const T*)B, (T*)C, bias, M, N, K, tiles_n, 0, 0,
+            0);
+      else
+        gemm_nt_8p_k<T, T, false><<<grid, 512, 0, s>>>(
+            (const T*)A, (const T*)B, (T*)C, nullptr, M, N, K, tiles_n, 0,
+            0, 0);
+    });
+    return;
+  }
   if (use_v2 && M >= 1024 && N >= 128 && K % 32 == 0 && !out_f32) {
     int tiles_m = (M + 255) / 256, tiles_n = (N + 127) / 128;
     dim3 grid(tiles_m * tiles_n);
