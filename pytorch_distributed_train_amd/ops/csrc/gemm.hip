@@ -378,6 +378,120 @@ __global__ __launch_bounds__(512) void gemm_nt_8p_k(
   }
 }
 
+
+// ---- 4-slot BK=32 ring: 256x256 tile, 8 waves, staging
+// runs THREE K-tiles ahead so the boundary wait is a counted
+// s_waitcnt vmcnt(8) — loads stay in flight across every barrier, never
+// drained to 0 (the documented ~20%% stall of the 2-slot schedule).
+template <typename T, typename OutT, bool BIAS>
+__global__ __launch_bounds__(512) void gemm_nt_r4_k(
+    const T* __restrict__ A, const T* __restrict__ B, OutT* __restrict__ C,
+    const float* __restrict__ bias, int M, int N, int K, int tiles_n,
+    long batch_sa, long batch_sb, long batch_sc) {
+  constexpr int BOFF = 256 * 32 * 2;  // 16 KiB per operand per slot
+  __shared__ __attribute__((aligned(16))) char lds[4][2 * BOFF];
+
+  A += (long)blockIdx.y * batch_sa;
+  B += (long)blockIdx.y * batch_sb;
+  C += (long)blockIdx.y * batch_sc;
+
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  int tm = bid / tiles_n, tn = bid % tiles_n;
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  int wm = wave >> 2, wn = wave & 3;  // 2 x 4
+  int m0 = tm * 256, n0 = tn * 256;
+  int ktiles = K / 32;
+
+  int row_in = lane >> 2, c4 = lane & 3;  // 64-B rows: 4 chunks of 16 B
+
+  // one K32 tile of A+B: 2 glds per thread (A: 16 KiB, B: 16 KiB)
+  auto stage_all = [&](int kt, int slot) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int row = (wave * 2 + i) * 16 + row_in;
+      int grow = m0 + row;
+      if (grow > M - 1) grow = M - 1;
+      const char* g = reinterpret_cast<const char*>(A) +
+                      ((long)grow * K + kt * 32) * 2 +
+                      ((c4 ^ (row_in & 3)) * 16);
+      glds16(g, (char*)lds[slot] + (wave * 2 + i) * 1024);
+    }
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int row = (wave * 2 + i) * 16 + row_in;
+      int grow = n0 + row;
+      if (grow > N - 1) grow = N - 1;
+      const char* g = reinterpret_cast<const char*>(B) +
+                      ((long)grow * K + kt * 32) * 2 +
+                      ((c4 ^ (row_in & 3)) * 16);
+      glds16(g, (char*)lds[slot] + BOFF + (wave * 2 + i) * 1024);
+    }
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  stage_all(0, 0);
+  if (ktiles > 1) stage_all(1, 1);
+  if (ktiles > 2) stage_all(2, 2);
+
+  for (int kt = 0; kt < ktiles; ++kt) {
+    int slot = kt & 3;
+    // slots kt+1, kt+2 may stay in flight: 2 slots x 4 glds/thread = 8
+    if (kt + 2 < ktiles)
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    else if (kt + 1 < ktiles)
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (kt + 3 < ktiles) stage_all(kt + 3, (kt + 3) & 3);
+    const char* la = lds[slot];
+    const char* lb = lds[slot] + BOFF;
+    s16x8 bf[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      bf[j] = frag_read64(lb, wn * 64 + j * 16 + (lane & 15), lane >> 4);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      s16x8 af[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+        af[i] = frag_read64(la, wm * 128 + q * 32 + i * 16 + (lane & 15),
+                            lane >> 4);
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[q * 2 + i][j] =
+              MfmaOp<T>::run(af[i], bf[j], acc[q * 2 + i][j]);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+  }
+
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    long rbase = m0 + wm * 128 + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = n0 + wn * 64 + j * 16 + (lane & 15);
+      if (col >= N) continue;
+      float badd = BIAS ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = rbase + r;
+        if (row < M)
+          C[(long)row * N + col] = from_f32<OutT>(acc[i][j][r] + badd);
+      }
+    }
+  }
+}
+
 // column sum: db[n] = sum_m dy[m][n]  (linear bias grad)
 template <typename T>
 __global__ void colsum_f32_k(const T* __restrict__ dy, float* __restrict__ db,
@@ -556,6 +670,23 @@ void pdta_gemm_nt(const void* A, const void* B, void* C, const float* bias,
   if (use_v2 < 0) use_v2 = getenv("PDTA_GEMM_V2") != nullptr;
   static int use_8p = -1;
   if (use_8p < 0) use_8p = getenv("PDTA_GEMM_8P") != nullptr;
+  static int use_r4 = -1;
+  if (use_r4 < 0) use_r4 = getenv("PDTA_GEMM_R4") != nullptr;
+  if (use_r4 && M >= 1024 && N >= 256 && K % 32 == 0 && !out_f32) {
+    int tiles_m = (M + 255) / 256, tiles_n = (N + 255) / 256;
+    dim3 grid(tiles_m * tiles_n);
+    PDTA_DISPATCH_DTYPE16(dt, T, {
+      if (bias)
+        gemm_nt_r4_k<T, T, true><<<grid, 512, 0, s>>>(
+            (const T*)A, (const T*)B, (T*)C, bias, M, N, K, tiles_n, 0, 0,
+            0);
+      else
+        gemm_nt_r4_k<T, T, false><<<grid, 512, 0, s>>>(
+            (const T*)A, (const T*)B, (T*)C, nullptr, M, N, K, tiles_n, 0,
+            0, 0);
+    });
+    return;
+  }
   if (use_8p && M >= 1024 && N >= 256 && K % 64 == 0 && !out_f32) {
     int tiles_m = (M + 255) / 256, tiles_n = (N + 255) / 256;
     dim3 grid(tiles_m * tiles_n);
