@@ -46,8 +46,9 @@ def bench(M, N, K, iters=30):
 
 
 def main():
-    mode = "8P" if os.environ.get("PDTA_GEMM_8P") else (
-        "V2" if os.environ.get("PDTA_GEMM_V2") else "base")
+    mode = ("R4" if os.environ.get("PDTA_GEMM_R4") else
+            "8P" if os.environ.get("PDTA_GEMM_8P") else
+            "V2" if os.environ.get("PDTA_GEMM_V2") else "base")
     print(f"schedule: {mode}")
     for shape in SHAPES:
         bench(*shape)
