@@ -379,6 +379,12 @@ __global__ __launch_bounds__(512) void gemm_nt_8p_k(
 }
 
 
+// Rows are 64 B (32 K-elems); swizzle XOR uses the row's 4 chunks.
+DEV_INLINE s16x8 frag_read64(const char* lds, int row, int chunk) {
+  return *reinterpret_cast<const s16x8*>(
+      lds + row * 64 + ((chunk ^ (row & 3)) * 16));
+}
+
 // ---- 4-slot BK=32 ring: 256x256 tile, 8 waves, staging
 // runs THREE K-tiles ahead so the boundary wait is a counted
 // s_waitcnt vmcnt(8) — loads stay in flight across every barrier, never
@@ -508,11 +514,6 @@ __global__ void colsum_f32_k(const T* __restrict__ dy, float* __restrict__ db,
 // Occupancy experiment (docs/DESIGN.md §6): acc 64 VGPR/thread and
 // 2x24 KiB ring slots -> 3 blocks/CU by LDS, ~4 waves/SIMD by VGPR
 // (the 256x256/BK=64 kernel is pinned at 2 waves/SIMD and 52% WAIT).
-// Rows are 64 B (32 K-elems); swizzle XOR uses the row's 4 chunks.
-DEV_INLINE s16x8 frag_read64(const char* lds, int row, int chunk) {
-  return *reinterpret_cast<const s16x8*>(
-      lds + row * 64 + ((chunk ^ (row & 3)) * 16));
-}
 
 template <typename T, typename OutT, bool BIAS>
 __global__ __launch_bounds__(512) void gemm_nt_v2_k(
