@@ -693,15 +693,27 @@ __global__ __launch_bounds__(TPB) void igemm_wrw_k(
   // stage 64 rows of dy [row][64 kout] and gathered x [row][64 c]; the
   // source address carries the chunk^=(row&7) swizzle matching the
   // tr_b16 fragment reads (rule 21: glds dest is lane-linear).
-  // incremental row->(n,p,q) decode for the gather (divisions only once)
-  int g_q[2], g_p[2], g_n[2];
+  // incremental row->(n,p,q) decode AND incremental byte addresses: the
+  // per-tile 64-bit (((n*H+h)*W+w)*C) chains were ~1/3 of this kernel's
+  // 18-VALU-per-MFMA issue load (PMC) — all replaced by
+  // precomputed per-advance address deltas with bounded carries
+  int g_q[2], g_p[2];
+  long g_adr[2];  // byte offset of x[n][h][w][c_t*64] (h,w may be OOB)
+  int g_h[2], g_w[2];
+  const long dq_b = (long)stride * C * 2;          // q += 1
+  const long dp_b = (long)stride * W * C * 2;      // p += 1
+  const long dn_b = (long)H * W * C * 2;           // n += 1
   if (GATHER) {
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
       long row = r0 + wave * 16 + i * 8 + row_in_chunk;
       g_q[i] = (int)(row % Q);
       g_p[i] = (int)((row / Q) % P);
-      g_n[i] = (int)(row / ((long)P * Q));
+      int n = (int)(row / ((long)P * Q));
+      g_h[i] = g_p[i] * stride - pad + r;
+      g_w[i] = g_q[i] * stride - pad + ss;
+      g_adr[i] = (long)n * dn_b + (long)g_h[i] * W * C * 2 +
+                 (long)g_w[i] * C * 2;
     }
   }
   int adv_dq = 64 % Q, adv_dp = 64 / Q;  // bounded carries per advance
@@ -712,35 +724,50 @@ __global__ __launch_bounds__(TPB) void igemm_wrw_k(
       g_q[i] += adv_dq;
       int cy = g_q[i] >= Q;
       g_q[i] -= cy ? Q : 0;
-      g_p[i] += adv_dp + cy;
+      int dp = adv_dp + cy;
+      g_p[i] += dp;
+      g_w[i] += (adv_dq - (cy ? Q : 0)) * stride;
+      g_h[i] += dp * stride;
+      g_adr[i] += (long)(adv_dq - (cy ? Q : 0)) * dq_b + (long)dp * dp_b;
       while (g_p[i] >= P) {
         g_p[i] -= P;
-        ++g_n[i];
+        g_h[i] -= P * stride;
+        g_adr[i] += dn_b - (long)P * dp_b;
       }
     }
   };
+  // dy addresses advance by a constant per tile; the swizzled column
+  // offset is per-i constant — both hoisted out of the loop
+  const char* gdy_p[2];
+  const char* gx_sw[2];
+  long dy_row[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    int ldsrow = wave * 16 + i * 8 + row_in_chunk;
+    int scol16 = col16 ^ (ldsrow & 7);
+    dy_row[i] = r0 + ldsrow;
+    gdy_p[i] = reinterpret_cast<const char*>(dy) +
+               (dy_row[i] * Kout + ko_t * 64) * 2 + scol16 * 16;
+    gx_sw[i] = reinterpret_cast<const char*>(x) +
+               ((long)c_t * 64) * 2 + scol16 * 16;
+  }
+  const long dy_step = (long)64 * Kout * 2;
   auto stage = [&](long rowbase, int buf) {
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
       int ldsrow = wave * 16 + i * 8 + row_in_chunk;
-      int scol16 = col16 ^ (ldsrow & 7);
-      long row = rowbase + ldsrow;
       const char* gdy;
       const char* gx;
-      if (row < r1) {
-        gdy = reinterpret_cast<const char*>(dy) +
-              (row * Kout + ko_t * 64) * 2 + scol16 * 16;
+      if (dy_row[i] < r1) {
+        gdy = gdy_p[i];
         if (GATHER) {
-          int h = g_p[i] * stride - pad + r, w = g_q[i] * stride - pad + ss;
-          if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
-            gx = reinterpret_cast<const char*>(x) +
-                 ((((long)g_n[i] * H + h) * W + w) * C + c_t * 64) * 2 +
-                 scol16 * 16;
+          if ((unsigned)g_h[i] < (unsigned)H &&
+              (unsigned)g_w[i] < (unsigned)W)
+            gx = gx_sw[i] + g_adr[i];
           else
             gx = reinterpret_cast<const char*>(zero_page);
         } else {
-          gx = reinterpret_cast<const char*>(x) +
-               (row * C + c_t * 64) * 2 + scol16 * 16;
+          gx = gx_sw[i] + dy_row[i] * (long)C * 2;
         }
       } else {
         gdy = reinterpret_cast<const char*>(zero_page);
@@ -748,6 +775,8 @@ __global__ __launch_bounds__(TPB) void igemm_wrw_k(
       }
       glds16(gdy, (char*)lds[buf][0] + (wave * 16 + i * 8) * 128);
       glds16(gx, (char*)lds[buf][1] + (wave * 16 + i * 8) * 128);
+      gdy_p[i] += dy_step;
+      dy_row[i] += 64;
     }
   };
 
@@ -761,6 +790,17 @@ __global__ __launch_bounds__(TPB) void igemm_wrw_k(
                           lds_addr_of(lds[2][0])};
   unsigned lx_base[3] = {lds_addr_of(lds[0][1]), lds_addr_of(lds[1][1]),
                          lds_addr_of(lds[2][1])};
+  // loop-invariant tr-read offsets (8 fragment pairs) precomputed once
+  unsigned a_off0[2][2], a_off1[2][2], b_off0[2][2], b_off1[2][2];
+#pragma unroll
+  for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      tr_off(kk * 32, wm * 32 + i * 16, lane, &a_off0[kk][i],
+             &a_off1[kk][i]);
+      tr_off(kk * 32, wn * 32 + i * 16, lane, &b_off0[kk][i],
+             &b_off1[kk][i]);
+    }
   int cur = 0;
   for (long rb = r0; rb < r1; rb += 64) {
     // wait for THIS tile's 4 glds (<=4 others stay in flight), then a raw
@@ -777,12 +817,14 @@ __global__ __launch_bounds__(TPB) void igemm_wrw_k(
     for (int kk = 0; kk < 2; ++kk) {
 #pragma unroll
       for (int i = 0; i < 2; ++i)
-        frag_read_tr_issue(ldy_base[cur], kk * 32, wm * 32 + i * 16, lane,
-                           &alo[kk][i], &ahi[kk][i]);
+        tr_issue_at(ldy_base[cur] + a_off0[kk][i],
+                    ldy_base[cur] + a_off1[kk][i], &alo[kk][i],
+                    &ahi[kk][i]);
 #pragma unroll
       for (int i = 0; i < 2; ++i)
-        frag_read_tr_issue(lx_base[cur], kk * 32, wn * 32 + i * 16, lane,
-                           &blo[kk][i], &bhi[kk][i]);
+        tr_issue_at(lx_base[cur] + b_off0[kk][i],
+                    lx_base[cur] + b_off1[kk][i], &blo[kk][i],
+                    &bhi[kk][i]);
     }
     tr_read_wait();
 #pragma unroll

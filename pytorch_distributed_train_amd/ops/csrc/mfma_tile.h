@@ -76,6 +76,33 @@ DEV_INLINE void frag_read_tr_issue(unsigned lds_base, int k0, int m0,
       : "v"(a0), "v"(a1));
 }
 
+// Offset-precompute split of frag_read_tr_issue: the (k0, m0, lane)
+// address math is loop-invariant in ring-buffered K-loops — compute the
+// in-slot byte offsets ONCE, then each tile's read is base+off (2 VALU)
+// instead of the full ~10-op recompute per fragment pair.
+DEV_INLINE void tr_off(int k0, int m0, int lane, unsigned* off0,
+                       unsigned* off1) {
+  int g = lane >> 4;
+  int row = (lane & 15) >> 2;
+  int seg = lane & 3;
+  int byte0 = m0 * 2 + seg * 8;
+  int chunk = byte0 >> 4, rem = byte0 & 15;
+  int ka = k0 + g * 8 + row;
+  int kb = ka + 4;
+  *off0 = ka * 128 + (((chunk ^ (ka & 7)) << 4) | rem);
+  *off1 = kb * 128 + (((chunk ^ (kb & 7)) << 4) | rem);
+}
+
+DEV_INLINE void tr_issue_at(unsigned a0, unsigned a1,
+                            unsigned long long* lo,
+                            unsigned long long* hi) {
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3"
+      : "=&v"(*lo), "=&v"(*hi)
+      : "v"(a0), "v"(a1));
+}
+
 // Stride-parameterized variant: [k][m] LDS image with ROW_BYTES-wide rows
 // (e.g. 512 B for 256-feature tiles). The XOR swizzle uses the FULL chunk
 // range of the row (k & (ROW_BYTES/16 - 1)): one ds_read_b64_tr_b16
