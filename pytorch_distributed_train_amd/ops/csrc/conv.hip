@@ -934,6 +934,229 @@ __global__ void wrw_reduce_k(const float* __restrict__ part,
   }
 }
 
+
+// ---- medium bwd-weight: 128x128 (ko x c) tile, 4 waves, 3-slot ring.
+// For Kout%128==0 && C%128==0 shapes below the big kernel's gate: the
+// 64x64 tile re-stages every dy row c_tiles(=C/64) times and every x row
+// ko_tiles times; this tile HALVES both (measured: the wrw family was
+// 21%% of the ResNet-50 step, gather-staging bound). Partials land as
+// 64x64 granules in wrw_reduce_k's layout (4 granules per block).
+template <typename T, bool GATHER>
+__global__ __launch_bounds__(TPB) void igemm_wrw_mid_k(
+    const T* __restrict__ x, const T* __restrict__ dy,
+    float* __restrict__ part, const T* __restrict__ zero_page, int Nn,
+    int H, int W, int C, int P, int Q, int Kout, int R, int S, int stride,
+    int pad, long rows_per_chunk, int ko_tiles, int c_tiles,
+    long tot_granules) {
+  // slots: dy [64 rows][128 ko] + x [64 rows][128 c], 256-B rows
+  __shared__ __attribute__((aligned(16))) char lds[3][2][64 * 128 * 2];
+
+  int ko_t2 = ko_tiles / 2, c_t2 = c_tiles / 2;
+  int bid = blockIdx.x;
+  int tap = bid / (ko_t2 * c_t2);
+  int rem = bid % (ko_t2 * c_t2);
+  int ko_t = rem / c_t2, c_t = rem % c_t2;  // in 128-units
+  int r = tap / S, ss = tap % S;
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  int wm = wave >> 1, wn = wave & 1;  // 2x2 waves over 128ko x 128c
+
+  long NPQ = (long)Nn * P * Q;
+  long r0 = (long)blockIdx.y * rows_per_chunk;
+  long r1 = min(NPQ, r0 + rows_per_chunk);
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  int row4 = lane >> 4;       // 4 rows per glds (256-B rows, 16 chunks)
+  int chunk16 = lane & 15;
+
+  long g_adr[4];
+  int g_h[4], g_w[4];
+  const long dq_b = (long)stride * C * 2;
+  const long dp_b = (long)stride * W * C * 2;
+  const long dn_b = (long)H * W * C * 2;
+  int gq4[4], gp4[4];
+  if (GATHER) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      long row = r0 + wave * 16 + i * 4 + row4;
+      gq4[i] = (int)(row % Q);
+      gp4[i] = (int)((row / Q) % P);
+      int n = (int)(row / ((long)P * Q));
+      g_h[i] = gp4[i] * stride - pad + r;
+      g_w[i] = gq4[i] * stride - pad + ss;
+      g_adr[i] = (long)n * dn_b + (long)g_h[i] * W * C * 2 +
+                 (long)g_w[i] * C * 2;
+    }
+  }
+  int adv_dq = 64 % Q, adv_dp = 64 / Q;
+  auto advance = [&]() {
+    if (!GATHER) return;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      gq4[i] += adv_dq;
+      int cy = gq4[i] >= Q;
+      gq4[i] -= cy ? Q : 0;
+      int dp = adv_dp + cy;
+      gp4[i] += dp;
+      g_w[i] += (adv_dq - (cy ? Q : 0)) * stride;
+      g_h[i] += dp * stride;
+      g_adr[i] += (long)(adv_dq - (cy ? Q : 0)) * dq_b + (long)dp * dp_b;
+      while (gp4[i] >= P) {
+        gp4[i] -= P;
+        g_h[i] -= P * stride;
+        g_adr[i] += dn_b - (long)P * dp_b;
+      }
+    }
+  };
+  const char* gdy_p[4];
+  const char* gx_sw[4];
+  long dy_row[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int ldsrow = wave * 16 + i * 4 + row4;
+    int scol = chunk16 ^ (ldsrow & 15);
+    dy_row[i] = r0 + ldsrow;
+    gdy_p[i] = reinterpret_cast<const char*>(dy) +
+               (dy_row[i] * Kout + (long)ko_t * 128) * 2 + scol * 16;
+    gx_sw[i] = reinterpret_cast<const char*>(x) +
+               ((long)c_t * 128) * 2 + scol * 16;
+  }
+  const long dy_step = (long)64 * Kout * 2;
+  auto stage = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const char* gdy;
+      const char* gx;
+      if (dy_row[i] < r1) {
+        gdy = gdy_p[i];
+        if (GATHER) {
+          if ((unsigned)g_h[i] < (unsigned)H &&
+              (unsigned)g_w[i] < (unsigned)W)
+            gx = gx_sw[i] + g_adr[i];
+          else
+            gx = reinterpret_cast<const char*>(zero_page);
+        } else {
+          gx = gx_sw[i] + dy_row[i] * (long)C * 2;
+        }
+      } else {
+        gdy = reinterpret_cast<const char*>(zero_page);
+        gx = reinterpret_cast<const char*>(zero_page);
+      }
+      // glds dest is the 4-row group base (lane-linear fill)
+      glds16(gdy, (char*)lds[buf][0] + (wave * 16 + i * 4) * 256);
+      glds16(gx, (char*)lds[buf][1] + (wave * 16 + i * 4) * 256);
+      gdy_p[i] += dy_step;
+      dy_row[i] += 64;
+    }
+  };
+
+  stage(0);
+  if (r0 + 64 < r1) {
+    advance();
+    stage(1);
+  }
+  unsigned ldy_base[3] = {lds_addr_of(lds[0][0]), lds_addr_of(lds[1][0]),
+                          lds_addr_of(lds[2][0])};
+  unsigned lx_base[3] = {lds_addr_of(lds[0][1]), lds_addr_of(lds[1][1]),
+                         lds_addr_of(lds[2][1])};
+  unsigned a_off0[2][2], a_off1[2][2], b_off0[2][2], b_off1[2][2];
+#pragma unroll
+  for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      tr_off_s<256>(kk * 32, wm * 64 + i * 16 + 0, lane, &a_off0[kk][i],
+                    &a_off1[kk][i]);
+      tr_off_s<256>(kk * 32, wn * 64 + i * 16 + 0, lane, &b_off0[kk][i],
+                    &b_off1[kk][i]);
+    }
+  // each wave computes 64ko x 64c = (2 a-frag groups of 32) x ... the
+  // 4x4 acc spans 64x64 via two 16-row a-frags and two 16-col b-frags
+  // per kk, with a second (i+2, j+2) pass offset by 32 rows/cols
+  unsigned a_off0b[2][2], a_off1b[2][2], b_off0b[2][2], b_off1b[2][2];
+#pragma unroll
+  for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      tr_off_s<256>(kk * 32, wm * 64 + 32 + i * 16, lane, &a_off0b[kk][i],
+                    &a_off1b[kk][i]);
+      tr_off_s<256>(kk * 32, wn * 64 + 32 + i * 16, lane, &b_off0b[kk][i],
+                    &b_off1b[kk][i]);
+    }
+  int cur = 0;
+  for (long rb = r0; rb < r1; rb += 64) {
+    if (rb + 64 < r1)
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    unsigned long long alo[2][4], ahi[2][4], blo[2][4], bhi[2][4];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        tr_issue_at(ldy_base[cur] + a_off0[kk][i],
+                    ldy_base[cur] + a_off1[kk][i], &alo[kk][i],
+                    &ahi[kk][i]);
+        tr_issue_at(ldy_base[cur] + a_off0b[kk][i],
+                    ldy_base[cur] + a_off1b[kk][i], &alo[kk][i + 2],
+                    &ahi[kk][i + 2]);
+        tr_issue_at(lx_base[cur] + b_off0[kk][i],
+                    lx_base[cur] + b_off1[kk][i], &blo[kk][i],
+                    &bhi[kk][i]);
+        tr_issue_at(lx_base[cur] + b_off0b[kk][i],
+                    lx_base[cur] + b_off1b[kk][i], &blo[kk][i + 2],
+                    &bhi[kk][i + 2]);
+      }
+    }
+    tr_read_wait();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      s16x8 af[4], bf[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        af[i] = tr_pair_to_frag(alo[kk][i], ahi[kk][i]);
+        bf[i] = tr_pair_to_frag(blo[kk][i], bhi[kk][i]);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = MfmaOp<T>::run(af[i], bf[j], acc[i][j]);
+    }
+    __builtin_amdgcn_s_barrier();
+    if (rb + 128 < r1) {
+      advance();
+      stage((cur + 2) % 3);
+    }
+    cur = (cur + 1) % 3;
+  }
+
+  // Wave (wm,wn) owns rows wm*64..+63, cols wn*64..+63 of the 128x128
+  // block = exactly ONE 64x64 granule: (ko64, c64) = (ko_t*2+wm, c_t*2+wn)
+  {
+    long g64 = (long)tap * ko_tiles * c_tiles +
+               (long)(ko_t * 2 + wm) * c_tiles + (c_t * 2 + wn);
+    float* out = part + ((long)blockIdx.y * tot_granules + g64) * 4096;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int ihalf = (i >= 2);     // af index: 0,1 -> +0; 2,3 -> +32
+      int ko_local = ihalf * 32 + (i & 1) * 16 + (lane >> 4) * 4;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int jhalf = (j >= 2);
+        int c_local = jhalf * 32 + (j & 1) * 16 + (lane & 15);
+#pragma unroll
+        for (int rg = 0; rg < 4; ++rg)
+          out[(ko_local + rg) * 64 + c_local] = acc[i][j][rg];
+      }
+    }
+  }
+}
+
 // ---- big bwd-weight: 256x256 (ko x c) tile, 8 waves, 2-slot ring -------
 // For Kout % 256 == 0 && C % 256 == 0 (including the linear-dW TN GEMMs
 // routed through this path): the 64x64-tile kernel re-stages every row
@@ -1297,6 +1520,22 @@ void pdta_conv_bwd_weight(const void* x, const void* dy, float* dw,
             (const T*)x, (const T*)dy, part, (const T*)zero_page, Nn, H, W,
             C, P, Q, Kout, R, S, stride, pad, rows_per_chunk, ko_tiles,
             c_tiles, mq.mul, mq.shift, mpq.mul, mpq.shift);
+    });
+  } else if (Kout % 128 == 0 && C % 128 == 0) {
+    // 128x128 tile: half the re-staged dy/x bytes of the 64x64 kernel
+    long blocks = (long)R * S * (ko_tiles / 2) * (c_tiles / 2);
+    dim3 grid((unsigned)blocks, (unsigned)chunks);
+    PDTA_DISPATCH_DTYPE16(dt, T, {
+      if (identity)
+        igemm_wrw_mid_k<T, false><<<grid, TPB, 0, s>>>(
+            (const T*)x, (const T*)dy, part, (const T*)zero_page, Nn, H, W,
+            C, P, Q, Kout, R, S, stride, pad, rows_per_chunk, ko_tiles,
+            c_tiles, tiles);
+      else
+        igemm_wrw_mid_k<T, true><<<grid, TPB, 0, s>>>(
+            (const T*)x, (const T*)dy, part, (const T*)zero_page, Nn, H, W,
+            C, P, Q, Kout, R, S, stride, pad, rows_per_chunk, ko_tiles,
+            c_tiles, tiles);
     });
   } else {
     dim3 grid((unsigned)tiles, (unsigned)chunks);

@@ -93,6 +93,23 @@ DEV_INLINE void tr_off(int k0, int m0, int lane, unsigned* off0,
   *off1 = kb * 128 + (((chunk ^ (kb & 7)) << 4) | rem);
 }
 
+// ROW_BYTES variant of tr_off (full-chunk-range XOR, matching
+// frag_read_tr_issue_s<ROW_BYTES>)
+template <int ROW_BYTES>
+DEV_INLINE void tr_off_s(int k0, int m0, int lane, unsigned* off0,
+                         unsigned* off1) {
+  constexpr int KMASK = ROW_BYTES / 16 - 1;
+  int g = lane >> 4;
+  int row = (lane & 15) >> 2;
+  int seg = lane & 3;
+  int byte0 = m0 * 2 + seg * 8;
+  int chunk = byte0 >> 4, rem = byte0 & 15;
+  int ka = k0 + g * 8 + row;
+  int kb = ka + 4;
+  *off0 = ka * ROW_BYTES + (((chunk ^ (ka & KMASK)) << 4) | rem);
+  *off1 = kb * ROW_BYTES + (((chunk ^ (kb & KMASK)) << 4) | rem);
+}
+
 DEV_INLINE void tr_issue_at(unsigned a0, unsigned a1,
                             unsigned long long* lo,
                             unsigned long long* hi) {
