@@ -151,6 +151,10 @@ class Conv2dNHWC(torch.autograd.Function):
                     ctx.dx_sink.t = dx
                     dx = None
             if need_dw:
+                import os as _os
+                if x.shape[3] % 64 != 0 and _os.environ.get("PDTA_DEBUG_STEM"):
+                    print("stem wrw xcol cached:", xcol is not None,
+                          flush=True)
                 if ctx.gslot is not None:
                     # direct-grad: write straight into the flat slice and
                     # return None (no AccumulateGrad add); tell the
