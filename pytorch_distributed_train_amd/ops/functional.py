@@ -115,8 +115,12 @@ class Conv2dNHWC(torch.autograd.Function):
         # (measured: ~53 FillFunctor launches/step on ResNet-50)
         ctx.set_materialize_grads(False)
         if x.is_cuda:
-            # stem (C % 64 != 0): cache the im2col image for bwd-weight
-            want_xcol = x.shape[3] % 64 != 0 and torch.is_grad_enabled()
+            # stem (C % 64 != 0): cache the im2col image for bwd-weight.
+            # NOTE grad mode is always DISABLED inside a custom Function's
+            # forward, so gate on the weight's requires_grad instead (a
+            # torch.is_grad_enabled() gate here is always False — measured
+            # as a full extra im2col pass per step)
+            want_xcol = x.shape[3] % 64 != 0 and weight.requires_grad
             out = ext().conv2d_fwd(x, wt, stride, pad, want_bn_stats,
                                    want_xcol)
             xcol = out[-1] if want_xcol else None
@@ -151,10 +155,6 @@ class Conv2dNHWC(torch.autograd.Function):
                     ctx.dx_sink.t = dx
                     dx = None
             if need_dw:
-                import os as _os
-                if x.shape[3] % 64 != 0 and _os.environ.get("PDTA_DEBUG_STEM"):
-                    print("stem wrw xcol cached:", xcol is not None,
-                          flush=True)
                 if ctx.gslot is not None:
                     # direct-grad: write straight into the flat slice and
                     # return None (no AccumulateGrad add); tell the
