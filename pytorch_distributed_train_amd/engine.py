@@ -56,7 +56,12 @@ class Trainer:
 
         self.comm = make_communicator(self.env, self.device, args.backend)
         self._bn_comm = None
-        if args.sync_bn and self.env.world_size > 1:
+        from .parallel.comm import RcclComm
+        if args.sync_bn and (self.env.world_size > 1
+                             or isinstance(self.comm, RcclComm)):
+            # ws-1 RCCL (PDTA_RCCL_WS1) also converts: the packed-stats
+            # allgather and [sum_dy‖sum_dy_xmu‖count] allreduce then run
+            # the REAL enqueue path on a single-GPU box (config #4 audit)
             self._bn_comm = self.comm.dup()
             self.model = convert_sync_batchnorm(self.model, self._bn_comm)
 
