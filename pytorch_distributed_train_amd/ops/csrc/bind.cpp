@@ -30,6 +30,13 @@ void pdta_bn_bwd_elemt(const void*, const void*, const void*, void*,
                        const float*, const float*, const float*,
                        const float*, float, long, int, int, const void*,
                        hipStream_t);
+void pdta_syncbn_pack(const float*, const float*, float, float*, int,
+                      hipStream_t);
+void pdta_syncbn_combine(const float*, int, int, float, float, float*,
+                         float*, float*, float*, float*, hipStream_t);
+void pdta_syncbn_pack2(const float*, const float*, float, float*, int,
+                       hipStream_t);
+void pdta_syncbn_div(float*, int, hipStream_t);
 void pdta_maxpool_fwd(const void*, void*, uint8_t*, int, int, int, int, int,
                       int, int, int, int, int, hipStream_t);
 void pdta_maxpool_bwd(const void*, const uint8_t*, void*, int, int, int, int,
@@ -289,6 +296,49 @@ std::vector<at::Tensor> bn_bwd_elemt(at::Tensor x, at::Tensor dy,
                     cur_stream());
   if (want_dres) return {dx, dres};
   return {dx};
+}
+
+
+// ---- SyncBN fused pack/combine ------------------------------------------
+at::Tensor syncbn_pack(at::Tensor mean, at::Tensor var, double n) {
+  CHECK_IN(mean);
+  int C = mean.numel();
+  auto out = at::empty({2 * C + 1}, mean.options());
+  pdta_syncbn_pack(mean.data_ptr<float>(), var.contiguous().data_ptr<float>(),
+                   (float)n, out.data_ptr<float>(), C, cur_stream());
+  return out;
+}
+
+std::vector<at::Tensor> syncbn_combine(at::Tensor gathered, int64_t ws,
+                                       int64_t C, double eps,
+                                       double momentum,
+                                       c10::optional<at::Tensor> running_mean,
+                                       c10::optional<at::Tensor> running_var) {
+  CHECK_IN(gathered);
+  auto opts = gathered.options();
+  auto mean = at::empty({C}, opts), invstd = at::empty({C}, opts);
+  auto total = at::empty({1}, opts);
+  pdta_syncbn_combine(
+      gathered.data_ptr<float>(), ws, C, (float)eps, (float)momentum,
+      mean.data_ptr<float>(), invstd.data_ptr<float>(),
+      running_mean.has_value() ? running_mean->data_ptr<float>() : nullptr,
+      running_var.has_value() ? running_var->data_ptr<float>() : nullptr,
+      total.data_ptr<float>(), cur_stream());
+  return {mean, invstd, total};
+}
+
+at::Tensor syncbn_pack2(at::Tensor a, at::Tensor b, double n) {
+  CHECK_IN(a);
+  int C = a.numel();
+  auto out = at::empty({2 * C + 1}, a.options());
+  pdta_syncbn_pack2(a.data_ptr<float>(), b.contiguous().data_ptr<float>(),
+                    (float)n, out.data_ptr<float>(), C, cur_stream());
+  return out;
+}
+
+void syncbn_div(at::Tensor packed, int64_t C) {
+  CHECK_IN(packed);
+  pdta_syncbn_div(packed.data_ptr<float>(), C, cur_stream());
 }
 
 // ---- pooling -------------------------------------------------------------
@@ -868,6 +918,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("weight"), py::arg("sum_dy"), py::arg("sum_dy_xmu"),
         py::arg("count"), py::arg("want_dres") = false,
         py::arg("bias") = c10::nullopt, py::arg("dy2") = c10::nullopt);
+  m.def("syncbn_pack", &syncbn_pack);
+  m.def("syncbn_combine", &syncbn_combine, py::arg("gathered"),
+        py::arg("ws"), py::arg("C"), py::arg("eps"), py::arg("momentum"),
+        py::arg("running_mean") = c10::nullopt,
+        py::arg("running_var") = c10::nullopt);
+  m.def("syncbn_pack2", &syncbn_pack2);
+  m.def("syncbn_div", &syncbn_div);
   m.def("maxpool_fwd", &maxpool_fwd);
   m.def("maxpool_bwd", &maxpool_bwd);
   m.def("avgpool_global_fwd", &avgpool_global_fwd);

@@ -452,6 +452,71 @@ static inline dim3 ew_grid2(long rows, int C) {
   return dim3(cgb, (unsigned)rblocks);
 }
 
+
+// ---- SyncBN pack/combine: the per-BN python op chain
+// (mul/add/cat/slice ~12 launches per layer per step) fused into single
+// kernels around the packed [sum ‖ sumsq ‖ count] collective.
+__global__ void syncbn_pack_k(const float* __restrict__ mean,
+                              const float* __restrict__ var, float n,
+                              float* __restrict__ out, int C) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < C) {
+    float m = mean[i];
+    out[i] = m * n;
+    out[C + i] = (var[i] + m * m) * n;
+  }
+  if (i == 0) out[2 * C] = n;
+}
+
+// gathered [ws][2C+1] -> global mean + invstd (+ running stats update,
+// count-weighted unbiased var) + total count
+__global__ void syncbn_combine_k(const float* __restrict__ g, int ws,
+                                 int C, float eps, float momentum,
+                                 float* __restrict__ mean,
+                                 float* __restrict__ invstd,
+                                 float* __restrict__ running_mean,
+                                 float* __restrict__ running_var,
+                                 float* __restrict__ total_out) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= C) return;
+  int stride = 2 * C + 1;
+  float s = 0.f, ss = 0.f, n = 0.f;
+  for (int r = 0; r < ws; ++r) {
+    const float* row = g + (long)r * stride;
+    s += row[i];
+    ss += row[C + i];
+    n += row[2 * C];
+  }
+  float m = s / n;
+  float v = fmaxf(ss / n - m * m, 0.f);
+  mean[i] = m;
+  invstd[i] = rsqrtf(v + eps);
+  if (running_mean) {
+    float unb = v * (n / fmaxf(n - 1.f, 1.f));
+    running_mean[i] = running_mean[i] * (1.f - momentum) + m * momentum;
+    running_var[i] = running_var[i] * (1.f - momentum) + unb * momentum;
+  }
+  if (i == 0 && total_out) *total_out = n;
+}
+
+// backward: [sum_dy ‖ sum_dy_xmu ‖ n] pack, and post-allreduce in-place
+// normalize (first 2C /= last element)
+__global__ void syncbn_pack2_k(const float* __restrict__ a,
+                               const float* __restrict__ b, float n,
+                               float* __restrict__ out, int C) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < C) {
+    out[i] = a[i];
+    out[C + i] = b[i];
+  }
+  if (i == 0) out[2 * C] = n;
+}
+
+__global__ void syncbn_div_k(float* __restrict__ p, int C) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < 2 * C) p[i] /= p[2 * C];
+}
+
 }  // namespace
 
 extern "C" {
@@ -624,6 +689,30 @@ void pdta_bn_bwd_elemt(const void* x, const void* dy, const void* relu_y,
           C / 8);
     }
   });
+}
+
+
+void pdta_syncbn_pack(const float* mean, const float* var, float n,
+                      float* out, int C, hipStream_t s) {
+  syncbn_pack_k<<<(C + 255) / 256, 256, 0, s>>>(mean, var, n, out, C);
+}
+
+void pdta_syncbn_combine(const float* g, int ws, int C, float eps,
+                         float momentum, float* mean, float* invstd,
+                         float* running_mean, float* running_var,
+                         float* total_out, hipStream_t s) {
+  syncbn_combine_k<<<(C + 255) / 256, 256, 0, s>>>(
+      g, ws, C, eps, momentum, mean, invstd, running_mean, running_var,
+      total_out);
+}
+
+void pdta_syncbn_pack2(const float* a, const float* b, float n, float* out,
+                       int C, hipStream_t s) {
+  syncbn_pack2_k<<<(C + 255) / 256, 256, 0, s>>>(a, b, n, out, C);
+}
+
+void pdta_syncbn_div(float* p, int C, hipStream_t s) {
+  syncbn_div_k<<<(2 * C + 255) / 256, 256, 0, s>>>(p, C);
 }
 
 }  // extern "C"

@@ -299,14 +299,12 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
                 # [sum_dy ‖ sum_dy_xmu ‖ local_count] in ONE all-reduce;
                 # normalizing by the reduced count on-device keeps the
                 # path host-sync-free AND exact for ragged rank batches
+                # (fused pack/div kernels: no cat/full/slice launches)
                 c = sum_dy.numel()
                 local_n = float(x.shape[0] * x.shape[1] * x.shape[2])
-                packed = torch.cat([sum_dy, sum_dy_xmu,
-                                    torch.full((1,), local_n,
-                                               dtype=torch.float32,
-                                               device=sum_dy.device)])
+                packed = ext().syncbn_pack2(sum_dy, sum_dy_xmu, local_n)
                 ctx.comm.allreduce_sum(packed).sync_stream()
-                packed[:2 * c] /= packed[2 * c]
+                ext().syncbn_div(packed, c)
                 sum_dy, sum_dy_xmu = packed[:c], packed[c:2 * c]
                 n = 1.0
             if dy_comb is not None:
