@@ -41,6 +41,20 @@ class LayerNorm(nn.Module):
                              gslot_b=(getattr(self, "_gslot_b", None)
                                       if on_direct else None))
 
+    def forward_fused(self, a, b):
+        """(ln(a+b), a+b) — fused residual join (AddLayerNormFn); the
+        pair-threaded encoder keeps the sum single-consumer so no
+        autograd fork add exists."""
+        if b is None:
+            return self.forward(a), a
+        on_direct = a.is_cuda and self.training
+        return OF.add_layer_norm(
+            a, b, self.weight, self.bias, self.eps,
+            gslot_w=(getattr(self, "_gslot_w", None)
+                     if on_direct else None),
+            gslot_b=(getattr(self, "_gslot_b", None)
+                     if on_direct else None))
+
 
 class MultiheadSelfAttention(nn.Module):
     """torchvision-compatible parameter layout: in_proj_weight [3d, d],
@@ -93,12 +107,18 @@ class EncoderBlock(nn.Module):
         self.mlp = nn.Sequential(
             Linear(dim, mlp_dim), GELU(), nn.Identity(), Linear(mlp_dim, dim))
 
-    def forward(self, x):
-        b, s, d = x.shape
-        x = x + self.self_attention(self.ln_1(x))
-        h = self.ln_2(x).reshape(b * s, d)
+    def forward(self, pair):
+        # (residual, branch) pair threading: each block receives its
+        # input UNSUMMED; ln_1 performs the previous join, ln_2 the
+        # attention join, and the mlp branch is handed onward unsummed
+        u, v = pair if isinstance(pair, tuple) else (pair, None)
+        b, s, d = u.shape
+        y1, x1 = self.ln_1.forward_fused(u, v)
+        a = self.self_attention(y1)
+        y2, x2 = self.ln_2.forward_fused(x1, a)
+        h = y2.reshape(b * s, d)
         h = self.mlp[3](self.mlp[1](self.mlp[0](h))).view(b, s, d)
-        return x + h
+        return (x2, h)
 
 
 class Encoder(nn.Module):
@@ -122,8 +142,9 @@ class Encoder(nn.Module):
             x = xp
         else:
             x = x + self.pos_embedding.to(x.dtype)
-        x = self.layers(x)
-        return self.ln(x)
+        u, v = self.layers((x, None))
+        y, _ = self.ln.forward_fused(u, v)
+        return y
 
 
 class VisionTransformer(nn.Module):

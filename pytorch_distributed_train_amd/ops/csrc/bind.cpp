@@ -73,11 +73,12 @@ void pdta_stem_im2col(const void*, void*, int, int, int, int, int, int,
 void pdta_gemm_nt_batched(const void*, const void*, void*, int, int, int,
                           int, long, long, long, int, int, hipStream_t);
 void pdta_ln_fwd(const void*, void*, const float*, const float*, float*,
-                 float*, long, int, float, int, hipStream_t);
+                 float*, long, int, float, int, const void*, void*,
+                 hipStream_t);
 int pdta_ln_chunks(long);
 void pdta_ln_bwd(const void*, const void*, void*, const float*,
                  const float*, const float*, float*, float*, float*, long,
-                 int, int, hipStream_t);
+                 int, int, const void*, hipStream_t);
 void pdta_gelu_fwd(const void*, void*, long, int, hipStream_t);
 void pdta_gelu_bwd(const void*, const void*, void*, long, int, hipStream_t);
 void pdta_softmax_fwd(const float*, void*, long, int, int, float, int,
@@ -655,7 +656,9 @@ at::Tensor gemm_nt_batched(at::Tensor a, at::Tensor b, bool out_f32) {
 }
 
 std::vector<at::Tensor> ln_fwd(at::Tensor x, at::Tensor w, at::Tensor b,
-                               double eps) {
+                               double eps,
+                               c10::optional<at::Tensor> res) {
+  // res: fused residual join — returns {ln(x+res), mean, invstd, x+res}
   CHECK_IN(x);
   int d = x.size(-1);
   long rows = x.numel() / d;
@@ -663,15 +666,24 @@ std::vector<at::Tensor> ln_fwd(at::Tensor x, at::Tensor w, at::Tensor b,
   auto opts = x.options().dtype(at::kFloat);
   auto mean = at::empty({rows}, opts), invstd = at::empty({rows}, opts);
   auto wf = w.to(at::kFloat).contiguous(), bf = b.to(at::kFloat).contiguous();
+  at::Tensor sum;
+  if (res.has_value()) {
+    TORCH_CHECK(res->is_contiguous() && res->sizes() == x.sizes(),
+                "ln_fwd res mismatch");
+    sum = at::empty_like(x);
+  }
   pdta_ln_fwd(x.data_ptr(), y.data_ptr(), wf.data_ptr<float>(),
               bf.data_ptr<float>(), mean.data_ptr<float>(),
               invstd.data_ptr<float>(), rows, d, eps, dt_of(x),
-              cur_stream());
+              res.has_value() ? res->data_ptr() : nullptr,
+              res.has_value() ? sum.data_ptr() : nullptr, cur_stream());
+  if (res.has_value()) return {y, mean, invstd, sum};
   return {y, mean, invstd};
 }
 
 std::vector<at::Tensor> ln_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
-                               at::Tensor mean, at::Tensor invstd) {
+                               at::Tensor mean, at::Tensor invstd,
+                               c10::optional<at::Tensor> dxadd) {
   CHECK_IN(x);
   CHECK_IN(dy);
   int d = x.size(-1);
@@ -683,10 +695,14 @@ std::vector<at::Tensor> ln_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
   auto part = at::empty({(long)(chunks + 16) * 2 * d}, opts);
   auto dw = at::empty({d}, opts), db = at::empty({d}, opts);
   auto wf = w.to(at::kFloat).contiguous();
+  if (dxadd.has_value())
+    TORCH_CHECK(dxadd->is_contiguous() && dxadd->sizes() == x.sizes(),
+                "ln_bwd dxadd mismatch");
   pdta_ln_bwd(x.data_ptr(), dy.data_ptr(), dx.data_ptr(),
               wf.data_ptr<float>(), mean.data_ptr<float>(),
               invstd.data_ptr<float>(), part.data_ptr<float>(),
               dw.data_ptr<float>(), db.data_ptr<float>(), rows, d, dt_of(x),
+              dxadd.has_value() ? dxadd->data_ptr() : nullptr,
               cur_stream());
   return {dx, dw, db};
 }
@@ -955,8 +971,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("out") = c10::nullopt);
   m.def("gemm_nt_batched", &gemm_nt_batched, py::arg("a"), py::arg("b"),
         py::arg("out_f32") = false);
-  m.def("ln_fwd", &ln_fwd);
-  m.def("ln_bwd", &ln_bwd);
+  m.def("ln_fwd", &ln_fwd, py::arg("x"), py::arg("w"), py::arg("b"),
+        py::arg("eps"), py::arg("res") = c10::nullopt);
+  m.def("ln_bwd", &ln_bwd, py::arg("x"), py::arg("dy"), py::arg("w"),
+        py::arg("mean"), py::arg("invstd"),
+        py::arg("dxadd") = c10::nullopt);
   m.def("gelu_fwd", &gelu_fwd);
   m.def("gelu_bwd", &gelu_bwd);
   m.def("attn_softmax_fwd", &attn_softmax_fwd);

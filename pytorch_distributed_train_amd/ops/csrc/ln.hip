@@ -14,13 +14,17 @@ constexpr int TPB = 256;  // 4 waves -> 4 rows per block
 // exactly: a fixed max-size version cost 128 VGPRs of dead partials and
 // single-wave occupancy — 6x off the HBM traffic floor in rocprof).
 // The row is loaded ONCE into registers and reused for the apply pass.
-template <typename T, int NG>
+template <typename T, int NG, bool RES = false>
 __global__ void ln_fwd_k(const T* __restrict__ x, T* __restrict__ y,
                          const float* __restrict__ w,
                          const float* __restrict__ b,
                          float* __restrict__ mean_out,
                          float* __restrict__ invstd_out, long rows, int d,
-                         float eps) {
+                         float eps, const T* __restrict__ res = nullptr,
+                         T* __restrict__ sum_out = nullptr) {
+  // RES: fused residual join ln(x + res), writing the (rounded) sum out
+  // too — replaces the ViT block's separate add kernel AND its autograd
+  // fork (the sum has exactly one consumer downstream)
   long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
   if (row >= rows) return;
   int lane = threadIdx.x & 63;
@@ -32,6 +36,13 @@ __global__ void ln_fwd_k(const T* __restrict__ x, T* __restrict__ y,
     int c = lane * 8 + g * 512;
     if (c < d) {
       xv[g] = load8(xr + c);
+      if (RES) {
+        Vec8<T> rv = load8(res + row * d + c);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          xv[g].v[j] = from_f32<T>(to_f32(xv[g].v[j]) + to_f32(rv.v[j]));
+        store8(sum_out + row * d + c, xv[g]);
+      }
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float f = to_f32(xv[g].v[j]);
@@ -78,13 +89,14 @@ __global__ void ln_fwd_k(const T* __restrict__ x, T* __restrict__ y,
 // per chunk, written to part [chunk][2d]; ln_bwd_final reduces chunks
 // with 64-col x 4-slice blocks (a d-threads-only finalize was one wave
 // chip-wide and chunk-serial).
-template <typename T, int NG>
+template <typename T, int NG, bool DXADD = false>
 __global__ void ln_bwd_k(const T* __restrict__ x, const T* __restrict__ dy,
                          T* __restrict__ dx, const float* __restrict__ w,
                          const float* __restrict__ mean,
                          const float* __restrict__ invstd,
                          float* __restrict__ part, long rows, int d,
-                         long rows_per_chunk) {
+                         long rows_per_chunk,
+                         const T* __restrict__ dxadd = nullptr) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* red = reinterpret_cast<float*>(smem);  // [4 waves][2d]
 
@@ -142,12 +154,18 @@ __global__ void ln_bwd_k(const T* __restrict__ x, const T* __restrict__ dy,
     for (int g = 0; g < NG; ++g) {
       int c = lane * 8 + g * 512;
       if (c < d) {
-        Vec8<T> o;
+        Vec8<T> o, av;
+        if (DXADD) av = load8(dxadd + row * d + c);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float gg = to_f32(gv[g].v[j]);
           float xh = (to_f32(xv[g].v[j]) - mu) * is;
-          o.v[j] = from_f32<T>(is * (gg * wreg[g][j] - s1 - xh * s2));
+          float oo = is * (gg * wreg[g][j] - s1 - xh * s2);
+          // DXADD: the downstream pass-through gradient of the fused
+          // residual sum, folded into the dx store (kills the autograd
+          // accumulation add at the fork)
+          if (DXADD) oo += to_f32(av.v[j]);
+          o.v[j] = from_f32<T>(oo);
           adw[g][j] += gg * xh;
           adb[g][j] += gg;
         }
@@ -461,12 +479,17 @@ extern "C" {
 
 void pdta_ln_fwd(const void* x, void* y, const float* w, const float* b,
                  float* mean, float* invstd, long rows, int d, float eps,
-                 int dt, hipStream_t s) {
+                 int dt, const void* res, void* sum_out, hipStream_t s) {
   long blocks = (rows + 3) / 4;
   PDTA_DISPATCH_DTYPE16(dt, T, {
     PDTA_LN_DISPATCH_NG(d, NG, {
-      ln_fwd_k<T, NG><<<(unsigned)blocks, TPB, 0, s>>>(
-          (const T*)x, (T*)y, w, b, mean, invstd, rows, d, eps);
+      if (res)
+        ln_fwd_k<T, NG, true><<<(unsigned)blocks, TPB, 0, s>>>(
+            (const T*)x, (T*)y, w, b, mean, invstd, rows, d, eps,
+            (const T*)res, (T*)sum_out);
+      else
+        ln_fwd_k<T, NG><<<(unsigned)blocks, TPB, 0, s>>>(
+            (const T*)x, (T*)y, w, b, mean, invstd, rows, d, eps);
     });
   });
 }
@@ -481,15 +504,20 @@ int pdta_ln_chunks(long rows) {
 void pdta_ln_bwd(const void* x, const void* dy, void* dx, const float* w,
                  const float* mean, const float* invstd, float* part,
                  float* dw, float* db, long rows, int d, int dt,
-                 hipStream_t s) {
+                 const void* dxadd, hipStream_t s) {
   int chunks = pdta_ln_chunks(rows);
   long rpc = (rows + chunks - 1) / chunks;
   int lds = 4 * 2 * d * sizeof(float);
   PDTA_DISPATCH_DTYPE16(dt, T, {
     PDTA_LN_DISPATCH_NG(d, NG, {
-      ln_bwd_k<T, NG><<<chunks, TPB, lds, s>>>(
-          (const T*)x, (const T*)dy, (T*)dx, w, mean, invstd, part, rows, d,
-          rpc);
+      if (dxadd)
+        ln_bwd_k<T, NG, true><<<chunks, TPB, lds, s>>>(
+            (const T*)x, (const T*)dy, (T*)dx, w, mean, invstd, part, rows,
+            d, rpc, (const T*)dxadd);
+      else
+        ln_bwd_k<T, NG><<<chunks, TPB, lds, s>>>(
+            (const T*)x, (const T*)dy, (T*)dx, w, mean, invstd, part, rows,
+            d, rpc);
     });
   });
   // two-level finalize for long chunk lists (part carries 16 extra rows —

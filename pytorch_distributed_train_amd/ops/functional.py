@@ -722,6 +722,74 @@ def layer_norm(x, weight, bias, eps: float = 1e-6, gslot_w=None,
     return LayerNormFn.apply(x, weight, bias, eps, gslot_w, gslot_b)
 
 
+
+class AddLayerNormFn(torch.autograd.Function):
+    """Fused residual join + LayerNorm: (y, s) = (ln(a+b), a+b).
+
+    The ViT block threads (residual, branch) PAIRS between sublayers so
+    the sum has exactly ONE consumer — no autograd fork add exists in
+    either direction. Backward folds the pass-through gradient ds into
+    the dx store (DXADD kernel variant) and returns the SAME combined
+    gradient for both inputs."""
+
+    @staticmethod
+    def forward(ctx, a, b, weight, bias, eps, gslot_w=None, gslot_b=None):
+        ctx.set_materialize_grads(False)
+        ctx.gslot_w, ctx.gslot_b = gslot_w, gslot_b
+        if a.is_cuda:
+            y, mean, invstd, s = ext().ln_fwd(a, weight, bias, eps,
+                                              res=b.contiguous())
+        else:
+            s = (a.float() + b.float()).to(a.dtype)
+            s32 = s.float()
+            mean = s32.mean(dim=-1)
+            var = s32.var(dim=-1, unbiased=False)
+            invstd = (var + eps).rsqrt()
+            y = ((s32 - mean.unsqueeze(-1)) * invstd.unsqueeze(-1)
+                 * weight.float() + bias.float()).to(a.dtype)
+        ctx.save_for_backward(s, weight, mean, invstd)
+        return y, s
+
+    @staticmethod
+    def backward(ctx, dy, ds):
+        s, weight, mean, invstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        if s.is_cuda:
+            dx, dw, db = ext().ln_bwd(
+                s, dy, weight, mean, invstd,
+                dxadd=ds.contiguous() if ds is not None else None)
+            if ctx.gslot_w is not None:
+                ctx.gslot_w[0].copy_(dw)
+                if ctx.gslot_w[1] is not None:
+                    ctx.gslot_w[1]()
+                dw = None
+            if ctx.gslot_b is not None:
+                ctx.gslot_b[0].copy_(db)
+                if ctx.gslot_b[1] is not None:
+                    ctx.gslot_b[1]()
+                db = None
+        else:
+            d = s.shape[-1]
+            s32, g32, w32 = s.float(), dy.float(), weight.float()
+            xhat = (s32 - mean.unsqueeze(-1)) * invstd.unsqueeze(-1)
+            gw = g32 * w32
+            s1 = gw.mean(dim=-1, keepdim=True)
+            s2 = (gw * xhat).mean(dim=-1, keepdim=True)
+            dx = (invstd.unsqueeze(-1) * (gw - s1 - xhat * s2))
+            if ds is not None:
+                dx = dx + ds.float()
+            dx = dx.to(s.dtype)
+            dims = tuple(range(s.dim() - 1))
+            dw = (g32 * xhat).sum(dim=dims)
+            db = g32.sum(dim=dims)
+        return dx, dx, dw, db, None, None, None
+
+
+def add_layer_norm(a, b, weight, bias, eps: float = 1e-6, gslot_w=None,
+                   gslot_b=None):
+    return AddLayerNormFn.apply(a, b, weight, bias, eps, gslot_w, gslot_b)
+
+
 class GELUFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x):
