@@ -20,11 +20,12 @@ void pdta_bn_stats(const void*, float*, float*, float*, float*, float*,
 int pdta_bn_chunks(long, int);
 void pdta_bn_apply(const void*, void*, const void*, const float*,
                    const float*, const float*, const float*, long, int, int,
-                   int, hipStream_t);
+                   int, unsigned char*, hipStream_t);
 void pdta_bn_bwd_reduce(const void*, const void*, const void*,
                         const float*, const float*, const float*,
                         const float*, int, float*, float*, float*, long,
-                        int, int, const void*, void*, hipStream_t);
+                        int, int, const void*, void*, const unsigned char*,
+                        hipStream_t);
 void pdta_bn_bwd_elemt(const void*, const void*, const void*, void*,
                        void*, const float*, const float*, int,
                        const float*, const float*, const float*,
@@ -190,9 +191,13 @@ std::vector<at::Tensor> bn_stats_train(at::Tensor x, at::Tensor running_mean,
   return {mean, invstd};
 }
 
-at::Tensor bn_apply(at::Tensor x, at::Tensor weight, at::Tensor bias,
-                    at::Tensor mean, at::Tensor invstd, bool relu,
-                    c10::optional<at::Tensor> res) {
+std::vector<at::Tensor> bn_apply_v(at::Tensor x, at::Tensor weight,
+                                   at::Tensor bias, at::Tensor mean,
+                                   at::Tensor invstd, bool relu,
+                                   c10::optional<at::Tensor> res,
+                                   bool want_mask) {
+  // want_mask (joins only): also return the 1-bit relu mask (1 byte per
+  // 8 elems) the backward reduce consumes instead of re-reading y
   CHECK_IN(x);
   int C = x.size(3);
   long rows = x.numel() / C;
@@ -200,12 +205,26 @@ at::Tensor bn_apply(at::Tensor x, at::Tensor weight, at::Tensor bias,
   auto w = weight.to(at::kFloat).contiguous();
   auto b = bias.to(at::kFloat).contiguous();
   auto m = mean.contiguous(), is = invstd.contiguous();
+  at::Tensor mask;
+  unsigned char* mp = nullptr;
+  if (want_mask) {
+    TORCH_CHECK(relu && res.has_value(), "bn_apply mask: joins only");
+    mask = at::empty({rows * C / 8}, x.options().dtype(at::kByte));
+    mp = mask.data_ptr<unsigned char>();
+  }
   pdta_bn_apply(x.data_ptr(), y.data_ptr(),
                 res.has_value() ? res->data_ptr() : nullptr,
                 w.data_ptr<float>(), b.data_ptr<float>(),
                 m.data_ptr<float>(), is.data_ptr<float>(), rows, C,
-                relu ? 1 : 0, dt_of(x), cur_stream());
-  return y;
+                relu ? 1 : 0, dt_of(x), mp, cur_stream());
+  if (want_mask) return {y, mask};
+  return {y};
+}
+
+at::Tensor bn_apply(at::Tensor x, at::Tensor weight, at::Tensor bias,
+                    at::Tensor mean, at::Tensor invstd, bool relu,
+                    c10::optional<at::Tensor> res) {
+  return bn_apply_v(x, weight, bias, mean, invstd, relu, res, false)[0];
 }
 
 std::vector<at::Tensor> bn_bwd_reduce(at::Tensor x, at::Tensor dy,
@@ -214,7 +233,8 @@ std::vector<at::Tensor> bn_bwd_reduce(at::Tensor x, at::Tensor dy,
                                       c10::optional<at::Tensor> invstd,
                                       c10::optional<at::Tensor> weight,
                                       c10::optional<at::Tensor> bias,
-                                      c10::optional<at::Tensor> dy2) {
+                                      c10::optional<at::Tensor> dy2,
+                                      c10::optional<at::Tensor> mbits) {
   // mask modes: relu_y given -> read it; else invstd/weight/bias given ->
   // recompute mask (gamma*xhat+beta > 0), skipping the relu_y pass;
   // else no mask
@@ -235,8 +255,8 @@ std::vector<at::Tensor> bn_bwd_reduce(at::Tensor x, at::Tensor dy,
     wf = weight->to(at::kFloat).contiguous();
     bf = bias->to(at::kFloat).contiguous();
   }
-  TORCH_CHECK(!dy2.has_value() || mask == 1,
-              "bn_bwd_reduce: dy2 requires the stored-relu mask");
+  TORCH_CHECK(!dy2.has_value() || mask == 1 || mbits.has_value(),
+              "bn_bwd_reduce: dy2 requires a relu mask");
   if (dy2.has_value())
     TORCH_CHECK(dy2->is_contiguous() && dy2->sizes() == dy.sizes() &&
                 dy2->scalar_type() == dy.scalar_type(),
@@ -251,6 +271,8 @@ std::vector<at::Tensor> bn_bwd_reduce(at::Tensor x, at::Tensor dy,
                      sum_dy_xmu.data_ptr<float>(), rows, C, dt_of(x),
                      dy2.has_value() ? dy2->data_ptr() : nullptr,
                      dy2.has_value() ? dres.data_ptr() : nullptr,
+                     mbits.has_value() ? mbits->data_ptr<unsigned char>()
+                                       : nullptr,
                      cur_stream());
   if (dy2.has_value()) return {sum_dy, sum_dy_xmu, dres};
   return {sum_dy, sum_dy_xmu};
@@ -922,13 +944,18 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("add_relu_fwd", &add_relu_fwd);
   m.def("bn_stats", &bn_stats);
   m.def("bn_stats_train", &bn_stats_train);
+  m.def("bn_apply_v", &bn_apply_v, py::arg("x"), py::arg("weight"),
+        py::arg("bias"), py::arg("mean"), py::arg("invstd"),
+        py::arg("relu"), py::arg("res") = c10::nullopt,
+        py::arg("want_mask") = false);
   m.def("bn_apply", &bn_apply, py::arg("x"), py::arg("weight"),
         py::arg("bias"), py::arg("mean"), py::arg("invstd"),
         py::arg("relu"), py::arg("res") = c10::nullopt);
   m.def("bn_bwd_reduce", &bn_bwd_reduce, py::arg("x"), py::arg("dy"),
         py::arg("relu_y"), py::arg("mean"),
         py::arg("invstd") = c10::nullopt, py::arg("weight") = c10::nullopt,
-        py::arg("bias") = c10::nullopt, py::arg("dy2") = c10::nullopt);
+        py::arg("bias") = c10::nullopt, py::arg("dy2") = c10::nullopt,
+        py::arg("mbits") = c10::nullopt);
   m.def("bn_bwd_elemt", &bn_bwd_elemt, py::arg("x"), py::arg("dy"),
         py::arg("relu_y"), py::arg("mean"), py::arg("invstd"),
         py::arg("weight"), py::arg("sum_dy"), py::arg("sum_dy_xmu"),

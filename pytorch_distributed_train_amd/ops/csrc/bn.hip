@@ -109,7 +109,11 @@ __global__ void bn_apply_k(const T* __restrict__ x, T* __restrict__ y,
                            const float* __restrict__ bias,
                            const float* __restrict__ mean,
                            const float* __restrict__ invstd, long rows,
-                           int c8) {  // c8 = C/8
+                           int c8,  // c8 = C/8
+                           unsigned char* __restrict__ mask_out = nullptr) {
+  // mask_out (joins only): 1 bit per element, bit j of byte (r*C+c0)/8 =
+  // (rounded output > 0) — the backward reduce reads 1/16 the bytes of
+  // the stored relu output it replaces
   int cpg = min(c8, TPB);
   int rpar = TPB / cpg;
   int cg = blockIdx.x * cpg + (int)(threadIdx.x % cpg);
@@ -149,6 +153,16 @@ __global__ void bn_apply_k(const T* __restrict__ x, T* __restrict__ y,
       va.v[j] = from_f32<T>(o0);
       vb.v[j] = from_f32<T>(o1);
     }
+    if (RELU && RES && mask_out) {
+      unsigned char m0 = 0, m1 = 0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        m0 |= (to_f32(va.v[j]) > 0.f) << j;
+        m1 |= (to_f32(vb.v[j]) > 0.f) << j;
+      }
+      mask_out[(r * C + c0) >> 3] = m0;
+      mask_out[(r2 * C + c0) >> 3] = m1;
+    }
     store8(y + r * C + c0, va);
     store8(y + r2 * C + c0, vb);
   }
@@ -163,6 +177,12 @@ __global__ void bn_apply_k(const T* __restrict__ x, T* __restrict__ y,
       if (RELU) o = fmaxf(o, 0.f);
       v.v[j] = from_f32<T>(o);
     }
+    if (RELU && RES && mask_out) {
+      unsigned char m = 0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) m |= (to_f32(v.v[j]) > 0.f) << j;
+      mask_out[(r * C + c0) >> 3] = m;
+    }
     store8(y + r * C + c0, v);
   }
 }
@@ -172,7 +192,9 @@ __global__ void bn_apply_k(const T* __restrict__ x, T* __restrict__ y,
 //    (read the stored relu output — required for residual joins, where the
 //    pre-activation includes the identity branch). MASK 2: recompute the
 //    mask as gamma*xhat+beta > 0 from x — one whole HBM pass over relu_y
-//    never happens (exact up to bf16 boundary rounding of y) -------------
+//    never happens (exact up to bf16 boundary rounding of y).
+//    MASK 3: 1-bit mask emitted by the join's bn_apply (mbits) — same
+//    semantics as MASK 1 at 1/16 the mask bytes ------------------------
 template <typename T, int MASK, bool DY2 = false>
 __global__ void bn_bwd_part_k(const T* __restrict__ x,
                               const T* __restrict__ dy,
@@ -184,7 +206,9 @@ __global__ void bn_bwd_part_k(const T* __restrict__ x,
                               float* __restrict__ part, long rows, int C,
                               int rows_per_block,
                               const T* __restrict__ dy2 = nullptr,
-                              T* __restrict__ dres_out = nullptr) {
+                              T* __restrict__ dres_out = nullptr,
+                              const unsigned char* __restrict__ mbits =
+                                  nullptr) {
   // DY2 also WRITES the masked combined dy (== the residual-path
   // gradient dres): the elemt pass then reads that single tensor with no
   // mask and no dres store — 8 HBM passes for the whole join backward
@@ -238,6 +262,13 @@ __global__ void bn_bwd_part_k(const T* __restrict__ x,
       g2 = to_f32(relu_y[(r + 2 * rpar) * C + c]) > 0.f ? g2 : 0.f;
       g3 = to_f32(relu_y[(r + 3 * rpar) * C + c]) > 0.f ? g3 : 0.f;
     }
+    if (MASK == 3) {
+      int bit = c & 7;
+      g0 = (mbits[(r * C + c) >> 3] >> bit) & 1 ? g0 : 0.f;
+      g1 = (mbits[((r + rpar) * C + c) >> 3] >> bit) & 1 ? g1 : 0.f;
+      g2 = (mbits[((r + 2 * rpar) * C + c) >> 3] >> bit) & 1 ? g2 : 0.f;
+      g3 = (mbits[((r + 3 * rpar) * C + c) >> 3] >> bit) & 1 ? g3 : 0.f;
+    }
     if (MASK == 2) {
       g0 = k1 * (v0 - m) + b0 > 0.f ? g0 : 0.f;
       g1 = k1 * (v1 - m) + b0 > 0.f ? g1 : 0.f;
@@ -259,6 +290,7 @@ __global__ void bn_bwd_part_k(const T* __restrict__ x,
     float v = to_f32(x[r * C + c]);
     if (MASK == 1) g = to_f32(relu_y[r * C + c]) > 0.f ? g : 0.f;
     if (MASK == 2) g = k1 * (v - m) + b0 > 0.f ? g : 0.f;
+    if (MASK == 3) g = (mbits[(r * C + c) >> 3] >> (c & 7)) & 1 ? g : 0.f;
     if constexpr (DY2) dres_out[r * C + c] = from_f32<T>(g);
     s += g;
     sx += g * (v - m);
@@ -579,14 +611,14 @@ int pdta_bn_chunks(long rows, int C) {
 void pdta_bn_apply(const void* x, void* y, const void* res,
                    const float* weight, const float* bias, const float* mean,
                    const float* invstd, long rows, int C, int relu, int dt,
-                   hipStream_t s) {
+                   unsigned char* mask_out, hipStream_t s) {
   dim3 g = ew_grid2(rows, C);
   PDTA_DISPATCH_DTYPE(dt, T, {
     if (res) {
       if (relu)
         bn_apply_k<T, true, true><<<g, TPB, 0, s>>>(
             (const T*)x, (T*)y, (const T*)res, weight, bias, mean, invstd,
-            rows, C / 8);
+            rows, C / 8, mask_out);
       else
         bn_apply_k<T, false, true><<<g, TPB, 0, s>>>(
             (const T*)x, (T*)y, (const T*)res, weight, bias, mean, invstd,
@@ -611,11 +643,22 @@ void pdta_bn_bwd_reduce(const void* x, const void* dy, const void* relu_y,
                         const float* weight, const float* bias, int mask,
                         float* part, float* sum_dy,
                         float* sum_dy_xmu, long rows, int C, int dt,
-                        const void* dy2, void* dres_out, hipStream_t s) {
+                        const void* dy2, void* dres_out,
+                        const unsigned char* mbits, hipStream_t s) {
   int rpb, chunks;
   dim3 g = reduce_grid(rows, C, &rpb, &chunks);
   PDTA_DISPATCH_DTYPE(dt, T, {
-    if (mask == 1 && dy2)
+    if (mbits && dy2)
+      // 1-bit join mask + two-dy: 1/16 the mask bytes of MASK 1
+      bn_bwd_part_k<T, 3, true><<<g, TPB, 0, s>>>(
+          (const T*)x, (const T*)dy, nullptr, mean, nullptr,
+          nullptr, nullptr, part, rows, C, rpb, (const T*)dy2,
+          (T*)dres_out, mbits);
+    else if (mbits)
+      bn_bwd_part_k<T, 3><<<g, TPB, 0, s>>>(
+          (const T*)x, (const T*)dy, nullptr, mean, nullptr, nullptr,
+          nullptr, part, rows, C, rpb, nullptr, nullptr, mbits);
+    else if (mask == 1 && dy2)
       // two-dy: residual-fork gradient summed on load, masked combined
       // dy written out as dres
       bn_bwd_part_k<T, 1, true><<<g, TPB, 0, s>>>(

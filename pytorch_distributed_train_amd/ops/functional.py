@@ -223,9 +223,16 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
         # element count (all_reduce of [sum_dy, sum_dy_xmu]).
         # `residual` fuses the ResNet join: y = relu(bn(x) + residual) in
         # one pass (requires fuse_relu).
+        mask = None
         if x.is_cuda:
-            y = ext().bn_apply(x, weight, bias, mean, invstd, fuse_relu,
-                               residual)
+            if fuse_relu and residual is not None:
+                # join: also emit the 1-bit relu mask (1/16 the backward
+                # mask bytes of re-reading y)
+                y, mask = ext().bn_apply_v(x, weight, bias, mean, invstd,
+                                           fuse_relu, residual, True)
+            else:
+                y = ext().bn_apply(x, weight, bias, mean, invstd,
+                                   fuse_relu, residual)
         else:
             x32 = x.float()
             xhat = (x32 - mean) * invstd
@@ -242,12 +249,12 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
         ctx.dres_sink = dres_sink
         ctx.dy2_source = dy2_source
         ctx.gslot_w, ctx.gslot_b = gslot_w, gslot_b
-        ctx.save_for_backward(x, weight, bias, mean, invstd, y)
+        ctx.save_for_backward(x, weight, bias, mean, invstd, y, mask)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, weight, bias, mean, invstd, y = ctx.saved_tensors
+        x, weight, bias, mean, invstd, y, mask = ctx.saved_tensors
         fuse_relu = ctx.fuse_relu
         dy = dy.contiguous()
         dy2 = None
@@ -268,10 +275,10 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
             relu_y = y if (fuse_relu and ctx.has_residual) else None
             premask = fuse_relu and not ctx.has_residual
             red = ext().bn_bwd_reduce(
-                x, dy, relu_y, mean,
+                x, dy, relu_y if mask is None else None, mean,
                 invstd=invstd if premask else None,
                 weight=weight if premask else None,
-                bias=bias if premask else None, dy2=dy2)
+                bias=bias if premask else None, dy2=dy2, mbits=mask)
             sum_dy, sum_dy_xmu = red[0], red[1]
             dy_comb = None
             if dy2 is not None:
