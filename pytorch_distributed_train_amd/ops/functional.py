@@ -225,9 +225,13 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
         # one pass (requires fuse_relu).
         mask = None
         if x.is_cuda:
-            if fuse_relu and residual is not None:
-                # join: also emit the 1-bit relu mask (1/16 the backward
-                # mask bytes of re-reading y)
+            import os as _os
+            if fuse_relu and residual is not None \
+                    and _os.environ.get("PDTA_BN_BITMASK"):
+                # 1-bit join relu mask: NEGATIVE result at B=256 (-2%
+                # step: the byte loads are 8x redundant across lanes and
+                # the emit costs more than the saved y pass) — kept
+                # opt-in for re-evaluation at other shapes
                 y, mask = ext().bn_apply_v(x, weight, bias, mean, invstd,
                                            fuse_relu, residual, True)
             else:
