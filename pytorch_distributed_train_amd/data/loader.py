@@ -2,10 +2,13 @@
 
 The reference uses torch DataLoader with worker processes + pinned memory
 (SURVEY.md §2b row 14). For synthetic data the cost is batch assembly
-(host randn), not disk IO, so the MI355X-native loader uses a background
-prefetch thread pool (torch ops release the GIL) filling a bounded queue
-of pinned batches; the trainer overlaps H2D copy with compute via
-non_blocking copies. workers=0 falls back to synchronous assembly.
+(host randn), not disk IO, so the MI355X-native loader defaults to a
+background prefetch thread pool (torch ops release the GIL) filling a
+bounded queue of pinned batches; the trainer overlaps H2D copy with
+compute via non_blocking copies. worker_mode="process" gives torch-
+DataLoader-style worker PROCESSES (fork, shared-memory tensor transport)
+for datasets whose assembly holds the GIL. workers=0 falls back to
+synchronous assembly.
 """
 from __future__ import annotations
 
@@ -15,9 +18,27 @@ import threading
 import torch
 
 
+def _proc_worker(dataset, batch_size, task_q, out_q):
+    """Worker-process loop: pull (index, idx-list) tasks, collate, ship
+    the batch back through a shared-memory queue. A None task stops."""
+    while True:
+        task = task_q.get()
+        if task is None:
+            return
+        bi, idxs = task
+        imgs = []
+        labels = torch.empty(len(idxs), dtype=torch.long)
+        for j, i in enumerate(idxs):
+            img, lab = dataset[i]
+            imgs.append(img)
+            labels[j] = lab
+        out_q.put((bi, torch.stack(imgs), labels))
+
+
 class Loader:
     def __init__(self, dataset, sampler, batch_size: int, pin_memory: bool = False,
-                 drop_last: bool = True, workers: int = 0, prefetch: int = 4):
+                 drop_last: bool = True, workers: int = 0, prefetch: int = 4,
+                 worker_mode: str = "thread"):
         self.dataset = dataset
         self.sampler = sampler
         self.batch_size = batch_size
@@ -25,6 +46,7 @@ class Loader:
         self.drop_last = drop_last
         self.workers = workers
         self.prefetch = max(prefetch, 2)
+        self.worker_mode = worker_mode
 
     def __len__(self) -> int:
         n = len(self.sampler)
@@ -60,7 +82,48 @@ class Loader:
             for idxs in self._batches():
                 yield self._collate(idxs)
             return
-        yield from self._iter_prefetch()
+        if self.worker_mode == "process":
+            yield from self._iter_procs()
+        else:
+            yield from self._iter_prefetch()
+
+    def _iter_procs(self):
+        """torch-DataLoader-style worker processes (reference row 2b#14):
+        fork context, tensors return through shared memory; batches are
+        re-ordered so every rank sees the identical sequence; pinning
+        happens in the parent (pinned pages don't survive pickling)."""
+        import torch.multiprocessing as mp
+        ctx = mp.get_context("fork")
+        task_q = ctx.Queue()
+        out_q = ctx.Queue()
+        batches = list(self._batches())
+        for bi, idxs in enumerate(batches):
+            task_q.put((bi, idxs))
+        for _ in range(self.workers):
+            task_q.put(None)
+        procs = [ctx.Process(target=_proc_worker,
+                             args=(self.dataset, self.batch_size, task_q,
+                                   out_q), daemon=True)
+                 for _ in range(self.workers)]
+        for p in procs:
+            p.start()
+        results = {}
+        try:
+            for next_bi in range(len(batches)):
+                while next_bi not in results:
+                    bi, x, labels = out_q.get()
+                    results[bi] = (x, labels)
+                x, labels = results.pop(next_bi)
+                if self.pin_memory:
+                    x = x.pin_memory()
+                    labels = labels.pin_memory()
+                yield x, labels
+        finally:
+            for p in procs:
+                if p.is_alive():
+                    p.terminate()
+            for p in procs:
+                p.join(timeout=5.0)
 
     def _iter_prefetch(self):
         batch_q: "queue.Queue" = queue.Queue()

@@ -48,7 +48,12 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--num-samples", type=int, default=0,
                    help="synthetic dataset size (0 = default per dataset)")
     p.add_argument("--workers", type=int, default=0,
-                   help="data loader worker processes (0 = synchronous)")
+                   help="data loader workers (0 = synchronous)")
+    p.add_argument("--worker-mode", default="thread",
+                   choices=("thread", "process"),
+                   help="prefetch via GIL-releasing threads (synthetic "
+                        "default) or torch-DataLoader-style worker "
+                        "processes")
     p.add_argument("--bucket-cap-mb", type=int, default=50,
                    help="DDP gradient bucket size; tuned for 7-link xGMI "
                         "(fewer, larger buckets than NVSwitch defaults)")

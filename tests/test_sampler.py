@@ -78,3 +78,19 @@ def test_prefetch_loader_matches_sync():
     for (x1, y1), (x2, y2) in zip(sync, pref):
         assert torch.equal(x1, x2)
         assert torch.equal(y1, y2)
+
+
+def test_process_worker_loader_matches_sync():
+    """worker_mode=process (torch-DataLoader-style worker processes,
+    2b#14 parity): identical batch sequence to the synchronous loader."""
+    from pytorch_distributed_train_amd.data import (DistributedSampler,
+                                                    Loader, make_dataset)
+    import torch
+    ds = make_dataset("synthetic-cifar10", 64, seed=3)
+    s1 = DistributedSampler(ds, 1, 0, shuffle=True, seed=5)
+    s2 = DistributedSampler(ds, 1, 0, shuffle=True, seed=5)
+    sync = list(Loader(ds, s1, 8, workers=0))
+    proc = list(Loader(ds, s2, 8, workers=3, worker_mode="process"))
+    assert len(sync) == len(proc) > 0
+    for (xa, ya), (xb, yb) in zip(sync, proc):
+        assert torch.equal(xa, xb) and torch.equal(ya, yb)

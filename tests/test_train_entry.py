@@ -10,6 +10,8 @@ import torch
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import train as train_mod  # repo-root entrypoint
 
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
 
 def _argv(tmp_path, extra=()):
     return ["--model", "resnet18", "--data", "synthetic-cifar10",
@@ -52,3 +54,24 @@ def test_lr_schedules():
         "--lr", "1.0", "--epochs", "10", "--lr-schedule", "cosine"])
     assert abs(train_mod.lr_at(a, 0) - 1.0) < 1e-9
     assert train_mod.lr_at(a, 9) < 0.05
+
+
+def test_bench_under_torchrun_cpu(tmp_path):
+    """The driver launches bench.py via `python -m torch.distributed.run
+    --nnodes=1 --nproc-per-node N --master-addr 127.0.0.1 bench.py
+    --gpus N ...` — verify that exact invocation produces an n_gpus=N
+    record (CPU/gloo here; RCCL on the GPU node)."""
+    import json
+    import subprocess
+    import sys
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29741", "bench.py", "--gpus", "2",
+         "--steps", "1", "--warmup", "0", "--model", "resnet18",
+         "--batch-size", "2"],
+        cwd=REPO, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    rec = json.loads(line)
+    assert rec["n_gpus"] == 2 and rec["config"]["parallelism"] == "dp2"

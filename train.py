@@ -93,7 +93,7 @@ def main(argv=None):
                                  shuffle=True, seed=args.seed)
     loader = Loader(dataset, sampler, args.batch_size,
                     pin_memory=(trainer.device.type == "cuda"),
-                    workers=args.workers)
+                    workers=args.workers, worker_mode=args.worker_mode)
 
     start_epoch = 0
     if args.resume:
