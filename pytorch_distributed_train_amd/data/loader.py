@@ -99,8 +99,9 @@ class Loader:
         batches = list(self._batches())
         for bi, idxs in enumerate(batches):
             task_q.put((bi, idxs))
-        for _ in range(self.workers):
-            task_q.put(None)
+        # no stop sentinels: a worker that exits before the consumer maps
+        # its shared-memory FDs would EOF the queue — workers idle on
+        # task_q.get() until the finally terminates them
         procs = [ctx.Process(target=_proc_worker,
                              args=(self.dataset, self.batch_size, task_q,
                                    out_q), daemon=True)
