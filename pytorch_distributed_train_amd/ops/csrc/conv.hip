@@ -1343,6 +1343,53 @@ __global__ void stem_im2col_k(const T* __restrict__ x, T* __restrict__ out,
   }
 }
 
+// v2: for [r][s][c]-ordered patches over NHWC input, the S*Cin elements
+// of one tap-row are CONTIGUOUS in x — copy them as a run instead of
+// decoding every element (the v1 per-element decode ran 7.5x off the
+// write-traffic floor). Task R zeroes the rsc..rsc_pad tail (the padded
+// GEMM K columns must be 0, not garbage: 0*NaN poisons the output).
+template <typename T>
+__global__ void stem_im2col2_k(const T* __restrict__ x, T* __restrict__ out,
+                               int Nn, int H, int W, int Cin, int P, int Q,
+                               int R, int S, int stride, int pad,
+                               int rsc_pad, unsigned mq, int sq,
+                               unsigned mpq, int spq) {
+  int L = S * Cin;
+  int rsc = R * S * Cin;
+  int tasks = R + 1;
+  long total = (long)Nn * P * Q * tasks;
+  long item = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long gstride = (long)gridDim.x * blockDim.x;
+  for (; item < total; item += gstride) {
+    int task = (int)(item % tasks);
+    unsigned row = (unsigned)(item / tasks);
+    unsigned npq = mdiv(row, mpq, spq);
+    unsigned rem = row - npq * (unsigned)(P * Q);
+    unsigned p = mdiv(rem, mq, sq);
+    unsigned q = rem - p * (unsigned)Q;
+    T* orow = out + (long)row * rsc_pad;
+    if (task == R) {
+      for (int e = rsc; e < rsc_pad; ++e) orow[e] = from_f32<T>(0.f);
+      continue;
+    }
+    int h = (int)p * stride - pad + task;
+    int w0 = (int)q * stride - pad;
+    T* dst = orow + task * L;
+    if ((unsigned)h < (unsigned)H && w0 >= 0 && w0 + S <= W) {
+      const T* src = x + (((long)npq * H + h) * W + w0) * Cin;
+      for (int e = 0; e < L; ++e) dst[e] = src[e];
+    } else {
+      for (int s2 = 0; s2 < S; ++s2) {
+        int w = w0 + s2;
+        bool ok = (unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W;
+        const T* src = x + (((long)npq * H + h) * W + w) * Cin;
+        for (int c = 0; c < Cin; ++c)
+          dst[s2 * Cin + c] = ok ? src[c] : from_f32<T>(0.f);
+      }
+    }
+  }
+}
+
 }  // namespace
 
 extern "C" {
@@ -1463,53 +1510,6 @@ void pdta_conv_bwd_data(const void* dy, const void* w2, void* dx,
 #undef PDTA_BWD_SMALL
 }
 
-
-// v2: for [r][s][c]-ordered patches over NHWC input, the S*Cin elements
-// of one tap-row are CONTIGUOUS in x — copy them as a run instead of
-// decoding every element (the v1 per-element decode ran 7.5x off the
-// write-traffic floor). Task R zeroes the rsc..rsc_pad tail (the padded
-// GEMM K columns must be 0, not garbage: 0*NaN poisons the output).
-template <typename T>
-__global__ void stem_im2col2_k(const T* __restrict__ x, T* __restrict__ out,
-                               int Nn, int H, int W, int Cin, int P, int Q,
-                               int R, int S, int stride, int pad,
-                               int rsc_pad, unsigned mq, int sq,
-                               unsigned mpq, int spq) {
-  int L = S * Cin;
-  int rsc = R * S * Cin;
-  int tasks = R + 1;
-  long total = (long)Nn * P * Q * tasks;
-  long item = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  long gstride = (long)gridDim.x * blockDim.x;
-  for (; item < total; item += gstride) {
-    int task = (int)(item % tasks);
-    unsigned row = (unsigned)(item / tasks);
-    unsigned npq = mdiv(row, mpq, spq);
-    unsigned rem = row - npq * (unsigned)(P * Q);
-    unsigned p = mdiv(rem, mq, sq);
-    unsigned q = rem - p * (unsigned)Q;
-    T* orow = out + (long)row * rsc_pad;
-    if (task == R) {
-      for (int e = rsc; e < rsc_pad; ++e) orow[e] = from_f32<T>(0.f);
-      continue;
-    }
-    int h = (int)p * stride - pad + task;
-    int w0 = (int)q * stride - pad;
-    T* dst = orow + task * L;
-    if ((unsigned)h < (unsigned)H && w0 >= 0 && w0 + S <= W) {
-      const T* src = x + (((long)npq * H + h) * W + w0) * Cin;
-      for (int e = 0; e < L; ++e) dst[e] = src[e];
-    } else {
-      for (int s2 = 0; s2 < S; ++s2) {
-        int w = w0 + s2;
-        bool ok = (unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W;
-        const T* src = x + (((long)npq * H + h) * W + w) * Cin;
-        for (int c = 0; c < Cin; ++c)
-          dst[s2 * Cin + c] = ok ? src[c] : from_f32<T>(0.f);
-      }
-    }
-  }
-}
 
 int pdta_wrw_use_big(int Nn, int P, int Q, int Kout, int C, int R, int S) {
   if (Kout % 256 != 0 || C % 256 != 0) return 0;
