@@ -233,6 +233,109 @@ __global__ void bn_bwd_part_k(const T* __restrict__ x,
   // 4 rows per iteration: keeps 8-12 loads in flight per thread (the
   // rolled loop was ~2x off the traffic floor, latency-bound)
   long r = r0;
+  // DY2: 8 rows per iteration — the 4-stream two-dy variant was
+  // ~2x off its traffic floor at 4 rows in flight
+  if (DY2)
+    for (; r + 7 * rpar < r_end; r += 8 * rpar) {
+    float g0 = to_f32(dy[r * C + c]);
+    float g1 = to_f32(dy[(r + rpar) * C + c]);
+    float g2 = to_f32(dy[(r + 2 * rpar) * C + c]);
+    float g3 = to_f32(dy[(r + 3 * rpar) * C + c]);
+    float h0, h1, h2, h3;
+    if constexpr (DY2) {
+      // issue all four dy2 loads before any arithmetic consumes them
+      h0 = to_f32(dy2[r * C + c]);
+      h1 = to_f32(dy2[(r + rpar) * C + c]);
+      h2 = to_f32(dy2[(r + 2 * rpar) * C + c]);
+      h3 = to_f32(dy2[(r + 3 * rpar) * C + c]);
+    }
+    float v0 = to_f32(x[r * C + c]);
+    float v1 = to_f32(x[(r + rpar) * C + c]);
+    float v2 = to_f32(x[(r + 2 * rpar) * C + c]);
+    float v3 = to_f32(x[(r + 3 * rpar) * C + c]);
+    if constexpr (DY2) {
+      g0 += h0;
+      g1 += h1;
+      g2 += h2;
+      g3 += h3;
+    }
+    if (MASK == 1) {
+      g0 = to_f32(relu_y[r * C + c]) > 0.f ? g0 : 0.f;
+      g1 = to_f32(relu_y[(r + rpar) * C + c]) > 0.f ? g1 : 0.f;
+      g2 = to_f32(relu_y[(r + 2 * rpar) * C + c]) > 0.f ? g2 : 0.f;
+      g3 = to_f32(relu_y[(r + 3 * rpar) * C + c]) > 0.f ? g3 : 0.f;
+    }
+    if (MASK == 3) {
+      int bit = c & 7;
+      g0 = (mbits[(r * C + c) >> 3] >> bit) & 1 ? g0 : 0.f;
+      g1 = (mbits[((r + rpar) * C + c) >> 3] >> bit) & 1 ? g1 : 0.f;
+      g2 = (mbits[((r + 2 * rpar) * C + c) >> 3] >> bit) & 1 ? g2 : 0.f;
+      g3 = (mbits[((r + 3 * rpar) * C + c) >> 3] >> bit) & 1 ? g3 : 0.f;
+    }
+    if (MASK == 2) {
+      g0 = k1 * (v0 - m) + b0 > 0.f ? g0 : 0.f;
+      g1 = k1 * (v1 - m) + b0 > 0.f ? g1 : 0.f;
+      g2 = k1 * (v2 - m) + b0 > 0.f ? g2 : 0.f;
+      g3 = k1 * (v3 - m) + b0 > 0.f ? g3 : 0.f;
+    }
+    if constexpr (DY2) {
+      dres_out[r * C + c] = from_f32<T>(g0);
+      dres_out[(r + rpar) * C + c] = from_f32<T>(g1);
+      dres_out[(r + 2 * rpar) * C + c] = from_f32<T>(g2);
+      dres_out[(r + 3 * rpar) * C + c] = from_f32<T>(g3);
+    }
+    s += g0 + g1 + g2 + g3;
+    sx += g0 * (v0 - m) + g1 * (v1 - m) + g2 * (v2 - m) + g3 * (v3 - m);
+      float g0b = to_f32(dy[(r + 4 * rpar) * C + c]);
+    float g1b = to_f32(dy[(r + 5 * rpar) * C + c]);
+    float g2b = to_f32(dy[(r + 6 * rpar) * C + c]);
+    float g3b = to_f32(dy[(r + 7 * rpar) * C + c]);
+    float h0b, h1b, h2b, h3b;
+    if constexpr (DY2) {
+      // issue all four dy2 loads before any arithmetic consumes them
+      h0b = to_f32(dy2[(r + 4 * rpar) * C + c]);
+      h1b = to_f32(dy2[(r + 5 * rpar) * C + c]);
+      h2b = to_f32(dy2[(r + 6 * rpar) * C + c]);
+      h3b = to_f32(dy2[(r + 7 * rpar) * C + c]);
+    }
+    float v0b = to_f32(x[(r + 4 * rpar) * C + c]);
+    float v1b = to_f32(x[(r + 5 * rpar) * C + c]);
+    float v2b = to_f32(x[(r + 6 * rpar) * C + c]);
+    float v3b = to_f32(x[(r + 7 * rpar) * C + c]);
+    if constexpr (DY2) {
+      g0b += h0b;
+      g1b += h1b;
+      g2b += h2b;
+      g3b += h3b;
+    }
+    if (MASK == 1) {
+      g0b = to_f32(relu_y[(r + 4 * rpar) * C + c]) > 0.f ? g0b : 0.f;
+      g1b = to_f32(relu_y[(r + 5 * rpar) * C + c]) > 0.f ? g1b : 0.f;
+      g2b = to_f32(relu_y[(r + 6 * rpar) * C + c]) > 0.f ? g2b : 0.f;
+      g3b = to_f32(relu_y[(r + 7 * rpar) * C + c]) > 0.f ? g3b : 0.f;
+    }
+    if (MASK == 3) {
+      int bit = c & 7;
+      g0b = (mbits[((r + 4 * rpar) * C + c) >> 3] >> bit) & 1 ? g0b : 0.f;
+      g1b = (mbits[((r + 5 * rpar) * C + c) >> 3] >> bit) & 1 ? g1b : 0.f;
+      g2b = (mbits[((r + 6 * rpar) * C + c) >> 3] >> bit) & 1 ? g2b : 0.f;
+      g3b = (mbits[((r + 7 * rpar) * C + c) >> 3] >> bit) & 1 ? g3b : 0.f;
+    }
+    if (MASK == 2) {
+      g0b = k1 * (v0b - m) + b0 > 0.f ? g0b : 0.f;
+      g1b = k1 * (v1b - m) + b0 > 0.f ? g1b : 0.f;
+      g2b = k1 * (v2b - m) + b0 > 0.f ? g2b : 0.f;
+      g3b = k1 * (v3b - m) + b0 > 0.f ? g3b : 0.f;
+    }
+    if constexpr (DY2) {
+      dres_out[(r + 4 * rpar) * C + c] = from_f32<T>(g0b);
+      dres_out[(r + 5 * rpar) * C + c] = from_f32<T>(g1b);
+      dres_out[(r + 6 * rpar) * C + c] = from_f32<T>(g2b);
+      dres_out[(r + 7 * rpar) * C + c] = from_f32<T>(g3b);
+    }
+    s += g0b + g1b + g2b + g3b;
+    sx += g0b * (v0b - m) + g1b * (v1b - m) + g2b * (v2b - m) + g3b * (v3b - m);
+      }
   for (; r + 3 * rpar < r_end; r += 4 * rpar) {
     float g0 = to_f32(dy[r * C + c]);
     float g1 = to_f32(dy[(r + rpar) * C + c]);
