@@ -9,7 +9,7 @@ from __future__ import annotations
 
 import argparse
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 
 MODEL_CHOICES = ("resnet18", "resnet34", "resnet50", "resnet101",
