@@ -118,7 +118,13 @@ class Trainer:
         module whose backward can write its parameter gradient straight
         into flat storage — no AccumulateGrad add kernels (161 small
         launches/step on ResNet-50), no zero+add HBM round trip. The
-        notify keeps the DDP bucket accounting exact (param_ready)."""
+        notify keeps the DDP bucket accounting exact (param_ready).
+
+        ASSUMES each wired parameter contributes exactly ONE gradient
+        per step (true for every model here: no weight sharing). A
+        double contribution would overwrite instead of accumulate — the
+        reducer raises loudly on a duplicate arrival, so a future
+        weight-shared model fails fast rather than training wrong."""
         from .models.layers import (BatchNorm2dNHWC, Conv2dNHWC, Linear)
         from .models.vit import LayerNorm as VitLayerNorm
 
