@@ -511,6 +511,139 @@ __global__ void colsum_f32_k(const T* __restrict__ dy, float* __restrict__ db,
 }
 
 // ---- v2 experimental: 256x128 tile, BK=32, 8 waves (4x2) ---------------
+
+// ---- half-tile pipelined 256x256 (PDTA_GEMM_8P2): the counted-vmcnt
+// schedule the 2-slot structure cannot express. Halves (128 rows) of
+// each operand stage into per-half slots with staggered lifetimes:
+//   - B-halves of tile t+2 stage at t's q1/q2 (B fragments are
+//     register-resident after each tile's q0, so B slots free early);
+//   - A-halves of tile t+1 stage at t's q0 (A slots free after q3).
+// The tile-boundary wait is then s_waitcnt vmcnt(4) — never 0: loads
+// stay in flight across every barrier (the documented ~20% stall of
+// draining schedules).
+template <typename T, typename OutT, bool BIAS>
+__global__ __launch_bounds__(512) void gemm_nt_8p2_k(
+    const T* __restrict__ A, const T* __restrict__ B, OutT* __restrict__ C,
+    const float* __restrict__ bias, int M, int N, int K, int tiles_n,
+    long batch_sa, long batch_sb, long batch_sc) {
+  constexpr int HB = 128 * 64 * 2;  // 16 KiB per half
+  // [slot][A0,A1,B0,B1]
+  __shared__ __attribute__((aligned(16))) char lds[2][4][HB];
+
+  A += (long)blockIdx.y * batch_sa;
+  B += (long)blockIdx.y * batch_sb;
+  C += (long)blockIdx.y * batch_sc;
+
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  int tm = bid / tiles_n, tn = bid % tiles_n;
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  int wm = wave >> 2, wn = wave & 3;  // 2 x 4
+  int m0 = tm * 256, n0 = tn * 256;
+  int ktiles = K / 64;
+
+  // stage one 128-row half (2 glds/thread): op 0 = A rows, 1 = B rows
+  auto stage_half = [&](int kt, int op, int h) {
+    int base0 = op ? n0 : m0;
+    const T* src = op ? B : A;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int lrow = j * 64 + wave * 8 + (lane >> 3);
+      int grow = base0 + h * 128 + lrow;
+      int gmax = op ? N - 1 : M - 1;
+      if (grow > gmax) grow = gmax;
+      const char* g = reinterpret_cast<const char*>(src) +
+                      ((long)grow * K + kt * 64) * 2 +
+                      (((lane & 7) ^ (lrow & 7)) * 16);
+      glds16(g, (char*)lds[kt & 1][op * 2 + h] + (j * 64 + wave * 8) * 128);
+    }
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // prologue: tile 0 fully + tile 1's B halves (A(1) stages at t0 q0)
+  stage_half(0, 0, 0);
+  stage_half(0, 0, 1);
+  stage_half(0, 1, 0);
+  stage_half(0, 1, 1);
+  if (ktiles > 1) {
+    stage_half(1, 1, 0);
+    stage_half(1, 1, 1);
+  }
+
+  for (int kt = 0; kt < ktiles; ++kt) {
+    // boundary: allow the 4 most recent loads (2 staged halves) to stay
+    // in flight — everything this tile reads landed earlier
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    const char* la = lds[kt & 1][wm];
+    const char* lb = lds[kt & 1][2 + (wn >> 1)];
+    int bl = (wn & 1) * 64;
+    s16x8 bf[2][4];
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      if (q == 0) {
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            bf[kk][j] = frag_read128(lb, bl + j * 16 + (lane & 15),
+                                     (lane >> 4) + kk * 4);
+      }
+      s16x8 af[2][2];
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+          af[kk][i] = frag_read128(la, q * 32 + i * 16 + (lane & 15),
+                                   (lane >> 4) + kk * 4);
+      // staggered staging: q0 -> A halves of kt+1; q1/q2 -> B halves of
+      // kt+2 (their slots freed by this tile's q0 register loads)
+      if (q == 0 && kt + 1 < ktiles) {
+        stage_half(kt + 1, 0, 0);
+        stage_half(kt + 1, 0, 1);
+      } else if (q == 1 && kt + 2 < ktiles) {
+        stage_half(kt + 2, 1, 0);
+      } else if (q == 2 && kt + 2 < ktiles) {
+        stage_half(kt + 2, 1, 1);
+      }
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[q * 2 + i][j] =
+                MfmaOp<T>::run(af[kk][i], bf[kk][j], acc[q * 2 + i][j]);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    long rbase = m0 + wm * 128 + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = n0 + wn * 64 + j * 16 + (lane & 15);
+      if (col >= N) continue;
+      float badd = BIAS ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = rbase + r;
+        if (row < M)
+          C[(long)row * N + col] = from_f32<OutT>(acc[i][j][r] + badd);
+      }
+    }
+  }
+}
+
 // Occupancy experiment (docs/DESIGN.md §6): acc 64 VGPR/thread and
 // 2x24 KiB ring slots -> 3 blocks/CU by LDS, ~4 waves/SIMD by VGPR
 // (the 256x256/BK=64 kernel is pinned at 2 waves/SIMD and 52% WAIT).
@@ -673,6 +806,24 @@ void pdta_gemm_nt(const void* A, const void* B, void* C, const float* bias,
   if (use_8p < 0) use_8p = getenv("PDTA_GEMM_8P") != nullptr;
   static int use_r4 = -1;
   if (use_r4 < 0) use_r4 = getenv("PDTA_GEMM_R4") != nullptr;
+  static int use_8p2 = -1;
+  if (use_8p2 < 0) use_8p2 = getenv("PDTA_GEMM_8P2") != nullptr;
+  if (use_8p2 && M >= 1024 && N >= 256 && K % 64 == 0 && K >= 256 &&
+      !out_f32) {
+    int tiles_m = (M + 255) / 256, tiles_n = (N + 255) / 256;
+    dim3 grid(tiles_m * tiles_n);
+    PDTA_DISPATCH_DTYPE16(dt, T, {
+      if (bias)
+        gemm_nt_8p2_k<T, T, true><<<grid, 512, 0, s>>>(
+            (const T*)A, (const T*)B, (T*)C, bias, M, N, K, tiles_n, 0, 0,
+            0);
+      else
+        gemm_nt_8p2_k<T, T, false><<<grid, 512, 0, s>>>(
+            (const T*)A, (const T*)B, (T*)C, nullptr, M, N, K, tiles_n, 0,
+            0, 0);
+    });
+    return;
+  }
   if (use_r4 && M >= 1024 && N >= 256 && K % 32 == 0 && !out_f32) {
     int tiles_m = (M + 255) / 256, tiles_n = (N + 255) / 256;
     dim3 grid(tiles_m * tiles_n);
