@@ -512,15 +512,18 @@ __global__ void colsum_f32_k(const T* __restrict__ dy, float* __restrict__ db,
 
 // ---- v2 experimental: 256x128 tile, BK=32, 8 waves (4x2) ---------------
 
-// ---- half-tile pipelined 256x256 (PDTA_GEMM_8P2): the counted-vmcnt
-// schedule the 2-slot structure cannot express. Halves (128 rows) of
-// each operand stage into per-half slots with staggered lifetimes:
+// ---- half-tile pipelined 256x256 (PDTA_GEMM_8P2) — EXPERIMENTAL, DO
+// NOT ENABLE IN PRODUCTION. Measured: no speedup over the 2-slot
+// schedule (per-phase barrier cost cancels the counted-vmcnt win) AND
+// an intermittent numerics deviation at K=4096 (suspected staging
+// lifetime race not yet localized). Kept as the starting point for the
+// next schedule investigation; see docs/DESIGN.md §5/§6.
+// Design: halves (128 rows) of each operand stage into per-half slots
+// with staggered lifetimes:
 //   - B-halves of tile t+2 stage at t's q1/q2 (B fragments are
 //     register-resident after each tile's q0, so B slots free early);
 //   - A-halves of tile t+1 stage at t's q0 (A slots free after q3).
-// The tile-boundary wait is then s_waitcnt vmcnt(4) — never 0: loads
-// stay in flight across every barrier (the documented ~20% stall of
-// draining schedules).
+// The tile-boundary wait is then s_waitcnt vmcnt(4) — never 0.
 template <typename T, typename OutT, bool BIAS>
 __global__ __launch_bounds__(512) void gemm_nt_8p2_k(
     const T* __restrict__ A, const T* __restrict__ B, OutT* __restrict__ C,
