@@ -234,7 +234,10 @@ std::vector<at::Tensor> bn_bwd_reduce(at::Tensor x, at::Tensor dy,
                                       c10::optional<at::Tensor> weight,
                                       c10::optional<at::Tensor> bias,
                                       c10::optional<at::Tensor> dy2,
-                                      c10::optional<at::Tensor> mbits) {
+                                      c10::optional<at::Tensor> mbits,
+                                      c10::optional<at::Tensor> sum_dy_out) {
+  // sum_dy_out: direct-grad destination (the BN bias flat slice) — the
+  // finalize kernel writes dbias in place, no D2D copy afterwards
   // mask modes: relu_y given -> read it; else invstd/weight/bias given ->
   // recompute mask (gamma*xhat+beta > 0), skipping the relu_y pass;
   // else no mask
@@ -245,7 +248,10 @@ std::vector<at::Tensor> bn_bwd_reduce(at::Tensor x, at::Tensor dy,
   auto opts = x.options().dtype(at::kFloat);
   int chunks = pdta_bn_chunks(rows, C);
   auto part = at::empty({(long)chunks * 2 * C}, opts);
-  auto sum_dy = at::empty({C}, opts), sum_dy_xmu = at::empty({C}, opts);
+  at::Tensor sum_dy = sum_dy_out.has_value()
+                          ? sum_dy_out->view({C})
+                          : at::empty({C}, opts);
+  auto sum_dy_xmu = at::empty({C}, opts);
   int mask = relu_y.has_value() ? 1 : (bias.has_value() ? 2 : 0);
   at::Tensor dres;  // two-dy: masked combined dy, written by the kernel
   if (dy2.has_value()) dres = at::empty_like(dy);
@@ -474,13 +480,16 @@ at::Tensor gemm_nt(at::Tensor a, at::Tensor b,
   return y;
 }
 
-at::Tensor colsum_f32(at::Tensor dy) {
+at::Tensor colsum_f32(at::Tensor dy, c10::optional<at::Tensor> out) {
   CHECK_IN(dy);
   int N = dy.size(1);
   int ws = pdta_colsum_ws(N);
   auto opts = dy.options().dtype(at::kFloat);
   // scalar fallback (ws == 0) atomics into db and needs it zeroed
-  auto db = ws ? at::empty({N}, opts) : at::zeros({N}, opts);
+  at::Tensor db = out.has_value()
+                      ? (ws ? out->view({N})
+                            : out->view({N}).zero_())
+                      : (ws ? at::empty({N}, opts) : at::zeros({N}, opts));
   at::Tensor part;
   float* pp = nullptr;
   if (ws) {
@@ -705,7 +714,9 @@ std::vector<at::Tensor> ln_fwd(at::Tensor x, at::Tensor w, at::Tensor b,
 
 std::vector<at::Tensor> ln_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
                                at::Tensor mean, at::Tensor invstd,
-                               c10::optional<at::Tensor> dxadd) {
+                               c10::optional<at::Tensor> dxadd,
+                               c10::optional<at::Tensor> dw_out,
+                               c10::optional<at::Tensor> db_out) {
   CHECK_IN(x);
   CHECK_IN(dy);
   int d = x.size(-1);
@@ -715,7 +726,10 @@ std::vector<at::Tensor> ln_bwd(at::Tensor x, at::Tensor dy, at::Tensor w,
   int chunks = pdta_ln_chunks(rows);
   // +16 rows: scratch for the kernel-side two-level finalize
   auto part = at::empty({(long)(chunks + 16) * 2 * d}, opts);
-  auto dw = at::empty({d}, opts), db = at::empty({d}, opts);
+  at::Tensor dw = dw_out.has_value() ? dw_out->view({d})
+                                     : at::empty({d}, opts);
+  at::Tensor db = db_out.has_value() ? db_out->view({d})
+                                     : at::empty({d}, opts);
   auto wf = w.to(at::kFloat).contiguous();
   if (dxadd.has_value())
     TORCH_CHECK(dxadd->is_contiguous() && dxadd->sizes() == x.sizes(),
@@ -955,7 +969,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("relu_y"), py::arg("mean"),
         py::arg("invstd") = c10::nullopt, py::arg("weight") = c10::nullopt,
         py::arg("bias") = c10::nullopt, py::arg("dy2") = c10::nullopt,
-        py::arg("mbits") = c10::nullopt);
+        py::arg("mbits") = c10::nullopt,
+        py::arg("sum_dy_out") = c10::nullopt);
   m.def("bn_bwd_elemt", &bn_bwd_elemt, py::arg("x"), py::arg("dy"),
         py::arg("relu_y"), py::arg("mean"), py::arg("invstd"),
         py::arg("weight"), py::arg("sum_dy"), py::arg("sum_dy_xmu"),
@@ -981,7 +996,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_ce_bwd", &softmax_ce_bwd);
   m.def("gemm_nt", &gemm_nt, py::arg("a"), py::arg("b"),
         py::arg("bias") = c10::nullopt, py::arg("out_f32") = false);
-  m.def("colsum_f32", &colsum_f32);
+  m.def("colsum_f32", &colsum_f32, py::arg("dy"),
+        py::arg("out") = py::none());
   m.def("conv2d_fwd", &conv2d_fwd, py::arg("x"), py::arg("wt"),
         py::arg("stride"), py::arg("pad"),
         py::arg("want_bn_stats") = false, py::arg("want_xcol") = false);
@@ -1002,7 +1018,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("eps"), py::arg("res") = c10::nullopt);
   m.def("ln_bwd", &ln_bwd, py::arg("x"), py::arg("dy"), py::arg("w"),
         py::arg("mean"), py::arg("invstd"),
-        py::arg("dxadd") = c10::nullopt);
+        py::arg("dxadd") = c10::nullopt,
+        py::arg("dw_out") = c10::nullopt,
+        py::arg("db_out") = c10::nullopt);
   m.def("gelu_fwd", &gelu_fwd);
   m.def("gelu_bwd", &gelu_bwd);
   m.def("attn_softmax_fwd", &attn_softmax_fwd);

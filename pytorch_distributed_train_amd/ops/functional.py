@@ -278,11 +278,17 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
             # re-reads y (one whole HBM pass saved per kernel)
             relu_y = y if (fuse_relu and ctx.has_residual) else None
             premask = fuse_relu and not ctx.has_residual
+            # direct-grad: dbias == sum_dy — have the finalize kernel
+            # write it straight into the flat slice (no D2D copy); with a
+            # comm the slice still holds the LOCAL sums (correct: DDP
+            # averages param grads) and the packed allreduce reads it
+            b_view = ctx.gslot_b[0] if ctx.gslot_b is not None else None
             red = ext().bn_bwd_reduce(
                 x, dy, relu_y if mask is None else None, mean,
                 invstd=invstd if premask else None,
                 weight=weight if premask else None,
-                bias=bias if premask else None, dy2=dy2, mbits=mask)
+                bias=bias if premask else None, dy2=dy2, mbits=mask,
+                sum_dy_out=b_view)
             sum_dy, sum_dy_xmu = red[0], red[1]
             dy_comb = None
             if dy2 is not None:
@@ -300,7 +306,7 @@ class BatchNorm2dNHWCFn(torch.autograd.Function):
             else:
                 dweight = sum_dy_xmu * invstd
             if ctx.gslot_b is not None and x.is_cuda:
-                ctx.gslot_b[0].copy_(sum_dy)
+                # sum_dy IS the flat slice (written by the reduce)
                 if ctx.gslot_b[1] is not None:
                     ctx.gslot_b[1]()
                 dbias = None
@@ -603,9 +609,10 @@ class LinearFn(torch.autograd.Function):
                         ctx.gslot_w[1]()
                     dw = None
             if ctx.has_bias and ctx.needs_input_grad[3]:
-                db = ext().colsum_f32(dy)                     # [M,N] -> [N]
+                db = ext().colsum_f32(
+                    dy, out=(ctx.gslot_b[0] if ctx.gslot_b is not None
+                             else None))                      # [M,N] -> [N]
                 if ctx.gslot_b is not None:
-                    ctx.gslot_b[0].copy_(db)
                     if ctx.gslot_b[1] is not None:
                         ctx.gslot_b[1]()
                     db = None
@@ -703,14 +710,15 @@ class LayerNormFn(torch.autograd.Function):
         x, weight, mean, invstd = ctx.saved_tensors
         dy = dy.contiguous()
         if x.is_cuda:
-            dx, dw, db = ext().ln_bwd(x, dy, weight, mean, invstd)
+            dx, dw, db = ext().ln_bwd(
+                x, dy, weight, mean, invstd,
+                dw_out=ctx.gslot_w[0] if ctx.gslot_w is not None else None,
+                db_out=ctx.gslot_b[0] if ctx.gslot_b is not None else None)
             if ctx.gslot_w is not None:
-                ctx.gslot_w[0].copy_(dw)
                 if ctx.gslot_w[1] is not None:
                     ctx.gslot_w[1]()
                 dw = None
             if ctx.gslot_b is not None:
-                ctx.gslot_b[0].copy_(db)
                 if ctx.gslot_b[1] is not None:
                     ctx.gslot_b[1]()
                 db = None
@@ -768,14 +776,14 @@ class AddLayerNormFn(torch.autograd.Function):
         if s.is_cuda:
             dx, dw, db = ext().ln_bwd(
                 s, dy, weight, mean, invstd,
-                dxadd=ds.contiguous() if ds is not None else None)
+                dxadd=ds.contiguous() if ds is not None else None,
+                dw_out=ctx.gslot_w[0] if ctx.gslot_w is not None else None,
+                db_out=ctx.gslot_b[0] if ctx.gslot_b is not None else None)
             if ctx.gslot_w is not None:
-                ctx.gslot_w[0].copy_(dw)
                 if ctx.gslot_w[1] is not None:
                     ctx.gslot_w[1]()
                 dw = None
             if ctx.gslot_b is not None:
-                ctx.gslot_b[0].copy_(db)
                 if ctx.gslot_b[1] is not None:
                     ctx.gslot_b[1]()
                 db = None
