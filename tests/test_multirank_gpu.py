@@ -272,3 +272,30 @@ def test_vit_ddp_2ranks(tmp_path):
     r1 = torch.load(tmp_path / "vit1.pt", weights_only=False)
     assert torch.allclose(r0["grad"], r1["grad"], atol=0, rtol=0)
     assert r0["loss"] == r0["loss"] and abs(r0["loss"]) < 100
+
+
+def test_train_entry_ws2_eval_ckpt_resume_gpu(tmp_path):
+    """Full train.py at ws=2 on hardware: 2 epochs with eval + rank-0
+    checkpoint, then a resumed epoch — the complete user-facing flow of
+    the reference, end to end on the GPU."""
+    env = dict(os.environ)
+    for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE", "MASTER_ADDR",
+              "MASTER_PORT"):
+        env.pop(k, None)
+    base = [sys.executable, "-m", "pytorch_distributed_train_amd.launch",
+            "--nproc_per_node=2", f"--master_port={_free_port()}",
+            "--max-restarts=0", os.path.join(REPO, "train.py"),
+            "--model", "resnet18", "--batch-size", "16",
+            "--num-samples", "64", "--steps-per-epoch", "2",
+            "--eval-freq", "1", "--eval-batches", "1",
+            "--checkpoint-dir", str(tmp_path), "--checkpoint-freq", "1"]
+    r = subprocess.run(base + ["--epochs", "2"], cwd=REPO, env=env,
+                       capture_output=True, text=True, timeout=420)
+    assert r.returncode == 0, r.stderr[-2000:]
+    ckpt = tmp_path / "resnet18_checkpoint.pt"
+    assert ckpt.exists()
+    r2 = subprocess.run(base + ["--epochs", "3", "--resume", str(ckpt)],
+                        cwd=REPO, env=env, capture_output=True, text=True,
+                        timeout=420)
+    assert r2.returncode == 0, r2.stderr[-2000:]
+    assert "resumed from" in r2.stdout
