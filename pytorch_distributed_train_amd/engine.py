@@ -127,6 +127,7 @@ class Trainer:
         weight-shared model fails fast rather than training wrong."""
         from .models.layers import (BatchNorm2dNHWC, Conv2dNHWC, Linear)
         from .models.vit import LayerNorm as VitLayerNorm
+        from .models.vit import MultiheadSelfAttention
 
         def slot(p):
             i = self.flat.index_by_param.get(id(p))
@@ -149,6 +150,9 @@ class Trainer:
                 m._gslot_w = slot(m.weight)
                 if m.bias is not None:
                     m._gslot_b = slot(m.bias)
+            elif isinstance(m, MultiheadSelfAttention):
+                m._gslot_inw = slot(m.in_proj_weight)
+                m._gslot_inb = slot(m.in_proj_bias)
 
     def _flatten_nbt(self):
         from .models.layers import BatchNorm2dNHWC

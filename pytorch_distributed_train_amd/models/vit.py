@@ -77,8 +77,13 @@ class MultiheadSelfAttention(nn.Module):
         if packed is None or packed.dtype != x.dtype:
             packed = self.in_proj_weight.to(x.dtype)
             packed_t = None
+        on_direct = x.is_cuda and self.training
         qkv = OF.linear(x.reshape(b * s, d), self.in_proj_weight, packed,
-                        self.in_proj_bias, packed_t)
+                        self.in_proj_bias, packed_t,
+                        gslot_w=(getattr(self, "_gslot_inw", None)
+                                 if on_direct else None),
+                        gslot_b=(getattr(self, "_gslot_inb", None)
+                                 if on_direct else None))
         scale = 1.0 / (hd ** 0.5)
         if x.is_cuda:
             # zero-copy path: q/k/v read in place, grads scattered in place
