@@ -299,3 +299,70 @@ def test_train_entry_ws2_eval_ckpt_resume_gpu(tmp_path):
                         timeout=420)
     assert r2.returncode == 0, r2.stderr[-2000:]
     assert "resumed from" in r2.stdout
+
+
+def _ragged_gpu_worker(rank, world, port, out_dir):
+    os.environ.update(RANK=str(rank), LOCAL_RANK=str(rank),
+                      WORLD_SIZE=str(world), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    from pytorch_distributed_train_amd.flags import DistEnv
+    from pytorch_distributed_train_amd.parallel.comm import make_communicator
+    from pytorch_distributed_train_amd.parallel.sync_bn import (
+        SyncBatchNorm2dNHWC)
+    env = DistEnv.from_env()
+    comm = make_communicator(env, "cuda", "auto")  # staged on 1 GPU
+    C = 64
+    g = torch.Generator().manual_seed(5)
+    x_full = torch.randn(8, 8, 8, C, generator=g)
+    w_full = torch.randn(8, 8, 8, C, generator=g)
+    sl = slice(0, 6) if rank == 0 else slice(6, 8)  # RAGGED 6 vs 2
+    torch.manual_seed(3)
+    bn = SyncBatchNorm2dNHWC(C).cuda()
+    bn.comm = comm
+    bn.train()
+    x = x_full[sl].cuda().bfloat16().requires_grad_(True)
+    y = bn(x)
+    (y.float() * w_full[sl].cuda()).sum().backward()
+    torch.cuda.synchronize()
+    torch.save({"y": y.detach().float().cpu(), "dx": x.grad.float().cpu(),
+                "rm": bn.running_mean.cpu(), "rv": bn.running_var.cpu()},
+               os.path.join(out_dir, f"rgpu{rank}.pt"))
+    comm.barrier()
+
+
+def test_syncbn_gpu_fast_path_ragged(tmp_path):
+    """The GPU SyncBN fast path (fused pack/combine + reduced count) with
+    UNEQUAL per-rank batches, against a CPU fp32 single-process oracle of
+    the concatenated batch."""
+    from pytorch_distributed_train_amd.models.layers import BatchNorm2dNHWC
+    port = _free_port()
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_ragged_gpu_worker,
+                         args=(r, 2, port, str(tmp_path)))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(300)
+        if p.is_alive():
+            p.terminate()
+            pytest.fail("ragged gpu worker hung")
+        assert p.exitcode == 0, f"worker rc={p.exitcode}"
+    r0 = torch.load(tmp_path / "rgpu0.pt", weights_only=False)
+    r1 = torch.load(tmp_path / "rgpu1.pt", weights_only=False)
+    C = 64
+    g = torch.Generator().manual_seed(5)
+    x_full = torch.randn(8, 8, 8, C, generator=g)
+    w_full = torch.randn(8, 8, 8, C, generator=g)
+    torch.manual_seed(3)
+    bn = BatchNorm2dNHWC(C)
+    bn.train()
+    x = x_full.clone().requires_grad_(True)
+    y = bn(x)
+    (y * w_full).sum().backward()
+    got_y = torch.cat([r0["y"], r1["y"]])
+    got_dx = torch.cat([r0["dx"], r1["dx"]])
+    assert float((got_y - y.detach()).norm() / y.detach().norm()) < 0.02
+    assert float((got_dx - x.grad).norm() / x.grad.norm()) < 0.03
+    assert torch.allclose(r0["rm"], r1["rm"], atol=1e-6)
+    assert torch.allclose(r0["rm"], bn.running_mean, atol=1e-2)
