@@ -647,6 +647,122 @@ __global__ __launch_bounds__(512) void gemm_nt_8p2_k(
   }
 }
 
+
+// ---- B-direct 256x256 (PDTA_GEMM_BD): only A is glds-staged; B
+// fragments are read straight from global memory (per-lane dwordx4 at
+// the natural [row][K] layout — no transpose needed for "nt"). The
+// 256-col B tile is a few hundred KB and is re-read by every tm block
+// of its column panel, so those reads hit the XCD's L2; in exchange the
+// kernel stages HALF the bytes through glds — the measured bound of the
+// 2-slot schedule (occupancy probe) — and halves its LDS
+// footprint (64 KiB -> 3 blocks/CU by LDS).
+template <typename T, typename OutT, bool BIAS>
+__global__ __launch_bounds__(512) void gemm_nt_bd_k(
+    const T* __restrict__ A, const T* __restrict__ B, OutT* __restrict__ C,
+    const float* __restrict__ bias, int M, int N, int K, int tiles_n,
+    long batch_sa, long batch_sb, long batch_sc) {
+  constexpr int BOFF = 256 * 64 * 2;  // A slot bytes
+  __shared__ __attribute__((aligned(16))) char lds[2][BOFF];
+
+  A += (long)blockIdx.y * batch_sa;
+  B += (long)blockIdx.y * batch_sb;
+  C += (long)blockIdx.y * batch_sc;
+
+  int bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  int tm = bid / tiles_n, tn = bid % tiles_n;
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  int wm = wave >> 2, wn = wave & 3;  // 2 x 4
+  int m0 = tm * 256, n0 = tn * 256;
+  int ktiles = K / 64;
+
+  int row_in_chunk = lane >> 3;
+  int col16 = lane & 7;
+
+  auto stage_a = [&](int kt, int slot) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int row = (wave * 4 + i) * 8 + row_in_chunk;
+      int grow = m0 + row;
+      if (grow > M - 1) grow = M - 1;
+      const char* g = reinterpret_cast<const char*>(A) +
+                      ((long)grow * K + kt * 64) * 2 +
+                      (col16 ^ (row & 7)) * 16;
+      glds16(g, (char*)lds[slot] + (wave * 4 + i) * 8 * 128);
+    }
+  };
+
+  // B fragment rows for this wave (clamped once)
+  long brow[4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    int r = n0 + wn * 64 + j * 16 + (lane & 15);
+    if (r > N - 1) r = N - 1;
+    brow[j] = (long)r * K;
+  }
+  int bk = (lane >> 4) * 8;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  stage_a(0, 0);
+
+  for (int kt = 0; kt < ktiles; ++kt) {
+    int slot = kt & 1;
+    // B fragments for this K-tile: plain vector loads (L2-resident)
+    s16x8 bf[2][4];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        bf[kk][j] = *reinterpret_cast<const s16x8*>(
+            B + brow[j] + kt * 64 + kk * 32 + bk);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (kt + 1 < ktiles) stage_a(kt + 1, slot ^ 1);
+    const char* la = lds[slot];
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      s16x8 af[2][2];
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+          af[kk][i] = frag_read128(
+              la, wm * 128 + q * 32 + i * 16 + (lane & 15),
+              (lane >> 4) + kk * 4);
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[q * 2 + i][j] =
+                MfmaOp<T>::run(af[kk][i], bf[kk][j], acc[q * 2 + i][j]);
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    long rbase = m0 + wm * 128 + i * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = n0 + wn * 64 + j * 16 + (lane & 15);
+      if (col >= N) continue;
+      float badd = BIAS ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = rbase + r;
+        if (row < M)
+          C[(long)row * N + col] = from_f32<OutT>(acc[i][j][r] + badd);
+      }
+    }
+  }
+}
+
 // Occupancy experiment (docs/DESIGN.md §6): acc 64 VGPR/thread and
 // 2x24 KiB ring slots -> 3 blocks/CU by LDS, ~4 waves/SIMD by VGPR
 // (the 256x256/BK=64 kernel is pinned at 2 waves/SIMD and 52% WAIT).
@@ -811,6 +927,23 @@ void pdta_gemm_nt(const void* A, const void* B, void* C, const float* bias,
   if (use_r4 < 0) use_r4 = getenv("PDTA_GEMM_R4") != nullptr;
   static int use_8p2 = -1;
   if (use_8p2 < 0) use_8p2 = getenv("PDTA_GEMM_8P2") != nullptr;
+  static int use_bd = -1;
+  if (use_bd < 0) use_bd = getenv("PDTA_GEMM_BD") != nullptr;
+  if (use_bd && M >= 1024 && N >= 256 && K % 64 == 0 && !out_f32) {
+    int tiles_m = (M + 255) / 256, tiles_n = (N + 255) / 256;
+    dim3 grid(tiles_m * tiles_n);
+    PDTA_DISPATCH_DTYPE16(dt, T, {
+      if (bias)
+        gemm_nt_bd_k<T, T, true><<<grid, 512, 0, s>>>(
+            (const T*)A, (const T*)B, (T*)C, bias, M, N, K, tiles_n, 0, 0,
+            0);
+      else
+        gemm_nt_bd_k<T, T, false><<<grid, 512, 0, s>>>(
+            (const T*)A, (const T*)B, (T*)C, nullptr, M, N, K, tiles_n, 0,
+            0, 0);
+    });
+    return;
+  }
   if (use_8p2 && M >= 1024 && N >= 256 && K % 64 == 0 && K >= 256 &&
       !out_f32) {
     int tiles_m = (M + 255) / 256, tiles_n = (N + 255) / 256;
