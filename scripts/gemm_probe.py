@@ -46,7 +46,9 @@ def bench(M, N, K, iters=30):
 
 
 def main():
-    mode = ("R4" if os.environ.get("PDTA_GEMM_R4") else
+    mode = ("BD" if os.environ.get("PDTA_GEMM_BD") else
+            "8P2" if os.environ.get("PDTA_GEMM_8P2") else
+            "R4" if os.environ.get("PDTA_GEMM_R4") else
             "8P" if os.environ.get("PDTA_GEMM_8P") else
             "V2" if os.environ.get("PDTA_GEMM_V2") else "base")
     print(f"schedule: {mode}")
