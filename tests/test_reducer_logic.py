@@ -77,3 +77,111 @@ def test_training_loss_decreases_cpu():
     y = torch.randint(0, 10, (8,), generator=g)
     losses = [float(t.train_step(x, y)) for _ in range(8)]
     assert losses[-1] < losses[0] * 0.8, losses  # same batch memorized
+
+
+def test_param_ready_duplicate_raises():
+    """Direct-grad safety: a second arrival for the same param in one
+    iteration must raise (a weight-shared model would otherwise train
+    on silently-overwritten gradients)."""
+    import pytest as _pytest
+    import torch as T
+    from pytorch_distributed_train_amd.parallel.flat import FlatParams
+    from pytorch_distributed_train_amd.parallel.ddp import GradReducer
+
+    class FakeComm:
+        rank, world_size = 0, 2
+
+        def allreduce_avg(self, t, stream=None):
+            from pytorch_distributed_train_amd.parallel.comm import Work
+            return Work()
+
+        def broadcast(self, t, src=0, stream=None):
+            from pytorch_distributed_train_amd.parallel.comm import Work
+            return Work()
+
+    T.manual_seed(0)
+    m = T.nn.Linear(8, 8)
+    flat = FlatParams(m.named_parameters())
+    red = GradReducer(flat, FakeComm(), bucket_cap_mb=1)
+    red.prepare()
+    red.param_ready(0)
+    with _pytest.raises(RuntimeError, match="arrived twice"):
+        red.param_ready(0)
+
+
+def test_take_over_arrival_removes_hook():
+    """After take_over_arrival, only the manual notify counts the param
+    (torch fires post-accumulate hooks even for None grads)."""
+    import torch as T
+    from pytorch_distributed_train_amd.parallel.flat import FlatParams
+    from pytorch_distributed_train_amd.parallel.ddp import GradReducer
+
+    class FakeComm:
+        rank, world_size = 0, 2
+        calls = 0
+
+        def allreduce_avg(self, t, stream=None):
+            FakeComm.calls += 1
+            from pytorch_distributed_train_amd.parallel.comm import Work
+            return Work()
+
+        def broadcast(self, t, src=0, stream=None):
+            from pytorch_distributed_train_amd.parallel.comm import Work
+            return Work()
+
+    T.manual_seed(0)
+    m = T.nn.Linear(8, 8)
+    flat = FlatParams(m.named_parameters())
+    red = GradReducer(flat, FakeComm(), bucket_cap_mb=1)
+    for i in range(flat.num_params):
+        red.take_over_arrival(i)
+    red.prepare()
+    x = T.randn(2, 8)
+    m(x).sum().backward()        # hooks gone: no arrivals counted
+    assert all(b.pending == len(b.param_ids) for b in red.buckets)
+    for i in range(flat.num_params):
+        red.param_ready(i)       # manual notifies complete the buckets
+    assert all(b.pending == 0 for b in red.buckets)
+    red.finalize()
+    assert FakeComm.calls >= 1
+
+
+def test_launch_order_enforced_ascending():
+    """An early-ready later bucket is deferred until its predecessors
+    launch (identical collective order on every rank)."""
+    import torch as T
+    from pytorch_distributed_train_amd.parallel.flat import FlatParams
+    from pytorch_distributed_train_amd.parallel.ddp import GradReducer
+
+    order = []
+
+    class FakeComm:
+        rank, world_size = 0, 2
+
+        def allreduce_avg(self, t, stream=None):
+            order.append(t.data_ptr())
+            from pytorch_distributed_train_amd.parallel.comm import Work
+            return Work()
+
+        def broadcast(self, t, src=0, stream=None):
+            from pytorch_distributed_train_amd.parallel.comm import Work
+            return Work()
+
+    T.manual_seed(0)
+    m = T.nn.Sequential(T.nn.Linear(512, 512), T.nn.Linear(512, 512))
+    flat = FlatParams(m.named_parameters())
+    red = GradReducer(flat, FakeComm(), bucket_cap_mb=1)
+    assert len(red.buckets) >= 2
+    red.prepare()
+    # complete the LAST bucket first: nothing may launch yet
+    last = red.buckets[-1]
+    for pid in last.param_ids:
+        red.param_ready(pid)
+    assert order == [] and last.ready
+    # then earlier buckets: everything launches, ascending by start
+    for b in red.buckets[:-1]:
+        for pid in b.param_ids:
+            red.param_ready(pid)
+    starts = [red.flat.grad[b.start:b.end].data_ptr()
+              for b in red.buckets]
+    assert order == starts
