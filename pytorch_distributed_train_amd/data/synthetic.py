@@ -11,22 +11,40 @@ import torch
 
 
 class SyntheticImageDataset:
+    """Deterministic per-index samples. Images come from a lazily-built
+    POOL of `pool_size` unique tensors (index -> pool[idx % pool_size]):
+    generating a fresh 3x224x224 randn per __getitem__ cost ~1 ms and
+    made the synchronous loader the training bottleneck (~880 img/s on a
+    GPU that trains at 8000+); the pool turns collation into memcpys.
+    Labels stay per-index seeded, and index -> data is still
+    deterministic, so DDP-vs-single-process equality tests hold."""
+
     def __init__(self, num_samples: int, image_shape, num_classes: int,
-                 seed: int = 0):
+                 seed: int = 0, pool_size: int = 256):
         self.num_samples = num_samples
         self.image_shape = tuple(image_shape)
         self.num_classes = num_classes
         self.seed = seed
+        self.pool_size = min(pool_size, num_samples) or 1
+        self._pool = None
 
     def __len__(self) -> int:
         return self.num_samples
 
+    def _ensure_pool(self):
+        if self._pool is None:
+            g = torch.Generator()
+            g.manual_seed(self.seed * 7_777_777 + 13)
+            self._pool = torch.randn((self.pool_size,) + self.image_shape,
+                                     generator=g)
+        return self._pool
+
     def __getitem__(self, idx: int):
+        pool = self._ensure_pool()
         g = torch.Generator()
         g.manual_seed(self.seed * 1_000_003 + idx)
-        img = torch.randn(self.image_shape, generator=g)
         label = int(torch.randint(0, self.num_classes, (1,), generator=g))
-        return img, label
+        return pool[idx % self.pool_size], label
 
 
 _PRESETS = {
