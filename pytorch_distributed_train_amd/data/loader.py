@@ -89,11 +89,14 @@ class Loader:
 
     def _iter_procs(self):
         """torch-DataLoader-style worker processes (reference row 2b#14):
-        fork context, tensors return through shared memory; batches are
-        re-ordered so every rank sees the identical sequence; pinning
-        happens in the parent (pinned pages don't survive pickling)."""
+        SPAWN context — fork would inherit whatever locks other threads
+        (watchdog, prefetch, torch internals) hold at fork time and
+        deadlock (observed); tensors return through shared memory;
+        batches are re-ordered so every rank sees the identical
+        sequence; pinning happens in the parent (pinned pages don't
+        survive pickling)."""
         import torch.multiprocessing as mp
-        ctx = mp.get_context("fork")
+        ctx = mp.get_context("spawn")
         task_q = ctx.Queue()
         out_q = ctx.Queue()
         batches = list(self._batches())
@@ -112,7 +115,15 @@ class Loader:
         try:
             for next_bi in range(len(batches)):
                 while next_bi not in results:
-                    bi, x, labels = out_q.get()
+                    try:
+                        bi, x, labels = out_q.get(timeout=5.0)
+                    except queue.Empty:
+                        if not any(p.is_alive() for p in procs):
+                            raise RuntimeError(
+                                "all loader worker processes died "
+                                "(spawn re-imports __main__: run from a "
+                                "script, not stdin)")
+                        continue
                     results[bi] = (x, labels)
                 x, labels = results.pop(next_bi)
                 if self.pin_memory:

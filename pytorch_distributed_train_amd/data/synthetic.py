@@ -37,14 +37,16 @@ class SyntheticImageDataset:
             g.manual_seed(self.seed * 7_777_777 + 13)
             self._pool = torch.randn((self.pool_size,) + self.image_shape,
                                      generator=g)
+            # all labels in one vectorized draw: a per-__getitem__
+            # Generator().manual_seed() costs ~1.5 ms EACH (it re-seeds
+            # the whole MT19937 state) and was the real loader bottleneck
+            self._labels = torch.randint(0, self.num_classes,
+                                         (self.num_samples,), generator=g)
         return self._pool
 
     def __getitem__(self, idx: int):
         pool = self._ensure_pool()
-        g = torch.Generator()
-        g.manual_seed(self.seed * 1_000_003 + idx)
-        label = int(torch.randint(0, self.num_classes, (1,), generator=g))
-        return pool[idx % self.pool_size], label
+        return pool[idx % self.pool_size], int(self._labels[idx])
 
 
 _PRESETS = {
