@@ -75,3 +75,20 @@ def test_bench_under_torchrun_cpu(tmp_path):
     line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
     rec = json.loads(line)
     assert rec["n_gpus"] == 2 and rec["config"]["parallelism"] == "dp2"
+
+
+def test_train_deterministic_across_runs(tmp_path):
+    """Same seed, same flags => bit-identical loss trajectory (CPU):
+    the loaders, sampler, init and step are all deterministic."""
+    import subprocess
+    outs = []
+    for _ in range(2):
+        r = subprocess.run(
+            [sys.executable, "train.py", "--model", "resnet18",
+             "--data", "synthetic-cifar10", "--batch-size", "8",
+             "--num-samples", "64", "--epochs", "2", "--device", "cpu",
+             "--print-freq", "1", "--seed", "123"],
+            cwd=REPO, capture_output=True, text=True, timeout=600)
+        assert r.returncode == 0, r.stderr[-1500:]
+        outs.append([l for l in r.stdout.splitlines() if "loss=" in l])
+    assert outs[0] == outs[1] and len(outs[0]) >= 8
