@@ -90,5 +90,7 @@ def test_train_deterministic_across_runs(tmp_path):
              "--print-freq", "1", "--seed", "123"],
             cwd=REPO, capture_output=True, text=True, timeout=600)
         assert r.returncode == 0, r.stderr[-1500:]
-        outs.append([l for l in r.stdout.splitlines() if "loss=" in l])
+        import re
+        outs.append([re.sub(r"images_per_sec=[0-9.]+", "", l)
+                     for l in r.stdout.splitlines() if "loss=" in l])
     assert outs[0] == outs[1] and len(outs[0]) >= 8
