@@ -366,3 +366,21 @@ def test_syncbn_gpu_fast_path_ragged(tmp_path):
     assert float((got_dx - x.grad).norm() / x.grad.norm()) < 0.03
     assert torch.allclose(r0["rm"], r1["rm"], atol=1e-6)
     assert torch.allclose(r0["rm"], bn.running_mean, atol=1e-2)
+
+
+def test_bench_self_launch_4ranks_on_1gpu():
+    """4 ranks on one GPU (staged collectives, small buckets): stresses
+    collective ordering with more peers than the ws=2 tests."""
+    env = dict(os.environ)
+    for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE", "MASTER_ADDR",
+              "MASTER_PORT"):
+        env.pop(k, None)
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--gpus", "4", "--steps", "2",
+         "--warmup", "1", "--model", "resnet18", "--batch-size", "16",
+         "--bucket-cap-mb", "8"],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=420)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    rec = json.loads(line)
+    assert rec["n_gpus"] == 4 and rec["config"]["parallelism"] == "dp4"
