@@ -223,7 +223,11 @@ __global__ __launch_bounds__(512) void igemm_fwd_big_k(
 
   int row_in_chunk = lane >> 3;
   int col16 = lane & 7;
-  int a_n[4], a_h0[4], a_w0[4], a_swz[4];
+  // per-row bases precomputed once; per-stage addresses are base +
+  // uniform offset (the 3-mul 64-bit chain per row per stage was ~1/3
+  // of this kernel's 8.2-VALU-per-MFMA issue load — PMC r01)
+  int a_h0[4], a_w0[4];
+  long a_base[4], b_base[4];
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
     int row = (wave * 4 + i) * 8 + row_in_chunk;
@@ -233,10 +237,14 @@ __global__ __launch_bounds__(512) void igemm_fwd_big_k(
     unsigned rem = (unsigned)m - npq * (unsigned)(P * Q);
     unsigned pp = mdiv(rem, mq, sq);
     unsigned qq = rem - pp * (unsigned)Q;
-    a_n[i] = (int)npq;
     a_h0[i] = (int)pp * stride - pad;
     a_w0[i] = (int)qq * stride - pad;
-    a_swz[i] = col16 ^ (row & 7);
+    int swz = col16 ^ (row & 7);
+    a_base[i] = ((((long)npq * H + a_h0[i]) * W + a_w0[i]) * C) * 2 +
+                swz * 16;
+    int ko = n0 + row;
+    if (ko > Kout - 1) ko = Kout - 1;
+    b_base[i] = (long)ko * RSC * 2 + swz * 16;
   }
 
   int cchunks = C / 64;
@@ -245,26 +253,21 @@ __global__ __launch_bounds__(512) void igemm_fwd_big_k(
   auto stage_all = [&](int kt, int slot) {
     int tap = kt / cchunks, ct = kt % cchunks;
     int r = tap / S, ss = tap % S;
+    long off_a = ((long)(r * W + ss) * C + ct * 64) * 2;  // uniform
+    long off_b = ((long)tap * C + ct * 64) * 2;
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       int h = a_h0[i] + r, w = a_w0[i] + ss;
       const char* g;
       if ((unsigned)h < (unsigned)H && (unsigned)w < (unsigned)W)
-        g = reinterpret_cast<const char*>(x) +
-            ((((long)a_n[i] * H + h) * W + w) * C + ct * 64) * 2 +
-            a_swz[i] * 16;
+        g = reinterpret_cast<const char*>(x) + a_base[i] + off_a;
       else
         g = reinterpret_cast<const char*>(zero_page);
       glds16(g, (char*)lds[slot] + (wave * 4 + i) * 8 * 128);
     }
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      int row = (wave * 4 + i) * 8 + row_in_chunk;
-      int ko = n0 + row;
-      if (ko > Kout - 1) ko = Kout - 1;
-      const char* g = reinterpret_cast<const char*>(wt) +
-                      ((long)ko * RSC + tap * C + ct * 64) * 2 +
-                      (col16 ^ (row & 7)) * 16;
+      const char* g = reinterpret_cast<const char*>(wt) + b_base[i] + off_b;
       glds16(g, (char*)lds[slot] + BOFF + (wave * 4 + i) * 8 * 128);
     }
   };
