@@ -243,6 +243,8 @@ class RcclComm:
         return self._run(self._ext.rccl_broadcast, tensor, stream, src)
 
     def allgather_into(self, out, inp, stream=None) -> Work:
+        record_collective("rccl_allgather",
+                          inp.numel() * inp.element_size())
         self._ext.rccl_allgather(inp, out, self._comm,
                                  self._stream_ptr(stream))
         ev = torch.cuda.Event()
