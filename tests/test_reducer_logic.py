@@ -185,3 +185,38 @@ def test_launch_order_enforced_ascending():
     starts = [red.flat.grad[b.start:b.end].data_ptr()
               for b in red.buckets]
     assert order == starts
+
+
+def test_ddp_wrapper_api_parity():
+    """The torch-style wrapper (model = DDP(module, ...)) drives the
+    reducer through plain forward/backward — API parity surface for
+    users coming from nn.parallel.DistributedDataParallel."""
+    import torch as T
+    from pytorch_distributed_train_amd.parallel.flat import FlatParams
+    from pytorch_distributed_train_amd.parallel.ddp import (
+        DistributedDataParallel)
+
+    class FakeComm:
+        rank, world_size = 0, 2
+        reduced = 0
+
+        def allreduce_avg(self, t, stream=None):
+            FakeComm.reduced += 1
+            from pytorch_distributed_train_amd.parallel.comm import Work
+            return Work()
+
+        def broadcast(self, t, src=0, stream=None):
+            from pytorch_distributed_train_amd.parallel.comm import Work
+            return Work()
+
+    T.manual_seed(0)
+    m = T.nn.Sequential(T.nn.Linear(16, 16), T.nn.ReLU(),
+                        T.nn.Linear(16, 4))
+    flat = FlatParams(m.named_parameters())
+    ddp = DistributedDataParallel(m, flat, FakeComm(), bucket_cap_mb=1)
+    x = T.randn(4, 16)
+    out = ddp(x)
+    out.sum().backward()
+    ddp.reducer.finalize()
+    assert FakeComm.reduced >= 1
+    assert T.isfinite(flat.grad).all()
