@@ -69,8 +69,14 @@ class MultiheadSelfAttention(nn.Module):
         nn.init.xavier_uniform_(self.in_proj_weight)
         self.out_proj = Linear(dim, dim)
 
-    def forward(self, x, nvalid: int = SEQ):
+    def forward(self, x, nvalid: int | None = None):
+        # x [B, s, d] — the encoder runs PACKED (s = 197): only this
+        # module pads to the attention kernel's S=256 tile and slices
+        # back, so every GEMM/LN/GELU in the block skips the ~30% dead
+        # rows the old pad-once design carried through the whole encoder
         b, s, d = x.shape
+        if nvalid is None:
+            nvalid = s
         h, hd = self.num_heads, self.head_dim
         packed = getattr(self, "_pack_inproj", None)
         packed_t = getattr(self, "_pack_inproj_t", None)
@@ -86,8 +92,19 @@ class MultiheadSelfAttention(nn.Module):
                                  if on_direct else None))
         scale = 1.0 / (hd ** 0.5)
         if x.is_cuda:
+            sp = s
+            q3 = qkv
+            if s < SEQ_PAD:
+                import torch.nn.functional as TF
+                q3 = TF.pad(qkv.view(b, s, 3 * d),
+                            (0, 0, 0, SEQ_PAD - s)).reshape(
+                                b * SEQ_PAD, 3 * d)
+                sp = SEQ_PAD
             # zero-copy path: q/k/v read in place, grads scattered in place
-            out = OF.attention_qkv(qkv, b, s, h, hd, nvalid, scale)
+            out = OF.attention_qkv(q3, b, sp, h, hd, nvalid, scale)
+            out = out.reshape(b, sp, d)
+            if sp != s:
+                out = out[:, :s].contiguous()
             return self.out_proj(out.reshape(b * s, d)).view(b, s, d)
         qkv = qkv.view(b, s, 3, h, hd).permute(2, 0, 3, 1, 4)
         q = qkv[0].reshape(b * h, s, hd).contiguous()
@@ -139,14 +156,8 @@ class Encoder(nn.Module):
         self.layers = nn.Sequential(layers)
         self.ln = LayerNorm(dim)
 
-    def forward(self, x):  # x [B, SEQ, dim]
-        b, s, d = x.shape
-        if s < SEQ_PAD:  # pad token rows for GEMM-friendly shapes
-            xp = x.new_zeros(b, SEQ_PAD, d)
-            xp[:, :s] = x + self.pos_embedding.to(x.dtype)
-            x = xp
-        else:
-            x = x + self.pos_embedding.to(x.dtype)
+    def forward(self, x):  # x [B, SEQ, dim] — runs PACKED (no pre-pad)
+        x = x + self.pos_embedding.to(x.dtype)
         u, v = self.layers((x, None))
         y, _ = self.ln.forward_fused(u, v)
         return y
