@@ -92,15 +92,20 @@ class MultiheadSelfAttention(nn.Module):
                                  if on_direct else None))
         scale = 1.0 / (hd ** 0.5)
         if x.is_cuda:
-            sp = s
+            # zero-copy path: q/k/v read in place, grads scattered in
+            # place. The fused kernel consumes the PACKED [b, s=197, 3d]
+            # qkv directly (rows past nvalid are clamped/zeroed inside;
+            # attn.hip header) — no pad/slice copies anywhere.
+            if hd == 64 and 64 <= s <= SEQ_PAD:
+                out = OF.attention_qkv(qkv, b, s, h, hd, nvalid, scale)
+                return self.out_proj(out.reshape(b * s, d)).view(b, s, d)
+            # other shapes: pad to the decomposed path's 64-multiple
+            import torch.nn.functional as TF
+            sp = (s + 63) // 64 * 64
             q3 = qkv
-            if s < SEQ_PAD:
-                import torch.nn.functional as TF
+            if sp != s:
                 q3 = TF.pad(qkv.view(b, s, 3 * d),
-                            (0, 0, 0, SEQ_PAD - s)).reshape(
-                                b * SEQ_PAD, 3 * d)
-                sp = SEQ_PAD
-            # zero-copy path: q/k/v read in place, grads scattered in place
+                            (0, 0, 0, sp - s)).reshape(b * sp, 3 * d)
             out = OF.attention_qkv(q3, b, sp, h, hd, nvalid, scale)
             out = out.reshape(b, sp, d)
             if sp != s:

@@ -13,8 +13,15 @@
 // [b, S, 3·h·hd] qkv projection (row stride 3·h·hd) — no permute/copy
 // kernels anywhere, as in the decomposed path.
 //
-// Shape-specialized: S(padded) == 256, head_dim == 64 (ViT-B/16 and
-// anything padded to it); other shapes take the decomposed path.
+// Shape-specialized: head_dim == 64, sequence <= 256 (ViT-B/16); other
+// shapes take the decomposed path. qkv/out/dout/dqkv may be UNPADDED
+// (srows = actual rows per batch, e.g. 197): global row reads past
+// nvalid-1 are clamped to the last valid row (the garbage they produce
+// is annihilated — key cols >= nvalid are masked in the softmax, value
+// rows >= nvalid carry zero probability, query rows >= nvalid are
+// discarded at the store), row stores past nvalid write zeros, and the
+// internal probs/ds buffers stay padded to SP so the dK/dV batched
+// GEMMs can run K = SP against exactly-zero tail rows.
 #include "common.h"
 #include "mfma_tile.h"
 
@@ -26,8 +33,8 @@ constexpr int HD = 64;   // head dim
 template <typename T>
 __global__ __launch_bounds__(256) void attn_fused_fwd_k(
     const T* __restrict__ qkv, T* __restrict__ probs, T* __restrict__ out,
-    int h, int nvalid, float scale) {
-  // grid: (SP/64 query blocks, b*h). 4 waves; each owns 16 query rows.
+    int h, int srows, int nvalid, float scale) {
+  // grid: (ceil(srows/64) query blocks, b*h). 4 waves; 16 query rows each.
   __shared__ __attribute__((aligned(16))) char qtile[64 * 128];
   __shared__ __attribute__((aligned(16))) char kv[SP * 128];  // K, then V
   __shared__ __attribute__((aligned(16))) char pimg[64 * 512];
@@ -36,26 +43,29 @@ __global__ __launch_bounds__(256) void attn_fused_fwd_k(
   int bh = blockIdx.y;
   int batch = bh / h, head = bh % h;
   long d = (long)h * HD, ld3 = 3 * d;
-  const T* qbase = qkv + (long)batch * SP * ld3 + (long)head * HD;
+  const T* qbase = qkv + (long)batch * srows * ld3 + (long)head * HD;
   const T* kbase = qbase + d;
   const T* vbase = qbase + 2 * d;
   int q0 = blockIdx.x * 64;
 
-  // ---- stage Q [64][128B] and K [256][128B] (glds: 1 KiB = 8 rows each)
+  // ---- stage Q [64][128B] and K [256][128B] (glds: 1 KiB = 8 rows each);
+  //      global rows >= nvalid clamp to the last valid row (see header)
   int row_in = lane >> 3, c16 = lane & 7;
 #pragma unroll
   for (int i = 0; i < 2; ++i) {
     int row = (wave * 2 + i) * 8 + row_in;
-    const char* g =
-        reinterpret_cast<const char*>(qbase + (long)(q0 + row) * ld3) +
-        (c16 ^ (row & 7)) * 16;
+    int gr = q0 + row;
+    if (gr >= nvalid) gr = nvalid - 1;
+    const char* g = reinterpret_cast<const char*>(qbase + (long)gr * ld3) +
+                    (c16 ^ (row & 7)) * 16;
     glds16(g, qtile + (wave * 2 + i) * 1024);
   }
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
     int row = (wave * 8 + i) * 8 + row_in;
+    int gr = row >= nvalid ? nvalid - 1 : row;
     const char* g =
-        reinterpret_cast<const char*>(kbase + (long)row * ld3) +
+        reinterpret_cast<const char*>(kbase + (long)gr * ld3) +
         (c16 ^ (row & 7)) * 16;
     glds16(g, kv + (wave * 8 + i) * 1024);
   }
@@ -82,8 +92,9 @@ __global__ __launch_bounds__(256) void attn_fused_fwd_k(
 #pragma unroll
   for (int i = 0; i < 8; ++i) {  // stage V; its latency hides under softmax
     int row = (wave * 8 + i) * 8 + row_in;
+    int gr = row >= nvalid ? nvalid - 1 : row;
     const char* g =
-        reinterpret_cast<const char*>(vbase + (long)row * ld3) +
+        reinterpret_cast<const char*>(vbase + (long)gr * ld3) +
         (c16 ^ (row & 7)) * 16;
     glds16(g, kv + (wave * 8 + i) * 1024);
   }
@@ -135,7 +146,11 @@ __global__ __launch_bounds__(256) void attn_fused_fwd_k(
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int row = wave * 16 + (lane >> 4) * 4 + r;
-      T pv = from_f32<T>(acc[jn][r] * sum[r]);
+      // query rows past nvalid store ZERO probs — keeps the padded
+      // probs/ds buffers' tail rows exactly zero so the K=SP dK/dV
+      // GEMMs and the backward's P-tile reads annihilate clamp garbage
+      T pv = q0 + row < nvalid ? from_f32<T>(acc[jn][r] * sum[r])
+                               : from_f32<T>(0.f);
       *reinterpret_cast<T*>(pimg + row * 512 +
                             (((chunk ^ (row & 7)) << 4) | rem)) = pv;
       pg[(long)row * SP + col] = pv;
@@ -165,14 +180,16 @@ __global__ __launch_bounds__(256) void attn_fused_fwd_k(
                                 acc2[jn]);
   }
 
-  T* obase = out + (long)batch * SP * d + (long)head * HD;
+  T* obase = out + (long)batch * srows * d + (long)head * HD;
 #pragma unroll
   for (int jn = 0; jn < 4; ++jn) {
     int col = jn * 16 + colbase;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int row = q0 + wave * 16 + (lane >> 4) * 4 + r;
-      obase[(long)row * d + col] = from_f32<T>(acc2[jn][r]);
+      if (row < srows)
+        obase[(long)row * d + col] =
+            row < nvalid ? from_f32<T>(acc2[jn][r]) : from_f32<T>(0.f);
     }
   }
 }
@@ -186,7 +203,7 @@ template <typename T>
 __global__ __launch_bounds__(256) void attn_fused_bwd_ds_k(
     const T* __restrict__ qkv, const T* __restrict__ p,
     const T* __restrict__ dout, T* __restrict__ ds, T* __restrict__ dqkv,
-    int h, float scale) {
+    int h, int srows, int nvalid, float scale) {
   __shared__ __attribute__((aligned(16))) char dotile[64 * 128];
   __shared__ __attribute__((aligned(16))) char vimg[SP * 128];
   __shared__ __attribute__((aligned(16))) char dsimg[64 * 512];
@@ -196,26 +213,30 @@ __global__ __launch_bounds__(256) void attn_fused_bwd_ds_k(
   int batch = bh / h, head = bh % h;
   long d = (long)h * HD, ld3 = 3 * d;
   const T* vbase =
-      qkv + (long)batch * SP * ld3 + (long)head * HD + 2 * d;
+      qkv + (long)batch * srows * ld3 + (long)head * HD + 2 * d;
   const T* kbase = vbase - d;  // k plane of the interleaved qkv
-  const T* dobase = dout + (long)batch * SP * d + (long)head * HD;
-  T* dqbase = dqkv + (long)batch * SP * ld3 + (long)head * HD;
+  const T* dobase = dout + (long)batch * srows * d + (long)head * HD;
+  T* dqbase = dqkv + (long)batch * srows * ld3 + (long)head * HD;
   int q0 = blockIdx.x * 64;
 
+  // row reads past nvalid-1 clamp to the last valid row; the garbage is
+  // annihilated by the zero P rows/cols the forward stored (see header)
   int row_in = lane >> 3, c16 = lane & 7;
 #pragma unroll
   for (int i = 0; i < 2; ++i) {
     int row = (wave * 2 + i) * 8 + row_in;
-    const char* g =
-        reinterpret_cast<const char*>(dobase + (long)(q0 + row) * d) +
-        (c16 ^ (row & 7)) * 16;
+    int gr = q0 + row;
+    if (gr >= nvalid) gr = nvalid - 1;
+    const char* g = reinterpret_cast<const char*>(dobase + (long)gr * d) +
+                    (c16 ^ (row & 7)) * 16;
     glds16(g, dotile + (wave * 2 + i) * 1024);
   }
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
     int row = (wave * 8 + i) * 8 + row_in;
+    int gr = row >= nvalid ? nvalid - 1 : row;
     const char* g =
-        reinterpret_cast<const char*>(vbase + (long)row * ld3) +
+        reinterpret_cast<const char*>(vbase + (long)gr * ld3) +
         (c16 ^ (row & 7)) * 16;
     glds16(g, vimg + (wave * 8 + i) * 1024);
   }
@@ -263,8 +284,9 @@ __global__ __launch_bounds__(256) void attn_fused_bwd_ds_k(
 #pragma unroll
   for (int i = 0; i < 8; ++i) {  // stage K; hides under the dS writes
     int krow = (wave * 8 + i) * 8 + row_in;
+    int gr = krow >= nvalid ? nvalid - 1 : krow;
     const char* g =
-        reinterpret_cast<const char*>(kbase + (long)krow * ld3) +
+        reinterpret_cast<const char*>(kbase + (long)gr * ld3) +
         (c16 ^ (krow & 7)) * 16;
     glds16(g, vimg + (wave * 8 + i) * 1024);
   }
@@ -311,7 +333,9 @@ __global__ __launch_bounds__(256) void attn_fused_bwd_ds_k(
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int row = q0 + wave * 16 + (lane >> 4) * 4 + r;
-      dqbase[(long)row * ld3 + col] = from_f32<T>(acc2[jn][r]);
+      if (row < srows)
+        dqbase[(long)row * ld3 + col] =
+            row < nvalid ? from_f32<T>(acc2[jn][r]) : from_f32<T>(0.f);
     }
   }
 }
@@ -320,27 +344,34 @@ __global__ __launch_bounds__(256) void attn_fused_bwd_ds_k(
 
 extern "C" {
 
-// 1 if the fused kernel handles this shape (bind.cpp attn_fwd dispatch)
-int pdta_attn_fused_ok(int S, int hd) { return S == SP && hd == HD; }
+// 1 if the fused kernel handles this shape (bind.cpp attn_fwd dispatch).
+// S is the ACTUAL per-batch row count (may be unpadded, e.g. 197); the
+// kernel pads internally to SP.
+int pdta_attn_fused_ok(int S, int hd) {
+  return S <= SP && S >= 64 && hd == HD;
+}
 
 void pdta_attn_fused_fwd(const void* qkv, void* probs, void* out, int b,
-                         int h, int nvalid, float scale, int dt,
+                         int h, int srows, int nvalid, float scale, int dt,
                          hipStream_t s) {
+  // always SP/64 blocks: the tail query blocks must WRITE the padded
+  // probs rows (zeros) — the K=SP dK/dV GEMMs read all SP rows
   dim3 grid(SP / 64, b * h);
   PDTA_DISPATCH_DTYPE16(dt, T, {
     attn_fused_fwd_k<T><<<grid, 256, 0, s>>>(
-        (const T*)qkv, (T*)probs, (T*)out, h, nvalid, scale);
+        (const T*)qkv, (T*)probs, (T*)out, h, srows, nvalid, scale);
   });
 }
 
 void pdta_attn_fused_bwd_ds(const void* qkv, const void* p,
                             const void* dout, void* ds, void* dqkv, int b,
-                            int h, float scale, int dt, hipStream_t s) {
-  dim3 grid(SP / 64, b * h);
+                            int h, int srows, int nvalid, float scale,
+                            int dt, hipStream_t s) {
+  dim3 grid(SP / 64, b * h);  // ds tail rows must be written (zeros) too
   PDTA_DISPATCH_DTYPE16(dt, T, {
     attn_fused_bwd_ds_k<T><<<grid, 256, 0, s>>>(
         (const T*)qkv, (const T*)p, (const T*)dout, (T*)ds, (T*)dqkv, h,
-        scale);
+        srows, nvalid, scale);
   });
 }
 

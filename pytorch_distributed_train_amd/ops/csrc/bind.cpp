@@ -88,14 +88,15 @@ void pdta_softmax_bwd(const void*, const void*, void*, long, int, float,
                       int, hipStream_t);
 void pdta_gemm_bs(const void*, const void*, void*, int, int, int, int, int,
                   int, int, int, long, long, long, long, long, long, int,
-                  int, int, int, hipStream_t);
+                  int, int, int, int, int, hipStream_t);
 void pdta_pack_weights(const float*, void*, const long*, const long*, int,
                        long, int, hipStream_t);
 int pdta_attn_fused_ok(int, int);
-void pdta_attn_fused_fwd(const void*, void*, void*, int, int, int, float,
-                         int, hipStream_t);
+void pdta_attn_fused_fwd(const void*, void*, void*, int, int, int, int,
+                         float, int, hipStream_t);
 void pdta_attn_fused_bwd_ds(const void*, const void*, const void*, void*,
-                            void*, int, int, float, int, hipStream_t);
+                            void*, int, int, int, int, float, int,
+                            hipStream_t);
 }
 
 namespace {
@@ -795,31 +796,34 @@ std::vector<at::Tensor> attn_fwd(at::Tensor qkv, int64_t b, int64_t S,
   char* k = q + d * 2;   // 16-bit elements
   char* v = q + 2 * d * 2;
   if (pdta_attn_fused_ok(S, hd)) {
-    // one kernel: scores never touch HBM, P written once for backward
-    auto p = at::empty({b * h, S, S}, qkv.options());
+    // one kernel: scores never touch HBM, P written once for backward.
+    // qkv may be UNPADDED (S = 197): P is padded to SP = 256 internally
+    // (tail rows/cols exactly zero) so the dK/dV GEMMs run K = SP.
+    const long Sp = 256;  // attn.hip SP
+    auto p = at::empty({b * h, Sp, Sp}, qkv.options());
     auto out = at::empty({b, S, d}, qkv.options());
     pdta_attn_fused_fwd(qkv.data_ptr(), p.data_ptr(), out.data_ptr(), b, h,
-                        nvalid, (float)scale, dt, cur_stream());
+                        S, nvalid, (float)scale, dt, cur_stream());
     return {out, p};
   }
   auto scores = at::empty({b * h, S, S}, qkv.options().dtype(at::kFloat));
   pdta_gemm_bs(q, k, scores.data_ptr(), S, S, hd, ld, ld, S, b * h, h,
-               S * ld, hd, S * ld, hd, (long)h * S * S, S * S, 0, 0, 1, dt,
-               cur_stream());
+               S * ld, hd, S * ld, hd, (long)h * S * S, S * S, 0, 0, 1,
+               hd - 1, hd - 1, dt, cur_stream());
   auto p = at::empty({b * h, S, S}, qkv.options());
   pdta_softmax_fwd(scores.data_ptr<float>(), p.data_ptr(), b * h * S, S,
                    nvalid, scale, dt, cur_stream());
   auto out = at::empty({b, S, d}, qkv.options());
   pdta_gemm_bs(p.data_ptr(), v, out.data_ptr(), S, hd, S, S, ld, d, b * h,
                h, (long)h * S * S, S * S, S * ld, hd, S * d, hd, 0, 1, 0,
-               dt, cur_stream());
+               S - 1, S - 1, dt, cur_stream());
   return {out, p};
 }
 
 // returns d(qkv) [b, S, 3*h*hd] (written in place per head, zero copies)
 at::Tensor attn_bwd(at::Tensor qkv, at::Tensor p, at::Tensor dout,
                     int64_t b, int64_t S, int64_t h, int64_t hd,
-                    double scale) {
+                    int64_t nvalid, double scale) {
   CHECK_IN(qkv);
   CHECK_IN(p);
   CHECK_IN(dout);
@@ -828,37 +832,42 @@ at::Tensor attn_bwd(at::Tensor qkv, at::Tensor p, at::Tensor dout,
   char* q = (char*)qkv.data_ptr();
   char* k = q + d * 2;
   char* v = q + 2 * d * 2;
-  auto ds = at::empty({b * h, S, S}, qkv.options());
+  bool fused = pdta_attn_fused_ok(S, hd);
+  // fused path: p/ds are padded to SP=256 with exactly-zero tail
+  // rows/cols, so dK/dV run K = Sp and the unpadded Q/dout operands'
+  // clamped re-reads (kbrmax = S-1) are annihilated
+  long Sp = fused ? 256 : S;
+  auto ds = at::empty({b * h, Sp, Sp}, qkv.options());
   auto dqkv = at::empty({b, S, ld}, qkv.options());
   char* dq = (char*)dqkv.data_ptr();
   char* dk = dq + d * 2;
   char* dv = dq + 2 * d * 2;
-  if (pdta_attn_fused_ok(S, hd)) {
+  if (fused) {
     // one kernel: dP never touches HBM, dQ comes straight from the
     // in-register dS (K reuses the V LDS slab)
     pdta_attn_fused_bwd_ds(qkv.data_ptr(), p.data_ptr(), dout.data_ptr(),
-                           ds.data_ptr(), dqkv.data_ptr(), b, h,
+                           ds.data_ptr(), dqkv.data_ptr(), b, h, S, nvalid,
                            (float)scale, dt, cur_stream());
   } else {
     auto dp = at::empty({b * h, S, S}, qkv.options());
     pdta_gemm_bs(dout.data_ptr(), v, dp.data_ptr(), S, S, hd, d, ld, S,
                  b * h, h, S * d, hd, S * ld, hd, (long)h * S * S, S * S,
-                 0, 0, 0, dt, cur_stream());
+                 0, 0, 0, hd - 1, hd - 1, dt, cur_stream());
     pdta_softmax_bwd(p.data_ptr(), dp.data_ptr(), ds.data_ptr(), b * h * S,
                      S, scale, dt, cur_stream());
     // dq = ds @ K
     pdta_gemm_bs(ds.data_ptr(), k, dq, S, hd, S, S, ld, ld, b * h, h,
                  (long)h * S * S, S * S, S * ld, hd, S * ld, hd, 0, 1, 0,
-                 dt, cur_stream());
+                 S - 1, S - 1, dt, cur_stream());
   }
   // dk = ds^T @ Q
-  pdta_gemm_bs(ds.data_ptr(), q, dk, S, hd, S, S, ld, ld, b * h, h,
-               (long)h * S * S, S * S, S * ld, hd, S * ld, hd, 1, 1, 0, dt,
-               cur_stream());
+  pdta_gemm_bs(ds.data_ptr(), q, dk, S, hd, Sp, Sp, ld, ld, b * h, h,
+               (long)h * Sp * Sp, Sp * Sp, S * ld, hd, S * ld, hd, 1, 1, 0,
+               Sp - 1, S - 1, dt, cur_stream());
   // dv = P^T @ dOut
-  pdta_gemm_bs(p.data_ptr(), dout.data_ptr(), dv, S, hd, S, S, d, ld,
-               b * h, h, (long)h * S * S, S * S, S * d, hd, S * ld, hd, 1,
-               1, 0, dt, cur_stream());
+  pdta_gemm_bs(p.data_ptr(), dout.data_ptr(), dv, S, hd, Sp, Sp, d, ld,
+               b * h, h, (long)h * Sp * Sp, Sp * Sp, S * d, hd, S * ld, hd,
+               1, 1, 0, Sp - 1, S - 1, dt, cur_stream());
   return dqkv;
 }
 

@@ -24,7 +24,8 @@ template <typename T, typename OutT, bool TRA, bool TRB>
 __global__ __launch_bounds__(TPB) void gemm_bs_k(
     const T* __restrict__ A, const T* __restrict__ B, OutT* __restrict__ C,
     int M, int N, int K, int lda, int ldb, int ldc, int batH, long saB,
-    long saH, long sbB, long sbH, long scB, long scH, int tiles_n) {
+    long saH, long sbB, long sbH, long scB, long scH, int tiles_n,
+    int karmax, int kbrmax) {
   __shared__ __attribute__((aligned(16))) char lds[2][2][64 * 64 * 2];
 
   int bb = blockIdx.y / batH, hh = blockIdx.y % batH;
@@ -63,13 +64,17 @@ __global__ __launch_bounds__(TPB) void gemm_bs_k(
       glds16(gptr, dst + (wave * 16 + i * 8) * 128);
     }
   };
+  // karmax/kbrmax: last readable k-row of a TR operand (normally K-1).
+  // An UNPADDED operand (attention's Q/dOut when qkv is packed at 197
+  // rows) passes its real last row; the clamped re-reads contribute
+  // garbage that the other operand's exactly-zero tail rows annihilate.
   auto stage_tiles = [&](int k0, int buf) {
     if (TRA)
-      stage(A, lda, k0, K - 1, m0, lds[buf][0]);
+      stage(A, lda, k0, karmax, m0, lds[buf][0]);
     else
       stage(A, lda, m0, M - 1, k0, lds[buf][0]);
     if (TRB)
-      stage(B, ldb, k0, K - 1, n0, lds[buf][1]);
+      stage(B, ldb, k0, kbrmax, n0, lds[buf][1]);
     else
       stage(B, ldb, n0, N - 1, k0, lds[buf][1]);
   };
@@ -149,14 +154,14 @@ extern "C" {
 void pdta_gemm_bs(const void* A, const void* B, void* C, int M, int N,
                   int K, int lda, int ldb, int ldc, int batches, int batH,
                   long saB, long saH, long sbB, long sbH, long scB,
-                  long scH, int tra, int trb, int out_f32, int dt,
-                  hipStream_t s) {
+                  long scH, int tra, int trb, int out_f32, int karmax,
+                  int kbrmax, int dt, hipStream_t s) {
   int tiles_m = (M + 63) / 64, tiles_n = (N + 63) / 64;
   dim3 grid(tiles_m * tiles_n, batches);
 #define LAUNCH(TA, TB, OT)                                               \
   gemm_bs_k<T, OT, TA, TB><<<grid, TPB, 0, s>>>(                         \
       (const T*)A, (const T*)B, (OT*)C, M, N, K, lda, ldb, ldc, batH,    \
-      saB, saH, sbB, sbH, scB, scH, tiles_n)
+      saB, saH, sbB, sbH, scB, scH, tiles_n, karmax, kbrmax)
   PDTA_DISPATCH_DTYPE16(dt, T, {
     if (out_f32) {
       if (tra && trb) LAUNCH(true, true, float);

@@ -875,15 +875,16 @@ class AttentionQKVFn(torch.autograd.Function):
         qkv3 = qkv2d.view(b, s, 3 * h * hd)
         out, p = ext().attn_fwd(qkv3, b, s, h, hd, nvalid, scale)
         ctx.save_for_backward(qkv2d, p)
-        ctx.dims = (b, s, h, hd, scale)
+        ctx.dims = (b, s, h, hd, nvalid, scale)
         return out  # [b, s, h*hd]
 
     @staticmethod
     def backward(ctx, dout):
         qkv2d, p = ctx.saved_tensors
-        b, s, h, hd, scale = ctx.dims
+        b, s, h, hd, nvalid, scale = ctx.dims
         dqkv = ext().attn_bwd(qkv2d.view(b, s, 3 * h * hd), p,
-                              dout.contiguous(), b, s, h, hd, scale)
+                              dout.contiguous(), b, s, h, hd, nvalid,
+                              scale)
         return dqkv.view(b * s, 3 * h * hd), None, None, None, None, None, \
             None
 
