@@ -157,3 +157,41 @@ def test_attention_qkv_zero_copy_matches_permute_path():
     out_a.backward(g)
     out_ref.backward(g.clone())
     _close(qkv_a.grad, qkv_b.grad, rtol=3e-2, atol=3e-2, what="attn_qkv dqkv")
+
+
+def test_attention_qkv_shape_sweep_vs_fp32():
+    """The fused kernel accepts any 64 <= s <= 256 with head_dim 64 and
+    an UNPADDED qkv (attn.hip: clamped row reads, zero prob tails, K=SP
+    dK/dV against those tails). Sweep (s, nvalid) shapes and compare fwd
+    + d(qkv) against a pure fp32 torch reference with -inf key masking.
+    Rows >= nvalid must come back EXACTLY zero-gradient on both sides."""
+    from pytorch_distributed_train_amd.ops import functional as OF
+    b, h, hd = 2, 4, 64
+    d = h * hd
+    for i, (s, nv) in enumerate(
+            [(64, 64), (100, 97), (197, 197), (256, 197), (256, 256)]):
+        scale = 1.0 / hd ** 0.5
+        qkv = _rand(b * s, 3 * d, seed=40 + i)
+        qkv_a = qkv.clone().requires_grad_(True)
+        out = OF.attention_qkv(qkv_a, b, s, h, hd, nv, scale)
+        g = torch.zeros(b, s, d, device=DEV, dtype=qkv.dtype)
+        g[:, :nv] = _rand(b, nv, d, seed=60 + i)
+        out.backward(g.view(b, s, d))
+
+        ref = qkv.float().cpu().view(b, s, 3, h, hd).permute(
+            2, 0, 3, 1, 4).clone().requires_grad_(True)  # [3,b,h,s,hd]
+        q, k, v = ref[0], ref[1], ref[2]
+        sc = (q @ k.transpose(-1, -2)) * scale
+        sc[..., nv:] = float("-inf")
+        p = sc.softmax(-1)
+        o = (p @ v).permute(0, 2, 1, 3).reshape(b, s, d)
+        o.backward(g.float().cpu())
+
+        _close(out.view(b, s, d)[:, :nv], o[:, :nv],
+               what=f"fwd s={s} nv={nv}")
+        dref = ref.grad.permute(1, 3, 0, 2, 4).reshape(b * s, 3 * d)
+        _close(qkv_a.grad, dref, rtol=3e-2, atol=3e-2,
+               what=f"dqkv s={s} nv={nv}")
+        if nv < s:
+            tail = qkv_a.grad.view(b, s, 3 * d)[:, nv:]
+            assert tail.abs().max().item() == 0.0, f"tail s={s} nv={nv}"
