@@ -89,8 +89,15 @@ __global__ void ln_fwd_k(const T* __restrict__ x, T* __restrict__ y,
 // per chunk, written to part [chunk][2d]; ln_bwd_final reduces chunks
 // with 64-col x 4-slice blocks (a d-threads-only finalize was one wave
 // chip-wide and chunk-serial).
+// __launch_bounds__(TPB): without it the compiler budgets registers for
+// a hypothetical 1024-thread block (4 waves/SIMD floor -> 128-VGPR cap)
+// and SPILLS 112-124 B/lane of the register-cached row (NG=2, d=768) to
+// scratch EVERY row iteration — measured 104 us/call vs the ~45 us this
+// kernel's 4-5 streams cost at ln_fwd's 4 TB/s. Declaring the real
+// 256-thread shape lifts the cap and the spills vanish.
 template <typename T, int NG, bool DXADD = false>
-__global__ void ln_bwd_k(const T* __restrict__ x, const T* __restrict__ dy,
+__global__ __launch_bounds__(256) void ln_bwd_k(
+    const T* __restrict__ x, const T* __restrict__ dy,
                          T* __restrict__ dx, const float* __restrict__ w,
                          const float* __restrict__ mean,
                          const float* __restrict__ invstd,
