@@ -91,6 +91,8 @@ void pdta_gemm_bs(const void*, const void*, void*, int, int, int, int, int,
                   int, int, int, int, int, hipStream_t);
 void pdta_pack_weights(const float*, void*, const long*, const long*, int,
                        long, int, hipStream_t);
+void pdta_pack_transpose(const float*, void*, const long*, const long*,
+                         int, long, int, hipStream_t);
 int pdta_attn_fused_ok(int, int);
 void pdta_attn_fused_fwd(const void*, void*, void*, int, int, int, int,
                          float, int, hipStream_t);
@@ -880,6 +882,18 @@ void pack_weights(at::Tensor master, at::Tensor pool, at::Tensor desc,
                     desc.size(0), total_chunks, dt_of(pool), cur_stream());
 }
 
+// LDS-tiled [N][K] fp32 -> [K][N] dtype transpose segments (packs.hip
+// pack_transpose_k — the gather form is 16x read-amplified for large K)
+void pack_transpose(at::Tensor master, at::Tensor pool, at::Tensor desc,
+                    at::Tensor tprefix, int64_t total_tiles) {
+  CHECK_IN(master);
+  CHECK_IN(pool);
+  pdta_pack_transpose(master.data_ptr<float>(), pool.data_ptr(),
+                      desc.data_ptr<long>(), tprefix.data_ptr<long>(),
+                      desc.size(0), total_tiles, dt_of(pool),
+                      cur_stream());
+}
+
 // ---- RCCL ----------------------------------------------------------------
 #define RCCL_CHECK(cmd)                                                  \
   do {                                                                   \
@@ -1037,6 +1051,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
   m.def("pack_weights", &pack_weights);
+  m.def("pack_transpose", &pack_transpose);
   m.def("rccl_unique_id", &rccl_unique_id);
   m.def("rccl_comm_init", &rccl_comm_init);
   m.def("rccl_allreduce_avg", &rccl_allreduce_avg);

@@ -68,6 +68,65 @@ __global__ void pack_weights_k(const float* __restrict__ master,
   }
 }
 
+// LDS-tiled transpose for the kind-3 (cast_t, [N][K] fp32 -> [K][N]
+// dtype) segments: the gather form above reads ONE 4-B element per 64-B
+// cache line when K is large (16x read amplification — measured 1.03 ms
+// /step for ViT-B, ~0.7 TB/s). Here a block loads a 64x64 fp32 tile
+// with coalesced float4 row reads, transposes through LDS (stride-65
+// padding: conflict-free column reads), and writes 32-B dtype row
+// chunks. desc_t: int64 [nseg][4] = {src_off, dst_off, N, K};
+// tprefix: int64 [nseg] cumulative tile counts (inclusive).
+template <typename T>
+__global__ __launch_bounds__(256) void pack_transpose_k(
+    const float* __restrict__ master, T* __restrict__ pool,
+    const long* __restrict__ desc, const long* __restrict__ tprefix,
+    int nseg, long total_tiles) {
+  __shared__ float lds[64][65];
+  for (long tile = blockIdx.x; tile < total_tiles; tile += gridDim.x) {
+    int lo = 0, hi = nseg - 1;
+    while (lo < hi) {
+      int mid = (lo + hi) >> 1;
+      if (tprefix[mid] > tile) hi = mid;
+      else lo = mid + 1;
+    }
+    const long* d = desc + (long)lo * 4;
+    long src = d[0], dst = d[1], N = d[2], K = d[3];
+    long base = (lo == 0 ? 0 : tprefix[lo - 1]);
+    long local = tile - base;
+    long tiles_k = (K + 63) >> 6;
+    long n0 = (local / tiles_k) * 64, k0 = (local % tiles_k) * 64;
+
+    int row = threadIdx.x >> 2, c16 = (threadIdx.x & 3) * 16;
+    long n = n0 + row;
+    if (n < N) {
+      const float* srow = master + src + n * K + k0;
+#pragma unroll
+      for (int j = 0; j < 16; ++j)
+        lds[row][c16 + j] = (k0 + c16 + j < K) ? srow[c16 + j] : 0.f;
+    }
+    __syncthreads();
+    long k = k0 + row;
+    if (k < K) {
+      T* drow = pool + dst + k * N + n0;
+      Vec8<T> v;
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        int nn = c16 + h * 8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          v.v[j] = from_f32<T>(lds[nn + j][row]);
+        if (n0 + nn + 7 < N) {
+          store8(drow + nn, v);
+        } else {
+          for (int j = 0; j < 8 && n0 + nn + j < N; ++j)
+            drow[nn + j] = v.v[j];
+        }
+      }
+    }
+    __syncthreads();  // lds reused by the next tile
+  }
+}
+
 }  // namespace
 
 extern "C" {
@@ -78,6 +137,16 @@ void pdta_pack_weights(const float* master, void* pool, const long* desc,
   PDTA_DISPATCH_DTYPE16(dt, T, {
     pack_weights_k<T><<<ew_grid(total_chunks, 256, 1), 256, 0, s>>>(
         master, (T*)pool, desc, prefix, nseg, total_chunks);
+  });
+}
+
+void pdta_pack_transpose(const float* master, void* pool, const long* desc,
+                         const long* tprefix, int nseg, long total_tiles,
+                         int dt, hipStream_t s) {
+  long blocks = total_tiles < 16384 ? total_tiles : 16384;
+  PDTA_DISPATCH_DTYPE16(dt, T, {
+    pack_transpose_k<T><<<(int)blocks, 256, 0, s>>>(
+        master, (T*)pool, desc, tprefix, nseg, total_tiles);
   });
 }
 

@@ -25,6 +25,8 @@ class PackManager:
         base_ptr = flat.master.data_ptr()
 
         segs = []   # (kind, src_off, dst_off, K, C, R, S, numel)
+        tsegs = []  # kind-3 routed to the LDS-tiled transpose kernel:
+                    # (src_off, dst_off, N, K) for [N][K] fp32 -> [K][N]
         assign = []  # (module, attr, dst_off, shape)
         dst = 0
 
@@ -32,7 +34,15 @@ class PackManager:
             nonlocal dst
             src_off = (param.data_ptr() - base_ptr) // 4
             numel = param.numel()
-            segs.append((kind, src_off, dst, *shape_dims, numel))
+            if kind == 3 and shape_dims[1] % 8 == 0:
+                # shape_dims = (K, N, 1, 1) with master [N][K]; the
+                # gather form reads one 4-B element per cache line at
+                # large K (16x amplification) — use the tiled kernel.
+                # N % 8 keeps every dst row 16-B aligned for store8;
+                # others (none today) stay on the gather form.
+                tsegs.append((src_off, dst, shape_dims[1], shape_dims[0]))
+            else:
+                segs.append((kind, src_off, dst, *shape_dims, numel))
             assign.append((mod, attr, dst, view_shape))
             dst += _align8(numel)
 
@@ -63,7 +73,15 @@ class PackManager:
         prefix = torch.tensor(chunks, dtype=torch.long,
                               device=device).cumsum(0)
         self.prefix = prefix.contiguous()
-        self.total_chunks = int(prefix[-1]) if len(chunks) else 0
+        self.total_chunks = int(prefix[-1]) if chunks else 0
+        self.tdesc = torch.tensor([list(s) for s in tsegs],
+                                  dtype=torch.long, device=device)
+        tiles = [((n + 63) // 64) * ((k + 63) // 64)
+                 for (_, _, n, k) in tsegs]
+        tprefix = torch.tensor(tiles, dtype=torch.long,
+                               device=device).cumsum(0)
+        self.tprefix = tprefix.contiguous()
+        self.total_tiles = int(tprefix[-1]) if tiles else 0
 
         for mod, attr, off, shape in assign:
             n = 1
@@ -75,3 +93,6 @@ class PackManager:
         if self.total_chunks:
             ext().pack_weights(self.master, self.pool, self.desc,
                                self.prefix, self.total_chunks)
+        if self.total_tiles:
+            ext().pack_transpose(self.master, self.pool, self.tdesc,
+                                 self.tprefix, self.total_tiles)

@@ -195,3 +195,20 @@ def test_attention_qkv_shape_sweep_vs_fp32():
         if nv < s:
             tail = qkv_a.grad.view(b, s, 3 * d)[:, nv:]
             assert tail.abs().max().item() == 0.0, f"tail s={s} nv={nv}"
+
+
+def test_pack_transpose_kernel():
+    """LDS-tiled [N][K] fp32 -> [K][N] dtype transpose (packs.hip
+    pack_transpose_k) vs torch .t(), including non-64-multiple edges."""
+    for (n, k) in [(64, 64), (72, 100), (1000, 768), (2304, 768)]:
+        master = torch.randn(n * k, device=DEV, dtype=torch.float32)
+        pool = torch.full((k * n + 8,), float("nan"), device=DEV,
+                          dtype=torch.bfloat16)
+        desc = torch.tensor([[0, 0, n, k]], dtype=torch.long, device=DEV)
+        tiles = ((n + 63) // 64) * ((k + 63) // 64)
+        tprefix = torch.tensor([tiles], dtype=torch.long, device=DEV)
+        ext().pack_transpose(master, pool, desc, tprefix, tiles)
+        got = pool[:k * n].view(k, n).float()
+        ref = master.view(n, k).t().to(torch.bfloat16).float()
+        assert torch.equal(got, ref), (n, k)
+        assert pool[k * n:].isnan().all(), "wrote past the segment"
