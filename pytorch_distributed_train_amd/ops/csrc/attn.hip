@@ -259,19 +259,37 @@ __global__ __launch_bounds__(256) void attn_fused_bwd_ds_k(
     }
   }
 
-  // read this block's P tile (C layout), rowdot, dS
-  int colbase = lane & 15;
+  // stage this block's P tile [64][512B] into dsimg with glds16
+  // (coalesced; the direct C-layout read was 64 scalar 2-B global loads
+  // per lane), then read it back at the fragment layout from LDS. The
+  // XOR swizzle matches the dS store below, so logical chunk c of row r
+  // lives at LDS chunk c ^ (r & 7).
   const T* pg = p + (long)bh * SP * SP + (long)q0 * SP;
   T* dg = ds + (long)bh * SP * SP + (long)q0 * SP;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    int pr = wave * 8 + i;               // row pair 0..31 (2 rows/call)
+    int prow = pr * 2 + (lane >> 5);
+    const char* g = reinterpret_cast<const char*>(pg + (long)prow * SP) +
+                    (((lane & 31) ^ (prow & 7)) << 4);
+    glds16(g, dsimg + pr * 1024);
+  }
+  // no barrier: each wave stages exactly the 16 rows it reads below
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+  int colbase = lane & 15;
   float pv[16][4];
   float dot[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
   for (int jn = 0; jn < 16; ++jn) {
     int col = jn * 16 + colbase;
+    int byte = col * 2;
+    int chunk = byte >> 4, rem = byte & 15;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int row = wave * 16 + (lane >> 4) * 4 + r;
-      pv[jn][r] = to_f32(pg[(long)row * SP + col]);
+      pv[jn][r] = to_f32(*reinterpret_cast<const T*>(
+          dsimg + row * 512 + (((chunk ^ (row & 7)) << 4) | rem)));
       dot[r] += pv[jn][r] * acc[jn][r];
     }
   }
@@ -290,7 +308,8 @@ __global__ __launch_bounds__(256) void attn_fused_bwd_ds_k(
         (c16 ^ (krow & 7)) * 16;
     glds16(g, vimg + (wave * 8 + i) * 1024);
   }
-  // dS -> global (consumed by dK = dS^T·Q) and -> LDS A-image for dQ
+  // dS -> LDS A-image for dQ (global store is a vectorized flush below —
+  // the per-element form was 64 scalar 2-B stores per lane)
 #pragma unroll
   for (int jn = 0; jn < 16; ++jn) {
     int col = jn * 16 + colbase;
@@ -302,11 +321,22 @@ __global__ __launch_bounds__(256) void attn_fused_bwd_ds_k(
       T dsv = from_f32<T>(scale * pv[jn][r] * (acc[jn][r] - dot[r]));
       *reinterpret_cast<T*>(dsimg + row * 512 +
                             (((chunk ^ (row & 7)) << 4) | rem)) = dsv;
-      dg[(long)row * SP + col] = dsv;
     }
   }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // K staged
   __syncthreads();                                  // dS image visible
+  // flush dS tile to global (consumed by dK = dS^T·Q): 16-B chunks,
+  // un-swizzling the image rows
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    int pr = wave * 8 + i;
+    int prow = pr * 2 + (lane >> 5);
+    int c = lane & 31;  // logical chunk
+    uint4 v = *reinterpret_cast<const uint4*>(
+        dsimg + prow * 512 + ((c ^ (prow & 7)) << 4));
+    *reinterpret_cast<uint4*>(
+        reinterpret_cast<char*>(dg + (long)prow * SP) + c * 16) = v;
+  }
 
   // dQ = dS·K (K-dim = 256 keys; K^T fragments via ds_read_b64_tr_b16)
   f32x4 acc2[4];
