@@ -92,6 +92,15 @@ def build_parser() -> argparse.ArgumentParser:
                         "when distributed, off single-process — the "
                         "always-on ProcessGroupNCCL watchdog analog); 0 = "
                         "off")
+    p.add_argument("--torch-profile", type=int, default=0,
+                   help="capture a torch.profiler (Kineto/roctracer) "
+                        "trace of the first N steps of the first epoch; "
+                        "rank 0 writes a chrome trace to "
+                        "--torch-profile-out (SURVEY.md §5.1 step-level "
+                        "overlap visualization; kernel-level counters "
+                        "come from rocprofv3 instead)")
+    p.add_argument("--torch-profile-out",
+                   default="gpurun_out/torch_trace.json")
     p.add_argument("--fault-inject", default="",
                    help="step:N:rank:R -> raise at step N on rank R; "
                         "hang:step:N:rank:R -> block forever instead "

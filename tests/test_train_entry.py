@@ -94,3 +94,18 @@ def test_train_deterministic_across_runs(tmp_path):
         outs.append([re.sub(r"images_per_sec=[0-9.]+", "", l)
                      for l in r.stdout.splitlines() if "loss=" in l])
     assert outs[0] == outs[1] and len(outs[0]) >= 8
+
+
+def test_torch_profile_flag_writes_trace(tmp_path, monkeypatch):
+    """--torch-profile N captures the first N steps and rank 0 exports a
+    chrome trace (SURVEY.md §5.1 step-level overlap visualization)."""
+    import train as train_mod
+    out = tmp_path / "trace.json"
+    monkeypatch.chdir(tmp_path)
+    rc = train_mod.main([
+        "--model", "resnet18", "--data", "synthetic-cifar10",
+        "--num-samples", "64", "--batch-size", "16", "--epochs", "1",
+        "--steps-per-epoch", "4", "--device", "cpu",
+        "--torch-profile", "3", "--torch-profile-out", str(out)])
+    assert rc == 0
+    assert out.exists() and out.stat().st_size > 100

@@ -115,6 +115,33 @@ def main(argv=None):
         ckpt_path = os.path.join(args.checkpoint_dir,
                                  f"{args.model}_checkpoint.pt")
 
+    profiler = None
+    if args.torch_profile > 0 and start_epoch < args.epochs:
+        # step-level overlap trace (compute vs reducer-stream collectives)
+        # via torch.profiler -> Kineto -> roctracer; spans come from the
+        # record_function markers in engine.Trainer.train_step
+        from torch.profiler import (profile, schedule, ProfilerActivity,
+                                    tensorboard_trace_handler)  # noqa: F401
+        acts = [ProfilerActivity.CPU]
+        if torch.cuda.is_available():
+            acts.append(ProfilerActivity.CUDA)
+
+        def _save(prof):
+            if env.rank == 0:
+                out = args.torch_profile_out
+                os.makedirs(os.path.dirname(out) or ".", exist_ok=True)
+                prof.export_chrome_trace(out)
+                print(f"torch-profile trace -> {out}")
+
+        profiler = profile(
+            activities=acts,
+            schedule=schedule(wait=0, warmup=min(2, args.torch_profile - 1),
+                              active=max(1, args.torch_profile -
+                                         min(2, args.torch_profile - 1)),
+                              repeat=1),
+            on_trace_ready=_save)
+        profiler.__enter__()
+
     for epoch in range(start_epoch, args.epochs):
         trainer.epoch = epoch
         trainer.optimizer.lr = lr_at(args, epoch)
@@ -127,6 +154,11 @@ def main(argv=None):
             images = images.to(trainer.device, non_blocking=True)
             labels = labels.to(trainer.device, non_blocking=True)
             loss = trainer.train_step(images, labels)
+            if profiler is not None:
+                profiler.step()
+                if epoch == start_epoch and step + 1 >= args.torch_profile:
+                    profiler.__exit__(None, None, None)
+                    profiler = None
             ips = timer.tick(args.batch_size * env.world_size)
             if step % args.print_freq == 0:
                 logger.log(epoch=epoch, step=step, loss=float(loss),
@@ -143,6 +175,8 @@ def main(argv=None):
             save_checkpoint(ckpt_path, epoch + 1, trainer.global_step,
                             trainer.model, trainer.optimizer, trainer.scaler)
             logger.log(epoch=epoch, checkpoint=ckpt_path)
+    if profiler is not None:  # fewer steps ran than --torch-profile
+        profiler.__exit__(None, None, None)
     logger.close()
     return 0
 
