@@ -282,6 +282,87 @@ class RcclComm:
         return clone
 
 
+class DebugComm:
+    """`TORCH_DISTRIBUTED_DEBUG=DETAIL` analog (SURVEY.md §5.2 — torch's
+    ProcessGroupWrapper, c10d/ProcessGroupWrapper.hpp): with
+    PDTA_DEBUG_COLLECTIVES=1, every collective first cross-checks its
+    signature (op, dtype, numel, src/out-numel) across ranks over the
+    gloo control plane and raises a descriptive error on EVERY rank
+    instead of letting a mismatched collective deadlock RCCL or silently
+    corrupt. Debug tool: adds one host-synchronous gloo allgather per
+    collective — never enable for performance runs."""
+
+    _OPS = {"allreduce_avg": 1, "allreduce_sum": 2, "broadcast": 3,
+            "allgather_into": 4, "barrier": 5}
+
+    def __init__(self, inner, control):
+        self._inner = inner
+        self._control = control
+        self.rank = inner.rank
+        self.world_size = inner.world_size
+
+    @staticmethod
+    def _dt_code(dtype) -> int:
+        # deterministic across processes (str hash() is randomized)
+        name = str(dtype)
+        code = 0
+        for ch in name:
+            code = (code * 31 + ord(ch)) % (1 << 31)
+        return code
+
+    def _verify(self, op: str, a: int, b: int, c: int) -> None:
+        sig = torch.tensor([self._OPS[op], a, b, c], dtype=torch.long)
+        table = torch.zeros(self.world_size * 4, dtype=torch.long)
+        # flat out: gloo's all_gather chunks it into inp-shaped pieces
+        self._control.allgather_into(table, sig).wait()
+        rows = table.view(self.world_size, 4)
+        if not (rows == sig).all():
+            names = {v: k for k, v in self._OPS.items()}
+            desc = "; ".join(
+                f"rank {r}: op={names.get(int(rows[r][0]), '?')} "
+                f"dt={int(rows[r][1])} n={int(rows[r][2])} "
+                f"x={int(rows[r][3])}" for r in range(self.world_size))
+            raise RuntimeError(
+                f"[PDTA_DEBUG_COLLECTIVES] collective mismatch at rank "
+                f"{self.rank} issuing {op}: {desc}")
+
+    def allreduce_avg(self, tensor, stream=None) -> Work:
+        self._verify("allreduce_avg", self._dt_code(tensor.dtype),
+                     tensor.numel(), 0)
+        return self._inner.allreduce_avg(tensor, stream)
+
+    def allreduce_sum(self, tensor, stream=None) -> Work:
+        self._verify("allreduce_sum", self._dt_code(tensor.dtype),
+                     tensor.numel(), 0)
+        return self._inner.allreduce_sum(tensor, stream)
+
+    def broadcast(self, tensor, src: int = 0, stream=None) -> Work:
+        self._verify("broadcast", self._dt_code(tensor.dtype),
+                     tensor.numel(), src)
+        return self._inner.broadcast(tensor, src, stream)
+
+    def allgather_into(self, out, inp, stream=None) -> Work:
+        self._verify("allgather_into", self._dt_code(inp.dtype),
+                     inp.numel(), out.numel())
+        return self._inner.allgather_into(out, inp, stream)
+
+    def barrier(self) -> None:
+        self._verify("barrier", 0, 0, 0)
+        self._inner.barrier()
+
+    def abort(self) -> None:
+        self._inner.abort()
+
+    def dup(self):
+        return DebugComm(self._inner.dup(), self._control)
+
+
+def _maybe_debug(comm, control):
+    if os.environ.get("PDTA_DEBUG_COLLECTIVES") == "1":
+        return DebugComm(comm, control)
+    return comm
+
+
 def make_communicator(env: DistEnv, device, backend: str = "auto"):
     """backend: auto|rccl|gloo. auto = rccl on cuda devices, gloo on cpu."""
     if env.world_size <= 1:
@@ -306,10 +387,11 @@ def make_communicator(env: DistEnv, device, backend: str = "auto"):
               f"{torch.cuda.device_count()} GPUs — RCCL cannot share a "
               "device; using host-staged gloo collectives",
               file=sys.stderr)
-        return StagedGlooComm(env, backend="gloo")
+        staged = StagedGlooComm(env, backend="gloo")
+        return _maybe_debug(staged, staged.dup())
     control = TorchDistComm(env, backend="gloo")
     if backend == "gloo":
-        return control
+        return _maybe_debug(control, control)
     if backend == "rccl":
-        return RcclComm(env, control, dev)
+        return _maybe_debug(RcclComm(env, control, dev), control)
     raise ValueError(f"unknown backend {backend}")

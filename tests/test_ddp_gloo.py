@@ -287,3 +287,57 @@ def test_syncbn_ragged_batches_match_single_process(tmp_path):
     assert torch.allclose(r0["rm"], bn.running_mean, atol=1e-5)
     assert torch.allclose(r0["rv"], bn.running_var, atol=1e-5)
     assert torch.allclose(r0["rm"], r1["rm"], atol=1e-7)
+
+
+def _debug_comm_worker(rank, world, port, out_dir, mismatch):
+    os.environ.update(RANK=str(rank), LOCAL_RANK=str(rank),
+                      WORLD_SIZE=str(world), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port),
+                      PDTA_DEBUG_COLLECTIVES="1")
+    from pytorch_distributed_train_amd.flags import DistEnv
+    from pytorch_distributed_train_amd.parallel.comm import (
+        make_communicator, DebugComm)
+    env = DistEnv.from_env()
+    comm = make_communicator(env, torch.device("cpu"), backend="gloo")
+    assert isinstance(comm, DebugComm)
+    result = {"ok": None, "err": ""}
+    try:
+        # matching collectives pass verification and compute correctly
+        t = torch.full((4,), float(rank + 1))
+        comm.allreduce_sum(t).wait()
+        assert torch.equal(t, torch.full((4,), 3.0))
+        comm.barrier()
+        if mismatch:
+            # rank 1 issues a different numel -> BOTH ranks must raise a
+            # descriptive error instead of hanging/corrupting
+            bad = torch.zeros(8 if rank == 0 else 16)
+            comm.allreduce_sum(bad).wait()
+        result["ok"] = True
+    except RuntimeError as e:
+        result["ok"] = False
+        result["err"] = str(e)
+    torch.save(result, os.path.join(out_dir, f"dbg{rank}.pt"))
+
+
+def test_debug_collectives_cross_check(tmp_path):
+    """PDTA_DEBUG_COLLECTIVES=1 (TORCH_DISTRIBUTED_DEBUG=DETAIL analog,
+    SURVEY.md §5.2): matched collectives pass; a shape mismatch raises a
+    descriptive error on every rank instead of deadlocking."""
+    for mismatch in (False, True):
+        ctx = mp.get_context("spawn")
+        port = _free_port()
+        procs = [ctx.Process(target=_debug_comm_worker,
+                             args=(r, 2, port, str(tmp_path), mismatch))
+                 for r in range(2)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(120)
+            assert p.exitcode == 0
+        for r in range(2):
+            res = torch.load(tmp_path / f"dbg{r}.pt")
+            if mismatch:
+                assert res["ok"] is False, res
+                assert "collective mismatch" in res["err"]
+            else:
+                assert res["ok"] is True, res
