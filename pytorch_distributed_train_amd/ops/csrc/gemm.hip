@@ -73,6 +73,22 @@ __global__ __launch_bounds__(TPB) void gemm_nt_k(
   int m0 = tm * BM, n0 = tn * BN;
   int ktiles = K / BK;
 
+  // bias values hoisted so the loads fly under the K-loop (loading
+  // bias[col] inside the store loop was latency-exposed and redundant)
+  float badd[4] = {0.f, 0.f, 0.f, 0.f};
+  if (BIAS) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = n0 + wn * 64 + j * 16 + (lane & 15);
+      badd[j] = bias[col < N ? col : N - 1];
+      // pin NOW: otherwise the compiler sinks these loads into the
+      // epilogue and every store cluster leads with s_waitcnt vmcnt(0),
+      // which (vmcnt counts stores too) serializes the whole store
+      // stream — measured +28% on the small-K ViT shapes
+      asm volatile("" : "+v"(badd[j]));
+    }
+  }
+
   stage_tile<T>(A, K, m0, M - 1, lds[0][0], wave, lane);
   stage_tile<T>(B, K, n0, N - 1, lds[0][1], wave, lane);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -121,12 +137,11 @@ __global__ __launch_bounds__(TPB) void gemm_nt_k(
     for (int j = 0; j < 4; ++j) {
       int col = n0 + wn * 64 + j * 16 + (lane & 15);
       if (col >= N) continue;
-      float badd = BIAS ? bias[col] : 0.f;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int row = rbase + r;
         if (row < M)
-          C[(long)row * N + col] = from_f32<OutT>(acc[i][j][r] + badd);
+          C[(long)row * N + col] = from_f32<OutT>(acc[i][j][r] + badd[j]);
       }
     }
   }
@@ -192,6 +207,23 @@ __global__ __launch_bounds__(512) void gemm_nt_big_k(
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
+  // bias hoisted pre-K-loop: the in-store-loop bias[col] load was 32
+  // latency-exposed scalar loads per lane (8x redundant) — measured
+  // +26-139% on the small-K ViT shapes (gemm_probe PDTA_PROBE_BIAS)
+  float badd[4] = {0.f, 0.f, 0.f, 0.f};
+  if (BIAS) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = n0 + wn * 64 + j * 16 + (lane & 15);
+      badd[j] = bias[col < N ? col : N - 1];
+      // pin NOW: otherwise the compiler sinks these loads into the
+      // epilogue and every store cluster leads with s_waitcnt vmcnt(0),
+      // which (vmcnt counts stores too) serializes the whole store
+      // stream — measured +28% on the small-K ViT shapes
+      asm volatile("" : "+v"(badd[j]));
+    }
+  }
+
   stage_all(0, 0);
 
   for (int kt = 0; kt < ktiles; ++kt) {
@@ -241,12 +273,11 @@ __global__ __launch_bounds__(512) void gemm_nt_big_k(
     for (int j = 0; j < 4; ++j) {
       int col = n0 + wn * 64 + j * 16 + (lane & 15);
       if (col >= N) continue;
-      float badd = BIAS ? bias[col] : 0.f;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         long row = rbase + r;
         if (row < M)
-          C[(long)row * N + col] = from_f32<OutT>(acc[i][j][r] + badd);
+          C[(long)row * N + col] = from_f32<OutT>(acc[i][j][r] + badd[j]);
       }
     }
   }
