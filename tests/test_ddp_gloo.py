@@ -307,11 +307,18 @@ def _debug_comm_worker(rank, world, port, out_dir, mismatch):
         comm.allreduce_sum(t).wait()
         assert torch.equal(t, torch.full((4,), 3.0))
         comm.barrier()
-        if mismatch:
+        if mismatch == "numel":
             # rank 1 issues a different numel -> BOTH ranks must raise a
             # descriptive error instead of hanging/corrupting
             bad = torch.zeros(8 if rank == 0 else 16)
             comm.allreduce_sum(bad).wait()
+        elif mismatch == "op":
+            # ranks disagree on WHICH collective comes next — the
+            # classic deadlock DETAIL mode exists to catch
+            if rank == 0:
+                comm.allreduce_sum(torch.zeros(4)).wait()
+            else:
+                comm.broadcast(torch.zeros(4), src=0).wait()
         result["ok"] = True
     except RuntimeError as e:
         result["ok"] = False
@@ -323,7 +330,7 @@ def test_debug_collectives_cross_check(tmp_path):
     """PDTA_DEBUG_COLLECTIVES=1 (TORCH_DISTRIBUTED_DEBUG=DETAIL analog,
     SURVEY.md §5.2): matched collectives pass; a shape mismatch raises a
     descriptive error on every rank instead of deadlocking."""
-    for mismatch in (False, True):
+    for mismatch in (False, "numel", "op"):
         ctx = mp.get_context("spawn")
         port = _free_port()
         procs = [ctx.Process(target=_debug_comm_worker,
