@@ -34,7 +34,8 @@ template <typename T>
 __global__ __launch_bounds__(256) void attn_fused_fwd_k(
     const T* __restrict__ qkv, T* __restrict__ probs, T* __restrict__ out,
     int h, int srows, int nvalid, float scale) {
-  // grid: (ceil(srows/64) query blocks, b*h). 4 waves; 16 query rows each.
+  // grid: (SP/64 query blocks, b*h) — always the full padded range so
+  // the probs tail rows get written (zeros). 4 waves; 16 query rows each.
   __shared__ __attribute__((aligned(16))) char qtile[64 * 128];
   __shared__ __attribute__((aligned(16))) char kv[SP * 128];  // K, then V
   __shared__ __attribute__((aligned(16))) char pimg[64 * 512];
