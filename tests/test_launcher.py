@@ -122,3 +122,20 @@ def test_watchdog_exits_gang_on_hang(tmp_path):
     assert r.returncode != 0
     assert "[watchdog]" in r.stderr, r.stderr[-2000:]
     assert "flight recorder" in r.stderr
+
+
+def test_launcher_multinode_rank_offsets(tmp_path):
+    """--nnodes/--node_rank parity with torch.distributed.launch: this
+    node's workers get global RANK = node_rank * nproc + local_rank and
+    WORLD_SIZE = nnodes * nproc (only this node's gang actually spawns —
+    rendezvous to the other node is the workers' business)."""
+    r = _launch(OK_SCRIPT, tmp_path,
+                extra=("--nnodes=3", "--node_rank=2",
+                       "--master_addr=127.0.0.1"))
+    assert r.returncode == 0, r.stderr
+    import json
+    for local in (0, 1):
+        env = json.load(open(tmp_path / f"w{4 + local}.json"))
+        assert env["RANK"] == str(2 * 2 + local)
+        assert env["LOCAL_RANK"] == str(local)
+        assert env["WORLD_SIZE"] == "6"
