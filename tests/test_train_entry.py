@@ -109,3 +109,26 @@ def test_torch_profile_flag_writes_trace(tmp_path, monkeypatch):
         "--torch-profile", "3", "--torch-profile-out", str(out)])
     assert rc == 0
     assert out.exists() and out.stat().st_size > 100
+
+
+def test_bench_self_launch_2ranks_cpu(tmp_path):
+    """`python bench.py --gpus 2` with no rendezvous env must spawn 2
+    local ranks and print an n_gpus=2 record — CPU/gloo variant of the
+    driver contract so every round covers the bare invocation."""
+    import json
+    import subprocess
+    env = dict(os.environ)
+    for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE", "MASTER_ADDR",
+              "MASTER_PORT"):
+        env.pop(k, None)
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--gpus", "2", "--steps", "2",
+         "--warmup", "1", "--model", "resnet18", "--batch-size", "4"],
+        cwd=repo, env=env, capture_output=True, text=True, timeout=420)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    rec = json.loads(line)
+    assert rec["n_gpus"] == 2
+    assert rec["config"]["parallelism"] == "dp2"
+    assert rec["value"] > 0
